@@ -1,0 +1,318 @@
+"""Net: the layer DAG built from a NetParameter.
+
+Parity with /root/reference/src/caffe/net.cpp (Init walk with in-place
+detection and shared-param ownership, FilterNet/StateMeetsRule,
+ForwardFromTo/BackwardFromTo, CopyTrainedLayersFrom, ToProto). The InitPS
+table walk (net.cpp:253-363) has no analogue: parameters are plain device
+tensors, synchronized by RCCL collectives.
+
+Backward accepts a post-layer callback so the distributed solver can launch
+per-layer gradient all-reduce on a side HIP stream as soon as a layer's
+grads exist -- the stream/event realization of Poseidon's DWBP
+(solver.cpp:405-451) without host threads.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Callable, Dict, List, Optional
+
+import torch
+
+from .blob import Blob
+from .context import ctx
+from .insert_splits import insert_splits
+from .layer import Layer, create_layer
+from ..proto import Message, spec
+
+TRAIN = spec.ENUMS["Phase"]["TRAIN"]
+TEST = spec.ENUMS["Phase"]["TEST"]
+
+
+def state_meets_rule(state: Message, rule: Message) -> bool:
+    """NetStateRule check (net.cpp:423-468)."""
+    if rule.has("phase") and rule.phase != state.phase:
+        return False
+    if rule.has("min_level") and state.level < rule.min_level:
+        return False
+    if rule.has("max_level") and state.level > rule.max_level:
+        return False
+    stages = set(state.stage)
+    for s in rule.stage:
+        if s not in stages:
+            return False
+    for s in rule.not_stage:
+        if s in stages:
+            return False
+    return True
+
+
+def filter_net(param: Message, state: Message) -> Message:
+    out = Message("NetParameter")
+    for name in ("name", "input", "input_dim", "force_backward"):
+        if param.has(name):
+            v = getattr(param, name)
+            if name in ("input", "input_dim"):
+                getattr(out, name).extend(v)
+            else:
+                setattr(out, name, v)
+    for layer in param.layers:
+        inc, exc = list(layer.include), list(layer.exclude)
+        if inc and exc:
+            raise ValueError(f"layer {layer.name}: specify include OR exclude, not both")
+        keep = (any(state_meets_rule(state, r) for r in inc) if inc
+                else not any(state_meets_rule(state, r) for r in exc))
+        if keep:
+            out.layers.append(layer)
+    return out
+
+
+@dataclass
+class ParamSpec:
+    blob: Blob
+    layer_idx: int
+    param_idx: int
+    lr_mult: float
+    decay_mult: float
+    owner: int  # index into net.params of the owner (== own index if owner)
+    name: str = ""
+
+
+class Net:
+    def __init__(self, param: Message, phase: int = TRAIN,
+                 state: Optional[Message] = None, verbose: Optional[bool] = None):
+        st = Message("NetState", phase=phase)
+        if param.has("state"):
+            st.merge_from(param.state)
+            st.phase = phase
+        if state is not None:
+            st.merge_from(state)
+        self.phase = phase
+        filtered = filter_net(param, st)
+        self.param = insert_splits(filtered)
+        self.name = self.param.name or ""
+        self._verbose = ctx().is_root() if verbose is None else verbose
+
+        self.layers: List[Layer] = []
+        self.layer_names: List[str] = []
+        self.bottoms: List[List[Blob]] = []
+        self.tops: List[List[Blob]] = []
+        self.bottom_need_bwd: List[List[bool]] = []
+        self.layer_need_bwd: List[bool] = []
+        self.blobs: Dict[str, Blob] = {}
+        self.params: List[ParamSpec] = []
+        self._param_name_to_idx: Dict[str, int] = {}
+        self.output_blob_names: List[str] = []
+        self._loss_tops: List = []  # (layer_idx, top_idx, weight)
+
+        self._build()
+
+    # ------------------------------------------------------------------
+    def _build(self) -> None:
+        p = self.param
+        available = set()
+        blob_need: Dict[str, bool] = {}
+
+        # explicit net inputs (deploy-style nets)
+        dims = list(p.input_dim)
+        for i, name in enumerate(p.input):
+            b = Blob(tuple(dims[4 * i:4 * i + 4]), name=name)
+            self.blobs[name] = b
+            available.add(name)
+            blob_need[name] = False
+
+        force_bwd = bool(p.force_backward)
+
+        for li, lp in enumerate(p.layers):
+            layer = create_layer(lp, self.phase)
+            self.layers.append(layer)
+            self.layer_names.append(layer.name)
+
+            bottoms: List[Blob] = []
+            bneed: List[bool] = []
+            for bname in lp.bottom:
+                if bname not in available:
+                    raise ValueError(f"layer {layer.name}: unknown bottom {bname!r}")
+                bottoms.append(self.blobs[bname])
+                bneed.append(blob_need.get(bname, False) or force_bwd)
+
+            tops: List[Blob] = []
+            for ti, tname in enumerate(lp.top):
+                if ti < len(lp.bottom) and tname == lp.bottom[ti]:
+                    tops.append(self.blobs[tname])  # in-place
+                else:
+                    if tname in self.blobs:
+                        raise ValueError(f"duplicate top blob {tname!r}")
+                    b = Blob(name=tname)
+                    self.blobs[tname] = b
+                    tops.append(b)
+                    available.add(tname)
+
+            layer.setup(bottoms, tops)
+
+            # loss weights
+            weights = list(lp.loss_weight)
+            lw = []
+            for ti in range(len(tops)):
+                w = weights[ti] if ti < len(weights) else layer.auto_loss_weight(ti)
+                lw.append(w)
+                if w != 0.0:
+                    self._loss_tops.append((li, ti, w))
+            layer.loss_weights = lw
+
+            # need-backward propagation
+            has_lr = any(layer.blobs_lr(i) != 0.0 for i in range(len(layer.blobs)))
+            need = (any(bneed) or (has_lr and len(layer.blobs) > 0)
+                    or any(w != 0.0 for w in lw)) and len(lp.bottom) > 0
+            # data-ish layers (no bottoms) never run backward
+            for tname in lp.top:
+                blob_need[tname] = blob_need.get(tname, False) or need
+            self.layer_need_bwd.append(need)
+            self.bottoms.append(bottoms)
+            self.tops.append(tops)
+            self.bottom_need_bwd.append(bneed)
+
+            # parameter registration (+ sharing by name)
+            pnames = list(lp.param)
+            share_modes = list(lp.blob_share_mode)
+            for pi, pblob in enumerate(layer.blobs):
+                pname = pnames[pi] if pi < len(pnames) else ""
+                idx = len(self.params)
+                if pname and pname in self._param_name_to_idx:
+                    owner_idx = self._param_name_to_idx[pname]
+                    owner = self.params[owner_idx]
+                    mode = share_modes[pi] if pi < len(share_modes) else 0
+                    if mode == 0 and owner.blob.shape != pblob.shape:
+                        raise ValueError(
+                            f"shared param {pname!r}: shape mismatch "
+                            f"{owner.blob.shape} vs {pblob.shape} (STRICT)")
+                    if mode == 1 and owner.blob.count != pblob.count:
+                        raise ValueError(f"shared param {pname!r}: count mismatch")
+                    pblob.share_data(owner.blob)
+                    pblob.share_diff(owner.blob)
+                    self.params.append(ParamSpec(
+                        blob=pblob, layer_idx=li, param_idx=pi,
+                        lr_mult=layer.blobs_lr(pi),
+                        decay_mult=layer.weight_decay_mult(pi),
+                        owner=owner_idx, name=pname))
+                else:
+                    if pname:
+                        self._param_name_to_idx[pname] = idx
+                    self.params.append(ParamSpec(
+                        blob=pblob, layer_idx=li, param_idx=pi,
+                        lr_mult=layer.blobs_lr(pi),
+                        decay_mult=layer.weight_decay_mult(pi),
+                        owner=idx, name=pname))
+
+            for bname in lp.bottom:
+                pass  # blobs stay available (split layers made copies already)
+
+        consumed = {b for lp in p.layers for b in lp.bottom}
+        self.output_blob_names = [n for n in available if n not in consumed]
+
+        # Prune layers that do not contribute to any loss from backward
+        # (Caffe's layer_contributes_loss logic): ACCURACY/ARGMAX heads etc.
+        contributing_blobs = set()
+        loss_layers = {li for (li, ti, w) in self._loss_tops}
+        for li in range(len(p.layers) - 1, -1, -1):
+            lp = p.layers[li]
+            contributes = li in loss_layers or any(
+                t in contributing_blobs for t in lp.top)
+            if contributes:
+                contributing_blobs.update(lp.bottom)
+            else:
+                self.layer_need_bwd[li] = False
+
+    # ------------------------------------------------------------------
+    @property
+    def learnable_params(self) -> List[ParamSpec]:
+        return [ps for i, ps in enumerate(self.params) if ps.owner == i
+                and ps.lr_mult != 0.0]
+
+    def zero_param_diffs(self) -> None:
+        for i, ps in enumerate(self.params):
+            if ps.owner == i:
+                ps.blob.zero_diff()
+
+    def forward(self, start: int = 0, end: Optional[int] = None) -> float:
+        end = len(self.layers) if end is None else end
+        loss = 0.0
+        for i in range(start, end):
+            self.layers[i].forward(self.bottoms[i], self.tops[i])
+            for (li, ti, w) in self._loss_tops:
+                if li == i:
+                    loss += w * float(self.tops[i][ti].data.sum().item())
+        return loss
+
+    def forward_async(self) -> torch.Tensor:
+        """Forward pass without host synchronization: returns the loss as a
+        0-d device tensor (sum of weighted loss tops)."""
+        dev = ctx().torch_device
+        loss = torch.zeros((), dtype=torch.float32, device=dev)
+        marks = {}
+        for (li, ti, w) in self._loss_tops:
+            marks.setdefault(li, []).append((ti, w))
+        for i, layer in enumerate(self.layers):
+            layer.forward(self.bottoms[i], self.tops[i])
+            for (ti, w) in marks.get(i, []):
+                loss = loss + w * self.tops[i][ti].data.sum().to(torch.float32)
+        return loss
+
+    def backward(self, post_layer_cb: Optional[Callable[[int, Layer], None]] = None
+                 ) -> None:
+        """Run backward over all layers that need it.
+
+        post_layer_cb(layer_idx, layer) fires right after a layer's backward
+        -- the DWBP hook: by this point the layer's param diffs are final and
+        gradient comm may be enqueued while backprop continues below.
+        """
+        # seed loss-top diffs with their loss weights
+        for (li, ti, w) in self._loss_tops:
+            t = self.tops[li][ti]
+            t.diff.fill_(w)
+        for i in range(len(self.layers) - 1, -1, -1):
+            if not self.layer_need_bwd[i]:
+                continue
+            self.layers[i].backward(self.tops[i], self.bottom_need_bwd[i],
+                                    self.bottoms[i])
+            if post_layer_cb is not None and self.layers[i].blobs:
+                post_layer_cb(i, self.layers[i])
+
+    def clear_activation_diffs(self) -> None:
+        """Zero intermediate blob diffs between iterations (layers accumulate
+        into bottom diffs where blobs fan out via splits)."""
+        for b in self.blobs.values():
+            if b.has_diff():
+                b.zero_diff()
+
+    # -- checkpoint interop --------------------------------------------
+    def to_proto(self, write_diff: bool = False) -> Message:
+        out = Message.decode("NetParameter", self.param.encode())
+        for li, lp in enumerate(out.layers):
+            del lp.blobs[:]
+            for pblob in self.layers[li].blobs:
+                lp.blobs.append(pblob.to_proto(write_diff))
+        return out
+
+    def copy_trained_layers_from(self, src: Message) -> None:
+        """Load weights by layer name (net.cpp:908-950)."""
+        by_name = {l.name: l for l in self.layers}
+        for lp in src.layers:
+            layer = by_name.get(lp.name)
+            if layer is None:
+                continue
+            src_blobs = list(lp.blobs)
+            if not src_blobs:
+                continue
+            if len(src_blobs) != len(layer.blobs):
+                raise ValueError(
+                    f"layer {lp.name}: {len(src_blobs)} checkpoint blobs vs "
+                    f"{len(layer.blobs)} params")
+            for pb, sb in zip(layer.blobs, src_blobs):
+                pb.from_proto(sb, reshape=False)
+
+    def blob(self, name: str) -> Blob:
+        return self.blobs[name]
+
+    def has_layer_type(self, type_name: str) -> bool:
+        return any(l.type_name == type_name for l in self.layers)

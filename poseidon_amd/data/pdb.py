@@ -1,0 +1,102 @@
+"""PDB ("Poseidon DB"): a simple indexed record file of serialized Datum
+protos -- the MI355X build's dataset container, replacing the reference's
+LevelDB/LMDB backends (which need native libs this image does not ship).
+
+Layout (little-endian):
+    magic "PSDB" | u32 version | u64 nrecords
+    u64 offsets[nrecords+1]           (byte offsets of record starts; last = EOF)
+    record bytes (concatenated serialized Datum protos)
+"""
+
+from __future__ import annotations
+
+import struct
+from typing import Iterator, List
+
+import numpy as np
+
+from ..proto import Message
+
+_MAGIC = b"PSDB"
+_VERSION = 1
+
+
+class PDBWriter:
+    def __init__(self, path: str):
+        self.path = path
+        self._records: List[bytes] = []
+
+    def put(self, datum: Message) -> None:
+        self._records.append(datum.encode())
+
+    def put_raw(self, raw: bytes) -> None:
+        self._records.append(raw)
+
+    def close(self) -> None:
+        n = len(self._records)
+        header = _MAGIC + struct.pack("<IQ", _VERSION, n)
+        base = len(header) + 8 * (n + 1)
+        offsets = [base]
+        for r in self._records:
+            offsets.append(offsets[-1] + len(r))
+        with open(self.path, "wb") as f:
+            f.write(header)
+            f.write(struct.pack(f"<{n + 1}Q", *offsets))
+            for r in self._records:
+                f.write(r)
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+
+
+class PDBReader:
+    def __init__(self, path: str):
+        self.path = path
+        self._f = open(path, "rb")
+        header = self._f.read(16)
+        if header[:4] != _MAGIC:
+            raise ValueError(f"{path}: not a PDB file")
+        version, n = struct.unpack("<IQ", header[4:])
+        self.n = n
+        self.offsets = np.frombuffer(self._f.read(8 * (n + 1)), dtype="<u8")
+
+    def __len__(self) -> int:
+        return self.n
+
+    def get_raw(self, i: int) -> bytes:
+        start, end = int(self.offsets[i]), int(self.offsets[i + 1])
+        self._f.seek(start)
+        return self._f.read(end - start)
+
+    def get(self, i: int) -> Message:
+        return Message.decode("Datum", self.get_raw(i))
+
+    def __iter__(self) -> Iterator[Message]:
+        for i in range(self.n):
+            yield self.get(i)
+
+    def close(self) -> None:
+        self._f.close()
+
+
+def datum_to_array(datum: Message) -> np.ndarray:
+    """Decode a Datum into a float32 CHW array (io.cpp Datum semantics)."""
+    c, h, w = datum.channels, datum.height, datum.width
+    if datum.has("data") and len(datum.data):
+        arr = np.frombuffer(datum.data, dtype=np.uint8).astype(np.float32)
+    else:
+        arr = np.asarray(datum.float_data, dtype=np.float32)
+    return arr.reshape(c, h, w)
+
+
+def array_to_datum(arr: np.ndarray, label: int) -> Message:
+    d = Message("Datum", channels=arr.shape[0], height=arr.shape[1],
+                width=arr.shape[2], label=int(label))
+    if arr.dtype == np.uint8:
+        d.data = arr.tobytes()
+    else:
+        d.float_data = arr.astype(np.float32).ravel()
+    return d

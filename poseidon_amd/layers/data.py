@@ -1,0 +1,300 @@
+"""Data layers: DummyData, MemoryData, Data (PDB/LMDB), ImageData,
+HDF5Data/HDF5Output, WindowData.
+
+Parity: /root/reference/src/caffe/layers/{dummy_data,memory_data,data,
+image_data,hdf5_data,hdf5_output,window_data}_layer.cpp and
+base_data_layer.cpp (background prefetch thread + double buffer).
+
+Distributed sharding keeps the reference's striped semantics (SURVEY P4,
+data_layer.cpp:143-161): rank r of W ranks reads records r, r+W, r+2W, ...
+(here one process == one GPU, so thread striping collapses to rank striping).
+"""
+
+from __future__ import annotations
+
+import os
+import threading
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from ..core.blob import Blob
+from ..core.context import ctx
+from ..core.layer import Layer, register_layer
+from ..core import filler as fillers
+from ..data.pdb import PDBReader, datum_to_array
+from ..data.transformer import DataTransformer
+from ..proto import Message
+
+
+@register_layer("DUMMY_DATA")
+class DummyDataLayer(Layer):
+    exact_num_bottom = 0
+
+    def layer_setup(self, bottom, top) -> None:
+        dp = self.param.ensure("dummy_data_param")
+        n_top = len(top)
+        fparams = list(dp.data_filler)
+        if len(fparams) == 0:
+            fparams = [Message("FillerParameter", type="constant", value=0.0)]
+        if len(fparams) == 1:
+            fparams = fparams * n_top
+        self.fillers = fparams
+        # refill each forward only for non-constant fillers (dummy_data_layer.cpp)
+        self.refill = [fp.type != "constant" for fp in fparams]
+
+        def dim(lst, i):
+            vals = list(lst)
+            if not vals:
+                return None
+            return int(vals[i]) if len(vals) > 1 else int(vals[0])
+
+        self.shapes = []
+        for i in range(n_top):
+            shape = tuple(d for d in (dim(dp.num, i), dim(dp.channels, i),
+                                      dim(dp.height, i), dim(dp.width, i))
+                          if d is not None)
+            self.shapes.append(shape)
+        self._filled = False
+
+    def reshape(self, bottom, top) -> None:
+        for t, s in zip(top, self.shapes):
+            t.reshape(s)
+
+    def forward(self, bottom, top) -> None:
+        for i, t in enumerate(top):
+            if not self._filled or self.refill[i]:
+                fillers.fill(t, self.fillers[i])
+        self._filled = True
+
+    def backward(self, top, propagate_down, bottom) -> None:
+        pass
+
+
+@register_layer("MEMORY_DATA")
+class MemoryDataLayer(Layer):
+    exact_num_bottom = 0
+    exact_num_top = 2
+
+    def layer_setup(self, bottom, top) -> None:
+        mp = self.param.ensure("memory_data_param")
+        self.batch = int(mp.batch_size)
+        self.c, self.h, self.w = int(mp.channels), int(mp.height), int(mp.width)
+        self._data: Optional[torch.Tensor] = None
+        self._labels: Optional[torch.Tensor] = None
+        self._pos = 0
+
+    def add_data(self, data: torch.Tensor, labels: torch.Tensor) -> None:
+        assert data.shape[0] % self.batch == 0
+        self._data = data
+        self._labels = labels
+        self._pos = 0
+
+    # alias matching the reference's MemoryDataLayer::Reset
+    reset = add_data
+
+    def reshape(self, bottom, top) -> None:
+        top[0].reshape(self.batch, self.c, self.h, self.w)
+        top[1].reshape(self.batch)
+
+    def forward(self, bottom, top) -> None:
+        if self._data is None:
+            raise RuntimeError("MemoryDataLayer: call add_data() first")
+        n = self._data.shape[0]
+        idx = torch.arange(self._pos, self._pos + self.batch) % n
+        dev = ctx().torch_device
+        top[0].data = self._data[idx].to(dev, torch.float32)
+        top[1].data = self._labels[idx].to(dev, torch.float32)
+        self._pos = (self._pos + self.batch) % n
+
+    def backward(self, top, propagate_down, bottom) -> None:
+        pass
+
+
+class _PrefetchingDataLayer(Layer):
+    """Base for DB-backed layers: a host thread assembles the next batch
+    (decode + transform) while the GPU trains on the current one
+    (base_data_layer.cpp:56-105)."""
+
+    exact_num_bottom = 0
+
+    def _start_prefetch(self) -> None:
+        self._ready = threading.Event()
+        self._taken = threading.Event()
+        self._taken.set()
+        self._stop = False
+        self._batch = None
+        self._thread = threading.Thread(target=self._loop, daemon=True)
+        self._thread.start()
+
+    def _loop(self) -> None:
+        while not self._stop:
+            self._taken.wait()
+            self._taken.clear()
+            self._batch = self._load_batch()
+            self._ready.set()
+
+    def _next_batch(self):
+        self._ready.wait()
+        self._ready.clear()
+        batch = self._batch
+        self._taken.set()
+        return batch
+
+    def _load_batch(self):
+        raise NotImplementedError
+
+    def backward(self, top, propagate_down, bottom) -> None:
+        pass
+
+
+@register_layer("DATA")
+class DataLayer(_PrefetchingDataLayer):
+    """Reads Datum records from a PDB file (or LMDB if the lib is present).
+
+    Striped sharding: with shared_file_system (one DB for all ranks) rank r
+    starts at record r and strides by world_size; without it, each rank opens
+    source_<r> (per-client shards, the reference's source_k convention,
+    data_layer.cpp:232-261)."""
+
+    def layer_setup(self, bottom, top) -> None:
+        dp = self.param.ensure("data_param")
+        self.batch = int(dp.batch_size)
+        c = ctx()
+        source = dp.source
+        backend = dp.enum_name("backend") if dp.has("backend") else "LEVELDB"
+        shared = bool(dp.shared_file_system) or c.world_size == 1
+        if not shared:
+            source = f"{source}_{c.rank}"
+        if backend == "LMDB" and not os.path.isfile(source):
+            try:
+                import lmdb  # noqa: F401
+                raise NotImplementedError(
+                    "LMDB directory sources need the lmdb module wiring")
+            except ImportError as e:
+                raise RuntimeError(
+                    f"LMDB backend requested but lmdb module unavailable: {e}. "
+                    "Convert the dataset to PDB with tools/convert_dataset.py")
+        self.db = PDBReader(source)
+        self.stride = c.world_size if shared else 1
+        self.cursor = c.rank if shared else 0
+        rng = np.random.default_rng(c.seed + 131 * c.rank)
+        self.transform = DataTransformer(self.param.transform_param,
+                                         self.phase, rng)
+        first = datum_to_array(self.db.get(0))
+        sample = self.transform(first)
+        self.shape = sample.shape
+        self._start_prefetch()
+
+    def _load_batch(self):
+        data = np.empty((self.batch,) + self.shape, dtype=np.float32)
+        labels = np.empty((self.batch,), dtype=np.float32)
+        for i in range(self.batch):
+            d = self.db.get(self.cursor % len(self.db))
+            data[i] = self.transform(datum_to_array(d))
+            labels[i] = d.label or 0
+            self.cursor += self.stride
+        return data, labels
+
+    def reshape(self, bottom, top) -> None:
+        top[0].reshape((self.batch,) + self.shape)
+        if len(top) > 1:
+            top[1].reshape(self.batch)
+
+    def forward(self, bottom, top) -> None:
+        data, labels = self._next_batch()
+        dev = ctx().torch_device
+        top[0].data = torch.from_numpy(data).to(dev, non_blocking=True)
+        if len(top) > 1:
+            top[1].data = torch.from_numpy(labels).to(dev, non_blocking=True)
+
+
+@register_layer("IMAGE_DATA")
+class ImageDataLayer(_PrefetchingDataLayer):
+    """File-list image reader (image_data_layer.cpp). Uses PIL for decode;
+    rank-striped partitioning of the list (thread_global_idx semantics)."""
+
+    def layer_setup(self, bottom, top) -> None:
+        ip = self.param.ensure("image_data_param")
+        self.batch = int(ip.batch_size)
+        self.new_h, self.new_w = int(ip.new_height), int(ip.new_width)
+        c = ctx()
+        source = ip.source
+        shared = bool(ip.shared_file_system) or c.world_size == 1
+        if not shared:
+            source = f"{source}_{c.rank}"
+        with open(source) as f:
+            lines = [l.split() for l in f if l.strip()]
+        self.entries = [(p, int(lbl)) for p, lbl in lines]
+        if shared and c.world_size > 1:
+            self.entries = self.entries[c.rank::c.world_size]
+        if bool(ip.shuffle):
+            rng = np.random.default_rng(c.seed + c.rank)
+            rng.shuffle(self.entries)
+        self.cursor = 0
+        rng = np.random.default_rng(c.seed + 977 * c.rank)
+        self.transform = DataTransformer(self.param.transform_param,
+                                         self.phase, rng)
+        sample = self.transform(self._read(self.entries[0][0]))
+        self.shape = sample.shape
+        self._start_prefetch()
+
+    def _read(self, path: str) -> np.ndarray:
+        from PIL import Image
+        img = Image.open(path).convert("RGB")
+        if self.new_h and self.new_w:
+            img = img.resize((self.new_w, self.new_h))
+        arr = np.asarray(img, dtype=np.float32)  # HWC RGB
+        # Caffe stores BGR; match channel order for mean-file compat
+        return np.ascontiguousarray(arr[:, :, ::-1].transpose(2, 0, 1))
+
+    def _load_batch(self):
+        data = np.empty((self.batch,) + self.shape, dtype=np.float32)
+        labels = np.empty((self.batch,), dtype=np.float32)
+        for i in range(self.batch):
+            path, lbl = self.entries[self.cursor % len(self.entries)]
+            data[i] = self.transform(self._read(path))
+            labels[i] = lbl
+            self.cursor += 1
+        return data, labels
+
+    def reshape(self, bottom, top) -> None:
+        top[0].reshape((self.batch,) + self.shape)
+        top[1].reshape(self.batch)
+
+    def forward(self, bottom, top) -> None:
+        data, labels = self._next_batch()
+        dev = ctx().torch_device
+        top[0].data = torch.from_numpy(data).to(dev, non_blocking=True)
+        top[1].data = torch.from_numpy(labels).to(dev, non_blocking=True)
+
+
+@register_layer("HDF5_DATA")
+class HDF5DataLayer(Layer):
+    def layer_setup(self, bottom, top) -> None:
+        raise RuntimeError(
+            "HDF5_DATA requires h5py, which this image does not ship; "
+            "convert to PDB with tools/convert_dataset.py")
+
+    def reshape(self, bottom, top) -> None:
+        pass
+
+
+@register_layer("HDF5_OUTPUT")
+class HDF5OutputLayer(Layer):
+    def layer_setup(self, bottom, top) -> None:
+        raise RuntimeError("HDF5_OUTPUT requires h5py (not available)")
+
+    def reshape(self, bottom, top) -> None:
+        pass
+
+
+@register_layer("WINDOW_DATA")
+class WindowDataLayer(Layer):
+    def layer_setup(self, bottom, top) -> None:
+        raise NotImplementedError(
+            "WINDOW_DATA (R-CNN window sampling) is not wired yet")
+
+    def reshape(self, bottom, top) -> None:
+        pass

@@ -1,0 +1,225 @@
+"""Vision layers: Convolution, Pooling, LRN, Im2col.
+
+Parity: /root/reference/src/caffe/layers/{conv_layer,pooling_layer,lrn_layer,
+im2col_layer}.{cpp,cu}. All GPU math routes through ops.functional into
+hand-written CDNA4 kernels (implicit-GEMM MFMA conv, fused pool/LRN kernels).
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import torch
+import torch.nn.functional as F
+
+from ..core.blob import Blob
+from ..core.layer import Layer, register_layer
+from ..core import filler
+from ..ops import functional as ops
+from ..proto import Message
+
+
+def _pair(param, base: str, generic: str):
+    """Resolve Caffe's (kernel_size | kernel_h/kernel_w)-style params."""
+    h = getattr(param, f"{base}_h")
+    w = getattr(param, f"{base}_w")
+    if param.has(f"{base}_h") or param.has(f"{base}_w"):
+        return int(h), int(w)
+    g = getattr(param, generic)
+    if g is None:
+        raise ValueError(f"missing {generic} / {base}_h/w")
+    return int(g), int(g)
+
+
+@register_layer("CONVOLUTION")
+class ConvolutionLayer(Layer):
+    exact_num_top = None  # conv supports multiple bottom/top pairs
+
+    def layer_setup(self, bottom: List[Blob], top: List[Blob]) -> None:
+        cp = self.param.ensure("convolution_param")
+        self.kernel = _pair(cp, "kernel", "kernel_size")
+        self.stride = _pair(cp, "stride", "stride")
+        ph, pw = cp.pad_h, cp.pad_w
+        self.pad = (int(ph), int(pw)) if (cp.has("pad_h") or cp.has("pad_w")) \
+            else (int(cp.pad), int(cp.pad))
+        self.group = int(cp.group)
+        self.num_output = int(cp.num_output)
+        self.bias_term = bool(cp.bias_term)
+        channels = bottom[0].channels
+        assert channels % self.group == 0 and self.num_output % self.group == 0
+
+        dtype = bottom[0].dtype
+        w = Blob((self.num_output, channels // self.group,
+                  self.kernel[0], self.kernel[1]), dtype=dtype,
+                 name=f"{self.name}.weight")
+        filler.fill(w, cp.weight_filler if cp.has("weight_filler") else None)
+        self.blobs = [w]
+        if self.bias_term:
+            # Caffe conv bias blob is (1,1,1,Cout) (conv_layer.cpp:96-99)
+            b = Blob((1, 1, 1, self.num_output), dtype=dtype,
+                     name=f"{self.name}.bias")
+            filler.fill(b, cp.bias_filler if cp.has("bias_filler") else None)
+            self.blobs.append(b)
+
+    def reshape(self, bottom: List[Blob], top: List[Blob]) -> None:
+        n, c, h, w = bottom[0].shape
+        ho = ops.conv_out_size(h, self.kernel[0], self.pad[0], self.stride[0])
+        wo = ops.conv_out_size(w, self.kernel[1], self.pad[1], self.stride[1])
+        for t in top:
+            t.reshape(n, self.num_output, ho, wo)
+
+    def forward(self, bottom: List[Blob], top: List[Blob]) -> None:
+        w = self.blobs[0].data
+        b = self.blobs[1].data.view(-1) if self.bias_term else None
+        for bo, t in zip(bottom, top):
+            t.data = ops.conv2d_forward(bo.data, w, b, self.stride, self.pad,
+                                        self.group)
+
+    def backward(self, top: List[Blob], propagate_down: List[bool],
+                 bottom: List[Blob]) -> None:
+        w = self.blobs[0].data
+        for i, (bo, t) in enumerate(zip(bottom, top)):
+            dy = t.diff
+            self.blobs[0].diff.add_(ops.conv2d_backward_weight(
+                bo.data, dy, self.blobs[0].shape, self.stride, self.pad,
+                self.group))
+            if self.bias_term:
+                self.blobs[1].diff.view(-1).add_(dy.sum(dim=(0, 2, 3)))
+            if propagate_down[i]:
+                bo.diff = ops.conv2d_backward_input(
+                    w, dy, bo.shape, self.stride, self.pad, self.group)
+
+
+@register_layer("POOLING")
+class PoolingLayer(Layer):
+    exact_num_bottom = 1
+
+    def layer_setup(self, bottom, top) -> None:
+        pp = self.param.ensure("pooling_param")
+        self.kernel = _pair(pp, "kernel", "kernel_size")
+        self.stride = _pair(pp, "stride", "stride")
+        self.pad = (int(pp.pad_h), int(pp.pad_w)) \
+            if (pp.has("pad_h") or pp.has("pad_w")) else (int(pp.pad), int(pp.pad))
+        self.method = pp.enum_name("pool")
+        if self.method != "AVE":
+            assert self.pad == (0, 0) or self.method == "MAX"
+        self._mask = None
+
+    def reshape(self, bottom, top) -> None:
+        n, c, h, w = bottom[0].shape
+        ho = ops.pool_out_size(h, self.kernel[0], self.pad[0], self.stride[0])
+        wo = ops.pool_out_size(w, self.kernel[1], self.pad[1], self.stride[1])
+        top[0].reshape(n, c, ho, wo)
+        if len(top) > 1:
+            top[1].reshape(n, c, ho, wo)
+
+    def forward(self, bottom, top) -> None:
+        x = bottom[0].data
+        if self.method == "MAX":
+            y, mask = ops.pool_max_forward(x, self.kernel, self.stride, self.pad)
+            self._mask = mask
+            if len(top) > 1:
+                top[1].data = mask.to(x.dtype)
+        elif self.method == "AVE":
+            y = ops.pool_ave_forward(x, self.kernel, self.stride, self.pad)
+        else:  # STOCHASTIC
+            if self.phase == 0:
+                y, mask = ops.pool_stoch_forward_train(
+                    x, self.kernel, self.stride, self.pad)
+                self._mask = mask
+            else:
+                y = ops.pool_stoch_forward_test(
+                    x, self.kernel, self.stride, self.pad)
+        top[0].data = y
+
+    def backward(self, top, propagate_down, bottom) -> None:
+        if not propagate_down[0]:
+            return
+        dy = top[0].diff
+        if self.method == "AVE":
+            bottom[0].diff = ops.pool_ave_backward(
+                dy, bottom[0].shape, self.kernel, self.stride, self.pad)
+        else:  # MAX and STOCHASTIC both scatter via stored mask
+            bottom[0].diff = ops.pool_max_backward(dy, self._mask, bottom[0].shape)
+
+
+@register_layer("LRN")
+class LRNLayer(Layer):
+    exact_num_bottom = 1
+    exact_num_top = 1
+
+    def layer_setup(self, bottom, top) -> None:
+        lp = self.param.ensure("lrn_param")
+        self.size = int(lp.local_size)
+        assert self.size % 2 == 1, "LRN local_size must be odd"
+        self.alpha = float(lp.alpha)
+        self.beta = float(lp.beta)
+        self.region = lp.enum_name("norm_region")
+        self._scale = None
+        self._y = None
+        if self.region == "WITHIN_CHANNEL":
+            # composite: x^2 -> ave-pool(size) -> scale -> x * scale^-beta
+            self.pre_pad = (self.size - 1) // 2
+
+    def reshape(self, bottom, top) -> None:
+        top[0].reshape(bottom[0].shape)
+
+    def forward(self, bottom, top) -> None:
+        x = bottom[0].data
+        if self.region == "ACROSS_CHANNELS":
+            y, scale = ops.lrn_forward(x, self.size, self.alpha, self.beta)
+        else:
+            k = (self.size, self.size)
+            sq = x * x
+            avg = ops.pool_ave_forward(sq, k, (1, 1), (self.pre_pad, self.pre_pad))
+            scale = 1.0 + self.alpha * avg
+            y = x * scale.pow(-self.beta)
+        self._scale = scale
+        self._y = y
+        top[0].data = y
+
+    def backward(self, top, propagate_down, bottom) -> None:
+        if not propagate_down[0]:
+            return
+        x, dy = bottom[0].data, top[0].diff
+        if self.region == "ACROSS_CHANNELS":
+            bottom[0].diff = ops.lrn_backward(x, self._y, self._scale, dy,
+                                              self.size, self.alpha, self.beta)
+        else:
+            scale = self._scale
+            k = (self.size, self.size)
+            # dx = dy*scale^-b - 2*a*b/size^2-less: within-channel window avg
+            ratio = dy * self._y / scale
+            spread = ops.pool_ave_backward(ratio, bottom[0].shape, k, (1, 1),
+                                           (self.pre_pad, self.pre_pad))
+            bottom[0].diff = dy * scale.pow(-self.beta) \
+                - 2.0 * self.alpha * self.beta * x * spread
+
+
+@register_layer("IM2COL")
+class Im2colLayer(Layer):
+    exact_num_bottom = 1
+    exact_num_top = 1
+
+    def layer_setup(self, bottom, top) -> None:
+        cp = self.param.ensure("convolution_param")
+        self.kernel = _pair(cp, "kernel", "kernel_size")
+        self.stride = _pair(cp, "stride", "stride")
+        self.pad = (int(cp.pad), int(cp.pad))
+
+    def reshape(self, bottom, top) -> None:
+        n, c, h, w = bottom[0].shape
+        ho = ops.conv_out_size(h, self.kernel[0], self.pad[0], self.stride[0])
+        wo = ops.conv_out_size(w, self.kernel[1], self.pad[1], self.stride[1])
+        top[0].reshape(n, c * self.kernel[0] * self.kernel[1], ho * wo)
+
+    def forward(self, bottom, top) -> None:
+        top[0].data = F.unfold(bottom[0].data, self.kernel, padding=self.pad,
+                               stride=self.stride)
+
+    def backward(self, top, propagate_down, bottom) -> None:
+        if not propagate_down[0]:
+            return
+        n, c, h, w = bottom[0].shape
+        bottom[0].diff = F.fold(top[0].diff, (h, w), self.kernel,
+                                padding=self.pad, stride=self.stride)

@@ -1,0 +1,447 @@
+"""Op dispatch: hand-written HIP/CDNA4 kernels on GPU, plain fp32 torch on CPU.
+
+The CPU path is the numerics reference for every GPU kernel (tests compare
+the two). The GPU path REQUIRES the compiled in-tree extension
+(poseidon_amd/ops/_hip) -- it raises rather than silently falling back to
+eager torch, so a GPU run that passes is running our CDNA4 kernels.
+
+Semantics mirror the reference layer math:
+  conv      /root/reference/src/caffe/layers/conv_layer.{cpp,cu}
+  pooling   /root/reference/src/caffe/layers/pooling_layer.cu (Caffe ceil
+            geometry + far-edge-clipped AVE pool_size)
+  lrn       /root/reference/src/caffe/layers/lrn_layer.cu
+  softmax   /root/reference/src/caffe/layers/softmax_layer.cu
+  sgd       /root/reference/src/caffe/solver.cpp:815-892 (fused here)
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _ext():
+    global _EXT, _EXT_ERR
+    if _EXT is None and _EXT_ERR is None:
+        try:
+            from . import _backend
+            _EXT = _backend.load()
+        except Exception as e:  # noqa: BLE001
+            _EXT_ERR = str(e)
+    if _EXT is None:
+        raise RuntimeError(
+            "poseidon_amd HIP extension is not built/loadable but a GPU op was "
+            f"requested. Build it with `python setup.py build_ext --inplace` "
+            f"(PYTORCH_ROCM_ARCH=gfx950). Original error: {_EXT_ERR}")
+    return _EXT
+
+
+def ext_available() -> bool:
+    try:
+        _ext()
+        return True
+    except RuntimeError:
+        return False
+
+
+# ---------------------------------------------------------------------------
+# Geometry helpers (Caffe pooling/conv shape rules)
+# ---------------------------------------------------------------------------
+
+def conv_out_size(h: int, k: int, p: int, s: int) -> int:
+    return (h + 2 * p - k) // s + 1
+
+
+def pool_out_size(h: int, k: int, p: int, s: int) -> Tuple[int, bool]:
+    """Caffe pooled size: ceil((h+2p-k)/s)+1, minus 1 when the last window
+    would start in the padding (pooling_layer.cpp:64-77)."""
+    out = int(math.ceil((h + 2 * p - k) / s)) + 1
+    if p > 0 and (out - 1) * s >= h + p:
+        out -= 1
+    return out
+
+
+# ---------------------------------------------------------------------------
+# Convolution
+# ---------------------------------------------------------------------------
+
+def conv2d_forward(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor],
+                   stride: Tuple[int, int], pad: Tuple[int, int],
+                   groups: int) -> torch.Tensor:
+    if x.is_cuda:
+        return _ext().conv2d_forward(x, w, b, stride[0], stride[1],
+                                     pad[0], pad[1], groups)
+    return F.conv2d(x, w, b, stride=stride, padding=pad, groups=groups)
+
+
+def conv2d_backward_input(w: torch.Tensor, dy: torch.Tensor,
+                          x_shape, stride, pad, groups: int) -> torch.Tensor:
+    if dy.is_cuda:
+        return _ext().conv2d_backward_input(w, dy, list(x_shape), stride[0],
+                                            stride[1], pad[0], pad[1], groups)
+    return torch.nn.grad.conv2d_input(list(x_shape), w, dy, stride=stride,
+                                      padding=pad, groups=groups)
+
+
+def conv2d_backward_weight(x: torch.Tensor, dy: torch.Tensor,
+                           w_shape, stride, pad, groups: int) -> torch.Tensor:
+    if dy.is_cuda:
+        return _ext().conv2d_backward_weight(x, dy, list(w_shape), stride[0],
+                                             stride[1], pad[0], pad[1], groups)
+    return torch.nn.grad.conv2d_weight(x, list(w_shape), dy, stride=stride,
+                                       padding=pad, groups=groups)
+
+
+# ---------------------------------------------------------------------------
+# Inner product (FC): y[M,N] = x[M,K] @ w[N,K]^T + b
+# ---------------------------------------------------------------------------
+
+def linear_forward(x: torch.Tensor, w: torch.Tensor,
+                   b: Optional[torch.Tensor]) -> torch.Tensor:
+    if x.is_cuda:
+        return _ext().linear_forward(x, w, b)
+    y = x.matmul(w.t())
+    if b is not None:
+        y = y + b
+    return y
+
+
+def linear_backward(x: torch.Tensor, w: torch.Tensor, dy: torch.Tensor,
+                    need_dx: bool, need_dw: bool, has_bias: bool):
+    if dy.is_cuda:
+        return _ext().linear_backward(x, w, dy, need_dx, need_dw, has_bias)
+    dx = dy.matmul(w) if need_dx else None
+    dw = dy.t().matmul(x) if need_dw else None
+    db = dy.sum(dim=0) if has_bias else None
+    return dx, dw, db
+
+
+def gemm_at_b(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """a[M,N]^T-free outer-product GEMM used by SFB reconstruction:
+    returns a^T @ b for a[B,N], b[B,K] -> [N,K]."""
+    if a.is_cuda:
+        return _ext().gemm_at_b(a, b)
+    return a.t().matmul(b)
+
+
+# ---------------------------------------------------------------------------
+# Pooling
+# ---------------------------------------------------------------------------
+
+def _unfold_pad(x, k, s, p, pad_value: float, Ho: int, Wo: int):
+    """Pad so that all Ho/Wo Caffe ceil-mode windows exist (the last window
+    may extend past H+p), then unfold."""
+    H, W = x.shape[2], x.shape[3]
+    far_h = max(p[0], (Ho - 1) * s[0] + k[0] - H - p[0])
+    far_w = max(p[1], (Wo - 1) * s[1] + k[1] - W - p[1])
+    xp = F.pad(x, (p[1], far_w, p[0], far_h), value=pad_value)
+    return xp.unfold(2, k[0], s[0]).unfold(3, k[1], s[1])
+
+
+def pool_max_forward(x, k, s, p):
+    if x.is_cuda:
+        return _ext().pool_max_forward(x, k[0], k[1], s[0], s[1], p[0], p[1])
+    N, C, H, W = x.shape
+    Ho = pool_out_size(H, k[0], p[0], s[0])
+    Wo = pool_out_size(W, k[1], p[1], s[1])
+    win = _unfold_pad(x, k, s, p, float("-inf"), Ho, Wo)[:, :, :Ho, :Wo]
+    flat = win.contiguous().view(N, C, Ho, Wo, -1)
+    y, idx = flat.max(dim=-1)
+    # map window-local argmax to bottom (unpadded) flat index h*W + w
+    kh = idx // k[1]
+    kw = idx % k[1]
+    oh = torch.arange(Ho, device=x.device).view(1, 1, Ho, 1)
+    ow = torch.arange(Wo, device=x.device).view(1, 1, 1, Wo)
+    h = oh * s[0] - p[0] + kh
+    wdx = ow * s[1] - p[1] + kw
+    mask = (h * W + wdx).to(torch.int32)
+    return y, mask
+
+
+def pool_max_backward(dy, mask, x_shape):
+    if dy.is_cuda:
+        return _ext().pool_max_backward(dy, mask, list(x_shape))
+    N, C, H, W = x_shape
+    dx = torch.zeros(N, C, H * W, dtype=dy.dtype, device=dy.device)
+    dx.scatter_add_(2, mask.view(N, C, -1).long(), dy.view(N, C, -1))
+    return dx.view(N, C, H, W)
+
+
+def _ave_pool_sizes(H, W, k, s, p, Ho, Wo, device):
+    oh = torch.arange(Ho, device=device)
+    ow = torch.arange(Wo, device=device)
+    hstart = oh * s[0] - p[0]
+    wstart = ow * s[1] - p[1]
+    hend = torch.clamp(hstart + k[0], max=H + p[0])
+    wend = torch.clamp(wstart + k[1], max=W + p[1])
+    return ((hend - hstart).view(Ho, 1) * (wend - wstart).view(1, Wo)).float()
+
+
+def pool_ave_forward(x, k, s, p):
+    if x.is_cuda:
+        return _ext().pool_ave_forward(x, k[0], k[1], s[0], s[1], p[0], p[1])
+    N, C, H, W = x.shape
+    Ho = pool_out_size(H, k[0], p[0], s[0])
+    Wo = pool_out_size(W, k[1], p[1], s[1])
+    win = _unfold_pad(x, k, s, p, 0.0, Ho, Wo)[:, :, :Ho, :Wo]
+    ssum = win.contiguous().view(N, C, Ho, Wo, -1).sum(dim=-1)
+    sizes = _ave_pool_sizes(H, W, k, s, p, Ho, Wo, x.device).to(x.dtype)
+    return ssum / sizes
+
+
+def pool_ave_backward(dy, x_shape, k, s, p):
+    if dy.is_cuda:
+        return _ext().pool_ave_backward(dy, list(x_shape), k[0], k[1],
+                                        s[0], s[1], p[0], p[1])
+    N, C, H, W = x_shape
+    Ho, Wo = dy.shape[2], dy.shape[3]
+    sizes = _ave_pool_sizes(H, W, k, s, p, Ho, Wo, dy.device).to(dy.dtype)
+    g = (dy / sizes).view(N * C, 1, Ho, Wo)
+    # distribute each output's gradient over its (real-region) window = fold
+    Hp = max(H + 2 * p[0], (Ho - 1) * s[0] + k[0])
+    Wp = max(W + 2 * p[1], (Wo - 1) * s[1] + k[1])
+    cols = g.expand(N * C, k[0] * k[1], Ho, Wo).reshape(N * C, k[0] * k[1], Ho * Wo)
+    dxp = F.fold(cols, (Hp, Wp), (k[0], k[1]), stride=(s[0], s[1]))
+    dx = dxp[:, :, p[0]:p[0] + H, p[1]:p[1] + W]
+    return dx.reshape(N, C, H, W)
+
+
+def pool_stoch_forward_train(x, k, s, p, rand=None):
+    """Stochastic pooling (train): pick an element of each window with
+    probability proportional to its value (pooling_layer.cu:81-120)."""
+    if x.is_cuda:
+        return _ext().pool_stoch_forward_train(x, k[0], k[1], s[0], s[1],
+                                               p[0], p[1],
+                                               int(torch.randint(0, 2**31 - 1, ()).item()))
+    N, C, H, W = x.shape
+    Ho = pool_out_size(H, k[0], p[0], s[0])
+    Wo = pool_out_size(W, k[1], p[1], s[1])
+    win = _unfold_pad(x, k, s, p, 0.0, Ho, Wo)[:, :, :Ho, :Wo]
+    flat = win.contiguous().view(N, C, Ho, Wo, -1)
+    csum = flat.cumsum(dim=-1)
+    total = csum[..., -1:]
+    if rand is None:
+        rand = torch.rand(N, C, Ho, Wo, 1, device=x.device, dtype=x.dtype)
+    thresh = rand * total
+    idx = (csum < thresh).sum(dim=-1).clamp(max=k[0] * k[1] - 1)
+    y = flat.gather(-1, idx.unsqueeze(-1)).squeeze(-1)
+    kh = idx // k[1]
+    kw = idx % k[1]
+    oh = torch.arange(Ho, device=x.device).view(1, 1, Ho, 1)
+    ow = torch.arange(Wo, device=x.device).view(1, 1, 1, Wo)
+    h = (oh * s[0] - p[0] + kh).clamp(0, H - 1)
+    wdx = (ow * s[1] - p[1] + kw).clamp(0, W - 1)
+    mask = (h * W + wdx).to(torch.int32)
+    return y, mask
+
+
+def pool_stoch_forward_test(x, k, s, p):
+    if x.is_cuda:
+        return _ext().pool_stoch_forward_test(x, k[0], k[1], s[0], s[1], p[0], p[1])
+    N, C, H, W = x.shape
+    Ho = pool_out_size(H, k[0], p[0], s[0])
+    Wo = pool_out_size(W, k[1], p[1], s[1])
+    win = _unfold_pad(x, k, s, p, 0.0, Ho, Wo)[:, :, :Ho, :Wo]
+    flat = win.contiguous().view(N, C, Ho, Wo, -1)
+    num = (flat * flat).sum(-1)
+    den = flat.sum(-1)
+    return num / (den + torch.finfo(x.dtype).tiny)
+
+
+# ---------------------------------------------------------------------------
+# LRN (across channels)
+# ---------------------------------------------------------------------------
+
+def lrn_forward(x, size: int, alpha: float, beta: float):
+    if x.is_cuda:
+        return _ext().lrn_forward(x, size, alpha, beta)
+    N, C, H, W = x.shape
+    sq = (x * x).view(N, 1, C, H * W)
+    pad = (size - 1) // 2
+    # sliding sum over channel dim: conv with ones kernel
+    kernel = torch.ones(1, 1, size, 1, dtype=x.dtype, device=x.device)
+    ssum = F.conv2d(sq, kernel, padding=(size - 1 - pad, 0))
+    ssum = ssum[:, :, :C, :].view(N, C, H, W)
+    scale = 1.0 + (alpha / size) * ssum
+    y = x * scale.pow(-beta)
+    return y, scale
+
+
+def lrn_backward(x, y, scale, dy, size: int, alpha: float, beta: float):
+    if x.is_cuda:
+        return _ext().lrn_backward(x, y, scale, dy, size, alpha, beta)
+    N, C, H, W = x.shape
+    pad = (size - 1) // 2
+    ratio = (dy * y / scale).view(N, 1, C, H * W)
+    kernel = torch.ones(1, 1, size, 1, dtype=x.dtype, device=x.device)
+    acc = F.conv2d(ratio, kernel, padding=(pad, 0))[:, :, :C, :].view(N, C, H, W)
+    return dy * scale.pow(-beta) - (2.0 * alpha * beta / size) * x * acc
+
+
+# ---------------------------------------------------------------------------
+# Softmax + losses
+# ---------------------------------------------------------------------------
+
+def softmax_forward(x: torch.Tensor) -> torch.Tensor:
+    """Channel softmax per spatial position (softmax_layer.cu)."""
+    if x.is_cuda:
+        return _ext().softmax_forward(x)
+    return F.softmax(x, dim=1)
+
+
+def softmax_backward(y: torch.Tensor, dy: torch.Tensor) -> torch.Tensor:
+    if y.is_cuda:
+        return _ext().softmax_backward(y, dy)
+    dot = (dy * y).sum(dim=1, keepdim=True)
+    return (dy - dot) * y
+
+
+def softmax_loss_forward(logits: torch.Tensor, labels: torch.Tensor):
+    """Returns (mean NLL over batch as 0-d tensor, probs)."""
+    if logits.is_cuda:
+        return _ext().softmax_loss_forward(logits, labels)
+    prob = F.softmax(logits, dim=1)
+    n = logits.shape[0]
+    picked = prob[torch.arange(n), labels.long().view(-1)]
+    loss = -torch.log(torch.clamp(picked, min=torch.finfo(prob.dtype).tiny)).sum() / n
+    return loss, prob
+
+
+def softmax_loss_backward(prob: torch.Tensor, labels: torch.Tensor,
+                          loss_weight: float) -> torch.Tensor:
+    if prob.is_cuda:
+        return _ext().softmax_loss_backward(prob, labels, loss_weight)
+    n = prob.shape[0]
+    dx = prob.clone()
+    dx[torch.arange(n), labels.long().view(-1)] -= 1.0
+    return dx * (loss_weight / n)
+
+
+# ---------------------------------------------------------------------------
+# Elementwise / neuron ops
+# ---------------------------------------------------------------------------
+
+def relu_forward(x, negative_slope: float = 0.0):
+    if x.is_cuda:
+        return _ext().relu_forward(x, negative_slope)
+    return F.leaky_relu(x, negative_slope)
+
+
+def relu_backward(x, dy, negative_slope: float = 0.0):
+    if x.is_cuda:
+        return _ext().relu_backward(x, dy, negative_slope)
+    return torch.where(x > 0, dy, dy * negative_slope)
+
+
+def sigmoid_forward(x):
+    if x.is_cuda:
+        return _ext().sigmoid_forward(x)
+    return torch.sigmoid(x)
+
+
+def sigmoid_backward(y, dy):
+    if y.is_cuda:
+        return _ext().sigmoid_backward(y, dy)
+    return dy * y * (1 - y)
+
+
+def tanh_forward(x):
+    if x.is_cuda:
+        return _ext().tanh_forward(x)
+    return torch.tanh(x)
+
+
+def tanh_backward(y, dy):
+    if y.is_cuda:
+        return _ext().tanh_backward(y, dy)
+    return dy * (1 - y * y)
+
+
+def bnll_forward(x):
+    """log(1 + exp(x)), computed stably (bnll_layer.cu)."""
+    if x.is_cuda:
+        return _ext().bnll_forward(x)
+    return F.softplus(x)
+
+
+def bnll_backward(x, dy):
+    if x.is_cuda:
+        return _ext().bnll_backward(x, dy)
+    return dy * torch.sigmoid(x)
+
+
+def dropout_forward(x, ratio: float, seed: int, offset: int):
+    """Train-mode dropout. Returns (y, mask) where mask is uint8 keep-mask;
+    y = x * mask * 1/(1-ratio)."""
+    if x.is_cuda:
+        return _ext().dropout_forward(x, ratio, seed, offset)
+    scale = 1.0 / (1.0 - ratio)
+    g = torch.Generator(device="cpu").manual_seed(seed + offset)
+    mask = (torch.rand(x.shape, generator=g) >= ratio)
+    return x * mask.to(x.dtype) * scale, mask
+
+
+def dropout_backward(dy, mask, ratio: float):
+    if dy.is_cuda:
+        return _ext().dropout_backward(dy, mask, ratio)
+    return dy * mask.to(dy.dtype) * (1.0 / (1.0 - ratio))
+
+
+# ---------------------------------------------------------------------------
+# Fused SGD update (solver.cpp:858-892 collapsed to one kernel)
+#   hist = momentum*hist + local_rate*(grad + decay*w)
+#   w   -= hist
+# ---------------------------------------------------------------------------
+
+def sgd_update(w: torch.Tensor, grad: torch.Tensor, hist: torch.Tensor,
+               local_rate: float, momentum: float, decay: float) -> None:
+    if w.is_cuda:
+        _ext().sgd_update(w, grad, hist, local_rate, momentum, decay)
+        return
+    gw = grad if decay == 0.0 else grad + decay * w
+    hist.mul_(momentum).add_(gw, alpha=local_rate)
+    w.sub_(hist)
+
+
+def nesterov_update(w, grad, hist, local_rate: float, momentum: float,
+                    decay: float) -> None:
+    """update = (1+mu)*h_new - mu*h_old (solver.cpp:1013-1120)."""
+    if w.is_cuda:
+        _ext().nesterov_update(w, grad, hist, local_rate, momentum, decay)
+        return
+    gw = grad if decay == 0.0 else grad + decay * w
+    h_old = hist.clone()
+    hist.mul_(momentum).add_(gw, alpha=local_rate)
+    w.sub_((1 + momentum) * hist - momentum * h_old)
+
+
+def adagrad_update(w, grad, hist, local_rate: float, delta: float,
+                   decay: float) -> None:
+    """hist += g^2 ; w -= lr * g / (sqrt(hist)+delta) (solver.cpp:1240-1364)."""
+    if w.is_cuda:
+        _ext().adagrad_update(w, grad, hist, local_rate, delta, decay)
+        return
+    gw = grad if decay == 0.0 else grad + decay * w
+    hist.add_(gw * gw)
+    w.sub_(local_rate * gw / (hist.sqrt() + delta))
+
+
+# ---------------------------------------------------------------------------
+# Metrics
+# ---------------------------------------------------------------------------
+
+def accuracy(pred: torch.Tensor, labels: torch.Tensor, top_k: int = 1) -> torch.Tensor:
+    n = pred.shape[0]
+    flat = pred.view(n, -1)
+    if top_k == 1:
+        correct = (flat.argmax(dim=1) == labels.long().view(-1))
+    else:
+        topk = flat.topk(top_k, dim=1).indices
+        correct = (topk == labels.long().view(-1, 1)).any(dim=1)
+    return correct.to(torch.float32).sum() / n

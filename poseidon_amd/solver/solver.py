@@ -1,0 +1,339 @@
+"""Solvers: SGD / Nesterov / AdaGrad with the distributed training loop.
+
+Parity: /root/reference/src/caffe/solver.cpp --
+  Solve loop (:246-402), lr policies (:767-790), momentum/decay update rules
+  (:815-892 SGD, :1013-1120 Nesterov, :1240-1364 AdaGrad -- fused into one
+  HIP kernel each, ops.functional.sgd_update etc.), Snapshot (:632-667),
+  Restore (:670-696), Test (:552-628), PrintNetOutputs (:699-756).
+
+DWBP (:405-451) is realized as stream/event scheduling: Net.backward fires a
+callback per learnable layer; GradReducer all-reduces that layer's grads on
+a side HIP stream while backprop continues. SFB (:477-531) goes through
+solver/sfb.py. The PS Clock/GlobalBarrier protocol collapses to BSP
+(staleness=0, the reference's own recommended setting, docs/performance.md).
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from ..core.context import ctx
+from ..core.net import Net, TRAIN, TEST
+from ..core.blob import Blob
+from ..ops import functional as ops
+from ..parallel import comm
+from ..proto import (Message, read_proto_text, read_proto_binary,
+                     write_proto_binary)
+from .sfb import SFBReducer, enable_sfb
+
+
+def get_solver(param: Message, **kw) -> "SGDSolver":
+    t = param.enum_name("solver_type")
+    return {"SGD": SGDSolver, "NESTEROV": NesterovSolver,
+            "ADAGRAD": AdaGradSolver}[t](param, **kw)
+
+
+class SGDSolver:
+    def __init__(self, param: Message, use_sfb: Optional[bool] = None,
+                 verbose: Optional[bool] = None):
+        self.param = param
+        c = ctx()
+        self.verbose = c.is_root() if verbose is None else verbose
+        self.iter = 0
+
+        net_param = self._load_net_param(param)
+        train_state = Message("NetState", phase=TRAIN)
+        if param.has("train_state"):
+            train_state.merge_from(param.train_state)
+            train_state.phase = TRAIN
+        self.net = Net(net_param, phase=TRAIN, state=train_state)
+
+        # test nets (solver.cpp InitTestNets, simplified: net + test_state's)
+        self.test_nets: List[Net] = []
+        test_states = list(param.test_state)
+        n_tests = len(list(param.test_iter))
+        for ti in range(n_tests):
+            st = Message("NetState", phase=TEST)
+            if ti < len(test_states):
+                st.merge_from(test_states[ti])
+                st.phase = TEST
+            tn = Net(net_param, phase=TEST, state=st)
+            self._share_params(tn)
+            self.test_nets.append(tn)
+
+        # optimizer state
+        self.history: Dict[int, torch.Tensor] = {}
+        for i, ps in enumerate(self.net.params):
+            if ps.owner == i and ps.lr_mult != 0.0:
+                self.history[i] = torch.zeros_like(ps.blob.diff)
+
+        # distributed
+        self.distributed = comm.init_distributed()
+        if self.distributed:
+            comm.broadcast_params([ps for i, ps in enumerate(self.net.params)
+                                   if ps.owner == i])
+        self.reducer = comm.GradReducer(self.net) if self.distributed else None
+        want_sfb = c.use_sfb if use_sfb is None else use_sfb
+        self.sfb = None
+        if self.distributed and want_sfb:
+            marked = enable_sfb(self.net, c.world_size)
+            if marked:
+                self.sfb = SFBReducer(
+                    marked,
+                    self.reducer.comm_stream if self.reducer.use_stream else None)
+                if self.verbose:
+                    print(f"[poseidon] SFB active on: "
+                          f"{[l.name for l in marked]}", flush=True)
+
+        self._net_outputs_rows: List[List[float]] = []
+        self._t0 = time.time()
+
+    # ------------------------------------------------------------------
+    @staticmethod
+    def _load_net_param(param: Message) -> Message:
+        if param.has("net_param"):
+            return param.net_param
+        if param.has("train_net_param"):
+            return param.train_net_param
+        path = param.net if param.has("net") else param.train_net
+        if not path:
+            raise ValueError("solver specifies no train net")
+        return read_proto_text(path, "NetParameter")
+
+    def _share_params(self, other: Net) -> None:
+        by_name = {l.name: l for l in self.net.layers}
+        for layer in other.layers:
+            src = by_name.get(layer.name)
+            if src is None or not src.blobs:
+                continue
+            for pb, sb in zip(layer.blobs, src.blobs):
+                pb.share_data(sb)
+
+    # -- learning rate policies (solver.cpp:767-790) --------------------
+    def get_learning_rate(self) -> float:
+        p = self.param
+        policy = p.lr_policy or "fixed"
+        base = float(p.base_lr)
+        if policy == "fixed":
+            return base
+        if policy == "step":
+            return base * (p.gamma ** (self.iter // p.stepsize))
+        if policy == "exp":
+            return base * (p.gamma ** self.iter)
+        if policy == "inv":
+            return base * (1.0 + p.gamma * self.iter) ** (-p.power)
+        if policy == "poly":
+            return base * (1.0 - self.iter / float(p.max_iter)) ** p.power
+        raise ValueError(f"unknown lr policy {policy!r}")
+
+    # -- one fused update per param -------------------------------------
+    def _apply_update(self, idx: int, ps, rate: float) -> None:
+        local_rate = rate * ps.lr_mult
+        wd = float(self.param.weight_decay or 0.0) * ps.decay_mult
+        # PS semantics sum per-worker decay terms (each worker adds λW before
+        # pushing) -- fold into the post-reduce update:
+        if self.distributed:
+            wd *= ctx().world_size
+        mom = float(self.param.momentum or 0.0)
+        reg = self.param.regularization_type
+        grad = ps.blob.diff
+        if reg == "L1" and wd != 0.0:
+            grad = grad + wd * torch.sign(ps.blob.data)
+            wd = 0.0
+        ops.sgd_update(ps.blob.data, grad, self.history[idx],
+                       local_rate, mom, wd)
+
+    # ------------------------------------------------------------------
+    def forward_backward(self) -> torch.Tensor:
+        net = self.net
+        net.zero_param_diffs()
+        loss = net.forward_async()
+        if self.reducer:
+            self.reducer.begin_iter()
+
+            def cb(layer_idx, layer):
+                if self.sfb is not None:
+                    self.sfb.on_layer_done(layer)
+                self.reducer.on_layer_done(layer_idx, layer)
+            net.backward(post_layer_cb=cb)
+            self.reducer.flush()
+            self.reducer.wait()
+            if self.sfb is not None:
+                self.sfb.finish()
+        else:
+            net.backward()
+        return loss
+
+    def step(self, iters: int) -> float:
+        last_loss = 0.0
+        p = self.param
+        while iters > 0:
+            if (p.snapshot and self.iter > 0
+                    and self.iter % p.snapshot == 0):
+                self.snapshot()
+            if (p.test_interval and self.iter % p.test_interval == 0
+                    and (self.iter > 0 or bool(p.test_initialization))):
+                self.test_all()
+            loss_t = self.forward_backward()
+            rate = self.get_learning_rate()
+            for i, ps in enumerate(self.net.params):
+                if ps.owner == i and ps.lr_mult != 0.0:
+                    self._apply_update(i, ps, rate)
+            if p.display and self.iter % p.display == 0:
+                last_loss = float(loss_t.item())
+                self._display(last_loss, rate)
+            self.iter += 1
+            iters -= 1
+        return last_loss
+
+    def solve(self, resume_file: Optional[str] = None) -> None:
+        p = self.param
+        if resume_file:
+            self.restore(resume_file)
+        comm.barrier()
+        self.step(int(p.max_iter) - self.iter)
+        if bool(p.snapshot_after_train):
+            self.snapshot()
+        if p.test_interval:
+            self.test_all()
+        comm.barrier()
+
+    # ------------------------------------------------------------------
+    def _display(self, loss: float, rate: float) -> None:
+        c = ctx()
+        vals = torch.tensor([loss], dtype=torch.float64)
+        if self.distributed:
+            comm.allreduce_metrics(vals)
+            vals /= c.world_size
+        row = [float(self.iter), time.time() - self._t0, float(vals[0])]
+        self._net_outputs_rows.append(row)
+        if self.verbose:
+            print(f"[poseidon] iter {self.iter} loss {vals[0]:.6f} "
+                  f"lr {rate:.6g}", flush=True)
+
+    def test_all(self) -> List[Dict[str, float]]:
+        results = []
+        for ti, tn in enumerate(self.test_nets):
+            results.append(self.test(ti))
+        return results
+
+    def test(self, test_net_id: int = 0) -> Dict[str, float]:
+        """Divide test_iter across ranks (solver.cpp:552-628 divides across
+        clients x threads); aggregate outputs with an all-reduce."""
+        tn = self.test_nets[test_net_id]
+        total_iters = int(list(self.param.test_iter)[test_net_id])
+        c = ctx()
+        my_iters = total_iters // c.world_size
+        if c.rank < total_iters % c.world_size:
+            my_iters += 1
+        sums: Dict[str, float] = {}
+        for _ in range(my_iters):
+            tn.forward()
+            for name in tn.output_blob_names:
+                b = tn.blobs[name]
+                sums[name] = sums.get(name, 0.0) + float(b.data.sum().item())
+        names = sorted(sums)
+        vals = torch.tensor([sums[n] for n in names] or [0.0],
+                            dtype=torch.float64)
+        if self.distributed:
+            comm.allreduce_metrics(vals)
+        out: Dict[str, float] = {}
+        for i, n in enumerate(names):
+            avg = float(vals[i]) / total_iters
+            if self.verbose:
+                print(f"[poseidon] test net {test_net_id} {n} = {avg:.5f}",
+                      flush=True)
+            out[n] = avg
+        return out
+
+    # -- checkpoint / resume (solver.cpp:632-696) ------------------------
+    def snapshot(self) -> str:
+        prefix = self.param.snapshot_prefix or "poseidon"
+        model_path = f"{prefix}_iter_{self.iter}.caffemodel"
+        c = ctx()
+        if c.is_root():
+            net_proto = self.net.to_proto()
+            gid = 0
+            for lp in net_proto.layers:
+                for b in lp.blobs:
+                    b.blob_mode = "GLOBAL"
+                    b.global_id = gid
+                    gid += 1
+            write_proto_binary(net_proto, model_path)
+        state = Message("SolverState", iter=self.iter, learned_net=model_path)
+        for i, ps in enumerate(self.net.params):
+            if ps.owner == i and i in self.history:
+                h = self.history[i]
+                hb = Blob(tuple(h.shape), device=h.device)
+                hb.data = h
+                state.history.append(hb.to_proto())
+        state_path = f"{prefix}_iter_{self.iter}.solverstate.{c.rank}.0"
+        write_proto_binary(state, state_path)
+        if self.verbose:
+            print(f"[poseidon] snapshot -> {model_path}", flush=True)
+        return model_path
+
+    def restore(self, state_file: str) -> None:
+        c = ctx()
+        if not os.path.exists(state_file):
+            # fall back to rank 0's file (reference falls back to thread 0)
+            base = state_file.rsplit(".", 2)[0]
+            state_file = f"{base}.0.0"
+        state = read_proto_binary(state_file, "SolverState")
+        self.iter = int(state.iter)
+        hist_protos = list(state.history)
+        own = [i for i, ps in enumerate(self.net.params)
+               if ps.owner == i and i in self.history]
+        for i, hp in zip(own, hist_protos):
+            hb = Blob((), device=self.history[i].device)
+            hb.from_proto(hp)
+            self.history[i] = hb.data.view(self.history[i].shape).to(
+                self.history[i].dtype)
+        if state.has("learned_net") and os.path.exists(state.learned_net):
+            net_proto = read_proto_binary(state.learned_net, "NetParameter")
+            self.net.copy_trained_layers_from(net_proto)
+        if self.distributed:
+            comm.broadcast_params([ps for i, ps in enumerate(self.net.params)
+                                   if ps.owner == i])
+
+    def load_weights(self, weights_file: str) -> None:
+        """Finetuning entry (--weights, caffe_engine.cpp:277-282)."""
+        net_proto = read_proto_binary(weights_file, "NetParameter")
+        self.net.copy_trained_layers_from(net_proto)
+        if self.distributed:
+            comm.broadcast_params([ps for i, ps in enumerate(self.net.params)
+                                   if ps.owner == i])
+
+    def write_net_outputs(self, path: str) -> None:
+        """CSV of display-time metrics (PrintNetOutputs, solver.cpp:699-756)."""
+        if not ctx().is_root():
+            return
+        with open(path + ".netoutputs", "w") as f:
+            f.write("iter,time,loss\n")
+            for row in self._net_outputs_rows:
+                f.write(",".join(f"{v:.6f}" for v in row) + "\n")
+
+
+class NesterovSolver(SGDSolver):
+    def _apply_update(self, idx: int, ps, rate: float) -> None:
+        local_rate = rate * ps.lr_mult
+        wd = float(self.param.weight_decay or 0.0) * ps.decay_mult
+        if self.distributed:
+            wd *= ctx().world_size
+        mom = float(self.param.momentum or 0.0)
+        ops.nesterov_update(ps.blob.data, ps.blob.diff, self.history[idx],
+                            local_rate, mom, wd)
+
+
+class AdaGradSolver(SGDSolver):
+    def _apply_update(self, idx: int, ps, rate: float) -> None:
+        local_rate = rate * ps.lr_mult
+        wd = float(self.param.weight_decay or 0.0) * ps.decay_mult
+        if self.distributed:
+            wd *= ctx().world_size
+        ops.adagrad_update(ps.blob.data, ps.blob.diff, self.history[idx],
+                           local_rate, float(self.param.delta), wd)

@@ -1,0 +1,173 @@
+"""End-to-end training: solver loop, lr policies, snapshot/restore, and a
+convergence smoke test (the reference's acceptance style, SURVEY.md §4:
+example configs must reach expected accuracy; here synthetic separable data
+replaces MNIST since the image has no datasets)."""
+
+import math
+import os
+
+import numpy as np
+import pytest
+import torch
+
+import poseidon_amd as pa
+
+
+def _ip1(net):
+    return next(l for l in net.layers if l.name == "ip1")
+from poseidon_amd.core.net import Net, TRAIN
+from poseidon_amd.proto import Message, parse_text
+from poseidon_amd.solver.solver import SGDSolver, get_solver
+
+
+def _separable_net_param(batch=32, dim=20, classes=5, seed=3):
+    """MemoryData + 2-layer MLP; data = class prototypes + small noise."""
+    return parse_text("NetParameter", f"""
+        name: "toy"
+        layers {{ name: "data" type: MEMORY_DATA top: "data" top: "label"
+                 memory_data_param {{ batch_size: {batch} channels: {dim}
+                                      height: 1 width: 1 }} }}
+        layers {{ name: "ip1" type: INNER_PRODUCT bottom: "data" top: "ip1"
+                 blobs_lr: 1 blobs_lr: 2
+                 inner_product_param {{ num_output: 32
+                     weight_filler {{ type: "xavier" }} }} }}
+        layers {{ name: "relu1" type: RELU bottom: "ip1" top: "ip1" }}
+        layers {{ name: "ip2" type: INNER_PRODUCT bottom: "ip1" top: "ip2"
+                 blobs_lr: 1 blobs_lr: 2
+                 inner_product_param {{ num_output: {classes}
+                     weight_filler {{ type: "xavier" }} }} }}
+        layers {{ name: "loss" type: SOFTMAX_LOSS bottom: "ip2" bottom: "label"
+                 top: "loss" }}
+        layers {{ name: "acc" type: ACCURACY bottom: "ip2" bottom: "label"
+                 top: "acc" }}
+    """)
+
+
+def _toy_dataset(n=256, dim=20, classes=5, seed=3):
+    g = torch.Generator().manual_seed(seed)
+    protos = torch.randn(classes, dim, generator=g) * 2.0
+    labels = torch.randint(0, classes, (n,), generator=g)
+    data = protos[labels] + 0.3 * torch.randn(n, dim, generator=g)
+    return data.view(n, dim, 1, 1), labels.float()
+
+
+def _solver_param(**kw):
+    defaults = dict(base_lr=0.1, lr_policy="fixed", momentum=0.9,
+                    weight_decay=0.0005, max_iter=60, display=0, snapshot=0)
+    defaults.update(kw)
+    sp = Message("SolverParameter")
+    for k, v in defaults.items():
+        setattr(sp, k, v)
+    return sp
+
+
+def test_sgd_converges_on_separable_data():
+    pa.init(device="cpu", seed=7)
+    sp = _solver_param()
+    sp.net_param = _separable_net_param()
+    solver = SGDSolver(sp, verbose=False)
+    data, labels = _toy_dataset()
+    solver.net.layers[0].add_data(data, labels)
+    first_loss = float(solver.net.forward())
+    solver.step(150)
+    solver.net.forward()
+    acc = float(solver.net.blobs["acc"].data)
+    loss = float(solver.net.blobs["loss"].data)
+    assert loss < first_loss * 0.2, (first_loss, loss)
+    assert acc > 0.95, acc
+
+
+@pytest.mark.parametrize("stype", ["NESTEROV", "ADAGRAD"])
+def test_other_solvers_converge(stype):
+    pa.init(device="cpu", seed=7)
+    sp = _solver_param(base_lr=0.5 if stype == "ADAGRAD" else 0.05)
+    sp.solver_type = stype
+    sp.net_param = _separable_net_param()
+    solver = get_solver(sp, verbose=False)
+    data, labels = _toy_dataset()
+    solver.net.layers[0].add_data(data, labels)
+    first_loss = float(solver.net.forward())
+    solver.step(150)
+    solver.net.forward()
+    assert float(solver.net.blobs["loss"].data) < first_loss * 0.5
+
+
+def test_lr_policies():
+    pa.init(device="cpu", seed=1)
+    sp = _solver_param(base_lr=0.1, lr_policy="inv", gamma=0.0001, power=0.75)
+    sp.net_param = _separable_net_param()
+    s = SGDSolver(sp, verbose=False)
+    assert s.get_learning_rate() == pytest.approx(0.1)
+    s.iter = 1000
+    assert s.get_learning_rate() == pytest.approx(0.1 * (1.1 ** -0.75))
+    for policy, extra, iter_, want in [
+        ("fixed", {}, 500, 0.1),
+        ("step", {"gamma": 0.5, "stepsize": 100}, 250, 0.1 * 0.25),
+        ("exp", {"gamma": 0.99}, 10, 0.1 * 0.99 ** 10),
+        ("poly", {"power": 2.0, "max_iter": 1000}, 500, 0.1 * 0.25),
+    ]:
+        sp2 = _solver_param(base_lr=0.1, lr_policy=policy, **extra)
+        sp2.net_param = _separable_net_param()
+        s2 = SGDSolver(sp2, verbose=False)
+        s2.iter = iter_
+        assert s2.get_learning_rate() == pytest.approx(want), policy
+
+
+def test_snapshot_restore(tmp_path):
+    pa.init(device="cpu", seed=11)
+    sp = _solver_param(snapshot_prefix=str(tmp_path / "toy"))
+    sp.net_param = _separable_net_param()
+    solver = SGDSolver(sp, verbose=False)
+    data, labels = _toy_dataset(n=32)  # one batch -> cursor-invariant restore
+    solver.net.layers[0].add_data(data, labels)
+    solver.step(20)
+    path = solver.snapshot()
+    assert os.path.exists(path)
+    w_before = _ip1(solver.net).blobs[0].data.clone()
+    h_before = {i: h.clone() for i, h in solver.history.items()}
+
+    solver2 = SGDSolver(sp, verbose=False)
+    solver2.net.layers[0].add_data(data, labels)
+    solver2.restore(str(tmp_path / "toy") + "_iter_20.solverstate.0.0")
+    assert solver2.iter == 20
+    assert torch.allclose(_ip1(solver2.net).blobs[0].data, w_before)
+    for i, h in solver2.history.items():
+        assert torch.allclose(h, h_before[i], atol=1e-6)
+    # deterministic continuation: same data order -> same next step
+    solver.step(5)
+    solver2.step(5)
+    assert torch.allclose(_ip1(solver.net).blobs[0].data,
+                          _ip1(solver2.net).blobs[0].data, atol=1e-5)
+
+
+def test_finetune_load_weights(tmp_path):
+    pa.init(device="cpu", seed=13)
+    sp = _solver_param(snapshot_prefix=str(tmp_path / "ft"))
+    sp.net_param = _separable_net_param()
+    solver = SGDSolver(sp, verbose=False)
+    data, labels = _toy_dataset()
+    solver.net.layers[0].add_data(data, labels)
+    solver.step(10)
+    model = solver.snapshot()
+
+    solver2 = SGDSolver(sp, verbose=False)
+    solver2.load_weights(model)
+    assert torch.allclose(_ip1(solver2.net).blobs[0].data,
+                          _ip1(solver.net).blobs[0].data)
+
+
+def test_test_net_evaluation():
+    pa.init(device="cpu", seed=17)
+    sp = _solver_param(max_iter=50)
+    sp.test_iter.append(4)
+    sp.test_interval = 1000  # only explicit test() calls
+    sp.test_initialization = False
+    sp.net_param = _separable_net_param()
+    solver = SGDSolver(sp, verbose=False)
+    data, labels = _toy_dataset()
+    solver.net.layers[0].add_data(data, labels)
+    solver.test_nets[0].layers[0].add_data(data, labels)
+    solver.step(150)
+    res = solver.test(0)
+    assert res["acc"] > 0.9, res
+    assert res["loss"] < 0.3, res
