@@ -71,23 +71,26 @@ class ConvolutionLayer(Layer):
     def forward(self, bottom: List[Blob], top: List[Blob]) -> None:
         w = self.blobs[0].data
         b = self.blobs[1].data.view(-1) if self.bias_term else None
+        self._colT = []
         for bo, t in zip(bottom, top):
-            t.data = ops.conv2d_forward(bo.data, w, b, self.stride, self.pad,
-                                        self.group)
+            y, colT = ops.conv2d_forward_ex(bo.data, w, b, self.stride,
+                                            self.pad, self.group)
+            t.data = y
+            self._colT.append(colT)
 
     def backward(self, top: List[Blob], propagate_down: List[bool],
                  bottom: List[Blob]) -> None:
         w = self.blobs[0].data
         for i, (bo, t) in enumerate(zip(bottom, top)):
             dy = t.diff
-            self.blobs[0].diff.add_(ops.conv2d_backward_weight(
-                bo.data, dy, self.blobs[0].shape, self.stride, self.pad,
-                self.group))
-            if self.bias_term:
-                self.blobs[1].diff.view(-1).add_(dy.sum(dim=(0, 2, 3)))
+            db = self.blobs[1].diff.view(-1) if self.bias_term else None
+            ops.conv2d_backward_weight_acc(
+                bo.data, self._colT[i], dy, self.blobs[0].diff, db,
+                self.stride, self.pad, self.group)
             if propagate_down[i]:
                 bo.diff = ops.conv2d_backward_input(
                     w, dy, bo.shape, self.stride, self.pad, self.group)
+        self._colT = []
 
 
 @register_layer("POOLING")
@@ -140,7 +143,9 @@ class PoolingLayer(Layer):
             bottom[0].diff = ops.pool_ave_backward(
                 dy, bottom[0].shape, self.kernel, self.stride, self.pad)
         else:  # MAX and STOCHASTIC both scatter via stored mask
-            bottom[0].diff = ops.pool_max_backward(dy, self._mask, bottom[0].shape)
+            bottom[0].diff = ops.pool_max_backward(
+                dy, self._mask, bottom[0].shape, self.kernel, self.stride,
+                self.pad)
 
 
 @register_layer("LRN")

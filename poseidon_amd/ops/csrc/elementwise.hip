@@ -1,0 +1,203 @@
+// Elementwise / neuron-layer kernels (grid-stride, float4-vectorized where
+// the tail allows). Replaces relu/sigmoid/tanh/bnll/dropout/threshold/power
+// CUDA kernels (reference src/caffe/layers/*_layer.cu) and the
+// math_functions.cu elementwise set.
+
+#include "ps_common.h"
+
+namespace ps {
+
+// ---------------------------------------------------------------------------
+// neuron fwd/bwd
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void relu_fwd_k(const T* x, T* y, int64_t n, float slope) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float v = to_f32(x[i]);
+    from_f32(v > 0.f ? v : v * slope, y[i]);
+  }
+}
+
+template <typename T>
+__global__ void relu_bwd_k(const T* x, const T* dy, T* dx, int64_t n, float slope) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float g = to_f32(dy[i]);
+    from_f32(to_f32(x[i]) > 0.f ? g : g * slope, dx[i]);
+  }
+}
+
+template <typename T>
+__global__ void sigmoid_fwd_k(const T* x, T* y, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    from_f32(1.0f / (1.0f + __expf(-to_f32(x[i]))), y[i]);
+}
+
+template <typename T>
+__global__ void sigmoid_bwd_k(const T* y, const T* dy, T* dx, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float yv = to_f32(y[i]);
+    from_f32(to_f32(dy[i]) * yv * (1.0f - yv), dx[i]);
+  }
+}
+
+template <typename T>
+__global__ void tanh_fwd_k(const T* x, T* y, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    from_f32(tanhf(to_f32(x[i])), y[i]);
+}
+
+template <typename T>
+__global__ void tanh_bwd_k(const T* y, const T* dy, T* dx, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float yv = to_f32(y[i]);
+    from_f32(to_f32(dy[i]) * (1.0f - yv * yv), dx[i]);
+  }
+}
+
+template <typename T>
+__global__ void bnll_fwd_k(const T* x, T* y, int64_t n) {
+  // log(1+e^x), stable (bnll_layer.cu)
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float v = to_f32(x[i]);
+    from_f32(v > 0 ? v + log1pf(__expf(-v)) : log1pf(__expf(v)), y[i]);
+  }
+}
+
+template <typename T>
+__global__ void bnll_bwd_k(const T* x, const T* dy, T* dx, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float v = to_f32(x[i]);
+    from_f32(to_f32(dy[i]) / (1.0f + __expf(-v)), dx[i]);
+  }
+}
+
+template <typename T>
+__global__ void dropout_fwd_k(const T* x, T* y, uint8_t* mask, int64_t n,
+                              float ratio, float scale, uint64_t seed,
+                              uint64_t offset) {
+  for (int64_t i4 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i4 * 4 < n; i4 += (int64_t)gridDim.x * blockDim.x) {
+    uint4 r = philox4(seed, offset, (uint32_t)i4);
+    uint32_t rv[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int64_t i = i4 * 4 + j;
+      if (i < n) {
+        bool keep = u32_to_uniform(rv[j]) >= ratio;
+        mask[i] = keep;
+        from_f32(keep ? to_f32(x[i]) * scale : 0.0f, y[i]);
+      }
+    }
+  }
+}
+
+template <typename T>
+__global__ void dropout_bwd_k(const T* dy, const uint8_t* mask, T* dx,
+                              int64_t n, float scale) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    from_f32(mask[i] ? to_f32(dy[i]) * scale : 0.0f, dx[i]);
+}
+
+// column-sum: out[c] += sum_r in[r][c] for in[R][C] row-major -- the bias
+// gradient for both linear (dy[M][N]) and NHWC conv (dy[(N*OH*OW)][C]).
+template <typename T>
+__global__ void colsum_k(const T* in, float* out, int64_t R, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float acc = 0.f;
+  for (int64_t r = 0; r < R; ++r) acc += to_f32(in[r * C + c]);
+  out[c] += acc;
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+#define PS_EW_LAUNCH(name, ...)                                         \
+  name<<<ew_grid(n), 256, 0, s>>>(__VA_ARGS__)
+
+extern "C" {
+
+#define PS_DEF_UNARY(opname, kern)                                            \
+  void ps_##opname##_f32(const float* a, float* b, int64_t n, hipStream_t s) {\
+    PS_EW_LAUNCH(kern<float>, a, b, n);                                       \
+  }                                                                           \
+  void ps_##opname##_bf16(const void* a, void* b, int64_t n, hipStream_t s) { \
+    PS_EW_LAUNCH(kern<__bf16>, (const __bf16*)a, (__bf16*)b, n);              \
+  }
+
+PS_DEF_UNARY(sigmoid_fwd, sigmoid_fwd_k)
+PS_DEF_UNARY(tanh_fwd, tanh_fwd_k)
+PS_DEF_UNARY(bnll_fwd, bnll_fwd_k)
+
+#define PS_DEF_BINARY(opname, kern)                                           \
+  void ps_##opname##_f32(const float* a, const float* b, float* c, int64_t n, \
+                         hipStream_t s) {                                     \
+    PS_EW_LAUNCH(kern<float>, a, b, c, n);                                    \
+  }                                                                           \
+  void ps_##opname##_bf16(const void* a, const void* b, void* c, int64_t n,   \
+                          hipStream_t s) {                                    \
+    PS_EW_LAUNCH(kern<__bf16>, (const __bf16*)a, (const __bf16*)b, (__bf16*)c, n); \
+  }
+
+PS_DEF_BINARY(sigmoid_bwd, sigmoid_bwd_k)
+PS_DEF_BINARY(tanh_bwd, tanh_bwd_k)
+PS_DEF_BINARY(bnll_bwd, bnll_bwd_k)
+
+void ps_relu_fwd_f32(const float* x, float* y, int64_t n, float slope, hipStream_t s) {
+  PS_EW_LAUNCH(relu_fwd_k<float>, x, y, n, slope);
+}
+void ps_relu_fwd_bf16(const void* x, void* y, int64_t n, float slope, hipStream_t s) {
+  PS_EW_LAUNCH(relu_fwd_k<__bf16>, (const __bf16*)x, (__bf16*)y, n, slope);
+}
+void ps_relu_bwd_f32(const float* x, const float* dy, float* dx, int64_t n,
+                     float slope, hipStream_t s) {
+  PS_EW_LAUNCH(relu_bwd_k<float>, x, dy, dx, n, slope);
+}
+void ps_relu_bwd_bf16(const void* x, const void* dy, void* dx, int64_t n,
+                      float slope, hipStream_t s) {
+  PS_EW_LAUNCH(relu_bwd_k<__bf16>, (const __bf16*)x, (const __bf16*)dy,
+               (__bf16*)dx, n, slope);
+}
+
+void ps_dropout_fwd_f32(const float* x, float* y, uint8_t* mask, int64_t n,
+                        float ratio, uint64_t seed, uint64_t offset, hipStream_t s) {
+  dropout_fwd_k<float><<<ew_grid((n + 3) / 4), 256, 0, s>>>(
+      x, y, mask, n, ratio, 1.0f / (1.0f - ratio), seed, offset);
+}
+void ps_dropout_fwd_bf16(const void* x, void* y, uint8_t* mask, int64_t n,
+                         float ratio, uint64_t seed, uint64_t offset, hipStream_t s) {
+  dropout_fwd_k<__bf16><<<ew_grid((n + 3) / 4), 256, 0, s>>>(
+      (const __bf16*)x, (__bf16*)y, mask, n, ratio, 1.0f / (1.0f - ratio),
+      seed, offset);
+}
+void ps_dropout_bwd_f32(const float* dy, const uint8_t* mask, float* dx,
+                        int64_t n, float ratio, hipStream_t s) {
+  PS_EW_LAUNCH(dropout_bwd_k<float>, dy, mask, dx, n, 1.0f / (1.0f - ratio));
+}
+void ps_dropout_bwd_bf16(const void* dy, const uint8_t* mask, void* dx,
+                         int64_t n, float ratio, hipStream_t s) {
+  PS_EW_LAUNCH(dropout_bwd_k<__bf16>, (const __bf16*)dy, mask, (__bf16*)dx, n,
+               1.0f / (1.0f - ratio));
+}
+
+void ps_colsum_f32(const float* in, float* out, int64_t R, int C, hipStream_t s) {
+  colsum_k<float><<<cdiv(C, 256), 256, 0, s>>>(in, out, R, C);
+}
+void ps_colsum_bf16(const void* in, float* out, int64_t R, int C, hipStream_t s) {
+  colsum_k<__bf16><<<cdiv(C, 256), 256, 0, s>>>((const __bf16*)in, out, R, C);
+}
+
+}  // extern "C"
+
+}  // namespace ps

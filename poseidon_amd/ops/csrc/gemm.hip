@@ -1,0 +1,230 @@
+// MFMA GEMM for gfx950 (CDNA4): the single compute primitive behind
+// InnerProduct, implicit-GEMM convolution (over im2col tiles), and the SFB
+// outer-product reconstruction.
+//
+// C[M,N] = alpha * op(A) @ op(B) + beta * C (+ bias[N])
+//   - op(A) is [M,K]: stored K-last ([M][lda], lda>=K) or K-major
+//     ([K][lda], lda>=M; staged with an on-the-fly transpose into LDS)
+//   - op(B) is [K,N]: stored as [N][ldb] K-last (the natural NT form) or
+//     [K][ldb] K-major (staged transposed)
+//   - batched via grid.z with element strides (stride 0 broadcasts weights)
+//
+// Structure (per /opt/skills/guides/cdna_hip_programming.md §5): 128x128
+// block tile, 256 threads = 4 waves in a 2x2 arrangement, each wave owns a
+// 64x64 sub-tile as 4x4 fragments of v_mfma_f32_16x16x32_bf16 (bf16, K
+// step 32) or v_mfma_f32_16x16x4_f32 (fp32, K step 4, exact f32 at the
+// 155 TF vector-rate ceiling). LDS tiles are K-contiguous with padded rows
+// (+16 B) to keep ds_read_b128 lane groups off a single bank.
+//
+// This is the correctness-first register-staged variant; the glds
+// (global_load_lds) pipelined variant is the planned fast path.
+//
+// Replaces the reference's cuBLAS call sites (math_functions.cu:16-65,
+// conv_layer.cu:25-121, inner_product_layer.cu:18-63).
+
+#include "ps_common.h"
+#include "ps_api.h"
+
+namespace ps {
+
+template <typename T> struct GemmTraits;
+
+template <> struct GemmTraits<float> {
+  static constexpr int BK = 16;      // K elems per LDS tile
+  static constexpr int KSTEP = 4;    // K per MFMA
+  static constexpr int VEC = 4;      // elems per staging vector load
+  static constexpr int RS = BK + 4;  // padded LDS row stride (elems)
+  using vec_t = f32x4;
+  using frag_t = float;  // one A/B element per lane
+  __device__ static inline f32x4 mfma(frag_t a, frag_t b, f32x4 acc) {
+    return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  }
+  __device__ static inline frag_t load_frag(const float* lds_row, int kk, int lane) {
+    return lds_row[kk + (lane >> 4)];
+  }
+};
+
+template <> struct GemmTraits<__bf16> {
+  static constexpr int BK = 64;
+  static constexpr int KSTEP = 32;
+  static constexpr int VEC = 8;
+  static constexpr int RS = BK + 8;  // 144 B rows: 16B-aligned, conflict-spread
+  typedef __attribute__((ext_vector_type(8))) __bf16 vec_t;
+  using frag_t = bf16x8;
+  __device__ static inline f32x4 mfma(frag_t a, frag_t b, f32x4 acc) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  __device__ static inline frag_t load_frag(const __bf16* lds_row, int kk, int lane) {
+    return *reinterpret_cast<const frag_t*>(&lds_row[kk + (lane >> 4) * 8]);
+  }
+};
+
+// Stage a [ROWS x BK] K-contiguous tile into LDS from a K-last source
+// (src[row][k], leading dim lda). Guarded, zero-filled out of range.
+template <typename T, int ROWS>
+__device__ inline void stage_klast(T* lds, const T* src, int64_t lda,
+                                   int row0, int rows_max, int k0, int K,
+                                   int tid) {
+  using TR = GemmTraits<T>;
+  constexpr int CK = TR::BK / TR::VEC;
+  using vec_t = typename TR::vec_t;
+  for (int c = tid; c < ROWS * CK; c += 256) {
+    int r = c / CK, kc = c % CK;
+    int gr = row0 + r, gk = k0 + kc * TR::VEC;
+    vec_t v = {};
+    if (gr < rows_max && gk < K) {
+      if (gk + TR::VEC <= K) {
+        v = *reinterpret_cast<const vec_t*>(&src[(int64_t)gr * lda + gk]);
+      } else {
+        for (int j = 0; j < TR::VEC; ++j)
+          if (gk + j < K) v[j] = src[(int64_t)gr * lda + gk + j];
+      }
+    }
+    *reinterpret_cast<vec_t*>(&lds[r * TR::RS + kc * TR::VEC]) = v;
+  }
+}
+
+// Stage the same tile from a K-major source (src[k][m], leading dim lda):
+// vector loads along m, scalar scatter into the K-contiguous LDS image.
+template <typename T, int ROWS>
+__device__ inline void stage_kmajor(T* lds, const T* src, int64_t lda,
+                                    int row0, int rows_max, int k0, int K,
+                                    int tid) {
+  using TR = GemmTraits<T>;
+  constexpr int CM = ROWS / TR::VEC;
+  using vec_t = typename TR::vec_t;
+  for (int c = tid; c < TR::BK * CM; c += 256) {
+    int kk = c / CM, mc = c % CM;
+    int gk = k0 + kk, gm = row0 + mc * TR::VEC;
+    vec_t v = {};
+    if (gk < K && gm < rows_max) {
+      if (gm + TR::VEC <= rows_max) {
+        v = *reinterpret_cast<const vec_t*>(&src[(int64_t)gk * lda + gm]);
+      } else {
+        for (int j = 0; j < TR::VEC; ++j)
+          if (gm + j < rows_max) v[j] = src[(int64_t)gk * lda + gm + j];
+      }
+    }
+    for (int j = 0; j < TR::VEC; ++j)
+      lds[(mc * TR::VEC + j) * TR::RS + kk] = v[j];
+  }
+}
+
+template <typename T, typename OUT, bool A_KLAST, bool B_KLAST, bool HAS_BIAS>
+__global__ __launch_bounds__(256)
+void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
+                 OUT* __restrict__ Cbase, const float* __restrict__ bias,
+                 int M, int N, int K,
+                 int64_t lda, int64_t ldb, int64_t ldc,
+                 int64_t strideA, int64_t strideB, int64_t strideC,
+                 float alpha, float beta) {
+  using TR = GemmTraits<T>;
+  constexpr int BM = 128, BN = 128, BK = TR::BK, RS = TR::RS;
+
+  const T* A = Abase + (int64_t)blockIdx.z * strideA;
+  const T* B = Bbase + (int64_t)blockIdx.z * strideB;
+  OUT* C = Cbase + (int64_t)blockIdx.z * strideC;
+
+  __shared__ __attribute__((aligned(16))) T a_lds[BM * RS];
+  __shared__ __attribute__((aligned(16))) T b_lds[BN * RS];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;           // 4 waves: 2x2
+  const int wm = (wid >> 1) * 64;     // wave row offset in block tile
+  const int wn = (wid & 1) * 64;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+
+  f32x4 acc[4][4] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    if (A_KLAST)
+      stage_klast<T, BM>(a_lds, A, lda, m0, M, k0, K, tid);
+    else
+      stage_kmajor<T, BM>(a_lds, A, lda, m0, M, k0, K, tid);
+    if (B_KLAST)
+      stage_klast<T, BN>(b_lds, B, ldb, n0, N, k0, K, tid);
+    else
+      stage_kmajor<T, BN>(b_lds, B, ldb, n0, N, k0, K, tid);
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += TR::KSTEP) {
+      typename TR::frag_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        a_frag[f] = TR::load_frag(&a_lds[(wm + f * 16 + (lane & 15)) * RS], kk, lane);
+        b_frag[f] = TR::load_frag(&b_lds[(wn + f * 16 + (lane & 15)) * RS], kk, lane);
+      }
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[fm][fn] = TR::mfma(a_frag[fm], b_frag[fn], acc[fm][fn]);
+    }
+    __syncthreads();
+  }
+
+  // Epilogue: C/D fragment map for 16x16 shapes: col = lane&15,
+  // row = (lane>>4)*4 + r (guide §3; dtype-independent on gfx950).
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col = n0 + wn + fn * 16 + (lane & 15);
+      if (col >= N) continue;
+      float bv = HAS_BIAS ? bias[col] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        int64_t idx = (int64_t)row * ldc + col;
+        float v = alpha * acc[fm][fn][r] + bv;
+        if (beta != 0.0f) v += beta * to_f32(C[idx]);
+        from_f32(v, C[idx]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+template <typename T, typename OUT>
+static void gemm_dispatch(const GemmArgs& g, hipStream_t s) {
+  dim3 grid(cdiv(g.N, 128), cdiv(g.M, 128), g.batch);
+  dim3 block(256);
+#define PS_GEMM_CASE(AK, BK_, HB)                                            \
+  gemm_kernel<T, OUT, AK, BK_, HB><<<grid, block, 0, s>>>(                   \
+      (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,        \
+      g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha, g.beta);
+  const bool hb = g.bias != nullptr;
+  if (g.a_klast && g.b_klast) {
+    if (hb) { PS_GEMM_CASE(true, true, true) } else { PS_GEMM_CASE(true, true, false) }
+  } else if (g.a_klast && !g.b_klast) {
+    if (hb) { PS_GEMM_CASE(true, false, true) } else { PS_GEMM_CASE(true, false, false) }
+  } else if (!g.a_klast && g.b_klast) {
+    if (hb) { PS_GEMM_CASE(false, true, true) } else { PS_GEMM_CASE(false, true, false) }
+  } else {
+    if (hb) { PS_GEMM_CASE(false, false, true) } else { PS_GEMM_CASE(false, false, false) }
+  }
+#undef PS_GEMM_CASE
+}
+
+extern "C" {
+
+void ps_gemm_f32(const GemmArgs* g, hipStream_t s) {
+  gemm_dispatch<float, float>(*g, s);
+}
+void ps_gemm_bf16_f32out(const GemmArgs* g, hipStream_t s) {
+  gemm_dispatch<__bf16, float>(*g, s);
+}
+void ps_gemm_bf16(const GemmArgs* g, hipStream_t s) {
+  gemm_dispatch<__bf16, __bf16>(*g, s);
+}
+
+}  // extern "C"
+
+}  // namespace ps

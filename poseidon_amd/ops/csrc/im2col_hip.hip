@@ -1,0 +1,177 @@
+#include "hip/hip_runtime.h"
+// im2col / col2im for NHWC activations, producing the TRANSPOSED column
+// matrix colT[(n,oh,ow)][k] with k = (g*KH*KW + kh*KW + kw)*Cg + cg --
+// K-contiguous rows that feed the NT MFMA GEMM directly (conv fwd/wgrad),
+// with per-group k-slices contiguous.
+//
+// Replaces reference src/caffe/util/im2col.cu:12-144 (which is NCHW and
+// column-major for cuBLAS). col2im is gather-form (no atomics), like the
+// reference's col2im_gpu_kernel.
+
+#include "ps_common_hip.h"
+#include "ps_api.h"
+
+namespace ps {
+
+// one thread per (np, g, khw); copies Cg contiguous channels
+template <typename T>
+__global__ void im2col_nhwc_k(const T* __restrict__ x, T* __restrict__ colT,
+                              ConvGeom g) {
+  const int Cg = g.C / g.G;
+  const int KHW = g.kh * g.kw;
+  const int Kcol = g.G * KHW * Cg;
+  const int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
+  int64_t total = NP * g.G * KHW;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int khw = i % KHW;
+    int64_t t = i / KHW;
+    int grp = t % g.G;
+    int64_t np = t / g.G;
+    int ow = np % g.Wo;
+    int64_t t2 = np / g.Wo;
+    int oh = t2 % g.Ho;
+    int n = t2 / g.Ho;
+    int kkh = khw / g.kw, kkw = khw % g.kw;
+    int ih = oh * g.sh - g.ph + kkh;
+    int iw = ow * g.sw - g.pw + kkw;
+    T* dst = colT + np * Kcol + ((int64_t)grp * KHW + khw) * Cg;
+    if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
+      const T* src = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C + grp * Cg;
+      for (int c = 0; c < Cg; ++c) dst[c] = src[c];
+    } else {
+      for (int c = 0; c < Cg; ++c) dst[c] = (T)0.0f;
+    }
+  }
+}
+
+// gather: one thread per (n, ih, iw, ci)
+template <typename T>
+__global__ void col2im_nhwc_k(const T* __restrict__ colT, T* __restrict__ dx,
+                              ConvGeom g) {
+  const int Cg = g.C / g.G;
+  const int KHW = g.kh * g.kw;
+  const int Kcol = g.G * KHW * Cg;
+  int64_t total = (int64_t)g.N * g.H * g.W * g.C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int ci = i % g.C;
+    int64_t t = i / g.C;
+    int iw = t % g.W; t /= g.W;
+    int ih = t % g.H;
+    int n = t / g.H;
+    int grp = ci / Cg, cg = ci % Cg;
+    float acc = 0.f;
+    for (int kkh = 0; kkh < g.kh; ++kkh) {
+      int oh_num = ih + g.ph - kkh;
+      if (oh_num < 0 || oh_num % g.sh) continue;
+      int oh = oh_num / g.sh;
+      if (oh >= g.Ho) continue;
+      for (int kkw = 0; kkw < g.kw; ++kkw) {
+        int ow_num = iw + g.pw - kkw;
+        if (ow_num < 0 || ow_num % g.sw) continue;
+        int ow = ow_num / g.sw;
+        if (ow >= g.Wo) continue;
+        int64_t np = ((int64_t)n * g.Ho + oh) * g.Wo + ow;
+        int k = (grp * KHW + kkh * g.kw + kkw) * Cg + cg;
+        acc += to_f32(colT[np * Kcol + k]);
+      }
+    }
+    from_f32(acc, dx[i]);
+  }
+}
+
+// NCHW [Co][Ci][kh][kw] -> khwc-per-group [Co][kh][kw][Cg] weight repack
+// (and its inverse for the weight gradient). Co already encodes the group.
+template <typename TI, typename TO>
+__global__ void weight_to_khwc_k(const TI* src, TO* dst, int Co, int Cig,
+                                 int KH, int KW) {
+  int64_t total = (int64_t)Co * Cig * KH * KW;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    // src index decomposition: [co][ci][kh][kw]
+    int kkw = i % KW;
+    int64_t t = i / KW;
+    int kkh = t % KH; t /= KH;
+    int ci = t % Cig;
+    int co = t / Cig;
+    int64_t dst_i = (((int64_t)co * KH + kkh) * KW + kkw) * Cig + ci;
+    from_f32(to_f32(src[i]), dst[dst_i]);
+  }
+}
+
+template <typename TI, typename TO>
+__global__ void weight_from_khwc_k(const TI* src, TO* dst, int Co, int Cig,
+                                   int KH, int KW, float beta) {
+  int64_t total = (int64_t)Co * Cig * KH * KW;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int kkw = i % KW;
+    int64_t t = i / KW;
+    int kkh = t % KH; t /= KH;
+    int ci = t % Cig;
+    int co = t / Cig;
+    int64_t src_i = (((int64_t)co * KH + kkh) * KW + kkw) * Cig + ci;
+    float v = to_f32(src[src_i]);
+    if (beta != 0.f) v += beta * to_f32(dst[i]);
+    from_f32(v, dst[i]);
+  }
+}
+
+// NCHW <-> NHWC activation converters (layout boundary with CPU/proto side)
+template <typename TI, typename TO>
+__global__ void nchw_to_nhwc_k(const TI* src, TO* dst, int N, int C, int H, int W) {
+  int64_t total = (int64_t)N * C * H * W;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    // i indexes NHWC dst
+    int c = i % C;
+    int64_t t = i / C;
+    int w = t % W; t /= W;
+    int h = t % H;
+    int n = t / H;
+    from_f32(to_f32(src[(((int64_t)n * C + c) * H + h) * W + w]), dst[i]);
+  }
+}
+
+extern "C" {
+
+void ps_im2col_nhwc_f32(const float* x, float* colT, const ConvGeom* g, hipStream_t s) {
+  int64_t total = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
+ hipLaunchKernelGGL(( im2col_nhwc_k<float>), dim3(ew_grid(total)), dim3(256), 0, s, x, colT, *g);
+}
+void ps_im2col_nhwc_bf16(const void* x, void* colT, const ConvGeom* g, hipStream_t s) {
+  int64_t total = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
+ hipLaunchKernelGGL(( im2col_nhwc_k<__bf16>), dim3(ew_grid(total)), dim3(256), 0, s, (const __bf16*)x,
+                                                       (__bf16*)colT, *g);
+}
+void ps_col2im_nhwc_f32(const float* colT, float* dx, const ConvGeom* g, hipStream_t s) {
+  int64_t total = (int64_t)g->N * g->H * g->W * g->C;
+ hipLaunchKernelGGL(( col2im_nhwc_k<float>), dim3(ew_grid(total)), dim3(256), 0, s, colT, dx, *g);
+}
+void ps_col2im_nhwc_bf16(const void* colT, void* dx, const ConvGeom* g, hipStream_t s) {
+  int64_t total = (int64_t)g->N * g->H * g->W * g->C;
+ hipLaunchKernelGGL(( col2im_nhwc_k<__bf16>), dim3(ew_grid(total)), dim3(256), 0, s, (const __bf16*)colT,
+                                                       (__bf16*)dx, *g);
+}
+void ps_weight_to_khwc_f32(const float* src, float* dst, int Co, int Cig,
+                           int KH, int KW, hipStream_t s) {
+ hipLaunchKernelGGL(( weight_to_khwc_k<float, float>)
+      , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, src, dst, Co, Cig, KH, KW);
+}
+void ps_weight_to_khwc_f32_bf16(const float* src, void* dst, int Co, int Cig,
+                                int KH, int KW, hipStream_t s) {
+ hipLaunchKernelGGL(( weight_to_khwc_k<float, __bf16>)
+      , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, src, (__bf16*)dst,
+                                                            Co, Cig, KH, KW);
+}
+void ps_weight_from_khwc_f32(const float* src, float* dst, int Co, int Cig,
+                             int KH, int KW, float beta, hipStream_t s) {
+ hipLaunchKernelGGL(( weight_from_khwc_k<float, float>)
+      , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, src, dst, Co, Cig,
+                                                            KH, KW, beta);
+}
+
+}  // extern "C"
+
+}  // namespace ps

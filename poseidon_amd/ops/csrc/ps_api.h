@@ -1,0 +1,122 @@
+// C API between the HIP kernel translation units and the torch bindings.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+struct GemmArgs {
+  const void* A;
+  const void* B;
+  void* C;
+  const float* bias;  // may be null
+  int M, N, K;
+  int64_t lda, ldb, ldc;
+  int64_t strideA, strideB, strideC;
+  int batch;
+  float alpha, beta;
+  bool a_klast, b_klast;
+};
+
+struct PoolGeom {
+  int N, C, H, W, Ho, Wo;
+  int kh, kw, sh, sw, ph, pw;
+};
+
+struct ConvGeom {
+  int N, C, H, W;
+  int Ho, Wo;
+  int kh, kw, sh, sw, ph, pw;
+  int G;
+};
+
+extern "C" {
+// gemm.hip
+void ps_gemm_f32(const GemmArgs* g, hipStream_t s);
+void ps_gemm_bf16_f32out(const GemmArgs* g, hipStream_t s);
+void ps_gemm_bf16(const GemmArgs* g, hipStream_t s);
+
+// elementwise.hip
+void ps_relu_fwd_f32(const float*, float*, int64_t, float, hipStream_t);
+void ps_relu_bwd_f32(const float*, const float*, float*, int64_t, float, hipStream_t);
+void ps_sigmoid_fwd_f32(const float*, float*, int64_t, hipStream_t);
+void ps_sigmoid_bwd_f32(const float*, const float*, float*, int64_t, hipStream_t);
+void ps_tanh_fwd_f32(const float*, float*, int64_t, hipStream_t);
+void ps_tanh_bwd_f32(const float*, const float*, float*, int64_t, hipStream_t);
+void ps_bnll_fwd_f32(const float*, float*, int64_t, hipStream_t);
+void ps_bnll_bwd_f32(const float*, const float*, float*, int64_t, hipStream_t);
+void ps_dropout_fwd_f32(const float*, float*, uint8_t*, int64_t, float,
+                        uint64_t, uint64_t, hipStream_t);
+void ps_dropout_bwd_f32(const float*, const uint8_t*, float*, int64_t, float,
+                        hipStream_t);
+void ps_colsum_f32(const float*, float*, int64_t, int, hipStream_t);
+void ps_relu_fwd_bf16(const void*, void*, int64_t, float, hipStream_t);
+void ps_relu_bwd_bf16(const void*, const void*, void*, int64_t, float, hipStream_t);
+void ps_sigmoid_fwd_bf16(const void*, void*, int64_t, hipStream_t);
+void ps_sigmoid_bwd_bf16(const void*, const void*, void*, int64_t, hipStream_t);
+void ps_tanh_fwd_bf16(const void*, void*, int64_t, hipStream_t);
+void ps_tanh_bwd_bf16(const void*, const void*, void*, int64_t, hipStream_t);
+void ps_bnll_fwd_bf16(const void*, void*, int64_t, hipStream_t);
+void ps_bnll_bwd_bf16(const void*, const void*, void*, int64_t, hipStream_t);
+void ps_dropout_fwd_bf16(const void*, void*, uint8_t*, int64_t, float,
+                         uint64_t, uint64_t, hipStream_t);
+void ps_dropout_bwd_bf16(const void*, const uint8_t*, void*, int64_t, float,
+                         hipStream_t);
+void ps_colsum_bf16(const void*, float*, int64_t, int, hipStream_t);
+
+// pool.hip
+void ps_maxpool_fwd_f32(const float*, float*, int*, const PoolGeom*, hipStream_t);
+void ps_maxpool_bwd_f32(const float*, const int*, float*, const PoolGeom*, hipStream_t);
+void ps_avepool_fwd_f32(const float*, float*, const PoolGeom*, hipStream_t);
+void ps_avepool_bwd_f32(const float*, float*, const PoolGeom*, hipStream_t);
+void ps_stochpool_fwd_train_f32(const float*, float*, int*, const PoolGeom*,
+                                uint64_t, hipStream_t);
+void ps_stochpool_fwd_test_f32(const float*, float*, const PoolGeom*, hipStream_t);
+void ps_maxpool_fwd_bf16(const void*, void*, int*, const PoolGeom*, hipStream_t);
+void ps_maxpool_bwd_bf16(const void*, const int*, void*, const PoolGeom*, hipStream_t);
+void ps_avepool_fwd_bf16(const void*, void*, const PoolGeom*, hipStream_t);
+void ps_avepool_bwd_bf16(const void*, void*, const PoolGeom*, hipStream_t);
+
+// lrn.hip
+void ps_lrn_fwd_f32(const float*, float*, float*, int64_t, int, int, float,
+                    float, hipStream_t);
+void ps_lrn_bwd_f32(const float*, const float*, const float*, const float*,
+                    float*, int64_t, int, int, float, float, hipStream_t);
+void ps_lrn_fwd_bf16(const void*, void*, float*, int64_t, int, int, float,
+                     float, hipStream_t);
+void ps_lrn_bwd_bf16(const void*, const void*, const float*, const void*,
+                     void*, int64_t, int, int, float, float, hipStream_t);
+
+// softmax.hip
+void ps_softmax_rows_f32(const float*, float*, int64_t, int, hipStream_t);
+void ps_softmax_bwd_rows_f32(const float*, const float*, float*, int64_t, int,
+                             hipStream_t);
+void ps_softmax_loss_fwd_f32(const float*, const float*, float*, float*,
+                             int64_t, int, hipStream_t);
+void ps_softmax_loss_bwd_f32(const float*, const float*, float*, int64_t, int,
+                             float, hipStream_t);
+void ps_softmax_rows_bf16(const void*, void*, int64_t, int, hipStream_t);
+void ps_softmax_bwd_rows_bf16(const void*, const void*, void*, int64_t, int,
+                              hipStream_t);
+void ps_softmax_loss_fwd_bf16(const void*, const void*, void*, float*,
+                              int64_t, int, hipStream_t);
+void ps_softmax_loss_bwd_bf16(const void*, const void*, void*, int64_t, int,
+                              float, hipStream_t);
+
+// sgd.hip
+void ps_sgd_update(float*, const float*, float*, int64_t, float, float, float,
+                   hipStream_t);
+void ps_nesterov_update(float*, const float*, float*, int64_t, float, float,
+                        float, hipStream_t);
+void ps_adagrad_update(float*, const float*, float*, int64_t, float, float,
+                       float, hipStream_t);
+void ps_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
+
+// im2col.hip
+void ps_im2col_nhwc_f32(const float*, float*, const ConvGeom*, hipStream_t);
+void ps_col2im_nhwc_f32(const float*, float*, const ConvGeom*, hipStream_t);
+void ps_im2col_nhwc_bf16(const void*, void*, const ConvGeom*, hipStream_t);
+void ps_col2im_nhwc_bf16(const void*, void*, const ConvGeom*, hipStream_t);
+void ps_weight_to_khwc_f32(const float*, float*, int, int, int, int, hipStream_t);
+void ps_weight_to_khwc_f32_bf16(const float*, void*, int, int, int, int, hipStream_t);
+void ps_weight_from_khwc_f32(const float*, float*, int, int, int, int, float,
+                             hipStream_t);
+}

@@ -71,13 +71,16 @@ def pool_out_size(h: int, k: int, p: int, s: int) -> Tuple[int, bool]:
 # Convolution
 # ---------------------------------------------------------------------------
 
-def conv2d_forward(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor],
-                   stride: Tuple[int, int], pad: Tuple[int, int],
-                   groups: int) -> torch.Tensor:
+def conv2d_forward_ex(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor],
+                      stride: Tuple[int, int], pad: Tuple[int, int],
+                      groups: int):
+    """Returns (y, colT_cache). colT is the im2col matrix on GPU (reused by
+    the backward GEMMs); None on CPU."""
     if x.is_cuda:
-        return _ext().conv2d_forward(x, w, b, stride[0], stride[1],
-                                     pad[0], pad[1], groups)
-    return F.conv2d(x, w, b, stride=stride, padding=pad, groups=groups)
+        y, colT = _ext().conv2d_forward_ex(x, w, b, stride[0], stride[1],
+                                           pad[0], pad[1], groups)
+        return y, colT
+    return F.conv2d(x, w, b, stride=stride, padding=pad, groups=groups), None
 
 
 def conv2d_backward_input(w: torch.Tensor, dy: torch.Tensor,
@@ -89,13 +92,19 @@ def conv2d_backward_input(w: torch.Tensor, dy: torch.Tensor,
                                       padding=pad, groups=groups)
 
 
-def conv2d_backward_weight(x: torch.Tensor, dy: torch.Tensor,
-                           w_shape, stride, pad, groups: int) -> torch.Tensor:
+def conv2d_backward_weight_acc(x: torch.Tensor, colT, dy: torch.Tensor,
+                               dw: torch.Tensor, db: Optional[torch.Tensor],
+                               stride, pad, groups: int) -> None:
+    """Accumulates into dw (NCHW) and db."""
     if dy.is_cuda:
-        return _ext().conv2d_backward_weight(x, dy, list(w_shape), stride[0],
-                                             stride[1], pad[0], pad[1], groups)
-    return torch.nn.grad.conv2d_weight(x, list(w_shape), dy, stride=stride,
-                                       padding=pad, groups=groups)
+        _ext().conv2d_backward_weight_acc(x, colT, dy, dw, db,
+                                          stride[0], stride[1], pad[0], pad[1],
+                                          groups)
+        return
+    dw.add_(torch.nn.grad.conv2d_weight(x, list(dw.shape), dy, stride=stride,
+                                        padding=pad, groups=groups))
+    if db is not None:
+        db.add_(dy.sum(dim=(0, 2, 3)))
 
 
 # ---------------------------------------------------------------------------
@@ -164,9 +173,10 @@ def pool_max_forward(x, k, s, p):
     return y, mask
 
 
-def pool_max_backward(dy, mask, x_shape):
+def pool_max_backward(dy, mask, x_shape, k=None, s_=None, p=None):
     if dy.is_cuda:
-        return _ext().pool_max_backward(dy, mask, list(x_shape))
+        return _ext().pool_max_backward(dy, mask, list(x_shape), k[0], k[1],
+                                        s_[0], s_[1], p[0], p[1])
     N, C, H, W = x_shape
     dx = torch.zeros(N, C, H * W, dtype=dy.dtype, device=dy.device)
     dx.scatter_add_(2, mask.view(N, C, -1).long(), dy.view(N, C, -1))
