@@ -53,13 +53,13 @@ class InnerProductLayer(Layer):
         top[0].reshape(self.M, self.N)
 
     def forward(self, bottom, top) -> None:
-        x = bottom[0].data.view(self.M, self.K)
+        x = bottom[0].data.reshape(self.M, self.K)
         b = self.blobs[1].data.view(-1) if self.bias_term else None
         top[0].data = ops.linear_forward(x, self.blobs[0].data.view(self.N, self.K), b)
 
     def backward(self, top, propagate_down, bottom) -> None:
-        x = bottom[0].data.view(self.M, self.K)
-        dy = top[0].diff.view(self.M, self.N)
+        x = bottom[0].data.reshape(self.M, self.K)
+        dy = top[0].diff.reshape(self.M, self.N)
         need_dw = not self.sfb_active
         dx, dw, db = ops.linear_backward(
             x, self.blobs[0].data.view(self.N, self.K), dy,
@@ -74,7 +74,7 @@ class InnerProductLayer(Layer):
         if self.bias_term:
             self.blobs[1].diff.view(-1).add_(db)
         if propagate_down[0]:
-            bottom[0].diff = dx.view(bottom[0].shape)
+            bottom[0].diff = dx.reshape(bottom[0].shape)
 
 
 @register_layer("CONCAT")
@@ -169,11 +169,11 @@ class FlattenLayer(Layer):
         top[0].reshape(bottom[0].num, bottom[0].count // bottom[0].num)
 
     def forward(self, bottom, top) -> None:
-        top[0].data = bottom[0].data.view(bottom[0].num, -1)
+        top[0].data = bottom[0].data.reshape(bottom[0].num, -1)
 
     def backward(self, top, propagate_down, bottom) -> None:
         if propagate_down[0]:
-            bottom[0].diff = top[0].diff.view(bottom[0].shape)
+            bottom[0].diff = top[0].diff.reshape(bottom[0].shape)
 
 
 @register_layer("ELTWISE")
@@ -305,7 +305,7 @@ class ArgMaxLayer(Layer):
         top[0].reshape(n, 2 if self.out_max_val else 1, self.top_k, 1)
 
     def forward(self, bottom, top) -> None:
-        x = bottom[0].data.view(bottom[0].num, -1)
+        x = bottom[0].data.reshape(bottom[0].num, -1)
         vals, idx = x.topk(self.top_k, dim=1)
         if self.out_max_val:
             top[0].data = torch.stack(

@@ -50,8 +50,8 @@ class SoftmaxLayer(Layer):
 @register_layer("SOFTMAX_LOSS")
 class SoftmaxWithLossLayer(LossLayer):
     def forward(self, bottom, top) -> None:
-        logits = bottom[0].data.view(bottom[0].num, -1)
-        labels = bottom[1].data.view(-1)
+        logits = bottom[0].data.reshape(bottom[0].num, -1)
+        labels = bottom[1].data.reshape(-1)
         loss, prob = ops.softmax_loss_forward(logits, labels)
         self._prob = prob
         self._labels = labels
@@ -64,14 +64,14 @@ class SoftmaxWithLossLayer(LossLayer):
         if propagate_down[0]:
             w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
             dx = ops.softmax_loss_backward(self._prob, self._labels, w)
-            bottom[0].diff = dx.view(bottom[0].shape)
+            bottom[0].diff = dx.reshape(bottom[0].shape)
 
 
 @register_layer("MULTINOMIAL_LOGISTIC_LOSS")
 class MultinomialLogisticLossLayer(LossLayer):
     def forward(self, bottom, top) -> None:
-        prob = bottom[0].data.view(bottom[0].num, -1)
-        labels = bottom[1].data.view(-1).long()
+        prob = bottom[0].data.reshape(bottom[0].num, -1)
+        labels = bottom[1].data.reshape(-1).long()
         n = prob.shape[0]
         picked = prob[torch.arange(n), labels].clamp(min=1e-20)
         self._cache = (prob, labels, picked)
@@ -85,7 +85,7 @@ class MultinomialLogisticLossLayer(LossLayer):
             w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
             dx = torch.zeros_like(prob)
             dx[torch.arange(n), labels] = -w / (picked * n)
-            bottom[0].diff = dx.view(bottom[0].shape)
+            bottom[0].diff = dx.reshape(bottom[0].shape)
 
 
 @register_layer("EUCLIDEAN_LOSS")
@@ -116,8 +116,8 @@ class HingeLossLayer(LossLayer):
         self.norm = hp.enum_name("norm") if hp is not None else "L1"
 
     def forward(self, bottom, top) -> None:
-        x = bottom[0].data.view(bottom[0].num, -1)
-        labels = bottom[1].data.view(-1).long()
+        x = bottom[0].data.reshape(bottom[0].num, -1)
+        labels = bottom[1].data.reshape(-1).long()
         n, k = x.shape
         margin = x.clone()
         rows = torch.arange(n)
@@ -143,7 +143,7 @@ class HingeLossLayer(LossLayer):
         else:
             g = 2.0 * margin
         g[rows, labels] *= -1
-        bottom[0].diff = (g * (w / n)).view(bottom[0].shape)
+        bottom[0].diff = (g * (w / n)).reshape(bottom[0].shape)
 
 
 @register_layer("SIGMOID_CROSS_ENTROPY_LOSS")
@@ -186,8 +186,8 @@ class InfogainLossLayer(LossLayer):
             self.H = torch.from_numpy(arr.reshape(dim, dim).copy())
 
     def forward(self, bottom, top) -> None:
-        prob = bottom[0].data.view(bottom[0].num, -1)
-        labels = bottom[1].data.view(-1).long()
+        prob = bottom[0].data.reshape(bottom[0].num, -1)
+        labels = bottom[1].data.reshape(-1).long()
         H = self.H.to(prob.device, prob.dtype) if self.H is not None \
             else bottom[2].data.view(prob.shape[1], prob.shape[1])
         n = prob.shape[0]
@@ -214,7 +214,7 @@ class ContrastiveLossLayer(LossLayer):
         self.margin = float(cp.margin) if cp is not None else 1.0
 
     def forward(self, bottom, top) -> None:
-        a = bottom[0].data.view(bottom[0].num, -1)
+        a = bottom[0].data.reshape(bottom[0].num, -1)
         b = bottom[1].data.view(bottom[1].num, -1)
         y = bottom[2].data.view(-1)
         diff = a - b

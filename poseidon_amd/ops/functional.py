@@ -448,7 +448,7 @@ def adagrad_update(w, grad, hist, local_rate: float, delta: float,
 
 def accuracy(pred: torch.Tensor, labels: torch.Tensor, top_k: int = 1) -> torch.Tensor:
     n = pred.shape[0]
-    flat = pred.view(n, -1)
+    flat = pred.reshape(n, -1)
     if top_k == 1:
         correct = (flat.argmax(dim=1) == labels.long().view(-1))
     else:
