@@ -1,0 +1,109 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: whole-node images/sec for AlexNet (default) /
+GoogLeNet / VGG-16 / CIFAR-quick on synthetic data with random-init weights
+(BASELINE.json metric). One rank per GPU over RCCL; launch N>1 via
+torch.distributed.run with --master-addr 127.0.0.1.
+
+    python bench.py --gpus 1 --steps 20 --warmup 5
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+        --master-addr 127.0.0.1 bench.py --gpus 8 --steps 20 --warmup 5
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+DEFAULT_BATCH = {  # per-GPU batch, matching the reference's training configs
+    "alexnet": 256,   # models/bvlc_alexnet/train_val.prototxt:10
+    "googlenet": 32,  # models/bvlc_googlenet/train_test.prototxt:9
+    "vgg16": 32,
+    "cifar10_quick": 100,
+    "lenet": 64,
+}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--model", default="alexnet",
+                    choices=list(DEFAULT_BATCH))
+    ap.add_argument("--batch", type=int, default=0, help="per-GPU batch")
+    ap.add_argument("--no-sfb", action="store_true")
+    args = ap.parse_args()
+
+    import poseidon_amd as pa
+    from poseidon_amd.models import zoo
+    from poseidon_amd.proto import Message
+    from poseidon_amd.solver.solver import SGDSolver
+    from poseidon_amd.parallel import comm
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    assert world == args.gpus or args.gpus == 1 or world == 1, \
+        f"WORLD_SIZE={world} vs --gpus={args.gpus}"
+    n_gpus = max(world, 1)
+
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    pa.init(device=device, seed=1234)
+    batch = args.batch or DEFAULT_BATCH[args.model]
+
+    sp = Message("SolverParameter", base_lr=0.01, lr_policy="fixed",
+                 momentum=0.9, weight_decay=0.0005, max_iter=1 << 30,
+                 display=0, snapshot=0)
+    sp.net_param = zoo.build_net(args.model, batch=batch)
+    solver = SGDSolver(sp, use_sfb=not args.no_sfb, verbose=False)
+
+    def sync():
+        if device == "cuda":
+            torch.cuda.synchronize()
+        comm.barrier()
+
+    solver.step(args.warmup)
+    sync()
+    t0 = time.perf_counter()
+    solver.step(args.steps)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if solver.distributed:
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t[0])
+
+    if rank == 0:
+        images = batch * n_gpus * args.steps
+        value = images / elapsed
+        out = {
+            "metric": "images/sec (whole node)",
+            "value": round(value, 2),
+            "unit": "images/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": batch * n_gpus,
+                "input": "3x227x227" if args.model == "alexnet" else "3x224x224",
+                "parallelism": f"dp{n_gpus}",
+                "sfb": not args.no_sfb and n_gpus > 1,
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()
