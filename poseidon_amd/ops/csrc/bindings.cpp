@@ -56,6 +56,23 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   g.batch = 1;
   g.alpha = alpha; g.beta = beta;
   g.a_klast = a_klast; g.b_klast = b_klast;
+  g.ws = nullptr;
+  g.splitk = 1;
+  // Split-K when the output tile grid cannot fill 256 CUs but K is deep
+  // (conv wgrad: M=Cout<=384, N=Kcol, K=N*OH*OW up to ~800k): target ~512
+  // workgroups, cap the f32 workspace at 256 MB.
+  int64_t tiles = (int64_t)((M + 127) / 128) * ((N + 127) / 128);
+  if (tiles < 96 && K >= 4096) {
+    int sk = (int)std::min<int64_t>(512 / tiles, (K + 2047) / 2048);
+    int64_t ws_elems = (int64_t)sk * M * N;
+    if (sk > 1 && ws_elems * 4 <= (256LL << 20)) {
+      g.splitk = sk;
+      Tensor ws = at::empty({ws_elems}, C.options().dtype(at::kFloat));
+      g.ws = ws.data_ptr<float>();
+      ps_gemm_f32(&g, stream());
+      return;
+    }
+  }
   ps_gemm_f32(&g, stream());
 }
 

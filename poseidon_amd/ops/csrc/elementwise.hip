@@ -110,13 +110,22 @@ __global__ void dropout_bwd_k(const T* dy, const uint8_t* mask, T* dx,
 
 // column-sum: out[c] += sum_r in[r][c] for in[R][C] row-major -- the bias
 // gradient for both linear (dy[M][N]) and NHWC conv (dy[(N*OH*OW)][C]).
+// Two-level: each block owns 256 columns x a slab of rows, accumulates in
+// registers, then one atomicAdd per column (device-scope atomics are cheap
+// and contention is gridDim.y-way -- guide §6 Guideline 12).
 template <typename T>
 __global__ void colsum_k(const T* in, float* out, int64_t R, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  int64_t rows_per = (R + gridDim.y - 1) / gridDim.y;
+  int64_t r0 = blockIdx.y * rows_per;
+  int64_t r1 = min(R, r0 + rows_per);
   float acc = 0.f;
-  for (int64_t r = 0; r < R; ++r) acc += to_f32(in[r * C + c]);
-  out[c] += acc;
+  for (int64_t r = r0; r < r1; ++r) acc += to_f32(in[r * C + c]);
+  if (gridDim.y == 1)
+    out[c] += acc;
+  else
+    atomicAdd(&out[c], acc);
 }
 
 // ---------------------------------------------------------------------------
@@ -191,11 +200,18 @@ void ps_dropout_bwd_bf16(const void* dy, const uint8_t* mask, void* dx,
                1.0f / (1.0f - ratio));
 }
 
+static inline dim3 colsum_grid(int64_t R, int C) {
+  int xb = cdiv(C, 256);
+  // enough row slabs to fill the chip (256 CUs want >> 256 workgroups)
+  int yb = 1;
+  while ((int64_t)xb * yb < 1024 && (R / yb) > 1024) yb *= 2;
+  return dim3(xb, yb);
+}
 void ps_colsum_f32(const float* in, float* out, int64_t R, int C, hipStream_t s) {
-  colsum_k<float><<<cdiv(C, 256), 256, 0, s>>>(in, out, R, C);
+  colsum_k<float><<<colsum_grid(R, C), 256, 0, s>>>(in, out, R, C);
 }
 void ps_colsum_bf16(const void* in, float* out, int64_t R, int C, hipStream_t s) {
-  colsum_k<__bf16><<<cdiv(C, 256), 256, 0, s>>>((const __bf16*)in, out, R, C);
+  colsum_k<__bf16><<<colsum_grid(R, C), 256, 0, s>>>((const __bf16*)in, out, R, C);
 }
 
 }  // extern "C"

@@ -110,20 +110,28 @@ __device__ inline void stage_kmajor(T* lds, const T* src, int64_t lda,
   }
 }
 
-template <typename T, typename OUT, bool A_KLAST, bool B_KLAST, bool HAS_BIAS>
+template <typename T, typename OUT, bool A_KLAST, bool B_KLAST, bool HAS_BIAS,
+          bool SPLITK>
 __global__ __launch_bounds__(256)
 void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
                  OUT* __restrict__ Cbase, const float* __restrict__ bias,
                  int M, int N, int K,
                  int64_t lda, int64_t ldb, int64_t ldc,
                  int64_t strideA, int64_t strideB, int64_t strideC,
-                 float alpha, float beta) {
+                 float alpha, float beta,
+                 float* __restrict__ ws, int kchunk) {
   using TR = GemmTraits<T>;
   constexpr int BM = 128, BN = 128, BK = TR::BK, RS = TR::RS;
 
-  const T* A = Abase + (int64_t)blockIdx.z * strideA;
-  const T* B = Bbase + (int64_t)blockIdx.z * strideB;
-  OUT* C = Cbase + (int64_t)blockIdx.z * strideC;
+  // split-K: grid.z indexes the K-slice (batch must be 1); otherwise batch
+  const T* A = Abase + (SPLITK ? 0 : (int64_t)blockIdx.z * strideA);
+  const T* B = Bbase + (SPLITK ? 0 : (int64_t)blockIdx.z * strideB);
+  OUT* C = Cbase + (SPLITK ? 0 : (int64_t)blockIdx.z * strideC);
+  int k_begin = 0, k_end = K;
+  if (SPLITK) {
+    k_begin = blockIdx.z * kchunk;
+    k_end = min(K, k_begin + kchunk);
+  }
 
   __shared__ __attribute__((aligned(16))) T a_lds[BM * RS];
   __shared__ __attribute__((aligned(16))) T b_lds[BN * RS];
@@ -138,15 +146,15 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
 
   f32x4 acc[4][4] = {};
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
     if (A_KLAST)
-      stage_klast<T, BM>(a_lds, A, lda, m0, M, k0, K, tid);
+      stage_klast<T, BM>(a_lds, A, lda, m0, M, k0, k_end, tid);
     else
-      stage_kmajor<T, BM>(a_lds, A, lda, m0, M, k0, K, tid);
+      stage_kmajor<T, BM>(a_lds, A, lda, m0, M, k0, k_end, tid);
     if (B_KLAST)
-      stage_klast<T, BN>(b_lds, B, ldb, n0, N, k0, K, tid);
+      stage_klast<T, BN>(b_lds, B, ldb, n0, N, k0, k_end, tid);
     else
-      stage_kmajor<T, BN>(b_lds, B, ldb, n0, N, k0, K, tid);
+      stage_kmajor<T, BN>(b_lds, B, ldb, n0, N, k0, k_end, tid);
     __syncthreads();
 
 #pragma unroll
@@ -179,12 +187,38 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
         if (row >= M) continue;
-        int64_t idx = (int64_t)row * ldc + col;
-        float v = alpha * acc[fm][fn][r] + bv;
-        if (beta != 0.0f) v += beta * to_f32(C[idx]);
-        from_f32(v, C[idx]);
+        if (SPLITK) {
+          // raw partial; alpha/beta/bias applied by the reduce kernel
+          ws[((int64_t)blockIdx.z * M + row) * N + col] = acc[fm][fn][r];
+        } else {
+          int64_t idx = (int64_t)row * ldc + col;
+          float v = alpha * acc[fm][fn][r] + bv;
+          if (beta != 0.0f) v += beta * to_f32(C[idx]);
+          from_f32(v, C[idx]);
+        }
       }
     }
+  }
+}
+
+// combine split-K partial slabs: C = alpha*sum_s ws[s] + bias + beta*C
+template <typename OUT, bool HAS_BIAS>
+__global__ void splitk_reduce_k(const float* __restrict__ ws,
+                                OUT* __restrict__ C,
+                                const float* __restrict__ bias,
+                                int M, int N, int64_t ldc, int splitk,
+                                float alpha, float beta) {
+  int64_t MN = (int64_t)M * N;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < MN;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float v = 0.f;
+    for (int s = 0; s < splitk; ++s) v += ws[s * MN + i];
+    int row = i / N, col = i % N;
+    v *= alpha;
+    if (HAS_BIAS) v += bias[col];
+    int64_t idx = (int64_t)row * ldc + col;
+    if (beta != 0.0f) v += beta * to_f32(C[idx]);
+    from_f32(v, C[idx]);
   }
 }
 
@@ -194,12 +228,37 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
 
 template <typename T, typename OUT>
 static void gemm_dispatch(const GemmArgs& g, hipStream_t s) {
-  dim3 grid(cdiv(g.N, 128), cdiv(g.M, 128), g.batch);
+  const bool sk = g.splitk > 1;
+  dim3 grid(cdiv(g.N, 128), cdiv(g.M, 128), sk ? g.splitk : g.batch);
   dim3 block(256);
+  int kchunk = 0;
+  if (sk) {
+    constexpr int BK = GemmTraits<T>::BK;
+    kchunk = cdiv(cdiv(g.K, g.splitk), BK) * BK;
+  }
 #define PS_GEMM_CASE(AK, BK_, HB)                                            \
-  gemm_kernel<T, OUT, AK, BK_, HB><<<grid, block, 0, s>>>(                   \
-      (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,        \
-      g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha, g.beta);
+  if (sk) {                                                                  \
+    gemm_kernel<T, OUT, AK, BK_, HB, true><<<grid, block, 0, s>>>(           \
+        (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,      \
+        g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha,       \
+        g.beta, (float*)g.ws, kchunk);                                       \
+    int64_t MN = (int64_t)g.M * g.N;                                         \
+    int64_t rb = cdiv64(MN, 256); if (rb > 2048) rb = 2048;            \
+    dim3 rg((unsigned)rb);                  \
+    if (HB)                                                                  \
+      splitk_reduce_k<OUT, true><<<rg, 256, 0, s>>>(                         \
+          (const float*)g.ws, (OUT*)g.C, g.bias, g.M, g.N, g.ldc, g.splitk,  \
+          g.alpha, g.beta);                                                  \
+    else                                                                     \
+      splitk_reduce_k<OUT, false><<<rg, 256, 0, s>>>(                        \
+          (const float*)g.ws, (OUT*)g.C, g.bias, g.M, g.N, g.ldc, g.splitk,  \
+          g.alpha, g.beta);                                                  \
+  } else {                                                                   \
+    gemm_kernel<T, OUT, AK, BK_, HB, false><<<grid, block, 0, s>>>(          \
+        (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,      \
+        g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha,       \
+        g.beta, nullptr, 0);                                                 \
+  }
   const bool hb = g.bias != nullptr;
   if (g.a_klast && g.b_klast) {
     if (hb) { PS_GEMM_CASE(true, true, true) } else { PS_GEMM_CASE(true, true, false) }

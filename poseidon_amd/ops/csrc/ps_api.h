@@ -14,6 +14,10 @@ struct GemmArgs {
   int batch;
   float alpha, beta;
   bool a_klast, b_klast;
+  // split-K (for small-tile huge-K GEMMs, e.g. conv wgrad): when splitk > 1,
+  // ws must hold splitk * M * N floats and batch must be 1.
+  void* ws;
+  int splitk;
 };
 
 struct PoolGeom {
