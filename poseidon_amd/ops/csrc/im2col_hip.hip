@@ -13,19 +13,23 @@
 
 namespace ps {
 
-// one thread per (np, g, khw); copies Cg contiguous channels
-template <typename T>
+// one thread per (np, g, khw, chunk-of-V-channels); V=16B vectors when the
+// group channel count allows (guide Guideline 13: always vectorize)
+template <typename T, int V>
 __global__ void im2col_nhwc_k(const T* __restrict__ x, T* __restrict__ colT,
                               ConvGeom g) {
+  typedef T vec_t __attribute__((ext_vector_type(V)));
   const int Cg = g.C / g.G;
+  const int CV = Cg / V;
   const int KHW = g.kh * g.kw;
   const int Kcol = g.G * KHW * Cg;
   const int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
-  int64_t total = NP * g.G * KHW;
+  int64_t total = NP * g.G * KHW * CV;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
-    int khw = i % KHW;
-    int64_t t = i / KHW;
+    int cv = i % CV;
+    int64_t t = i / CV;
+    int khw = t % KHW; t /= KHW;
     int grp = t % g.G;
     int64_t np = t / g.G;
     int ow = np % g.Wo;
@@ -35,13 +39,14 @@ __global__ void im2col_nhwc_k(const T* __restrict__ x, T* __restrict__ colT,
     int kkh = khw / g.kw, kkw = khw % g.kw;
     int ih = oh * g.sh - g.ph + kkh;
     int iw = ow * g.sw - g.pw + kkw;
-    T* dst = colT + np * Kcol + ((int64_t)grp * KHW + khw) * Cg;
+    T* dst = colT + np * Kcol + ((int64_t)grp * KHW + khw) * Cg + cv * V;
+    vec_t v = {};
     if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
-      const T* src = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C + grp * Cg;
-      for (int c = 0; c < Cg; ++c) dst[c] = src[c];
-    } else {
-      for (int c = 0; c < Cg; ++c) dst[c] = (T)0.0f;
+      const T* src = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C
+                     + grp * Cg + cv * V;
+      v = *reinterpret_cast<const vec_t*>(src);
     }
+    *reinterpret_cast<vec_t*>(dst) = v;
   }
 }
 
@@ -137,13 +142,22 @@ __global__ void nchw_to_nhwc_k(const TI* src, TO* dst, int N, int C, int H, int 
 extern "C" {
 
 void ps_im2col_nhwc_f32(const float* x, float* colT, const ConvGeom* g, hipStream_t s) {
-  int64_t total = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
- hipLaunchKernelGGL(( im2col_nhwc_k<float>), dim3(ew_grid(total)), dim3(256), 0, s, x, colT, *g);
+  int Cg = g->C / g->G;
+  int64_t base = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
+  if (Cg % 4 == 0)
+   hipLaunchKernelGGL(( im2col_nhwc_k<float, 4>), dim3(ew_grid(base * (Cg / 4))), dim3(256), 0, s, x, colT, *g);
+  else
+   hipLaunchKernelGGL(( im2col_nhwc_k<float, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, x, colT, *g);
 }
 void ps_im2col_nhwc_bf16(const void* x, void* colT, const ConvGeom* g, hipStream_t s) {
-  int64_t total = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
- hipLaunchKernelGGL(( im2col_nhwc_k<__bf16>), dim3(ew_grid(total)), dim3(256), 0, s, (const __bf16*)x,
-                                                       (__bf16*)colT, *g);
+  int Cg = g->C / g->G;
+  int64_t base = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
+  if (Cg % 8 == 0)
+   hipLaunchKernelGGL(( im2col_nhwc_k<__bf16, 8>), dim3(ew_grid(base * (Cg / 8))), dim3(256), 0, s, 
+        (const __bf16*)x, (__bf16*)colT, *g);
+  else
+   hipLaunchKernelGGL(( im2col_nhwc_k<__bf16, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, 
+        (const __bf16*)x, (__bf16*)colT, *g);
 }
 void ps_col2im_nhwc_f32(const float* colT, float* dx, const ConvGeom* g, hipStream_t s) {
   int64_t total = (int64_t)g->N * g->H * g->W * g->C;
