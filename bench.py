@@ -35,6 +35,9 @@ def main():
                     choices=list(DEFAULT_BATCH))
     ap.add_argument("--batch", type=int, default=0, help="per-GPU batch")
     ap.add_argument("--no-sfb", action="store_true")
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"],
+                    help="compute dtype (activations + GEMM inputs); "
+                         "master weights and gradient accumulation stay fp32")
     args = ap.parse_args()
 
     import poseidon_amd as pa
@@ -50,7 +53,9 @@ def main():
     n_gpus = max(world, 1)
 
     device = "cuda" if torch.cuda.is_available() else "cpu"
-    pa.init(device=device, seed=1234)
+    cd = torch.bfloat16 if (args.dtype == "bf16" and device == "cuda") \
+        else torch.float32
+    pa.init(device=device, seed=1234, compute_dtype=cd)
     batch = args.batch or DEFAULT_BATCH[args.model]
 
     sp = Message("SolverParameter", base_lr=0.01, lr_policy="fixed",
@@ -92,7 +97,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": args.dtype if device == "cuda" else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,

@@ -96,10 +96,11 @@ class Blob:
 
     @property
     def diff(self) -> torch.Tensor:
+        # diff dtype follows data: bf16 activations carry bf16 grads
+        # (bandwidth); parameter blobs are fp32 masters, so their grads
+        # accumulate in fp32.
         if self._diff is None:
-            d = self._data
-            dtype = torch.float32 if self.dtype in (torch.bfloat16, torch.float16) else self.dtype
-            self._diff = torch.zeros(self._shape, dtype=dtype, device=self.device)
+            self._diff = torch.zeros(self._shape, dtype=self.dtype, device=self.device)
         return self._diff
 
     @diff.setter

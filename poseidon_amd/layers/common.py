@@ -31,7 +31,7 @@ class InnerProductLayer(Layer):
         self.N = int(ip.num_output)
         self.bias_term = bool(ip.bias_term)
         self.K = bottom[0].count // bottom[0].num
-        dtype = bottom[0].dtype
+        dtype = torch.float64 if bottom[0].dtype == torch.float64 else torch.float32
         # Caffe IP weight blob is literally (1,1,N,K), bias (1,1,1,N)
         # (inner_product_layer.cpp:83-99) -- keep those shapes for
         # .caffemodel byte-compat; compute views them as (N,K)/(N,).

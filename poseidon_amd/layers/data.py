@@ -63,9 +63,14 @@ class DummyDataLayer(Layer):
             t.reshape(s)
 
     def forward(self, bottom, top) -> None:
+        cd = ctx().compute_dtype
         for i, t in enumerate(top):
             if not self._filled or self.refill[i]:
                 fillers.fill(t, self.fillers[i])
+                # data top (first) runs at the compute dtype; label-like tops
+                # stay fp32 (class ids > 256 are inexact in bf16)
+                if i == 0 and t.data.dtype != cd:
+                    t.data = t.data.to(cd)
         self._filled = True
 
     def backward(self, top, propagate_down, bottom) -> None:
@@ -103,9 +108,9 @@ class MemoryDataLayer(Layer):
             raise RuntimeError("MemoryDataLayer: call add_data() first")
         n = self._data.shape[0]
         idx = torch.arange(self._pos, self._pos + self.batch) % n
-        dev = ctx().torch_device
-        top[0].data = self._data[idx].to(dev, torch.float32)
-        top[1].data = self._labels[idx].to(dev, torch.float32)
+        c = ctx()
+        top[0].data = self._data[idx].to(c.torch_device, c.compute_dtype)
+        top[1].data = self._labels[idx].to(c.torch_device, torch.float32)
         self._pos = (self._pos + self.batch) % n
 
     def backward(self, top, propagate_down, bottom) -> None:
@@ -204,10 +209,12 @@ class DataLayer(_PrefetchingDataLayer):
 
     def forward(self, bottom, top) -> None:
         data, labels = self._next_batch()
-        dev = ctx().torch_device
-        top[0].data = torch.from_numpy(data).to(dev, non_blocking=True)
+        c = ctx()
+        top[0].data = torch.from_numpy(data).to(c.torch_device, c.compute_dtype,
+                                                non_blocking=True)
         if len(top) > 1:
-            top[1].data = torch.from_numpy(labels).to(dev, non_blocking=True)
+            top[1].data = torch.from_numpy(labels).to(c.torch_device,
+                                                      non_blocking=True)
 
 
 @register_layer("IMAGE_DATA")
@@ -265,9 +272,11 @@ class ImageDataLayer(_PrefetchingDataLayer):
 
     def forward(self, bottom, top) -> None:
         data, labels = self._next_batch()
-        dev = ctx().torch_device
-        top[0].data = torch.from_numpy(data).to(dev, non_blocking=True)
-        top[1].data = torch.from_numpy(labels).to(dev, non_blocking=True)
+        c = ctx()
+        top[0].data = torch.from_numpy(data).to(c.torch_device, c.compute_dtype,
+                                                non_blocking=True)
+        top[1].data = torch.from_numpy(labels).to(c.torch_device,
+                                                  non_blocking=True)
 
 
 @register_layer("HDF5_DATA")

@@ -8,6 +8,7 @@ power,absval,threshold}_layer.{cpp,cu}.
 from __future__ import annotations
 
 import torch
+import zlib
 
 from ..core.blob import Blob
 from ..core.layer import Layer, register_layer
@@ -78,7 +79,7 @@ class DropoutLayer(NeuronLayer):
         dp = self.param.dropout_param
         self.ratio = float(dp.dropout_ratio) if dp is not None else 0.5
         assert 0.0 <= self.ratio < 1.0
-        self.seed = ctx().seed * 7919 + (hash(self.name) & 0xFFFF)
+        self.seed = ctx().seed * 7919 + (zlib.crc32(self.name.encode()) & 0xFFFF)
         self.offset = 0
         self._mask = None
 

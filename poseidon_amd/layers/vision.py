@@ -48,7 +48,9 @@ class ConvolutionLayer(Layer):
         channels = bottom[0].channels
         assert channels % self.group == 0 and self.num_output % self.group == 0
 
-        dtype = bottom[0].dtype
+        # params are fp32 masters even when activations are bf16
+        # (f64 preserved for CPU finite-difference checks)
+        dtype = torch.float64 if bottom[0].dtype == torch.float64 else torch.float32
         w = Blob((self.num_output, channels // self.group,
                   self.kernel[0], self.kernel[1]), dtype=dtype,
                  name=f"{self.name}.weight")

@@ -2,6 +2,11 @@
 // are channels-last (NHWC physical); bindings normalize layouts and compose
 // the conv pipeline (im2col -> MFMA GEMM -> bias / col2im / weight repack).
 //
+// Dtypes: activations are fp32 or bf16 (the MFMA fast path); parameter
+// master copies are always fp32 -- conv/linear cast weights to bf16 shadows
+// per call when the activations are bf16, and weight/bias GRADIENTS are
+// produced in fp32 (bf16 GEMM with fp32 accumulate and fp32 output).
+//
 // These entry points are the ONLY GPU compute path -- ops/functional.py has
 // no eager-torch fallback on CUDA tensors, so a passing GPU test means these
 // kernels ran.
@@ -9,6 +14,9 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cmath>
 
 #include "ps_api.h"
 
@@ -18,37 +26,54 @@ using at::Tensor;
 
 hipStream_t stream() { return c10::hip::getCurrentHIPStream().stream(); }
 
+bool is_bf16(const Tensor& t) { return t.scalar_type() == at::kBFloat16; }
+
+void check_float_like(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.scalar_type() == at::kFloat || t.scalar_type() == at::kBFloat16,
+              name, ": expected float32 or bfloat16");
+  TORCH_CHECK(t.is_cuda(), name, ": expected device tensor");
+}
+
 Tensor cl4(const Tensor& t) {  // channels-last contiguous view of a 4D tensor
   TORCH_CHECK(t.dim() == 4, "expected 4D tensor");
   return t.contiguous(at::MemoryFormat::ChannelsLast);
 }
 
 // 2D [N*H*W, C] view of a channels-last 4D tensor (physical NHWC rows).
-// permute to logical NHWC first: that order IS contiguous for channels-last
-// storage, so reshape is a zero-copy view.
 Tensor rows2d(const Tensor& t_cl) {
   int64_t NP = t_cl.size(0) * t_cl.size(2) * t_cl.size(3);
   return t_cl.permute({0, 2, 3, 1}).reshape({NP, t_cl.size(1)});
 }
 
-void check_f32(const Tensor& t, const char* name) {
-  TORCH_CHECK(t.scalar_type() == at::kFloat, name, ": expected float32");
-  TORCH_CHECK(t.is_cuda(), name, ": expected device tensor");
+// fp32 master weight -> compute-dtype shadow (bf16 cast via our kernel)
+Tensor weight_shadow(const Tensor& w_f32, bool bf16) {
+  if (!bf16) return w_f32.contiguous();
+  auto wc = w_f32.contiguous();
+  Tensor out = at::empty(wc.sizes(), wc.options().dtype(at::kBFloat16));
+  ps_f32_to_bf16(wc.data_ptr<float>(), out.data_ptr(), wc.numel(), stream());
+  return out;
 }
 
 // ---------------------------------------------------------------------------
 // GEMM plumbing
 // ---------------------------------------------------------------------------
 
+// in_bf16: A/B are bf16; out_f32: C is fp32 (else C matches input dtype)
 void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
               const float* bias, int M, int N, int K,
               int64_t lda, int64_t ldb, int64_t ldc,
               int64_t a_off, int64_t b_off, int64_t c_off,
               bool a_klast, bool b_klast, float alpha, float beta) {
+  const bool in_bf16 = is_bf16(A);
+  TORCH_CHECK(is_bf16(B) == in_bf16, "gemm: A/B dtype mismatch");
+  const bool out_f32 = !is_bf16(C);
   GemmArgs g;
-  g.A = A.data_ptr<float>() + a_off;
-  g.B = B.data_ptr<float>() + b_off;
-  g.C = C.data_ptr<float>() + c_off;
+  g.A = in_bf16 ? (const void*)((const at::BFloat16*)A.data_ptr() + a_off)
+                : (const void*)(A.data_ptr<float>() + a_off);
+  g.B = in_bf16 ? (const void*)((const at::BFloat16*)B.data_ptr() + b_off)
+                : (const void*)(B.data_ptr<float>() + b_off);
+  g.C = out_f32 ? (void*)(C.data_ptr<float>() + c_off)
+                : (void*)((at::BFloat16*)C.data_ptr() + c_off);
   g.bias = bias;
   g.M = M; g.N = N; g.K = K;
   g.lda = lda; g.ldb = ldb; g.ldc = ldc;
@@ -59,30 +84,36 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   g.ws = nullptr;
   g.splitk = 1;
   // Split-K when the output tile grid cannot fill 256 CUs but K is deep
-  // (conv wgrad: M=Cout<=384, N=Kcol, K=N*OH*OW up to ~800k): target ~512
+  // (conv wgrad: M=Cout, N=Kcol, K=N*OH*OW up to ~800k): target ~512
   // workgroups, cap the f32 workspace at 256 MB.
   int64_t tiles = (int64_t)((M + 127) / 128) * ((N + 127) / 128);
+  Tensor ws;  // keep alive until launch returns
   if (tiles < 96 && K >= 4096) {
     int sk = (int)std::min<int64_t>(512 / tiles, (K + 2047) / 2048);
     int64_t ws_elems = (int64_t)sk * M * N;
     if (sk > 1 && ws_elems * 4 <= (256LL << 20)) {
       g.splitk = sk;
-      Tensor ws = at::empty({ws_elems}, C.options().dtype(at::kFloat));
+      ws = at::empty({ws_elems}, C.options().dtype(at::kFloat));
       g.ws = ws.data_ptr<float>();
-      ps_gemm_f32(&g, stream());
-      return;
     }
   }
-  ps_gemm_f32(&g, stream());
+  if (!in_bf16) {
+    TORCH_CHECK(out_f32, "f32 gemm must have f32 out");
+    ps_gemm_f32(&g, stream());
+  } else if (out_f32) {
+    ps_gemm_bf16_f32out(&g, stream());
+  } else {
+    ps_gemm_bf16(&g, stream());
+  }
 }
 
 // Generic exposed GEMM (tests): C[M,N] = op(A)@op(B); op via *_klast flags.
 Tensor gemm(const Tensor& A, const Tensor& B, int M, int N, int K,
             bool a_klast, bool b_klast) {
-  check_f32(A, "A"); check_f32(B, "B");
+  check_float_like(A, "A");
   auto Ac = A.contiguous();
   auto Bc = B.contiguous();
-  Tensor C = at::empty({M, N}, A.options());
+  Tensor C = at::empty({M, N}, A.options().dtype(at::kFloat));
   int64_t lda = a_klast ? K : M;
   int64_t ldb = b_klast ? K : N;
   run_gemm(Ac, Bc, C, nullptr, M, N, K, lda, ldb, N, 0, 0, 0,
@@ -96,9 +127,9 @@ Tensor gemm(const Tensor& A, const Tensor& B, int M, int N, int K,
 
 Tensor linear_forward(const Tensor& x, const Tensor& w,
                       const c10::optional<Tensor>& bias) {
-  check_f32(x, "x"); check_f32(w, "w");
+  check_float_like(x, "x");
   auto xc = x.contiguous();
-  auto wc = w.contiguous();
+  auto wc = weight_shadow(w, is_bf16(x));
   int M = xc.size(0), K = xc.size(1), N = wc.size(0);
   TORCH_CHECK(wc.size(1) == K, "linear: K mismatch");
   Tensor y = at::empty({M, N}, x.options());
@@ -115,9 +146,9 @@ Tensor linear_forward(const Tensor& x, const Tensor& w,
 std::vector<c10::optional<Tensor>> linear_backward(
     const Tensor& x, const Tensor& w, const Tensor& dy,
     bool need_dx, bool need_dw, bool has_bias) {
-  check_f32(dy, "dy");
+  check_float_like(dy, "dy");
   auto xc = x.contiguous();
-  auto wc = w.contiguous();
+  auto wc = weight_shadow(w, is_bf16(dy));
   auto dyc = dy.contiguous();
   int M = xc.size(0), K = xc.size(1), N = wc.size(0);
   c10::optional<Tensor> dx, dw, db;
@@ -129,27 +160,30 @@ std::vector<c10::optional<Tensor>> linear_backward(
     dx = t;
   }
   if (need_dw) {
-    Tensor t = at::empty({N, K}, w.options());
-    // dW[N,K] = dy^T[N,M] @ x[M,K]: contraction M; both K-major
+    Tensor t = at::empty({N, K}, x.options().dtype(at::kFloat));
+    // dW[N,K] = dy^T[N,M] @ x[M,K]: contraction M; both K-major; fp32 out
     run_gemm(dyc, xc, t, nullptr, N, K, M, N, K, K, 0, 0, 0, false, false,
              1.0f, 0.0f);
     dw = t;
   }
   if (has_bias) {
-    Tensor t = at::zeros({N}, x.options());
-    ps_colsum_f32(dyc.data_ptr<float>(), t.data_ptr<float>(), M, N, stream());
+    Tensor t = at::zeros({N}, x.options().dtype(at::kFloat));
+    if (is_bf16(dyc))
+      ps_colsum_bf16(dyc.data_ptr(), t.data_ptr<float>(), M, N, stream());
+    else
+      ps_colsum_f32(dyc.data_ptr<float>(), t.data_ptr<float>(), M, N, stream());
     db = t;
   }
   return {dx, dw, db};
 }
 
-// SFB reconstruction: dW[N,K] = a[M,N]^T @ b[M,K]
+// SFB reconstruction: dW[N,K] = a[M,N]^T @ b[M,K] -> fp32
 Tensor gemm_at_b(const Tensor& a, const Tensor& b) {
-  check_f32(a, "a"); check_f32(b, "b");
+  check_float_like(a, "a");
   auto ac = a.contiguous();
   auto bc = b.contiguous();
   int M = ac.size(0), N = ac.size(1), K = bc.size(1);
-  Tensor t = at::empty({N, K}, a.options());
+  Tensor t = at::empty({N, K}, a.options().dtype(at::kFloat));
   run_gemm(ac, bc, t, nullptr, N, K, M, N, K, K, 0, 0, 0, false, false,
            1.0f, 0.0f);
   return t;
@@ -159,7 +193,7 @@ Tensor gemm_at_b(const Tensor& a, const Tensor& b) {
 // Convolution (im2col + grouped MFMA GEMM, NHWC)
 // ---------------------------------------------------------------------------
 
-ConvGeom conv_geom(const Tensor& x_cl, int Co, int kh, int kw, int sh, int sw,
+ConvGeom conv_geom(const Tensor& x_cl, int kh, int kw, int sh, int sw,
                    int ph, int pw, int G) {
   ConvGeom g;
   g.N = x_cl.size(0); g.C = x_cl.size(1);
@@ -170,26 +204,37 @@ ConvGeom conv_geom(const Tensor& x_cl, int Co, int kh, int kw, int sh, int sw,
   return g;
 }
 
-// colT cache handle: conv forward returns (y, colT); backward reuses colT.
+// weights [Co,Cig,kh,kw] fp32 -> khwc [Co, kh*kw*Cig] in compute dtype
+Tensor weight_khwc(const Tensor& w, bool bf16) {
+  auto wc = w.contiguous();
+  int Co = wc.size(0), Cig = wc.size(1), kh = wc.size(2), kw = wc.size(3);
+  Tensor wk = at::empty({Co, (int64_t)kh * kw * Cig},
+                        wc.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
+  if (bf16)
+    ps_weight_to_khwc_f32_bf16(wc.data_ptr<float>(), wk.data_ptr(), Co, Cig,
+                               kh, kw, stream());
+  else
+    ps_weight_to_khwc_f32(wc.data_ptr<float>(), wk.data_ptr<float>(), Co, Cig,
+                          kh, kw, stream());
+  return wk;
+}
+
 std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
                                       const c10::optional<Tensor>& bias,
                                       int sh, int sw, int ph, int pw, int G) {
-  check_f32(x, "x"); check_f32(w, "w");
+  check_float_like(x, "x");
+  const bool bf16 = is_bf16(x);
   auto x_cl = cl4(x);
-  auto wc = w.contiguous();  // [Co, Cg, kh, kw] NCHW
-  int Co = wc.size(0), Cig = wc.size(1), kh = wc.size(2), kw = wc.size(3);
-  ConvGeom g = conv_geom(x_cl, Co, kh, kw, sh, sw, ph, pw, G);
+  int Co = w.size(0), Cig = w.size(1), kh = w.size(2), kw = w.size(3);
+  ConvGeom g = conv_geom(x_cl, kh, kw, sh, sw, ph, pw, G);
   TORCH_CHECK(Cig * G == g.C, "conv channel/group mismatch");
   int Cg = g.C / G;
-  int Kg = kh * kw * Cg;       // per-group contraction size
+  int Kg = kh * kw * Cg;
   int Kcol = G * Kg;
   int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
   int Cog = Co / G;
 
-  // weights -> khwc [Co, Kg]
-  Tensor wk = at::empty({Co, Kg}, w.options());
-  ps_weight_to_khwc_f32(wc.data_ptr<float>(), wk.data_ptr<float>(), Co, Cig,
-                        kh, kw, stream());
+  Tensor wk = weight_khwc(w, bf16);
 
   bool is_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0);
   Tensor colT;
@@ -197,8 +242,11 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
     colT = rows2d(x_cl);  // alias: x rows ARE the col rows
   } else {
     colT = at::empty({NP, (int64_t)Kcol}, x.options());
-    ps_im2col_nhwc_f32(x_cl.data_ptr<float>(), colT.data_ptr<float>(), &g,
-                       stream());
+    if (bf16)
+      ps_im2col_nhwc_bf16(x_cl.data_ptr(), colT.data_ptr(), &g, stream());
+    else
+      ps_im2col_nhwc_f32(x_cl.data_ptr<float>(), colT.data_ptr<float>(), &g,
+                         stream());
   }
 
   Tensor y = at::empty({g.N, Co, g.Ho, g.Wo},
@@ -217,18 +265,16 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
              /*c_off=*/(int64_t)grp * Cog,
              true, true, 1.0f, 0.0f);
   }
-  // run_gemm's bias pointer is pre-offset per group above; colT returned for
-  // the backward pass to reuse.
   return {y, colT};
 }
 
 Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
                              std::vector<int64_t> x_shape, int sh, int sw,
                              int ph, int pw, int G) {
-  check_f32(dy, "dy");
+  check_float_like(dy, "dy");
+  const bool bf16 = is_bf16(dy);
   auto dy_cl = cl4(dy);
-  auto wc = w.contiguous();
-  int Co = wc.size(0), Cig = wc.size(1), kh = wc.size(2), kw = wc.size(3);
+  int Co = w.size(0), kh = w.size(2), kw = w.size(3);
   Tensor dx = at::empty({x_shape[0], x_shape[1], x_shape[2], x_shape[3]},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
   ConvGeom g;
@@ -240,9 +286,7 @@ Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
   int Kcol = G * Kg;
   int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
 
-  Tensor wk = at::empty({Co, Kg}, w.options());
-  ps_weight_to_khwc_f32(wc.data_ptr<float>(), wk.data_ptr<float>(), Co, Cig,
-                        kh, kw, stream());
+  Tensor wk = weight_khwc(w, bf16);
 
   bool is_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0);
   Tensor dcolT = is_1x1 ? rows2d(dx)
@@ -257,18 +301,22 @@ Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
              /*c_off=*/(int64_t)grp * Kg,
              true, false, 1.0f, 0.0f);
   }
-  if (!is_1x1)
-    ps_col2im_nhwc_f32(dcolT.data_ptr<float>(), dx.data_ptr<float>(), &g,
-                       stream());
+  if (!is_1x1) {
+    if (bf16)
+      ps_col2im_nhwc_bf16(dcolT.data_ptr(), dx.data_ptr(), &g, stream());
+    else
+      ps_col2im_nhwc_f32(dcolT.data_ptr<float>(), dx.data_ptr<float>(), &g,
+                         stream());
+  }
   return dx;
 }
 
-// dW accumulated into dw_out (NCHW [Co,Cg,kh,kw]); db into db_out if given.
+// dW accumulated into dw_out (fp32 NCHW [Co,Cg,kh,kw]); db into db_out.
 void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
                                 const Tensor& dy, Tensor dw_out,
                                 c10::optional<Tensor> db_out,
                                 int sh, int sw, int ph, int pw, int G) {
-  check_f32(dy, "dy");
+  check_float_like(dy, "dy");
   auto dy_cl = cl4(dy);
   int Co = dw_out.size(0), Cig = dw_out.size(1);
   int kh = dw_out.size(2), kw = dw_out.size(3);
@@ -277,7 +325,8 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
   int64_t NP = (int64_t)dy_cl.size(0) * dy_cl.size(2) * dy_cl.size(3);
   int64_t Kcol = colT.size(1);
 
-  Tensor dwk = at::empty({Co, Kg}, dy.options());
+  // fp32 gradient accumulation regardless of activation dtype
+  Tensor dwk = at::empty({Co, Kg}, dy.options().dtype(at::kFloat));
   Tensor dy2 = rows2d(dy_cl);
   for (int grp = 0; grp < G; ++grp) {
     // dwk_g[Cog, Kg] = dy_g^T[Cog, NP] @ colT_g[NP, Kg]: contraction NP
@@ -288,12 +337,15 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
              /*c_off=*/(int64_t)grp * Cog * Kg,
              false, false, 1.0f, 0.0f);
   }
-  // accumulate into NCHW grad
   ps_weight_from_khwc_f32(dwk.data_ptr<float>(), dw_out.data_ptr<float>(),
                           Co, Cig, kh, kw, /*beta=*/1.0f, stream());
   if (db_out.has_value()) {
-    ps_colsum_f32(dy_cl.data_ptr<float>(), db_out->data_ptr<float>(), NP, Co,
-                  stream());
+    if (is_bf16(dy_cl))
+      ps_colsum_bf16(dy_cl.data_ptr(), db_out->data_ptr<float>(), NP, Co,
+                     stream());
+    else
+      ps_colsum_f32(dy_cl.data_ptr<float>(), db_out->data_ptr<float>(), NP,
+                    Co, stream());
   }
 }
 
@@ -319,7 +371,7 @@ PoolGeom pool_geom(const Tensor& x_cl, int kh, int kw, int sh, int sw,
 
 std::vector<Tensor> pool_max_forward(const Tensor& x, int kh, int kw, int sh,
                                      int sw, int ph, int pw) {
-  check_f32(x, "x");
+  check_float_like(x, "x");
   auto x_cl = cl4(x);
   PoolGeom g = pool_geom(x_cl, kh, kw, sh, sw, ph, pw);
   auto opts_cl = x.options().memory_format(at::MemoryFormat::ChannelsLast);
@@ -327,8 +379,12 @@ std::vector<Tensor> pool_max_forward(const Tensor& x, int kh, int kw, int sh,
   Tensor mask = at::empty({g.N, g.C, g.Ho, g.Wo},
                           x.options().dtype(at::kInt)
                               .memory_format(at::MemoryFormat::ChannelsLast));
-  ps_maxpool_fwd_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(),
-                     mask.data_ptr<int>(), &g, stream());
+  if (is_bf16(x))
+    ps_maxpool_fwd_bf16(x_cl.data_ptr(), y.data_ptr(), mask.data_ptr<int>(),
+                        &g, stream());
+  else
+    ps_maxpool_fwd_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(),
+                       mask.data_ptr<int>(), &g, stream());
   return {y, mask};
 }
 
@@ -343,19 +399,26 @@ Tensor pool_max_backward(const Tensor& dy, const Tensor& mask,
   g.Ho = dy_cl.size(2); g.Wo = dy_cl.size(3);
   Tensor dx = at::empty({g.N, g.C, g.H, g.W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
-  ps_maxpool_bwd_f32(dy_cl.data_ptr<float>(), mask_cl.data_ptr<int>(),
-                     dx.data_ptr<float>(), &g, stream());
+  if (is_bf16(dy))
+    ps_maxpool_bwd_bf16(dy_cl.data_ptr(), mask_cl.data_ptr<int>(),
+                        dx.data_ptr(), &g, stream());
+  else
+    ps_maxpool_bwd_f32(dy_cl.data_ptr<float>(), mask_cl.data_ptr<int>(),
+                       dx.data_ptr<float>(), &g, stream());
   return dx;
 }
 
 Tensor pool_ave_forward(const Tensor& x, int kh, int kw, int sh, int sw,
                         int ph, int pw) {
-  check_f32(x, "x");
+  check_float_like(x, "x");
   auto x_cl = cl4(x);
   PoolGeom g = pool_geom(x_cl, kh, kw, sh, sw, ph, pw);
   Tensor y = at::empty({g.N, g.C, g.Ho, g.Wo},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  ps_avepool_fwd_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(), &g, stream());
+  if (is_bf16(x))
+    ps_avepool_fwd_bf16(x_cl.data_ptr(), y.data_ptr(), &g, stream());
+  else
+    ps_avepool_fwd_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(), &g, stream());
   return y;
 }
 
@@ -368,54 +431,63 @@ Tensor pool_ave_backward(const Tensor& dy, std::vector<int64_t> x_shape,
   g.Ho = dy_cl.size(2); g.Wo = dy_cl.size(3);
   Tensor dx = at::empty({g.N, g.C, g.H, g.W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
-  ps_avepool_bwd_f32(dy_cl.data_ptr<float>(), dx.data_ptr<float>(), &g, stream());
+  if (is_bf16(dy))
+    ps_avepool_bwd_bf16(dy_cl.data_ptr(), dx.data_ptr(), &g, stream());
+  else
+    ps_avepool_bwd_f32(dy_cl.data_ptr<float>(), dx.data_ptr<float>(), &g,
+                       stream());
   return dx;
 }
 
 std::vector<Tensor> pool_stoch_forward_train(const Tensor& x, int kh, int kw,
                                              int sh, int sw, int ph, int pw,
                                              int64_t seed) {
-  check_f32(x, "x");
-  auto x_cl = cl4(x);
+  check_float_like(x, "x");
+  auto x_cl = cl4(x).to(at::kFloat);  // stochastic pooling runs fp32
   PoolGeom g = pool_geom(x_cl, kh, kw, sh, sw, ph, pw);
   Tensor y = at::empty({g.N, g.C, g.Ho, g.Wo},
-                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+                       x_cl.options().memory_format(at::MemoryFormat::ChannelsLast));
   Tensor mask = at::empty({g.N, g.C, g.Ho, g.Wo},
-                          x.options().dtype(at::kInt)
+                          x_cl.options().dtype(at::kInt)
                               .memory_format(at::MemoryFormat::ChannelsLast));
   ps_stochpool_fwd_train_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(),
                              mask.data_ptr<int>(), &g, (uint64_t)seed, stream());
-  return {y, mask};
+  return {y.to(x.scalar_type()), mask};
 }
 
 Tensor pool_stoch_forward_test(const Tensor& x, int kh, int kw, int sh,
                                int sw, int ph, int pw) {
-  check_f32(x, "x");
-  auto x_cl = cl4(x);
+  check_float_like(x, "x");
+  auto x_cl = cl4(x).to(at::kFloat);
   PoolGeom g = pool_geom(x_cl, kh, kw, sh, sw, ph, pw);
   Tensor y = at::empty({g.N, g.C, g.Ho, g.Wo},
-                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+                       x_cl.options().memory_format(at::MemoryFormat::ChannelsLast));
   ps_stochpool_fwd_test_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(), &g,
                             stream());
-  return y;
+  return y.to(x.scalar_type());
 }
 
 // ---------------------------------------------------------------------------
-// LRN
+// LRN (scale is fp32 in both paths)
 // ---------------------------------------------------------------------------
 
 std::vector<Tensor> lrn_forward(const Tensor& x, int size, double alpha,
                                 double beta) {
-  check_f32(x, "x");
+  check_float_like(x, "x");
   auto x_cl = cl4(x);
   int64_t rows = (int64_t)x_cl.size(0) * x_cl.size(2) * x_cl.size(3);
   int C = x_cl.size(1);
   auto opts_cl = x.options().memory_format(at::MemoryFormat::ChannelsLast);
   Tensor y = at::empty_like(x_cl, opts_cl);
-  Tensor scale = at::empty_like(x_cl, opts_cl);
-  ps_lrn_fwd_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(),
-                 scale.data_ptr<float>(), rows, C, size, (float)alpha,
-                 (float)beta, stream());
+  Tensor scale = at::empty_like(x_cl,
+      x.options().dtype(at::kFloat).memory_format(at::MemoryFormat::ChannelsLast));
+  if (is_bf16(x))
+    ps_lrn_fwd_bf16(x_cl.data_ptr(), y.data_ptr(), scale.data_ptr<float>(),
+                    rows, C, size, (float)alpha, (float)beta, stream());
+  else
+    ps_lrn_fwd_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(),
+                   scale.data_ptr<float>(), rows, C, size, (float)alpha,
+                   (float)beta, stream());
   return {y, scale};
 }
 
@@ -428,10 +500,15 @@ Tensor lrn_backward(const Tensor& x, const Tensor& y, const Tensor& scale,
   int64_t rows = (int64_t)x_cl.size(0) * x_cl.size(2) * x_cl.size(3);
   int C = x_cl.size(1);
   Tensor dx = at::empty_like(x_cl, x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  ps_lrn_bwd_f32(x_cl.data_ptr<float>(), y_cl.data_ptr<float>(),
-                 sc_cl.data_ptr<float>(), dy_cl.data_ptr<float>(),
-                 dx.data_ptr<float>(), rows, C, size, (float)alpha,
-                 (float)beta, stream());
+  if (is_bf16(x))
+    ps_lrn_bwd_bf16(x_cl.data_ptr(), y_cl.data_ptr(), sc_cl.data_ptr<float>(),
+                    dy_cl.data_ptr(), dx.data_ptr(), rows, C, size,
+                    (float)alpha, (float)beta, stream());
+  else
+    ps_lrn_bwd_f32(x_cl.data_ptr<float>(), y_cl.data_ptr<float>(),
+                   sc_cl.data_ptr<float>(), dy_cl.data_ptr<float>(),
+                   dx.data_ptr<float>(), rows, C, size, (float)alpha,
+                   (float)beta, stream());
   return dx;
 }
 
@@ -440,58 +517,72 @@ Tensor lrn_backward(const Tensor& x, const Tensor& y, const Tensor& scale,
 // ---------------------------------------------------------------------------
 
 Tensor softmax_forward(const Tensor& x) {
-  check_f32(x, "x");
+  check_float_like(x, "x");
+  Tensor xc, y;
+  int64_t rows;
+  int C;
   if (x.dim() == 4) {
-    auto x_cl = cl4(x);
-    int64_t rows = (int64_t)x_cl.size(0) * x_cl.size(2) * x_cl.size(3);
-    int C = x_cl.size(1);
-    Tensor y = at::empty_like(x_cl, x.options().memory_format(at::MemoryFormat::ChannelsLast));
-    ps_softmax_rows_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(), rows, C,
-                        stream());
-    return y;
+    xc = cl4(x);
+    rows = (int64_t)xc.size(0) * xc.size(2) * xc.size(3);
+    C = xc.size(1);
+    y = at::empty_like(xc, x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  } else {
+    xc = x.contiguous();
+    rows = xc.numel() / xc.size(-1);
+    C = xc.size(-1);
+    y = at::empty_like(xc);
   }
-  auto xc = x.contiguous();
-  int64_t rows = xc.numel() / xc.size(-1);
-  int C = xc.size(-1);
-  Tensor y = at::empty_like(xc);
-  ps_softmax_rows_f32(xc.data_ptr<float>(), y.data_ptr<float>(), rows, C,
-                      stream());
+  if (is_bf16(x))
+    ps_softmax_rows_bf16(xc.data_ptr(), y.data_ptr(), rows, C, stream());
+  else
+    ps_softmax_rows_f32(xc.data_ptr<float>(), y.data_ptr<float>(), rows, C,
+                        stream());
   return y;
 }
 
 Tensor softmax_backward(const Tensor& y, const Tensor& dy) {
+  Tensor yc, dyc, dx;
+  int64_t rows;
+  int C;
   if (y.dim() == 4) {
-    auto y_cl = cl4(y);
-    auto dy_cl = cl4(dy);
-    int64_t rows = (int64_t)y_cl.size(0) * y_cl.size(2) * y_cl.size(3);
-    int C = y_cl.size(1);
-    Tensor dx = at::empty_like(y_cl, y.options().memory_format(at::MemoryFormat::ChannelsLast));
-    ps_softmax_bwd_rows_f32(y_cl.data_ptr<float>(), dy_cl.data_ptr<float>(),
-                            dx.data_ptr<float>(), rows, C, stream());
-    return dx;
+    yc = cl4(y);
+    dyc = cl4(dy);
+    rows = (int64_t)yc.size(0) * yc.size(2) * yc.size(3);
+    C = yc.size(1);
+    dx = at::empty_like(yc, y.options().memory_format(at::MemoryFormat::ChannelsLast));
+  } else {
+    yc = y.contiguous();
+    dyc = dy.contiguous();
+    rows = yc.numel() / yc.size(-1);
+    C = yc.size(-1);
+    dx = at::empty_like(yc);
   }
-  auto yc = y.contiguous();
-  auto dyc = dy.contiguous();
-  int64_t rows = yc.numel() / yc.size(-1);
-  int C = yc.size(-1);
-  Tensor dx = at::empty_like(yc);
-  ps_softmax_bwd_rows_f32(yc.data_ptr<float>(), dyc.data_ptr<float>(),
-                          dx.data_ptr<float>(), rows, C, stream());
+  if (is_bf16(y))
+    ps_softmax_bwd_rows_bf16(yc.data_ptr(), dyc.data_ptr(), dx.data_ptr(),
+                             rows, C, stream());
+  else
+    ps_softmax_bwd_rows_f32(yc.data_ptr<float>(), dyc.data_ptr<float>(),
+                            dx.data_ptr<float>(), rows, C, stream());
   return dx;
 }
 
 std::vector<Tensor> softmax_loss_forward(const Tensor& logits,
                                          const Tensor& labels) {
-  check_f32(logits, "logits");
+  check_float_like(logits, "logits");
   auto xc = logits.contiguous();
   auto lc = labels.contiguous().to(at::kFloat);
   int64_t rows = xc.size(0);
   int C = xc.size(1);
   Tensor prob = at::empty_like(xc);
-  Tensor loss = at::zeros({}, xc.options());
-  ps_softmax_loss_fwd_f32(xc.data_ptr<float>(), lc.data_ptr<float>(),
-                          prob.data_ptr<float>(), loss.data_ptr<float>(),
-                          rows, C, stream());
+  Tensor loss = at::zeros({}, xc.options().dtype(at::kFloat));
+  if (is_bf16(xc))
+    ps_softmax_loss_fwd_bf16(xc.data_ptr(), lc.data_ptr<float>(),
+                             prob.data_ptr(), loss.data_ptr<float>(), rows, C,
+                             stream());
+  else
+    ps_softmax_loss_fwd_f32(xc.data_ptr<float>(), lc.data_ptr<float>(),
+                            prob.data_ptr<float>(), loss.data_ptr<float>(),
+                            rows, C, stream());
   loss.div_((double)rows);
   return {loss, prob};
 }
@@ -503,9 +594,14 @@ Tensor softmax_loss_backward(const Tensor& prob, const Tensor& labels,
   int64_t rows = pc.size(0);
   int C = pc.size(1);
   Tensor dx = at::empty_like(pc);
-  ps_softmax_loss_bwd_f32(pc.data_ptr<float>(), lc.data_ptr<float>(),
-                          dx.data_ptr<float>(), rows, C,
-                          (float)(loss_weight / rows), stream());
+  if (is_bf16(pc))
+    ps_softmax_loss_bwd_bf16(pc.data_ptr(), lc.data_ptr<float>(),
+                             dx.data_ptr(), rows, C,
+                             (float)(loss_weight / rows), stream());
+  else
+    ps_softmax_loss_bwd_f32(pc.data_ptr<float>(), lc.data_ptr<float>(),
+                            dx.data_ptr<float>(), rows, C,
+                            (float)(loss_weight / rows), stream());
   return dx;
 }
 
@@ -513,71 +609,97 @@ Tensor softmax_loss_backward(const Tensor& prob, const Tensor& labels,
 // neuron ops
 // ---------------------------------------------------------------------------
 
+Tensor any_contig(const Tensor& t) {
+  return (t.dim() == 4 && t.is_contiguous(at::MemoryFormat::ChannelsLast))
+             ? t : t.contiguous();
+}
+
 Tensor relu_forward(const Tensor& x, double slope) {
-  check_f32(x, "x");
-  auto xc = x.is_contiguous(at::MemoryFormat::ChannelsLast) ? x : x.contiguous();
+  check_float_like(x, "x");
+  auto xc = any_contig(x);
   Tensor y = at::empty_like(xc);
-  ps_relu_fwd_f32(xc.data_ptr<float>(), y.data_ptr<float>(), xc.numel(),
-                  (float)slope, stream());
+  if (is_bf16(x))
+    ps_relu_fwd_bf16(xc.data_ptr(), y.data_ptr(), xc.numel(), (float)slope,
+                     stream());
+  else
+    ps_relu_fwd_f32(xc.data_ptr<float>(), y.data_ptr<float>(), xc.numel(),
+                    (float)slope, stream());
   return y;
 }
 
 Tensor relu_backward(const Tensor& x, const Tensor& dy, double slope) {
-  auto xc = x.is_contiguous(at::MemoryFormat::ChannelsLast) ? x : x.contiguous();
-  auto dyc = dy.dim() == 4 && xc.is_contiguous(at::MemoryFormat::ChannelsLast)
+  auto xc = any_contig(x);
+  auto dyc = xc.dim() == 4 && xc.is_contiguous(at::MemoryFormat::ChannelsLast)
                  ? cl4(dy) : dy.contiguous();
   Tensor dx = at::empty_like(xc);
-  ps_relu_bwd_f32(xc.data_ptr<float>(), dyc.data_ptr<float>(),
-                  dx.data_ptr<float>(), xc.numel(), (float)slope, stream());
+  if (is_bf16(x))
+    ps_relu_bwd_bf16(xc.data_ptr(), dyc.data_ptr(), dx.data_ptr(), xc.numel(),
+                     (float)slope, stream());
+  else
+    ps_relu_bwd_f32(xc.data_ptr<float>(), dyc.data_ptr<float>(),
+                    dx.data_ptr<float>(), xc.numel(), (float)slope, stream());
   return dx;
 }
 
-#define PS_BIND_UNARY(pyname, fn)                                           \
+#define PS_BIND_UNARY(pyname, fn32, fn16)                                   \
   Tensor pyname(const Tensor& x) {                                          \
-    check_f32(x, #pyname);                                                  \
-    auto xc = x.is_contiguous(at::MemoryFormat::ChannelsLast) ? x           \
-                                                              : x.contiguous(); \
+    check_float_like(x, #pyname);                                           \
+    auto xc = any_contig(x);                                                \
     Tensor y = at::empty_like(xc);                                          \
-    fn(xc.data_ptr<float>(), y.data_ptr<float>(), xc.numel(), stream());    \
+    if (is_bf16(x))                                                         \
+      fn16(xc.data_ptr(), y.data_ptr(), xc.numel(), stream());              \
+    else                                                                    \
+      fn32(xc.data_ptr<float>(), y.data_ptr<float>(), xc.numel(), stream());\
     return y;                                                               \
   }
 
-#define PS_BIND_BINARY(pyname, fn)                                          \
+#define PS_BIND_BINARY(pyname, fn32, fn16)                                  \
   Tensor pyname(const Tensor& a, const Tensor& b) {                         \
-    auto ac = a.is_contiguous(at::MemoryFormat::ChannelsLast) ? a           \
-                                                              : a.contiguous(); \
-    auto bc = b.is_contiguous(at::MemoryFormat::ChannelsLast) ? b           \
-                                                              : b.contiguous(); \
+    auto ac = any_contig(a);                                                \
+    auto bc = any_contig(b);                                                \
     Tensor y = at::empty_like(ac);                                          \
-    fn(ac.data_ptr<float>(), bc.data_ptr<float>(), y.data_ptr<float>(),     \
-       ac.numel(), stream());                                               \
+    if (is_bf16(a))                                                         \
+      fn16(ac.data_ptr(), bc.data_ptr(), y.data_ptr(), ac.numel(), stream()); \
+    else                                                                    \
+      fn32(ac.data_ptr<float>(), bc.data_ptr<float>(), y.data_ptr<float>(), \
+           ac.numel(), stream());                                           \
     return y;                                                               \
   }
 
-PS_BIND_UNARY(sigmoid_forward, ps_sigmoid_fwd_f32)
-PS_BIND_BINARY(sigmoid_backward, ps_sigmoid_bwd_f32)
-PS_BIND_UNARY(tanh_forward, ps_tanh_fwd_f32)
-PS_BIND_BINARY(tanh_backward, ps_tanh_bwd_f32)
-PS_BIND_UNARY(bnll_forward, ps_bnll_fwd_f32)
-PS_BIND_BINARY(bnll_backward, ps_bnll_bwd_f32)
+PS_BIND_UNARY(sigmoid_forward, ps_sigmoid_fwd_f32, ps_sigmoid_fwd_bf16)
+PS_BIND_BINARY(sigmoid_backward, ps_sigmoid_bwd_f32, ps_sigmoid_bwd_bf16)
+PS_BIND_UNARY(tanh_forward, ps_tanh_fwd_f32, ps_tanh_fwd_bf16)
+PS_BIND_BINARY(tanh_backward, ps_tanh_bwd_f32, ps_tanh_bwd_bf16)
+PS_BIND_UNARY(bnll_forward, ps_bnll_fwd_f32, ps_bnll_fwd_bf16)
+PS_BIND_BINARY(bnll_backward, ps_bnll_bwd_f32, ps_bnll_bwd_bf16)
 
 std::vector<Tensor> dropout_forward(const Tensor& x, double ratio,
                                     int64_t seed, int64_t offset) {
-  check_f32(x, "x");
-  auto xc = x.is_contiguous(at::MemoryFormat::ChannelsLast) ? x : x.contiguous();
+  check_float_like(x, "x");
+  auto xc = any_contig(x);
   Tensor y = at::empty_like(xc);
   Tensor mask = at::empty(xc.sizes(), xc.options().dtype(at::kByte));
-  ps_dropout_fwd_f32(xc.data_ptr<float>(), y.data_ptr<float>(),
-                     mask.data_ptr<uint8_t>(), xc.numel(), (float)ratio,
-                     (uint64_t)seed, (uint64_t)offset, stream());
+  if (is_bf16(x))
+    ps_dropout_fwd_bf16(xc.data_ptr(), y.data_ptr(), mask.data_ptr<uint8_t>(),
+                        xc.numel(), (float)ratio, (uint64_t)seed,
+                        (uint64_t)offset, stream());
+  else
+    ps_dropout_fwd_f32(xc.data_ptr<float>(), y.data_ptr<float>(),
+                       mask.data_ptr<uint8_t>(), xc.numel(), (float)ratio,
+                       (uint64_t)seed, (uint64_t)offset, stream());
   return {y, mask};
 }
 
 Tensor dropout_backward(const Tensor& dy, const Tensor& mask, double ratio) {
-  auto dyc = dy.is_contiguous(at::MemoryFormat::ChannelsLast) ? dy : dy.contiguous();
+  auto dyc = any_contig(dy);
   Tensor dx = at::empty_like(dyc);
-  ps_dropout_bwd_f32(dyc.data_ptr<float>(), mask.data_ptr<uint8_t>(),
-                     dx.data_ptr<float>(), dyc.numel(), (float)ratio, stream());
+  if (is_bf16(dy))
+    ps_dropout_bwd_bf16(dyc.data_ptr(), mask.data_ptr<uint8_t>(),
+                        dx.data_ptr(), dyc.numel(), (float)ratio, stream());
+  else
+    ps_dropout_bwd_f32(dyc.data_ptr<float>(), mask.data_ptr<uint8_t>(),
+                       dx.data_ptr<float>(), dyc.numel(), (float)ratio,
+                       stream());
   return dx;
 }
 
@@ -638,5 +760,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_update", &sgd_update);
   m.def("nesterov_update", &nesterov_update);
   m.def("adagrad_update", &adagrad_update);
-  m.attr("compute_dtypes") = std::vector<std::string>{"float32"};
+  m.attr("compute_dtypes") = std::vector<std::string>{"float32", "bfloat16"};
 }

@@ -100,9 +100,9 @@ void ps_softmax_loss_bwd_f32(const float*, const float*, float*, int64_t, int,
 void ps_softmax_rows_bf16(const void*, void*, int64_t, int, hipStream_t);
 void ps_softmax_bwd_rows_bf16(const void*, const void*, void*, int64_t, int,
                               hipStream_t);
-void ps_softmax_loss_fwd_bf16(const void*, const void*, void*, float*,
+void ps_softmax_loss_fwd_bf16(const void*, const float*, void*, float*,
                               int64_t, int, hipStream_t);
-void ps_softmax_loss_bwd_bf16(const void*, const void*, void*, int64_t, int,
+void ps_softmax_loss_bwd_bf16(const void*, const float*, void*, int64_t, int,
                               float, hipStream_t);
 
 // sgd.hip

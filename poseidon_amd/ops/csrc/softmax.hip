@@ -125,20 +125,22 @@ void ps_softmax_loss_fwd_f32(const float* x, const float* labels, float* prob,
   softmax_loss_fwd_k<float, float><<<cdiv64(rows, 4), 256, 0, s>>>(
       x, labels, prob, loss_out, rows, C);
 }
-void ps_softmax_loss_fwd_bf16(const void* x, const void* labels, void* prob,
+void ps_softmax_loss_fwd_bf16(const void* x, const float* labels, void* prob,
                               float* loss_out, int64_t rows, int C, hipStream_t s) {
-  softmax_loss_fwd_k<__bf16, __bf16><<<cdiv64(rows, 4), 256, 0, s>>>(
-      (const __bf16*)x, (const __bf16*)labels, (__bf16*)prob, loss_out, rows, C);
+  // labels stay fp32: class indices above 256 are not exactly representable
+  // in bf16's 8-bit mantissa
+  softmax_loss_fwd_k<__bf16, float><<<cdiv64(rows, 4), 256, 0, s>>>(
+      (const __bf16*)x, labels, (__bf16*)prob, loss_out, rows, C);
 }
 void ps_softmax_loss_bwd_f32(const float* prob, const float* labels, float* dx,
                              int64_t rows, int C, float w, hipStream_t s) {
   softmax_loss_bwd_k<float, float><<<ew_grid(rows * C), 256, 0, s>>>(
       prob, labels, dx, rows, C, w);
 }
-void ps_softmax_loss_bwd_bf16(const void* prob, const void* labels, void* dx,
+void ps_softmax_loss_bwd_bf16(const void* prob, const float* labels, void* dx,
                               int64_t rows, int C, float w, hipStream_t s) {
-  softmax_loss_bwd_k<__bf16, __bf16><<<ew_grid(rows * C), 256, 0, s>>>(
-      (const __bf16*)prob, (const __bf16*)labels, (__bf16*)dx, rows, C, w);
+  softmax_loss_bwd_k<__bf16, float><<<ew_grid(rows * C), 256, 0, s>>>(
+      (const __bf16*)prob, labels, (__bf16*)dx, rows, C, w);
 }
 
 }  // extern "C"
