@@ -86,7 +86,9 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   // Split-K when the output tile grid cannot fill 256 CUs but K is deep
   // (conv wgrad: M=Cout, N=Kcol, K=N*OH*OW up to ~800k): target ~512
   // workgroups, cap the f32 workspace at 256 MB.
-  int64_t tiles = (int64_t)((M + 127) / 128) * ((N + 127) / 128);
+  int tbm, tbn;
+  ps_pick_gemm_tile(M, N, &tbm, &tbn);
+  int64_t tiles = (int64_t)((M + tbm - 1) / tbm) * ((N + tbn - 1) / tbn);
   Tensor ws;  // keep alive until launch returns
   if (tiles < 96 && K >= 4096) {
     int sk = (int)std::min<int64_t>(512 / tiles, (K + 2047) / 2048);

@@ -110,22 +110,36 @@ __global__ void dropout_bwd_k(const T* dy, const uint8_t* mask, T* dx,
 
 // column-sum: out[c] += sum_r in[r][c] for in[R][C] row-major -- the bias
 // gradient for both linear (dy[M][N]) and NHWC conv (dy[(N*OH*OW)][C]).
-// Two-level: each block owns 256 columns x a slab of rows, accumulates in
-// registers, then one atomicAdd per column (device-scope atomics are cheap
-// and contention is gridDim.y-way -- guide §6 Guideline 12).
+// Each block owns a row slab; within the block, 64-lane groups sweep 64
+// consecutive columns (coalesced) with 4 row-phases, LDS-reduce the 4
+// partials per column, then one atomicAdd per (block, column)
+// (guide §6 Guideline 12: partial-reduce first, few atomics).
 template <typename T>
-__global__ void colsum_k(const T* in, float* out, int64_t R, int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  int64_t rows_per = (R + gridDim.y - 1) / gridDim.y;
-  int64_t r0 = blockIdx.y * rows_per;
+__global__ void colsum_k(const T* __restrict__ in, float* __restrict__ out,
+                         int64_t R, int C) {
+  __shared__ float part[4][64];
+  const int lane = threadIdx.x & 63;
+  const int rg = threadIdx.x >> 6;  // 4 row groups
+  int64_t rows_per = (R + gridDim.x - 1) / gridDim.x;
+  int64_t r0 = (int64_t)blockIdx.x * rows_per;
   int64_t r1 = min(R, r0 + rows_per);
-  float acc = 0.f;
-  for (int64_t r = r0; r < r1; ++r) acc += to_f32(in[r * C + c]);
-  if (gridDim.y == 1)
-    out[c] += acc;
-  else
-    atomicAdd(&out[c], acc);
+  for (int c0 = 0; c0 < C; c0 += 64) {
+    int c = c0 + lane;
+    float acc = 0.f;
+    if (c < C)
+      for (int64_t r = r0 + rg; r < r1; r += 4)
+        acc += to_f32(in[r * C + c]);
+    part[rg][lane] = acc;
+    __syncthreads();
+    if (rg == 0 && c < C) {
+      float v = part[0][lane] + part[1][lane] + part[2][lane] + part[3][lane];
+      if (gridDim.x == 1)
+        out[c] += v;
+      else
+        atomicAdd(&out[c], v);
+    }
+    __syncthreads();
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -201,11 +215,11 @@ void ps_dropout_bwd_bf16(const void* dy, const uint8_t* mask, void* dx,
 }
 
 static inline dim3 colsum_grid(int64_t R, int C) {
-  int xb = cdiv(C, 256);
-  // enough row slabs to fill the chip (256 CUs want >> 256 workgroups)
-  int yb = 1;
-  while ((int64_t)xb * yb < 1024 && (R / yb) > 1024) yb *= 2;
-  return dim3(xb, yb);
+  // row slabs: fill the chip but keep >= ~64 rows per block
+  int64_t blocks = R / 64;
+  if (blocks < 1) blocks = 1;
+  if (blocks > 1024) blocks = 1024;
+  return dim3((unsigned)blocks);
 }
 void ps_colsum_f32(const float* in, float* out, int64_t R, int C, hipStream_t s) {
   colsum_k<float><<<colsum_grid(R, C), 256, 0, s>>>(in, out, R, C);

@@ -50,6 +50,51 @@ __global__ void im2col_nhwc_k(const T* __restrict__ x, T* __restrict__ colT,
   }
 }
 
+// Row-run variant for G==1: in NHWC, for a fixed kh the whole (kw, c) span
+// of a patch row is CONTIGUOUS in x (kw*C + c walks iw*C + c). One thread
+// per (np, kh) copies kw*C elems with a 16B-vector loop -- the fast path
+// for small-C first layers (conv1: C=3, kw*C=33) where per-channel chunks
+// cannot vectorize.
+template <typename T, int VBYTES>
+__global__ void im2col_nhwc_rowrun_k(const T* __restrict__ x,
+                                     T* __restrict__ colT, ConvGeom g) {
+  constexpr int V = VBYTES / sizeof(T);
+  // element-aligned only: run starts (np*Kcol, kh*RUN) are not 16B-aligned
+  // for odd K (conv1 Kcol=363); clang emits the widest legal loads
+  typedef T vec_t __attribute__((ext_vector_type(V), aligned(sizeof(T))));
+  const int RUN = g.kw * g.C;  // elems per (np, kh)
+  const int Kcol = g.kh * RUN;
+  const int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
+  int64_t total = NP * g.kh;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int kkh = i % g.kh;
+    int64_t np = i / g.kh;
+    int ow = np % g.Wo;
+    int64_t t2 = np / g.Wo;
+    int oh = t2 % g.Ho;
+    int n = t2 / g.Ho;
+    int ih = oh * g.sh - g.ph + kkh;
+    int iw0 = ow * g.sw - g.pw;
+    T* dst = colT + np * Kcol + (int64_t)kkh * RUN;
+    if (ih < 0 || ih >= g.H) {
+      for (int e = 0; e < RUN; ++e) dst[e] = (T)0.0f;
+      continue;
+    }
+    const T* src = x + (((int64_t)n * g.H + ih) * g.W + iw0) * g.C;
+    // valid elem range within the run: iw in [0, W)
+    int e_lo = iw0 < 0 ? -iw0 * g.C : 0;
+    int e_hi = min(RUN, (g.W - iw0) * g.C);
+    int e = 0;
+    for (; e < e_lo; ++e) dst[e] = (T)0.0f;
+    for (; e + V <= e_hi; e += V)
+      *reinterpret_cast<vec_t*>(&dst[e]) =
+          *reinterpret_cast<const vec_t*>(&src[e]);
+    for (; e < e_hi; ++e) dst[e] = src[e];
+    for (; e < RUN; ++e) dst[e] = (T)0.0f;
+  }
+}
+
 // gather: one thread per (n, ih, iw, ci)
 template <typename T>
 __global__ void col2im_nhwc_k(const T* __restrict__ colT, T* __restrict__ dx,
@@ -146,6 +191,10 @@ void ps_im2col_nhwc_f32(const float* x, float* colT, const ConvGeom* g, hipStrea
   int64_t base = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
   if (Cg % 4 == 0)
    hipLaunchKernelGGL(( im2col_nhwc_k<float, 4>), dim3(ew_grid(base * (Cg / 4))), dim3(256), 0, s, x, colT, *g);
+  else if (g->G == 1)
+   hipLaunchKernelGGL(( im2col_nhwc_rowrun_k<float, 16>)
+        , dim3(ew_grid((int64_t)g->N * g->Ho * g->Wo * g->kh)), dim3(256), 0, s, 
+            x, colT, *g);
   else
    hipLaunchKernelGGL(( im2col_nhwc_k<float, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, x, colT, *g);
 }
@@ -155,6 +204,10 @@ void ps_im2col_nhwc_bf16(const void* x, void* colT, const ConvGeom* g, hipStream
   if (Cg % 8 == 0)
    hipLaunchKernelGGL(( im2col_nhwc_k<__bf16, 8>), dim3(ew_grid(base * (Cg / 8))), dim3(256), 0, s, 
         (const __bf16*)x, (__bf16*)colT, *g);
+  else if (g->G == 1)
+   hipLaunchKernelGGL(( im2col_nhwc_rowrun_k<__bf16, 16>)
+        , dim3(ew_grid((int64_t)g->N * g->Ho * g->Wo * g->kh)), dim3(256), 0, s, 
+            (const __bf16*)x, (__bf16*)colT, *g);
   else
    hipLaunchKernelGGL(( im2col_nhwc_k<__bf16, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, 
         (const __bf16*)x, (__bf16*)colT, *g);

@@ -32,6 +32,29 @@ struct ConvGeom {
   int G;
 };
 
+// Tile selection shared by the launcher (grid) and bindings (split-K
+// workspace sizing). Prefers the tile with least padding waste; ties go to
+// the larger tile (more LDS reuse per byte staged).
+inline void ps_pick_gemm_tile(int M, int N, int* bm_out, int* bn_out) {
+  const int cand[4][2] = {{128, 128}, {128, 32}, {32, 128}, {64, 64}};
+  double best_util = -1.0;
+  int64_t best_area = 0;
+  for (int i = 0; i < 4; ++i) {
+    int bm = cand[i][0], bn = cand[i][1];
+    int64_t padded = (int64_t)((M + bm - 1) / bm) * bm
+                     * (int64_t)((N + bn - 1) / bn) * bn;
+    double util = (double)M * N / (double)padded;
+    int64_t area = (int64_t)bm * bn;
+    if (util > best_util * 1.05 ||
+        (util > best_util * 0.999 && area > best_area)) {
+      best_util = util;
+      best_area = area;
+      *bm_out = bm;
+      *bn_out = bn;
+    }
+  }
+}
+
 extern "C" {
 // gemm.hip
 void ps_gemm_f32(const GemmArgs* g, hipStream_t s);
