@@ -33,22 +33,22 @@ struct ConvGeom {
 };
 
 // Tile selection shared by the launcher (grid) and bindings (split-K
-// workspace sizing). Prefers the tile with least padding waste; ties go to
-// the larger tile (more LDS reuse per byte staged).
+// workspace sizing). Cost model: MFMA time on the PADDED output plus LDS
+// staging traffic (each A panel is re-staged once per N-tile and vice
+// versa). ALPHA ~= MACs the matrix cores retire per element the memory
+// system delivers (bf16: ~1.25e15 MAC/s vs ~3.2e12 elem/s -> ~400).
 inline void ps_pick_gemm_tile(int M, int N, int* bm_out, int* bn_out) {
   const int cand[4][2] = {{128, 128}, {128, 32}, {32, 128}, {64, 64}};
-  double best_util = -1.0;
-  int64_t best_area = 0;
+  const double ALPHA = 400.0;
+  double best = 0;
   for (int i = 0; i < 4; ++i) {
     int bm = cand[i][0], bn = cand[i][1];
-    int64_t padded = (int64_t)((M + bm - 1) / bm) * bm
-                     * (int64_t)((N + bn - 1) / bn) * bn;
-    double util = (double)M * N / (double)padded;
-    int64_t area = (int64_t)bm * bn;
-    if (util > best_util * 1.05 ||
-        (util > best_util * 0.999 && area > best_area)) {
-      best_util = util;
-      best_area = area;
+    int64_t tm = (M + bm - 1) / bm, tn = (N + bn - 1) / bn;
+    double flops = (double)(tm * bm) * (tn * bn) / ALPHA;  // (x K, common)
+    double staging = (double)tm * bm * tn + (double)tn * bn * tm;
+    double cost = flops + staging;
+    if (i == 0 || cost < best) {
+      best = cost;
       *bm_out = bm;
       *bn_out = bn;
     }
