@@ -60,56 +60,80 @@ template <> struct GemmTraits<__bf16> {
   }
 };
 
-// Stage a [ROWS x BK] K-contiguous tile into LDS from a K-last source
-// (src[row][k], leading dim lda). Guarded, zero-filled out of range.
-template <typename T, int ROWS>
-__device__ inline void stage_klast(T* lds, const T* src, int64_t lda,
-                                   int row0, int rows_max, int k0, int K,
-                                   int tid) {
+// Staging, split into {issue global loads -> regs} and {write regs -> LDS}
+// halves so the next tile's HBM latency hides under the current tile's MFMA
+// phase (guide T14 / Guideline 15: write AFTER the barrier, re-issue at
+// once). A [ROWS x BK] tile is ROWS*BK/(256*VEC) vectors per thread.
+// Guarded, zero-filled out of range.
+template <typename T, int ROWS, bool KLAST>
+struct Stager {
   using TR = GemmTraits<T>;
-  constexpr int CK = TR::BK / TR::VEC;
   using vec_t = typename TR::vec_t;
-  for (int c = tid; c < ROWS * CK; c += 256) {
-    int r = c / CK, kc = c % CK;
-    int gr = row0 + r, gk = k0 + kc * TR::VEC;
-    vec_t v = {};
-    if (gr < rows_max && gk < K) {
-      if (gk + TR::VEC <= K) {
-        v = *reinterpret_cast<const vec_t*>(&src[(int64_t)gr * lda + gk]);
-      } else {
-        for (int j = 0; j < TR::VEC; ++j)
-          if (gk + j < K) v[j] = src[(int64_t)gr * lda + gk + j];
-      }
-    }
-    *reinterpret_cast<vec_t*>(&lds[r * TR::RS + kc * TR::VEC]) = v;
-  }
-}
+  static constexpr int NV = ROWS * TR::BK / (256 * TR::VEC);
+  vec_t v[NV];
 
-// Stage the same tile from a K-major source (src[k][m], leading dim lda):
-// vector loads along m, scalar scatter into the K-contiguous LDS image.
-template <typename T, int ROWS>
-__device__ inline void stage_kmajor(T* lds, const T* src, int64_t lda,
-                                    int row0, int rows_max, int k0, int K,
-                                    int tid) {
-  using TR = GemmTraits<T>;
-  constexpr int CM = ROWS / TR::VEC;
-  using vec_t = typename TR::vec_t;
-  for (int c = tid; c < TR::BK * CM; c += 256) {
-    int kk = c / CM, mc = c % CM;
-    int gk = k0 + kk, gm = row0 + mc * TR::VEC;
-    vec_t v = {};
-    if (gk < K && gm < rows_max) {
-      if (gm + TR::VEC <= rows_max) {
-        v = *reinterpret_cast<const vec_t*>(&src[(int64_t)gk * lda + gm]);
-      } else {
-        for (int j = 0; j < TR::VEC; ++j)
-          if (gm + j < rows_max) v[j] = src[(int64_t)gk * lda + gm + j];
+  __device__ inline void load(const T* src, int64_t lda, int row0,
+                              int rows_max, int k0, int K, int tid) {
+    if (KLAST) {
+      constexpr int CK = TR::BK / TR::VEC;
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        int c = tid + i * 256;
+        int r = c / CK, kc = c % CK;
+        int gr = row0 + r, gk = k0 + kc * TR::VEC;
+        vec_t val = {};
+        if (gr < rows_max && gk < K) {
+          if (gk + TR::VEC <= K) {
+            val = *reinterpret_cast<const vec_t*>(&src[(int64_t)gr * lda + gk]);
+          } else {
+            for (int j = 0; j < TR::VEC; ++j)
+              if (gk + j < K) val[j] = src[(int64_t)gr * lda + gk + j];
+          }
+        }
+        v[i] = val;
+      }
+    } else {
+      constexpr int CM = ROWS / TR::VEC;
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        int c = tid + i * 256;
+        int kk = c / CM, mc = c % CM;
+        int gk = k0 + kk, gm = row0 + mc * TR::VEC;
+        vec_t val = {};
+        if (gk < K && gm < rows_max) {
+          if (gm + TR::VEC <= rows_max) {
+            val = *reinterpret_cast<const vec_t*>(&src[(int64_t)gk * lda + gm]);
+          } else {
+            for (int j = 0; j < TR::VEC; ++j)
+              if (gm + j < rows_max) val[j] = src[(int64_t)gk * lda + gm + j];
+          }
+        }
+        v[i] = val;
       }
     }
-    for (int j = 0; j < TR::VEC; ++j)
-      lds[(mc * TR::VEC + j) * TR::RS + kk] = v[j];
   }
-}
+
+  __device__ inline void write(T* lds, int tid) {
+    if (KLAST) {
+      constexpr int CK = TR::BK / TR::VEC;
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        int c = tid + i * 256;
+        int r = c / CK, kc = c % CK;
+        *reinterpret_cast<vec_t*>(&lds[r * TR::RS + kc * TR::VEC]) = v[i];
+      }
+    } else {
+      constexpr int CM = ROWS / TR::VEC;
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        int c = tid + i * 256;
+        int kk = c / CM, mc = c % CM;
+        for (int j = 0; j < TR::VEC; ++j)
+          lds[(mc * TR::VEC + j) * TR::RS + kk] = v[i][j];
+      }
+    }
+  }
+};
 
 // Tile geometry: BM x BN block tile, 4 waves arranged WGM x WGN, each wave
 // owns a (BM/WGM) x (BN/WGN) sub-tile as FM x FN fragments of 16x16.
@@ -138,8 +162,8 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
     k_end = min(K, k_begin + kchunk);
   }
 
-  __shared__ __attribute__((aligned(16))) T a_lds[BM * RS];
-  __shared__ __attribute__((aligned(16))) T b_lds[BN * RS];
+  __shared__ __attribute__((aligned(16))) T a_lds[2][BM * RS];
+  __shared__ __attribute__((aligned(16))) T b_lds[2][BN * RS];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -151,26 +175,41 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
 
   f32x4 acc[FM][FN] = {};
 
+  Stager<T, BM, A_KLAST> sa;
+  Stager<T, BN, B_KLAST> sb;
+  // prologue: tile 0 -> LDS[0]; issue tile 1 loads
+  sa.load(A, lda, m0, M, k_begin, k_end, tid);
+  sb.load(B, ldb, n0, N, k_begin, k_end, tid);
+  sa.write(a_lds[0], tid);
+  sb.write(b_lds[0], tid);
+  if (k_begin + BK < k_end) {
+    sa.load(A, lda, m0, M, k_begin + BK, k_end, tid);
+    sb.load(B, ldb, n0, N, k_begin + BK, k_end, tid);
+  }
+  __syncthreads();
+
+  int cur = 0;
   for (int k0 = k_begin; k0 < k_end; k0 += BK) {
-    if (A_KLAST)
-      stage_klast<T, BM>(a_lds, A, lda, m0, M, k0, k_end, tid);
-    else
-      stage_kmajor<T, BM>(a_lds, A, lda, m0, M, k0, k_end, tid);
-    if (B_KLAST)
-      stage_klast<T, BN>(b_lds, B, ldb, n0, N, k0, k_end, tid);
-    else
-      stage_kmajor<T, BN>(b_lds, B, ldb, n0, N, k0, k_end, tid);
-    __syncthreads();
+    // regs hold tile t+1: write it into the other LDS buffer, then issue
+    // tile t+2's loads so they fly during this tile's MFMA phase
+    if (k0 + BK < k_end) {
+      sa.write(a_lds[cur ^ 1], tid);
+      sb.write(b_lds[cur ^ 1], tid);
+      if (k0 + 2 * BK < k_end) {
+        sa.load(A, lda, m0, M, k0 + 2 * BK, k_end, tid);
+        sb.load(B, ldb, n0, N, k0 + 2 * BK, k_end, tid);
+      }
+    }
 
 #pragma unroll
     for (int kk = 0; kk < BK; kk += TR::KSTEP) {
       typename TR::frag_t a_frag[FM], b_frag[FN];
 #pragma unroll
       for (int f = 0; f < FM; ++f)
-        a_frag[f] = TR::load_frag(&a_lds[(wm + f * 16 + (lane & 15)) * RS], kk, lane);
+        a_frag[f] = TR::load_frag(&a_lds[cur][(wm + f * 16 + (lane & 15)) * RS], kk, lane);
 #pragma unroll
       for (int f = 0; f < FN; ++f)
-        b_frag[f] = TR::load_frag(&b_lds[(wn + f * 16 + (lane & 15)) * RS], kk, lane);
+        b_frag[f] = TR::load_frag(&b_lds[cur][(wn + f * 16 + (lane & 15)) * RS], kk, lane);
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
@@ -178,6 +217,7 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
           acc[fm][fn] = TR::mfma(a_frag[fm], b_frag[fn], acc[fm][fn]);
     }
     __syncthreads();
+    cur ^= 1;
   }
 
   // Epilogue: C/D fragment map for 16x16 shapes: col = lane&15,
