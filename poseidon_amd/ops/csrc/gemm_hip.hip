@@ -69,7 +69,8 @@ template <typename T, int ROWS, bool KLAST>
 struct Stager {
   using TR = GemmTraits<T>;
   using vec_t = typename TR::vec_t;
-  static constexpr int NV = ROWS * TR::BK / (256 * TR::VEC);
+  static constexpr int TV = ROWS * TR::BK / TR::VEC;  // total vectors in tile
+  static constexpr int NV = (TV + 255) / 256;         // per-thread (>=1)
   vec_t v[NV];
 
   __device__ inline void load(const T* src, int64_t lda, int row0,
@@ -79,6 +80,7 @@ struct Stager {
 #pragma unroll
       for (int i = 0; i < NV; ++i) {
         int c = tid + i * 256;
+        if (c >= TV) break;
         int r = c / CK, kc = c % CK;
         int gr = row0 + r, gk = k0 + kc * TR::VEC;
         vec_t val = {};
@@ -97,6 +99,7 @@ struct Stager {
 #pragma unroll
       for (int i = 0; i < NV; ++i) {
         int c = tid + i * 256;
+        if (c >= TV) break;
         int kk = c / CM, mc = c % CM;
         int gk = k0 + kk, gm = row0 + mc * TR::VEC;
         vec_t val = {};
@@ -119,6 +122,7 @@ struct Stager {
 #pragma unroll
       for (int i = 0; i < NV; ++i) {
         int c = tid + i * 256;
+        if (c >= TV) break;
         int r = c / CK, kc = c % CK;
         *reinterpret_cast<vec_t*>(&lds[r * TR::RS + kc * TR::VEC]) = v[i];
       }
@@ -127,6 +131,7 @@ struct Stager {
 #pragma unroll
       for (int i = 0; i < NV; ++i) {
         int c = tid + i * 256;
+        if (c >= TV) break;
         int kk = c / CM, mc = c % CM;
         for (int j = 0; j < TR::VEC; ++j)
           lds[(mc * TR::VEC + j) * TR::RS + kk] = v[i][j];
