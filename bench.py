@@ -35,6 +35,8 @@ def main():
                     choices=list(DEFAULT_BATCH))
     ap.add_argument("--batch", type=int, default=0, help="per-GPU batch")
     ap.add_argument("--no-sfb", action="store_true")
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph capture of the iteration")
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"],
                     help="compute dtype (activations + GEMM inputs); "
                          "master weights and gradient accumulation stay fp32")
@@ -63,6 +65,9 @@ def main():
                  display=0, snapshot=0)
     sp.net_param = zoo.build_net(args.model, batch=batch)
     solver = SGDSolver(sp, use_sfb=not args.no_sfb, verbose=False)
+    graphed = False
+    if not args.no_graph and device == "cuda":
+        graphed = solver.enable_graph()
 
     def sync():
         if device == "cuda":
@@ -105,6 +110,7 @@ def main():
                 "input": "3x227x227" if args.model == "alexnet" else "3x224x224",
                 "parallelism": f"dp{n_gpus}",
                 "sfb": not args.no_sfb and n_gpus > 1,
+                "hipgraph": graphed,
             },
         }
         print(json.dumps(out), flush=True)

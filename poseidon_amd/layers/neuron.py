@@ -81,13 +81,22 @@ class DropoutLayer(NeuronLayer):
         assert 0.0 <= self.ratio < 1.0
         self.seed = ctx().seed * 7919 + (zlib.crc32(self.name.encode()) & 0xFFFF)
         self.offset = 0
+        self._offset_dev = None
         self._mask = None
 
     def forward(self, bottom, top) -> None:
         if self.phase == 0:  # TRAIN
-            y, mask = ops.dropout_forward(bottom[0].data, self.ratio,
-                                          self.seed, self.offset)
-            self.offset += 1
+            x = bottom[0].data
+            if x.is_cuda:
+                if self._offset_dev is None:
+                    self._offset_dev = torch.zeros(1, dtype=torch.int64,
+                                                   device=x.device)
+                y, mask = ops.dropout_forward(x, self.ratio, self.seed, 0,
+                                              offset_dev=self._offset_dev)
+            else:
+                y, mask = ops.dropout_forward(x, self.ratio, self.seed,
+                                              self.offset)
+                self.offset += 1
             self._mask = mask
             top[0].data = y
         else:

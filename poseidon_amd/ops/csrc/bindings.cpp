@@ -692,6 +692,28 @@ std::vector<Tensor> dropout_forward(const Tensor& x, double ratio,
   return {y, mask};
 }
 
+// graph-replayable dropout: philox offset lives in device memory and is
+// incremented in-stream, so each replay draws a fresh mask
+std::vector<Tensor> dropout_forward_offdev(const Tensor& x, double ratio,
+                                           int64_t seed, Tensor offset_dev) {
+  check_float_like(x, "x");
+  auto xc = any_contig(x);
+  Tensor y = at::empty_like(xc);
+  Tensor mask = at::empty(xc.sizes(), xc.options().dtype(at::kByte));
+  if (is_bf16(x))
+    ps_dropout_fwd_bf16_offdev(xc.data_ptr(), y.data_ptr(),
+                               mask.data_ptr<uint8_t>(), xc.numel(),
+                               (float)ratio, (uint64_t)seed,
+                               offset_dev.data_ptr(), stream());
+  else
+    ps_dropout_fwd_f32_offdev(xc.data_ptr<float>(), y.data_ptr<float>(),
+                              mask.data_ptr<uint8_t>(), xc.numel(),
+                              (float)ratio, (uint64_t)seed,
+                              offset_dev.data_ptr(), stream());
+  ps_u64_inc(offset_dev.data_ptr(), stream());
+  return {y, mask};
+}
+
 Tensor dropout_backward(const Tensor& dy, const Tensor& mask, double ratio) {
   auto dyc = any_contig(dy);
   Tensor dx = at::empty_like(dyc);
@@ -713,6 +735,14 @@ void sgd_update(Tensor w, const Tensor& g, Tensor h, double lr, double mom,
                 double wd) {
   ps_sgd_update(w.data_ptr<float>(), g.data_ptr<float>(), h.data_ptr<float>(),
                 w.numel(), (float)lr, (float)mom, (float)wd, stream());
+}
+// graph-replayable: lr read from lr_dev[0] * lr_mult at kernel time
+void sgd_update_lrdev(Tensor w, const Tensor& g, Tensor h, double lr_mult,
+                      double mom, double wd, const Tensor& lr_dev) {
+  ps_sgd_update_lrdev(w.data_ptr<float>(), g.data_ptr<float>(),
+                      h.data_ptr<float>(), w.numel(), (float)lr_mult,
+                      (float)mom, (float)wd, lr_dev.data_ptr<float>(),
+                      stream());
 }
 void nesterov_update(Tensor w, const Tensor& g, Tensor h, double lr,
                      double mom, double wd) {
@@ -760,6 +790,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_forward", &dropout_forward);
   m.def("dropout_backward", &dropout_backward);
   m.def("sgd_update", &sgd_update);
+  m.def("sgd_update_lrdev", &sgd_update_lrdev);
+  m.def("dropout_forward_offdev", &dropout_forward_offdev);
   m.def("nesterov_update", &nesterov_update);
   m.def("adagrad_update", &adagrad_update);
   m.attr("compute_dtypes") = std::vector<std::string>{"float32", "bfloat16"};

@@ -81,10 +81,13 @@ __global__ void bnll_bwd_k(const T* x, const T* dy, T* dx, int64_t n) {
   }
 }
 
-template <typename T>
+// OFFDEV reads the philox offset from device memory so a hipGraph replay
+// draws fresh masks each iteration (the host-value form freezes under replay)
+template <typename T, bool OFFDEV>
 __global__ void dropout_fwd_k(const T* x, T* y, uint8_t* mask, int64_t n,
                               float ratio, float scale, uint64_t seed,
-                              uint64_t offset) {
+                              uint64_t offset, const unsigned long long* off_dev) {
+  if (OFFDEV) offset = off_dev[0];
   for (int64_t i4 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i4 * 4 < n; i4 += (int64_t)gridDim.x * blockDim.x) {
     uint4 r = philox4(seed, offset, (uint32_t)i4);
@@ -196,14 +199,28 @@ void ps_relu_bwd_bf16(const void* x, const void* dy, void* dx, int64_t n,
 
 void ps_dropout_fwd_f32(const float* x, float* y, uint8_t* mask, int64_t n,
                         float ratio, uint64_t seed, uint64_t offset, hipStream_t s) {
- hipLaunchKernelGGL(( dropout_fwd_k<float>), dim3(ew_grid((n + 3) / 4)), dim3(256), 0, s, 
-      x, y, mask, n, ratio, 1.0f / (1.0f - ratio), seed, offset);
+ hipLaunchKernelGGL(( dropout_fwd_k<float, false>), dim3(ew_grid((n + 3) / 4)), dim3(256), 0, s, 
+      x, y, mask, n, ratio, 1.0f / (1.0f - ratio), seed, offset, nullptr);
 }
 void ps_dropout_fwd_bf16(const void* x, void* y, uint8_t* mask, int64_t n,
                          float ratio, uint64_t seed, uint64_t offset, hipStream_t s) {
- hipLaunchKernelGGL(( dropout_fwd_k<__bf16>), dim3(ew_grid((n + 3) / 4)), dim3(256), 0, s, 
+ hipLaunchKernelGGL(( dropout_fwd_k<__bf16, false>), dim3(ew_grid((n + 3) / 4)), dim3(256), 0, s, 
       (const __bf16*)x, (__bf16*)y, mask, n, ratio, 1.0f / (1.0f - ratio),
-      seed, offset);
+      seed, offset, nullptr);
+}
+void ps_dropout_fwd_f32_offdev(const float* x, float* y, uint8_t* mask,
+                               int64_t n, float ratio, uint64_t seed,
+                               const void* off_dev, hipStream_t s) {
+ hipLaunchKernelGGL(( dropout_fwd_k<float, true>), dim3(ew_grid((n + 3) / 4)), dim3(256), 0, s, 
+      x, y, mask, n, ratio, 1.0f / (1.0f - ratio), seed, 0,
+      (const unsigned long long*)off_dev);
+}
+void ps_dropout_fwd_bf16_offdev(const void* x, void* y, uint8_t* mask,
+                                int64_t n, float ratio, uint64_t seed,
+                                const void* off_dev, hipStream_t s) {
+ hipLaunchKernelGGL(( dropout_fwd_k<__bf16, true>), dim3(ew_grid((n + 3) / 4)), dim3(256), 0, s, 
+      (const __bf16*)x, (__bf16*)y, mask, n, ratio, 1.0f / (1.0f - ratio),
+      seed, 0, (const unsigned long long*)off_dev);
 }
 void ps_dropout_bwd_f32(const float* dy, const uint8_t* mask, float* dx,
                         int64_t n, float ratio, hipStream_t s) {

@@ -136,6 +136,13 @@ void ps_nesterov_update(float*, const float*, float*, int64_t, float, float,
 void ps_adagrad_update(float*, const float*, float*, int64_t, float, float,
                        float, hipStream_t);
 void ps_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
+void ps_sgd_update_lrdev(float*, const float*, float*, int64_t, float, float,
+                         float, const float*, hipStream_t);
+void ps_u64_inc(void*, hipStream_t);
+void ps_dropout_fwd_f32_offdev(const float*, float*, uint8_t*, int64_t, float,
+                               uint64_t, const void*, hipStream_t);
+void ps_dropout_fwd_bf16_offdev(const void*, void*, uint8_t*, int64_t, float,
+                                uint64_t, const void*, hipStream_t);
 
 // im2col.hip
 void ps_im2col_nhwc_f32(const float*, float*, const ConvGeom*, hipStream_t);

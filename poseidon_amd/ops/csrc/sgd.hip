@@ -7,10 +7,15 @@
 
 namespace ps {
 
-// hist = mom*hist + lr*(g + wd*w); w -= hist
+__global__ void u64_inc_k(unsigned long long* p) { *p += 1; }
+
+// hist = mom*hist + lr*(g + wd*w); w -= hist. LRDEV reads lr from device
+// memory so a hipGraph replay can see per-iteration learning rates.
+template <bool LRDEV>
 __global__ void sgd_update_k(float* __restrict__ w, const float* __restrict__ g,
-                             float* __restrict__ h, int64_t n, float lr,
-                             float mom, float wd) {
+                             float* __restrict__ h, int64_t n, float lr_or_mult,
+                             float mom, float wd, const float* __restrict__ lr_dev) {
+  const float lr = LRDEV ? lr_dev[0] * lr_or_mult : lr_or_mult;
   int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
   for (; i + 4 <= n; i += stride) {
@@ -27,9 +32,11 @@ __global__ void sgd_update_k(float* __restrict__ w, const float* __restrict__ g,
   }
 }
 
+template <bool LRDEV>
 __global__ void sgd_update_tail_k(float* w, const float* g, float* h,
-                                  int64_t start, int64_t n, float lr,
-                                  float mom, float wd) {
+                                  int64_t start, int64_t n, float lr_or_mult,
+                                  float mom, float wd, const float* lr_dev) {
+  const float lr = LRDEV ? lr_dev[0] * lr_or_mult : lr_or_mult;
   int64_t i = start + threadIdx.x;
   if (i < n) {
     float hv = mom * h[i] + lr * (g[i] + wd * w[i]);
@@ -75,9 +82,28 @@ void ps_sgd_update(float* w, const float* g, float* h, int64_t n, float lr,
                    float mom, float wd, hipStream_t s) {
   int64_t nv = n / 4;
   if (nv > 0)
-    sgd_update_k<<<ew_grid(nv), 256, 0, s>>>(w, g, h, nv * 4, lr, mom, wd);
+    sgd_update_k<false><<<ew_grid(nv), 256, 0, s>>>(w, g, h, nv * 4, lr, mom,
+                                                    wd, nullptr);
   if (n % 4)
-    sgd_update_tail_k<<<1, 4, 0, s>>>(w, g, h, n & ~3LL, n, lr, mom, wd);
+    sgd_update_tail_k<false><<<1, 4, 0, s>>>(w, g, h, n & ~3LL, n, lr, mom,
+                                             wd, nullptr);
+}
+
+// lr = lr_dev[0] * lr_mult (lr_mult folds the per-param blobs_lr)
+void ps_sgd_update_lrdev(float* w, const float* g, float* h, int64_t n,
+                         float lr_mult, float mom, float wd,
+                         const float* lr_dev, hipStream_t s) {
+  int64_t nv = n / 4;
+  if (nv > 0)
+    sgd_update_k<true><<<ew_grid(nv), 256, 0, s>>>(w, g, h, nv * 4, lr_mult,
+                                                   mom, wd, lr_dev);
+  if (n % 4)
+    sgd_update_tail_k<true><<<1, 4, 0, s>>>(w, g, h, n & ~3LL, n, lr_mult,
+                                            mom, wd, lr_dev);
+}
+
+void ps_u64_inc(void* p, hipStream_t s) {
+  u64_inc_k<<<1, 1, 0, s>>>((unsigned long long*)p);
 }
 
 void ps_nesterov_update(float* w, const float* g, float* h, int64_t n,

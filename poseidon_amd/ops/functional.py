@@ -386,10 +386,13 @@ def bnll_backward(x, dy):
     return dy * torch.sigmoid(x)
 
 
-def dropout_forward(x, ratio: float, seed: int, offset: int):
+def dropout_forward(x, ratio: float, seed: int, offset, offset_dev=None):
     """Train-mode dropout. Returns (y, mask) where mask is uint8 keep-mask;
-    y = x * mask * 1/(1-ratio)."""
+    y = x * mask * 1/(1-ratio). On GPU, offset_dev (int64 device scalar)
+    makes the op hipGraph-replayable (fresh mask per replay)."""
     if x.is_cuda:
+        if offset_dev is not None:
+            return _ext().dropout_forward_offdev(x, ratio, seed, offset_dev)
         return _ext().dropout_forward(x, ratio, seed, offset)
     scale = 1.0 / (1.0 - ratio)
     g = torch.Generator(device="cpu").manual_seed(seed + offset)
@@ -410,9 +413,14 @@ def dropout_backward(dy, mask, ratio: float):
 # ---------------------------------------------------------------------------
 
 def sgd_update(w: torch.Tensor, grad: torch.Tensor, hist: torch.Tensor,
-               local_rate: float, momentum: float, decay: float) -> None:
+               local_rate: float, momentum: float, decay: float,
+               lr_dev=None) -> None:
     if w.is_cuda:
-        _ext().sgd_update(w, grad, hist, local_rate, momentum, decay)
+        if lr_dev is not None:
+            _ext().sgd_update_lrdev(w, grad, hist, local_rate, momentum,
+                                    decay, lr_dev)
+        else:
+            _ext().sgd_update(w, grad, hist, local_rate, momentum, decay)
         return
     gw = grad if decay == 0.0 else grad + decay * w
     hist.mul_(momentum).add_(gw, alpha=local_rate)

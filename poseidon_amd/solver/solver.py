@@ -92,6 +92,80 @@ class SGDSolver:
         self._net_outputs_rows: List[List[float]] = []
         self._t0 = time.time()
 
+        # hipGraph capture of the steady-state iteration (HIP streams and
+        # graphs instead of a tracing compiler): opt-in via enable_graph().
+        self._use_graph = False
+        self._graph = None
+        self._graph_loss = None
+        self._lr_dev: Optional[torch.Tensor] = None
+
+    # ------------------------------------------------------------------
+    # hipGraph iteration capture
+    # ------------------------------------------------------------------
+    def enable_graph(self) -> bool:
+        """Capture fwd+bwd+update as one hipGraph and replay per iteration
+        (collapses per-kernel launch gaps -- GoogLeNet's hundreds of small
+        kernels). Requires: GPU, SGD solver, static data layers (constant
+        DummyData), L2 regularization. Returns True if enabled."""
+        c = ctx()
+        if c.device != "cuda" or type(self) is not SGDSolver:
+            return False
+        if self.param.regularization_type == "L1":
+            return False
+        for layer in self.net.layers:
+            t = layer.type_name
+            if t in ("DATA", "IMAGE_DATA", "MEMORY_DATA", "WINDOW_DATA"):
+                return False  # per-iter host work cannot replay
+            if t == "DUMMY_DATA" and any(layer.refill):
+                layer.forward(self.net.bottoms[self.net.layers.index(layer)],
+                              self.net.tops[self.net.layers.index(layer)])
+                layer.refill = [False] * len(layer.refill)  # freeze fills
+        self._use_graph = True
+        return True
+
+    def _graph_body(self) -> torch.Tensor:
+        loss = self.forward_backward()
+        for i, ps in enumerate(self.net.params):
+            if ps.owner == i and ps.lr_mult != 0.0:
+                wd = float(self.param.weight_decay or 0.0) * ps.decay_mult
+                if self.distributed:
+                    wd *= ctx().world_size
+                ops.sgd_update(ps.blob.data, ps.blob.diff, self.history[i],
+                               ps.lr_mult, float(self.param.momentum or 0.0),
+                               wd, lr_dev=self._lr_dev)
+        return loss
+
+    def _capture_graph(self) -> None:
+        dev = ctx().torch_device
+        self._lr_dev = torch.zeros(1, dtype=torch.float32, device=dev)
+        self._lr_dev.fill_(self.get_learning_rate())
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):  # allocation warmup
+                self._graph_body()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._graph_loss = self._graph_body()
+        self._graph = g
+
+    def _step_graphed(self, iters: int) -> float:
+        p = self.param
+        last_loss = 0.0
+        if self._graph is None:
+            self._capture_graph()
+        while iters > 0:
+            self._lr_dev.fill_(self.get_learning_rate())
+            self._graph.replay()
+            if p.display and self.iter % p.display == 0:
+                last_loss = float(self._graph_loss.item())
+                self._display(last_loss, self.get_learning_rate())
+            self.iter += 1
+            iters -= 1
+        return last_loss
+
     # ------------------------------------------------------------------
     @staticmethod
     def _load_net_param(param: Message) -> Message:
@@ -169,6 +243,15 @@ class SGDSolver:
         return loss
 
     def step(self, iters: int) -> float:
+        if self._use_graph:
+            try:
+                return self._step_graphed(iters)
+            except Exception as e:  # noqa: BLE001 -- capture unsupported
+                if self.verbose:
+                    print(f"[poseidon] graph capture failed ({e}); "
+                          "falling back to eager", flush=True)
+                self._use_graph = False
+                self._graph = None
         last_loss = 0.0
         p = self.param
         while iters > 0:
