@@ -66,7 +66,8 @@ def main():
     sp.net_param = zoo.build_net(args.model, batch=batch)
     solver = SGDSolver(sp, use_sfb=not args.no_sfb, verbose=False)
     graphed = False
-    if not args.no_graph and device == "cuda":
+    if not args.no_graph and device == "cuda" and n_gpus == 1:
+        # (multi-rank RCCL graph capture untested on this pool; eager DWBP there)
         graphed = solver.enable_graph()
 
     def sync():
