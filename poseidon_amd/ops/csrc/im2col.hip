@@ -94,23 +94,27 @@ __global__ void im2col_nhwc_rowrun_k(const T* __restrict__ x,
   }
 }
 
-// gather: one thread per (n, ih, iw, ci)
-template <typename T>
+// gather: one thread per (n, ih, iw, V-chunk of channels) -- channel chunks
+// are contiguous in BOTH dcolT rows and NHWC dx, so loads/stores vectorize
+template <typename T, int V>
 __global__ void col2im_nhwc_k(const T* __restrict__ colT, T* __restrict__ dx,
                               ConvGeom g) {
+  typedef T vec_t __attribute__((ext_vector_type(V)));
+  typedef float facc_t __attribute__((ext_vector_type(V)));
   const int Cg = g.C / g.G;
+  const int CV = Cg / V;
   const int KHW = g.kh * g.kw;
   const int Kcol = g.G * KHW * Cg;
-  int64_t total = (int64_t)g.N * g.H * g.W * g.C;
+  int64_t total = (int64_t)g.N * g.H * g.W * g.G * CV;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
-    int ci = i % g.C;
-    int64_t t = i / g.C;
+    int cv = i % CV;
+    int64_t t = i / CV;
+    int grp = t % g.G; t /= g.G;
     int iw = t % g.W; t /= g.W;
     int ih = t % g.H;
     int n = t / g.H;
-    int grp = ci / Cg, cg = ci % Cg;
-    float acc = 0.f;
+    facc_t acc = {};
     for (int kkh = 0; kkh < g.kh; ++kkh) {
       int oh_num = ih + g.ph - kkh;
       if (oh_num < 0 || oh_num % g.sh) continue;
@@ -122,11 +126,17 @@ __global__ void col2im_nhwc_k(const T* __restrict__ colT, T* __restrict__ dx,
         int ow = ow_num / g.sw;
         if (ow >= g.Wo) continue;
         int64_t np = ((int64_t)n * g.Ho + oh) * g.Wo + ow;
-        int k = (grp * KHW + kkh * g.kw + kkw) * Cg + cg;
-        acc += to_f32(colT[np * Kcol + k]);
+        int k = (grp * KHW + kkh * g.kw + kkw) * Cg + cv * V;
+        vec_t v = *reinterpret_cast<const vec_t*>(&colT[np * Kcol + k]);
+#pragma unroll
+        for (int j = 0; j < V; ++j) acc[j] += to_f32(v[j]);
       }
     }
-    from_f32(acc, dx[i]);
+    vec_t out;
+#pragma unroll
+    for (int j = 0; j < V; ++j) from_f32(acc[j], out[j]);
+    *reinterpret_cast<vec_t*>(
+        &dx[(((int64_t)n * g.H + ih) * g.W + iw) * g.C + grp * Cg + cv * V]) = out;
   }
 }
 
@@ -212,13 +222,22 @@ void ps_im2col_nhwc_bf16(const void* x, void* colT, const ConvGeom* g, hipStream
         (const __bf16*)x, (__bf16*)colT, *g);
 }
 void ps_col2im_nhwc_f32(const float* colT, float* dx, const ConvGeom* g, hipStream_t s) {
-  int64_t total = (int64_t)g->N * g->H * g->W * g->C;
-  col2im_nhwc_k<float><<<ew_grid(total), 256, 0, s>>>(colT, dx, *g);
+  int Cg = g->C / g->G;
+  int64_t base = (int64_t)g->N * g->H * g->W * g->G;
+  if (Cg % 4 == 0)
+    col2im_nhwc_k<float, 4><<<ew_grid(base * (Cg / 4)), 256, 0, s>>>(colT, dx, *g);
+  else
+    col2im_nhwc_k<float, 1><<<ew_grid(base * Cg), 256, 0, s>>>(colT, dx, *g);
 }
 void ps_col2im_nhwc_bf16(const void* colT, void* dx, const ConvGeom* g, hipStream_t s) {
-  int64_t total = (int64_t)g->N * g->H * g->W * g->C;
-  col2im_nhwc_k<__bf16><<<ew_grid(total), 256, 0, s>>>((const __bf16*)colT,
-                                                       (__bf16*)dx, *g);
+  int Cg = g->C / g->G;
+  int64_t base = (int64_t)g->N * g->H * g->W * g->G;
+  if (Cg % 8 == 0)
+    col2im_nhwc_k<__bf16, 8><<<ew_grid(base * (Cg / 8)), 256, 0, s>>>(
+        (const __bf16*)colT, (__bf16*)dx, *g);
+  else
+    col2im_nhwc_k<__bf16, 1><<<ew_grid(base * Cg), 256, 0, s>>>(
+        (const __bf16*)colT, (__bf16*)dx, *g);
 }
 void ps_weight_to_khwc_f32(const float* src, float* dst, int Co, int Cig,
                            int KH, int KW, hipStream_t s) {
