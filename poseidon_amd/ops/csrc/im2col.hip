@@ -134,7 +134,11 @@ __global__ void col2im_nhwc_k(const T* __restrict__ colT, T* __restrict__ dx,
     }
     vec_t out;
 #pragma unroll
-    for (int j = 0; j < V; ++j) from_f32(acc[j], out[j]);
+    for (int j = 0; j < V; ++j) {
+      T o;
+      from_f32(acc[j], o);
+      out[j] = o;
+    }
     *reinterpret_cast<vec_t*>(
         &dx[(((int64_t)n * g.H + ih) * g.W + iw) * g.C + grp * Cg + cv * V]) = out;
   }
