@@ -288,20 +288,30 @@ Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
   int Kcol = G * Kg;
   int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
 
-  Tensor wk = weight_khwc(w, bf16);
+  // transposed khwc repack [G][Kg][Cog]: the dgrad GEMM becomes pure NT
+  // (both operands K-last) instead of a K-major-staged NN
+  auto wc2 = w.contiguous();
+  Tensor wkT = at::empty({(int64_t)G * Kg, (int64_t)Cog},
+                         w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
+  if (bf16)
+    ps_weight_to_khwc_tr_f32_bf16(wc2.data_ptr<float>(), wkT.data_ptr(),
+                                  Co, w.size(1), kh, kw, G, stream());
+  else
+    ps_weight_to_khwc_tr_f32(wc2.data_ptr<float>(), wkT.data_ptr<float>(),
+                             Co, w.size(1), kh, kw, G, stream());
 
   bool is_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0);
   Tensor dcolT = is_1x1 ? rows2d(dx)
                         : at::empty({NP, (int64_t)Kcol}, dy.options());
   Tensor dy2 = rows2d(dy_cl);
   for (int grp = 0; grp < G; ++grp) {
-    // dcolT_g[NP, Kg] = dy_g[NP, Cog] @ wk_g[Cog, Kg]: contraction Cog
-    run_gemm(dy2, wk, dcolT, nullptr,
+    // dcolT_g[NP, Kg] = dy_g[NP, Cog] @ wkT_g[Kg, Cog]^T: NT, contraction Cog
+    run_gemm(dy2, wkT, dcolT, nullptr,
              (int)NP, Kg, Cog,
-             /*lda=*/Co, /*ldb=*/Kg, /*ldc=*/Kcol,
-             /*a_off=*/(int64_t)grp * Cog, /*b_off=*/(int64_t)grp * Cog * Kg,
+             /*lda=*/Co, /*ldb=*/Cog, /*ldc=*/Kcol,
+             /*a_off=*/(int64_t)grp * Cog, /*b_off=*/(int64_t)grp * Kg * Cog,
              /*c_off=*/(int64_t)grp * Kg,
-             true, false, 1.0f, 0.0f);
+             true, true, 1.0f, 0.0f);
   }
   if (!is_1x1) {
     if (bf16)

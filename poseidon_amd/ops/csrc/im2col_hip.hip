@@ -164,6 +164,31 @@ __global__ void weight_to_khwc_k(const TI* src, TO* dst, int Co, int Cig,
   }
 }
 
+// NCHW [Co][Cig][kh][kw] -> TRANSPOSED khwc per group: [G][Kg][Cog] with
+// Kg = kh*kw*Cig rows and the group's output channels contiguous -- the
+// K-last B operand of the dgrad NT GEMM (dcol = dy @ W): avoids the
+// K-major scatter staging path entirely.
+template <typename TI, typename TO>
+__global__ void weight_to_khwc_tr_k(const TI* src, TO* dst, int Co, int Cig,
+                                    int KH, int KW, int G) {
+  const int Cog = Co / G;
+  const int Kg = KH * KW * Cig;
+  int64_t total = (int64_t)Co * Kg;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    // src index decomposition: [co][ci][kh][kw]
+    int kkw = i % KW;
+    int64_t t = i / KW;
+    int kkh = t % KH; t /= KH;
+    int ci = t % Cig;
+    int co = t / Cig;
+    int grp = co / Cog, cog = co % Cog;
+    int kg = (kkh * KW + kkw) * Cig + ci;
+    from_f32(to_f32(src[i]),
+             dst[((int64_t)grp * Kg + kg) * Cog + cog]);
+  }
+}
+
 template <typename TI, typename TO>
 __global__ void weight_from_khwc_k(const TI* src, TO* dst, int Co, int Cig,
                                    int KH, int KW, float beta) {
@@ -254,6 +279,18 @@ void ps_weight_to_khwc_f32_bf16(const float* src, void* dst, int Co, int Cig,
  hipLaunchKernelGGL(( weight_to_khwc_k<float, __bf16>)
       , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, src, (__bf16*)dst,
                                                             Co, Cig, KH, KW);
+}
+void ps_weight_to_khwc_tr_f32(const float* src, float* dst, int Co, int Cig,
+                              int KH, int KW, int G, hipStream_t s) {
+ hipLaunchKernelGGL(( weight_to_khwc_tr_k<float, float>)
+      , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, src, dst, Co, Cig,
+                                                            KH, KW, G);
+}
+void ps_weight_to_khwc_tr_f32_bf16(const float* src, void* dst, int Co, int Cig,
+                                   int KH, int KW, int G, hipStream_t s) {
+ hipLaunchKernelGGL(( weight_to_khwc_tr_k<float, __bf16>)
+      , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, 
+          src, (__bf16*)dst, Co, Cig, KH, KW, G);
 }
 void ps_weight_from_khwc_f32(const float* src, float* dst, int Co, int Cig,
                              int KH, int KW, float beta, hipStream_t s) {

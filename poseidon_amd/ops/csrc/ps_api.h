@@ -153,4 +153,8 @@ void ps_weight_to_khwc_f32(const float*, float*, int, int, int, int, hipStream_t
 void ps_weight_to_khwc_f32_bf16(const float*, void*, int, int, int, int, hipStream_t);
 void ps_weight_from_khwc_f32(const float*, float*, int, int, int, int, float,
                              hipStream_t);
+void ps_weight_to_khwc_tr_f32(const float*, float*, int, int, int, int, int,
+                              hipStream_t);
+void ps_weight_to_khwc_tr_f32_bf16(const float*, void*, int, int, int, int,
+                                   int, hipStream_t);
 }
