@@ -35,6 +35,7 @@ template <> struct GemmTraits<float> {
   static constexpr int KSTEP = 4;    // K per MFMA
   static constexpr int VEC = 4;      // elems per staging vector load
   static constexpr int RS = BK + 4;  // padded LDS row stride (elems)
+  static constexpr int GSH = 2;      // log2(VEC): k-group shift for swizzle
   using vec_t = f32x4;
   using frag_t = float;  // one A/B element per lane
   __device__ static inline f32x4 mfma(frag_t a, frag_t b, f32x4 acc) {
@@ -50,6 +51,7 @@ template <> struct GemmTraits<__bf16> {
   static constexpr int KSTEP = 32;
   static constexpr int VEC = 8;
   static constexpr int RS = BK + 8;  // 144 B rows: 16B-aligned, conflict-spread
+  static constexpr int GSH = 3;      // log2(VEC)
   typedef __attribute__((ext_vector_type(8))) __bf16 vec_t;
   using frag_t = bf16x8;
   __device__ static inline f32x4 mfma(frag_t a, frag_t b, f32x4 acc) {
@@ -59,6 +61,19 @@ template <> struct GemmTraits<__bf16> {
     return *reinterpret_cast<const frag_t*>(&lds_row[kk + (lane >> 4) * 8]);
   }
 };
+
+// Column offset inside an LDS tile row. K-major-staged tiles XOR the
+// k-group index with the row band so the block-transposed vector writes
+// spread across banks (write lanes hit 8 distinct banks instead of 1);
+// fragment reads apply the same XOR. K-last tiles stay linear.
+template <typename T, bool SWZ>
+__device__ inline int lds_col(int row, int kk) {
+  using TR = GemmTraits<T>;
+  if (!SWZ) return kk;
+  constexpr int GM = (TR::BK / TR::VEC) - 1;  // k-group mask
+  int kg = kk >> TR::GSH;
+  return (((kg ^ (row >> TR::GSH)) & GM) << TR::GSH) | (kk & (TR::VEC - 1));
+}
 
 // Staging, split into {issue global loads -> regs} and {write regs -> LDS}
 // halves so the next tile's HBM latency hides under the current tile's MFMA
@@ -71,7 +86,10 @@ struct Stager {
   using vec_t = typename TR::vec_t;
   static constexpr int TV = ROWS * TR::BK / TR::VEC;  // total vectors in tile
   static constexpr int NV = (TV + 255) / 256;         // per-thread (>=1)
-  vec_t v[NV];
+  // K-major path: VEC x VEC blocks, VEC vectors each
+  static constexpr int TB = ROWS * TR::BK / (TR::VEC * TR::VEC);
+  static constexpr int NVB = (TB + 255) / 256;
+  vec_t v[KLAST ? NV : NVB * TR::VEC];
 
   __device__ inline void load(const T* src, int64_t lda, int row0,
                               int rows_max, int k0, int K, int tid) {
@@ -95,23 +113,30 @@ struct Stager {
         v[i] = val;
       }
     } else {
-      constexpr int CM = ROWS / TR::VEC;
+      // K-major: each thread owns a VEC x VEC block (k-block kb, m-block mb)
+      // and loads VEC row-vectors along the contiguous m dim (coalesced
+      // across lanes: consecutive threads -> consecutive m-blocks)
+      constexpr int BLK_M = ROWS / TR::VEC;
 #pragma unroll
-      for (int i = 0; i < NV; ++i) {
+      for (int i = 0; i < NVB; ++i) {
         int c = tid + i * 256;
-        if (c >= TV) break;
-        int kk = c / CM, mc = c % CM;
-        int gk = k0 + kk, gm = row0 + mc * TR::VEC;
-        vec_t val = {};
-        if (gk < K && gm < rows_max) {
-          if (gm + TR::VEC <= rows_max) {
-            val = *reinterpret_cast<const vec_t*>(&src[(int64_t)gk * lda + gm]);
-          } else {
-            for (int j = 0; j < TR::VEC; ++j)
-              if (gm + j < rows_max) val[j] = src[(int64_t)gk * lda + gm + j];
+        if (c >= TB) break;
+        int kb = c / BLK_M, mb = c % BLK_M;
+        int gm = row0 + mb * TR::VEC;
+#pragma unroll
+        for (int j = 0; j < TR::VEC; ++j) {
+          int gk = k0 + kb * TR::VEC + j;
+          vec_t val = {};
+          if (gk < K && gm < rows_max) {
+            if (gm + TR::VEC <= rows_max) {
+              val = *reinterpret_cast<const vec_t*>(&src[(int64_t)gk * lda + gm]);
+            } else {
+              for (int e = 0; e < TR::VEC; ++e)
+                if (gm + e < rows_max) val[e] = src[(int64_t)gk * lda + gm + e];
+            }
           }
+          v[i * TR::VEC + j] = val;
         }
-        v[i] = val;
       }
     }
   }
@@ -127,14 +152,23 @@ struct Stager {
         *reinterpret_cast<vec_t*>(&lds[r * TR::RS + kc * TR::VEC]) = v[i];
       }
     } else {
-      constexpr int CM = ROWS / TR::VEC;
+      // register-transpose the VEC x VEC block, then VEC aligned vector
+      // writes to swizzled columns (conflict-free within the lane group)
+      constexpr int BLK_M = ROWS / TR::VEC;
 #pragma unroll
-      for (int i = 0; i < NV; ++i) {
+      for (int i = 0; i < NVB; ++i) {
         int c = tid + i * 256;
-        if (c >= TV) break;
-        int kk = c / CM, mc = c % CM;
-        for (int j = 0; j < TR::VEC; ++j)
-          lds[(mc * TR::VEC + j) * TR::RS + kk] = v[i][j];
+        if (c >= TB) break;
+        int kb = c / BLK_M, mb = c % BLK_M;
+#pragma unroll
+        for (int j = 0; j < TR::VEC; ++j) {  // j: m within the block
+          vec_t out;
+#pragma unroll
+          for (int e = 0; e < TR::VEC; ++e) out[e] = v[i * TR::VEC + e][j];
+          int row = mb * TR::VEC + j;
+          *reinterpret_cast<vec_t*>(
+              &lds[row * TR::RS + lds_col<T, true>(row, kb * TR::VEC)]) = out;
+        }
       }
     }
   }
@@ -210,11 +244,19 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
     for (int kk = 0; kk < BK; kk += TR::KSTEP) {
       typename TR::frag_t a_frag[FM], b_frag[FN];
 #pragma unroll
-      for (int f = 0; f < FM; ++f)
-        a_frag[f] = TR::load_frag(&a_lds[cur][(wm + f * 16 + (lane & 15)) * RS], kk, lane);
+      for (int f = 0; f < FM; ++f) {
+        int row = wm + f * 16 + (lane & 15);
+        int col = kk + (lane >> 4) * (TR::KSTEP / 4);
+        a_frag[f] = *reinterpret_cast<const typename TR::frag_t*>(
+            &a_lds[cur][row * RS + lds_col<T, !A_KLAST>(row, col)]);
+      }
 #pragma unroll
-      for (int f = 0; f < FN; ++f)
-        b_frag[f] = TR::load_frag(&b_lds[cur][(wn + f * 16 + (lane & 15)) * RS], kk, lane);
+      for (int f = 0; f < FN; ++f) {
+        int row = wn + f * 16 + (lane & 15);
+        int col = kk + (lane >> 4) * (TR::KSTEP / 4);
+        b_frag[f] = *reinterpret_cast<const typename TR::frag_t*>(
+            &b_lds[cur][row * RS + lds_col<T, !B_KLAST>(row, col)]);
+      }
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
