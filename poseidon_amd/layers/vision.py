@@ -124,7 +124,7 @@ class PoolingLayer(Layer):
             y, mask = ops.pool_max_forward(x, self.kernel, self.stride, self.pad)
             self._mask = mask
             if len(top) > 1:
-                top[1].data = mask.to(x.dtype)
+                top[1].data = self._mask_to_spatial(mask, x).to(x.dtype)
         elif self.method == "AVE":
             y = ops.pool_ave_forward(x, self.kernel, self.stride, self.pad)
         else:  # STOCHASTIC
@@ -136,6 +136,22 @@ class PoolingLayer(Layer):
                 y = ops.pool_stoch_forward_test(
                     x, self.kernel, self.stride, self.pad)
         top[0].data = y
+
+    def _mask_to_spatial(self, mask, x):
+        """GPU masks are u8 window-local argmax indices; Caffe's optional
+        mask top wants the bottom spatial index h*W + w."""
+        if not mask.is_cuda:
+            return mask
+        N, C, Ho, Wo = mask.shape
+        W = x.shape[3]
+        dev = mask.device
+        oh = torch.arange(Ho, device=dev).view(1, 1, Ho, 1)
+        ow = torch.arange(Wo, device=dev).view(1, 1, 1, Wo)
+        kh = mask.long() // self.kernel[1]
+        kw = mask.long() % self.kernel[1]
+        h = oh * self.stride[0] - self.pad[0] + kh
+        w = ow * self.stride[1] - self.pad[1] + kw
+        return h * W + w
 
     def backward(self, top, propagate_down, bottom) -> None:
         if not propagate_down[0]:

@@ -389,14 +389,14 @@ std::vector<Tensor> pool_max_forward(const Tensor& x, int kh, int kw, int sh,
   auto opts_cl = x.options().memory_format(at::MemoryFormat::ChannelsLast);
   Tensor y = at::empty({g.N, g.C, g.Ho, g.Wo}, opts_cl);
   Tensor mask = at::empty({g.N, g.C, g.Ho, g.Wo},
-                          x.options().dtype(at::kInt)
+                          x.options().dtype(at::kByte)
                               .memory_format(at::MemoryFormat::ChannelsLast));
   if (is_bf16(x))
-    ps_maxpool_fwd_bf16(x_cl.data_ptr(), y.data_ptr(), mask.data_ptr<int>(),
-                        &g, stream());
+    ps_maxpool_fwd_bf16(x_cl.data_ptr(), y.data_ptr(),
+                        mask.data_ptr<uint8_t>(), &g, stream());
   else
     ps_maxpool_fwd_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(),
-                       mask.data_ptr<int>(), &g, stream());
+                       mask.data_ptr<uint8_t>(), &g, stream());
   return {y, mask};
 }
 
@@ -412,10 +412,10 @@ Tensor pool_max_backward(const Tensor& dy, const Tensor& mask,
   Tensor dx = at::empty({g.N, g.C, g.H, g.W},
                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
   if (is_bf16(dy))
-    ps_maxpool_bwd_bf16(dy_cl.data_ptr(), mask_cl.data_ptr<int>(),
+    ps_maxpool_bwd_bf16(dy_cl.data_ptr(), mask_cl.data_ptr<uint8_t>(),
                         dx.data_ptr(), &g, stream());
   else
-    ps_maxpool_bwd_f32(dy_cl.data_ptr<float>(), mask_cl.data_ptr<int>(),
+    ps_maxpool_bwd_f32(dy_cl.data_ptr<float>(), mask_cl.data_ptr<uint8_t>(),
                        dx.data_ptr<float>(), &g, stream());
   return dx;
 }
@@ -460,10 +460,11 @@ std::vector<Tensor> pool_stoch_forward_train(const Tensor& x, int kh, int kw,
   Tensor y = at::empty({g.N, g.C, g.Ho, g.Wo},
                        x_cl.options().memory_format(at::MemoryFormat::ChannelsLast));
   Tensor mask = at::empty({g.N, g.C, g.Ho, g.Wo},
-                          x_cl.options().dtype(at::kInt)
+                          x_cl.options().dtype(at::kByte)
                               .memory_format(at::MemoryFormat::ChannelsLast));
   ps_stochpool_fwd_train_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(),
-                             mask.data_ptr<int>(), &g, (uint64_t)seed, stream());
+                             mask.data_ptr<uint8_t>(), &g, (uint64_t)seed,
+                             stream());
   return {y.to(x.scalar_type()), mask};
 }
 
@@ -512,15 +513,16 @@ Tensor lrn_backward(const Tensor& x, const Tensor& y, const Tensor& scale,
   int64_t rows = (int64_t)x_cl.size(0) * x_cl.size(2) * x_cl.size(3);
   int C = x_cl.size(1);
   Tensor dx = at::empty_like(x_cl, x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  Tensor ratio = at::empty({rows * C}, x.options().dtype(at::kFloat));
   if (is_bf16(x))
     ps_lrn_bwd_bf16(x_cl.data_ptr(), y_cl.data_ptr(), sc_cl.data_ptr<float>(),
-                    dy_cl.data_ptr(), dx.data_ptr(), rows, C, size,
-                    (float)alpha, (float)beta, stream());
+                    dy_cl.data_ptr(), dx.data_ptr(), ratio.data_ptr<float>(),
+                    rows, C, size, (float)alpha, (float)beta, stream());
   else
     ps_lrn_bwd_f32(x_cl.data_ptr<float>(), y_cl.data_ptr<float>(),
                    sc_cl.data_ptr<float>(), dy_cl.data_ptr<float>(),
-                   dx.data_ptr<float>(), rows, C, size, (float)alpha,
-                   (float)beta, stream());
+                   dx.data_ptr<float>(), ratio.data_ptr<float>(), rows, C,
+                   size, (float)alpha, (float)beta, stream());
   return dx;
 }
 

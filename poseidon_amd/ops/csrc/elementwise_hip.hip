@@ -121,26 +121,46 @@ __global__ void dropout_bwd_k(const T* dy, const uint8_t* mask, T* dx,
 template <typename T>
 __global__ void colsum_k(const T* __restrict__ in, float* __restrict__ out,
                          int64_t R, int C) {
-  __shared__ float part[4][64];
+  typedef T vec2 __attribute__((ext_vector_type(2)));
+  __shared__ float part[4][128];
   const int lane = threadIdx.x & 63;
   const int rg = threadIdx.x >> 6;  // 4 row groups
   int64_t rows_per = (R + gridDim.x - 1) / gridDim.x;
   int64_t r0 = (int64_t)blockIdx.x * rows_per;
   int64_t r1 = min(R, r0 + rows_per);
-  for (int c0 = 0; c0 < C; c0 += 64) {
-    int c = c0 + lane;
-    float acc = 0.f;
-    if (c < C)
-      for (int64_t r = r0 + rg; r < r1; r += 4)
-        acc += to_f32(in[r * C + c]);
-    part[rg][lane] = acc;
+  const bool v2 = (C % 2) == 0;
+  const int span = v2 ? 128 : 64;
+  for (int c0 = 0; c0 < C; c0 += span) {
+    float acc0 = 0.f, acc1 = 0.f;
+    if (v2) {
+      int c = c0 + lane * 2;
+      if (c + 1 < C || c < C) {
+        for (int64_t r = r0 + rg; r < r1; r += 4) {
+          vec2 v = *reinterpret_cast<const vec2*>(&in[r * C + c]);
+          acc0 += to_f32(v[0]);
+          acc1 += to_f32(v[1]);
+        }
+      }
+      part[rg][lane * 2] = acc0;
+      part[rg][lane * 2 + 1] = acc1;
+    } else {
+      int c = c0 + lane;
+      if (c < C)
+        for (int64_t r = r0 + rg; r < r1; r += 4)
+          acc0 += to_f32(in[r * C + c]);
+      part[rg][lane] = acc0;
+    }
     __syncthreads();
-    if (rg == 0 && c < C) {
-      float v = part[0][lane] + part[1][lane] + part[2][lane] + part[3][lane];
-      if (gridDim.x == 1)
-        out[c] += v;
-      else
-        atomicAdd(&out[c], v);
+    if (rg < 2) {
+      int idx = rg * 64 + lane;  // 128 partial slots
+      int c = c0 + idx;
+      if (idx < span && c < C) {
+        float v = part[0][idx] + part[1][idx] + part[2][idx] + part[3][idx];
+        if (gridDim.x == 1)
+          out[c] += v;
+        else
+          atomicAdd(&out[c], v);
+      }
     }
     __syncthreads();
   }

@@ -34,14 +34,22 @@ __global__ void lrn_fwd_k(const T* x, T* y, float* scale, int64_t rows, int C,
   }
 }
 
+// backward pass 1: ratio = dy*y/scale (one divide per element instead of
+// one per window entry)
 template <typename T>
-__global__ void lrn_bwd_k(const T* x, const T* y, const float* scale,
-                          const T* dy, T* dx, int64_t rows, int C, int size,
-                          float cache_ratio, float beta) {
+__global__ void lrn_ratio_k(const T* y, const float* scale, const T* dy,
+                            float* ratio, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x)
+    ratio[i] = to_f32(dy[i]) * to_f32(y[i]) / scale[i];
+}
+
+// backward pass 2: dx = dy*scale^-beta - cr * x * window_sum(ratio)
+template <typename T>
+__global__ void lrn_bwd_k(const T* x, const float* scale, const T* dy,
+                          const float* ratio, T* dx, int64_t rows, int C,
+                          int size, float cache_ratio, float beta) {
   const int pre = (size - 1) / 2;
-  // window of i contains j iff |i-j| pattern: j in [i-pre, i-pre+size).
-  // reverse: i in [j - (size-1-pre), j + pre] -- accumulate over j whose
-  // window covers c.
   const int post = size - 1 - pre;
   int64_t total = rows * C;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -51,12 +59,8 @@ __global__ void lrn_bwd_k(const T* x, const T* y, const float* scale,
     int64_t base = row * C;
     int c0 = max(c - post, 0), c1 = min(c + pre + 1, C);
     float acc = 0.f;
-    for (int j = c0; j < c1; ++j) {
-      int64_t jj = base + j;
-      acc += to_f32(dy[jj]) * to_f32(y[jj]) / scale[jj];
-    }
-    float sc = scale[i];
-    from_f32(to_f32(dy[i]) * __powf(sc, -beta)
+    for (int j = c0; j < c1; ++j) acc += ratio[base + j];
+    from_f32(to_f32(dy[i]) * __powf(scale[i], -beta)
                  - cache_ratio * to_f32(x[i]) * acc,
              dx[i]);
   }
@@ -75,17 +79,22 @@ void ps_lrn_fwd_bf16(const void* x, void* y, float* scale, int64_t rows,
       (const __bf16*)x, (__bf16*)y, scale, rows, C, size, alpha / size, beta);
 }
 void ps_lrn_bwd_f32(const float* x, const float* y, const float* scale,
-                    const float* dy, float* dx, int64_t rows, int C, int size,
-                    float alpha, float beta, hipStream_t s) {
+                    const float* dy, float* dx, float* ratio_ws, int64_t rows,
+                    int C, int size, float alpha, float beta, hipStream_t s) {
+  lrn_ratio_k<float><<<ew_grid(rows * C), 256, 0, s>>>(y, scale, dy, ratio_ws,
+                                                       rows * C);
   lrn_bwd_k<float><<<ew_grid(rows * C), 256, 0, s>>>(
-      x, y, scale, dy, dx, rows, C, size, 2.0f * alpha * beta / size, beta);
+      x, scale, dy, ratio_ws, dx, rows, C, size, 2.0f * alpha * beta / size,
+      beta);
 }
 void ps_lrn_bwd_bf16(const void* x, const void* y, const float* scale,
-                     const void* dy, void* dx, int64_t rows, int C, int size,
-                     float alpha, float beta, hipStream_t s) {
+                     const void* dy, void* dx, float* ratio_ws, int64_t rows,
+                     int C, int size, float alpha, float beta, hipStream_t s) {
+  lrn_ratio_k<__bf16><<<ew_grid(rows * C), 256, 0, s>>>(
+      (const __bf16*)y, scale, (const __bf16*)dy, ratio_ws, rows * C);
   lrn_bwd_k<__bf16><<<ew_grid(rows * C), 256, 0, s>>>(
-      (const __bf16*)x, (const __bf16*)y, scale, (const __bf16*)dy,
-      (__bf16*)dx, rows, C, size, 2.0f * alpha * beta / size, beta);
+      (const __bf16*)x, scale, (const __bf16*)dy, ratio_ws, (__bf16*)dx,
+      rows, C, size, 2.0f * alpha * beta / size, beta);
 }
 
 }  // extern "C"

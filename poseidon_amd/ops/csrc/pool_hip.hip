@@ -13,9 +13,12 @@
 
 namespace ps {
 
-// one thread per (n, oh, ow, c): mask stores the bottom spatial index h*W+w
+// one thread per (n, oh, ow, c): mask stores the WINDOW-LOCAL argmax index
+// kh*kw_w + kw as u8 (4x less mask traffic than a spatial int; the layer
+// converts to Caffe's bottom-spatial-index semantics if the optional mask
+// top is requested)
 template <typename T>
-__global__ void maxpool_fwd_k(const T* x, T* y, int* mask, PoolGeom g) {
+__global__ void maxpool_fwd_k(const T* x, T* y, uint8_t* mask, PoolGeom g) {
   int64_t total = (int64_t)g.N * g.Ho * g.Wo * g.C;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -24,24 +27,24 @@ __global__ void maxpool_fwd_k(const T* x, T* y, int* mask, PoolGeom g) {
     int ow = t % g.Wo; t /= g.Wo;
     int oh = t % g.Ho;
     int n = t / g.Ho;
-    int h0 = oh * g.sh - g.ph, w0 = ow * g.sw - g.pw;
-    int h1 = min(h0 + g.kh, g.H), w1 = min(w0 + g.kw, g.W);
-    h0 = max(h0, 0); w0 = max(w0, 0);
+    int hb = oh * g.sh - g.ph, wb = ow * g.sw - g.pw;
+    int h0 = max(hb, 0), w0 = max(wb, 0);
+    int h1 = min(hb + g.kh, g.H), w1 = min(wb + g.kw, g.W);
     float best = -3.4e38f;
-    int best_idx = h0 * g.W + w0;
+    int best_idx = (h0 - hb) * g.kw + (w0 - wb);
     for (int h = h0; h < h1; ++h)
       for (int w = w0; w < w1; ++w) {
         float v = to_f32(x[(((int64_t)n * g.H + h) * g.W + w) * g.C + c]);
-        if (v > best) { best = v; best_idx = h * g.W + w; }
+        if (v > best) { best = v; best_idx = (h - hb) * g.kw + (w - wb); }
       }
     from_f32(best, y[i]);
-    mask[i] = best_idx;
+    mask[i] = (uint8_t)best_idx;
   }
 }
 
 // gather: one thread per (n, h, w, c) scans covering windows
 template <typename T>
-__global__ void maxpool_bwd_k(const T* dy, const int* mask, T* dx, PoolGeom g) {
+__global__ void maxpool_bwd_k(const T* dy, const uint8_t* mask, T* dx, PoolGeom g) {
   int64_t total = (int64_t)g.N * g.H * g.W * g.C;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -50,7 +53,6 @@ __global__ void maxpool_bwd_k(const T* dy, const int* mask, T* dx, PoolGeom g) {
     int w = t % g.W; t /= g.W;
     int h = t % g.H;
     int n = t / g.H;
-    int me = h * g.W + w;
     int oh0 = (h + g.ph < g.kh) ? 0 : (h + g.ph - g.kh) / g.sh + 1;
     int oh1 = min((h + g.ph) / g.sh + 1, g.Ho);
     int ow0 = (w + g.pw < g.kw) ? 0 : (w + g.pw - g.kw) / g.sw + 1;
@@ -59,7 +61,8 @@ __global__ void maxpool_bwd_k(const T* dy, const int* mask, T* dx, PoolGeom g) {
     for (int oh = oh0; oh < oh1; ++oh)
       for (int ow = ow0; ow < ow1; ++ow) {
         int64_t oi = (((int64_t)n * g.Ho + oh) * g.Wo + ow) * g.C + c;
-        if (mask[oi] == me) acc += to_f32(dy[oi]);
+        int local = (h - (oh * g.sh - g.ph)) * g.kw + (w - (ow * g.sw - g.pw));
+        if ((int)mask[oi] == local) acc += to_f32(dy[oi]);
       }
     from_f32(acc, dx[i]);
   }
@@ -122,7 +125,7 @@ __global__ void avepool_bwd_k(const T* dy, T* dx, PoolGeom g) {
 // probability proportional to its (nonnegative) activation; test is the
 // activation-weighted average. mask reuses the maxpool backward.
 template <typename T>
-__global__ void stochpool_fwd_train_k(const T* x, T* y, int* mask, PoolGeom g,
+__global__ void stochpool_fwd_train_k(const T* x, T* y, uint8_t* mask, PoolGeom g,
                                       uint64_t seed) {
   int64_t total = (int64_t)g.N * g.Ho * g.Wo * g.C;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -144,17 +147,18 @@ __global__ void stochpool_fwd_train_k(const T* x, T* y, int* mask, PoolGeom g,
     float thresh = (r >> 8) * (1.0f / 16777216.0f) * sum;
     float cum = 0.f;
     float pick = 0.f;
-    int pick_idx = h0 * g.W + w0;
+    int hb = oh * g.sh - g.ph, wb = ow * g.sw - g.pw;
+    int pick_idx = (h0 - hb) * g.kw + (w0 - wb);
     bool done = false;
     for (int h = h0; h < h1 && !done; ++h)
       for (int w = w0; w < w1 && !done; ++w) {
         float v = to_f32(x[(((int64_t)n * g.H + h) * g.W + w) * g.C + c]);
         cum += v;
-        pick = v; pick_idx = h * g.W + w;
+        pick = v; pick_idx = (h - hb) * g.kw + (w - wb);
         if (cum >= thresh) done = true;
       }
     from_f32(pick, y[i]);
-    mask[i] = pick_idx;
+    mask[i] = (uint8_t)pick_idx;
   }
 }
 
@@ -187,20 +191,20 @@ extern "C" {
 #define PS_POOL_LAUNCH(kern, count, ...) \
  hipLaunchKernelGGL(( kern), dim3(ew_grid(count)), dim3(256), 0, s, __VA_ARGS__)
 
-void ps_maxpool_fwd_f32(const float* x, float* y, int* mask, const PoolGeom* g,
+void ps_maxpool_fwd_f32(const float* x, float* y, uint8_t* mask, const PoolGeom* g,
                         hipStream_t s) {
   PS_POOL_LAUNCH(maxpool_fwd_k<float>, (int64_t)g->N * g->Ho * g->Wo * g->C, x, y, mask, *g);
 }
-void ps_maxpool_fwd_bf16(const void* x, void* y, int* mask, const PoolGeom* g,
+void ps_maxpool_fwd_bf16(const void* x, void* y, uint8_t* mask, const PoolGeom* g,
                          hipStream_t s) {
   PS_POOL_LAUNCH(maxpool_fwd_k<__bf16>, (int64_t)g->N * g->Ho * g->Wo * g->C,
                  (const __bf16*)x, (__bf16*)y, mask, *g);
 }
-void ps_maxpool_bwd_f32(const float* dy, const int* mask, float* dx,
+void ps_maxpool_bwd_f32(const float* dy, const uint8_t* mask, float* dx,
                         const PoolGeom* g, hipStream_t s) {
   PS_POOL_LAUNCH(maxpool_bwd_k<float>, (int64_t)g->N * g->H * g->W * g->C, dy, mask, dx, *g);
 }
-void ps_maxpool_bwd_bf16(const void* dy, const int* mask, void* dx,
+void ps_maxpool_bwd_bf16(const void* dy, const uint8_t* mask, void* dx,
                          const PoolGeom* g, hipStream_t s) {
   PS_POOL_LAUNCH(maxpool_bwd_k<__bf16>, (int64_t)g->N * g->H * g->W * g->C,
                  (const __bf16*)dy, mask, (__bf16*)dx, *g);
@@ -219,7 +223,7 @@ void ps_avepool_bwd_bf16(const void* dy, void* dx, const PoolGeom* g, hipStream_
   PS_POOL_LAUNCH(avepool_bwd_k<__bf16>, (int64_t)g->N * g->H * g->W * g->C,
                  (const __bf16*)dy, (__bf16*)dx, *g);
 }
-void ps_stochpool_fwd_train_f32(const float* x, float* y, int* mask,
+void ps_stochpool_fwd_train_f32(const float* x, float* y, uint8_t* mask,
                                 const PoolGeom* g, uint64_t seed, hipStream_t s) {
   PS_POOL_LAUNCH(stochpool_fwd_train_k<float>, (int64_t)g->N * g->Ho * g->Wo * g->C,
                  x, y, mask, *g, seed);

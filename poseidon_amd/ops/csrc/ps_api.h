@@ -90,15 +90,15 @@ void ps_dropout_bwd_bf16(const void*, const uint8_t*, void*, int64_t, float,
 void ps_colsum_bf16(const void*, float*, int64_t, int, hipStream_t);
 
 // pool.hip
-void ps_maxpool_fwd_f32(const float*, float*, int*, const PoolGeom*, hipStream_t);
-void ps_maxpool_bwd_f32(const float*, const int*, float*, const PoolGeom*, hipStream_t);
+void ps_maxpool_fwd_f32(const float*, float*, uint8_t*, const PoolGeom*, hipStream_t);
+void ps_maxpool_bwd_f32(const float*, const uint8_t*, float*, const PoolGeom*, hipStream_t);
 void ps_avepool_fwd_f32(const float*, float*, const PoolGeom*, hipStream_t);
 void ps_avepool_bwd_f32(const float*, float*, const PoolGeom*, hipStream_t);
-void ps_stochpool_fwd_train_f32(const float*, float*, int*, const PoolGeom*,
+void ps_stochpool_fwd_train_f32(const float*, float*, uint8_t*, const PoolGeom*,
                                 uint64_t, hipStream_t);
 void ps_stochpool_fwd_test_f32(const float*, float*, const PoolGeom*, hipStream_t);
-void ps_maxpool_fwd_bf16(const void*, void*, int*, const PoolGeom*, hipStream_t);
-void ps_maxpool_bwd_bf16(const void*, const int*, void*, const PoolGeom*, hipStream_t);
+void ps_maxpool_fwd_bf16(const void*, void*, uint8_t*, const PoolGeom*, hipStream_t);
+void ps_maxpool_bwd_bf16(const void*, const uint8_t*, void*, const PoolGeom*, hipStream_t);
 void ps_avepool_fwd_bf16(const void*, void*, const PoolGeom*, hipStream_t);
 void ps_avepool_bwd_bf16(const void*, void*, const PoolGeom*, hipStream_t);
 
@@ -106,11 +106,11 @@ void ps_avepool_bwd_bf16(const void*, void*, const PoolGeom*, hipStream_t);
 void ps_lrn_fwd_f32(const float*, float*, float*, int64_t, int, int, float,
                     float, hipStream_t);
 void ps_lrn_bwd_f32(const float*, const float*, const float*, const float*,
-                    float*, int64_t, int, int, float, float, hipStream_t);
+                    float*, float*, int64_t, int, int, float, float, hipStream_t);
 void ps_lrn_fwd_bf16(const void*, void*, float*, int64_t, int, int, float,
                      float, hipStream_t);
 void ps_lrn_bwd_bf16(const void*, const void*, const float*, const void*,
-                     void*, int64_t, int, int, float, float, hipStream_t);
+                     void*, float*, int64_t, int, int, float, float, hipStream_t);
 
 // softmax.hip
 void ps_softmax_rows_f32(const float*, float*, int64_t, int, hipStream_t);
