@@ -24,6 +24,7 @@ from .context import ctx
 from .insert_splits import insert_splits
 from .layer import Layer, create_layer
 from ..proto import Message, spec
+from ..proto.upgrade import net_needs_upgrade, upgrade_v0_net
 
 TRAIN = spec.ENUMS["Phase"]["TRAIN"]
 TEST = spec.ENUMS["Phase"]["TEST"]
@@ -88,6 +89,8 @@ class Net:
         if state is not None:
             st.merge_from(state)
         self.phase = phase
+        if net_needs_upgrade(param):
+            param = upgrade_v0_net(param)  # legacy V0 prototxt support
         filtered = filter_net(param, st)
         self.param = insert_splits(filtered)
         self.name = self.param.name or ""
