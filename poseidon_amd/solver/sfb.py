@@ -68,6 +68,11 @@ class SFBReducer:
         flat = torch.cat([a.reshape(-1), b.reshape(-1)])
         out = torch.empty(W * flat.numel(), dtype=flat.dtype, device=flat.device)
         if self.comm_stream is not None:
+            # flat/out are allocated on the compute stream but used by the
+            # comm stream: record_stream stops the caching allocator from
+            # recycling them while the collective is in flight
+            flat.record_stream(self.comm_stream)
+            out.record_stream(self.comm_stream)
             ev = torch.cuda.Event()
             ev.record(torch.cuda.current_stream())
             with torch.cuda.stream(self.comm_stream):
