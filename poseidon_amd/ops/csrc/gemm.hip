@@ -61,6 +61,34 @@ template <> struct GemmTraits<__bf16> {
   }
 };
 
+// Direct global->LDS staging (glds): each wave-instruction moves 1 KiB
+// (64 lanes x 16 B) HBM -> LDS without a VGPR round trip
+// (__builtin_amdgcn_global_load_lds, width 16 -- guide §5 step 3: the
+// +67% width-4 -> width-16 lever). The LDS image is LINEAR [row][BK]
+// (the builtin writes wave-uniform-base + lane*16), so this path is used
+// for interior tiles of K-last operands only; fragment reads use the
+// unpadded BK stride.
+template <typename T, int ROWS>
+__device__ inline void stage_glds(T* lds, const T* __restrict__ src,
+                                  int64_t lda, int row0, int k0, int wid,
+                                  int lane) {
+  using TR = GemmTraits<T>;
+  constexpr int EPB = 16 / sizeof(T);            // elems per 16B lane-load
+  constexpr int LPR = TR::BK / EPB;              // lanes per row
+  constexpr int RPC = 64 / LPR;                  // rows per 1KB chunk
+  constexpr int CHUNKS = ROWS / RPC;
+  const int r_in = lane / LPR;
+  const int kc = (lane % LPR) * EPB;
+#pragma unroll
+  for (int ci = wid; ci < CHUNKS; ci += 4) {
+    const T* g = src + (int64_t)(row0 + ci * RPC + r_in) * lda + k0 + kc;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(lds + ci * (1024 / (int)sizeof(T))),
+        16, 0, 0);
+  }
+}
+
 // Column offset inside an LDS tile row. K-major-staged tiles XOR the
 // k-group index with the row band so the block-transposed vector writes
 // spread across banks (write lanes hit 8 distinct banks instead of 1);
@@ -213,6 +241,61 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
 
   f32x4 acc[FM][FN] = {};
 
+  // glds fast path: interior tile, K-span a multiple of BK, vector-aligned
+  // rows (guide §5: direct-to-LDS staging removes the VGPR round trip and
+  // its VALU address work; edge tiles keep the guarded register pipeline)
+  {
+    constexpr int EPB = 16 / (int)sizeof(T);
+    const int k_span = k_end - k_begin;
+    const bool glds_ok = A_KLAST && B_KLAST && (m0 + BM <= M) &&
+                         (n0 + BN <= N) && (k_span % BK) == 0 && k_span > 0 &&
+                         (lda % EPB) == 0 && (ldb % EPB) == 0 &&
+                         (k_begin % EPB) == 0 &&
+                         (((uintptr_t)A & 15) == 0) && (((uintptr_t)B & 15) == 0);
+    if (glds_ok) {
+      T* a_lin0 = a_lds[0];
+      T* b_lin0 = b_lds[0];
+      T* a_lin1 = a_lds[1];
+      T* b_lin1 = b_lds[1];
+      stage_glds<T, BM>(a_lin0, A, lda, m0, k_begin, wid, lane);
+      stage_glds<T, BN>(b_lin0, B, ldb, n0, k_begin, wid, lane);
+      __syncthreads();  // drains the in-flight glds (vmcnt 0) + barrier
+      int cur2 = 0;
+      for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+        if (k0 + BK < k_end) {
+          stage_glds<T, BM>(cur2 ? a_lin0 : a_lin1, A, lda, m0, k0 + BK, wid, lane);
+          stage_glds<T, BN>(cur2 ? b_lin0 : b_lin1, B, ldb, n0, k0 + BK, wid, lane);
+        }
+        const T* al = cur2 ? a_lin1 : a_lin0;
+        const T* bl = cur2 ? b_lin1 : b_lin0;
+#pragma unroll
+        for (int kk = 0; kk < BK; kk += TR::KSTEP) {
+          typename TR::frag_t a_frag[FM], b_frag[FN];
+#pragma unroll
+          for (int f = 0; f < FM; ++f) {
+            int row = wm + f * 16 + (lane & 15);
+            a_frag[f] = *reinterpret_cast<const typename TR::frag_t*>(
+                &al[row * BK + kk + (lane >> 4) * (TR::KSTEP / 4)]);
+          }
+#pragma unroll
+          for (int f = 0; f < FN; ++f) {
+            int row = wn + f * 16 + (lane & 15);
+            b_frag[f] = *reinterpret_cast<const typename TR::frag_t*>(
+                &bl[row * BK + kk + (lane >> 4) * (TR::KSTEP / 4)]);
+          }
+#pragma unroll
+          for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+            for (int fn = 0; fn < FN; ++fn)
+              acc[fm][fn] = TR::mfma(a_frag[fm], b_frag[fn], acc[fm][fn]);
+        }
+        __syncthreads();
+        cur2 ^= 1;
+      }
+      goto epilogue;
+    }
+  }
+  {
   Stager<T, BM, A_KLAST> sa;
   Stager<T, BN, B_KLAST> sb;
   // prologue: tile 0 -> LDS[0]; issue tile 1 loads
@@ -265,7 +348,9 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
     __syncthreads();
     cur ^= 1;
   }
+  }
 
+epilogue:
   // Epilogue: C/D fragment map for 16x16 shapes: col = lane&15,
   // row = (lane>>4)*4 + r (guide §3; dtype-independent on gfx950).
 #pragma unroll
