@@ -79,15 +79,32 @@ __device__ inline void stage_glds(T* lds, const T* __restrict__ src,
   constexpr int RPC = 64 / LPR;                  // rows per 1KB chunk
   constexpr int CHUNKS = ROWS / RPC;
   const int r_in = lane / LPR;
-  const int kc = (lane % LPR) * EPB;
+  const int slot = lane % LPR;
 #pragma unroll
   for (int ci = wid; ci < CHUNKS; ci += 4) {
-    const T* g = src + (int64_t)(row0 + ci * RPC + r_in) * lda + k0 + kc;
+    const int row = ci * RPC + r_in;
+    // rule 21: the LDS image is XOR-swizzled by swizzling the SOURCE
+    // address per lane (glds writes lane-linear); reads apply the same XOR.
+    // Spreads the 16-lane ds_read_b128 groups from an 8-way bank conflict
+    // (128 B linear rows) to <= 2-way.
+    const int kc = (slot ^ (row & (LPR - 1))) * EPB;
+    const T* g = src + (int64_t)(row0 + row) * lda + k0 + kc;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g,
         (__attribute__((address_space(3))) void*)(lds + ci * (1024 / (int)sizeof(T))),
         16, 0, 0);
   }
+}
+
+// matching read-side XOR for glds-staged tiles: element offset within a
+// linear [row][BK] image whose 16 B slots are swizzled by row
+template <typename T>
+__device__ inline int glds_col(int row, int col) {
+  using TR = GemmTraits<T>;
+  constexpr int EPB = 16 / (int)sizeof(T);
+  constexpr int LPR = TR::BK / EPB;
+  int slot = col / EPB;
+  return (slot ^ (row & (LPR - 1))) * EPB + (col % EPB);
 }
 
 // Column offset inside an LDS tile row. K-major-staged tiles XOR the
@@ -275,14 +292,16 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
 #pragma unroll
           for (int f = 0; f < FM; ++f) {
             int row = wm + f * 16 + (lane & 15);
+            int col = kk + (lane >> 4) * (TR::KSTEP / 4);
             a_frag[f] = *reinterpret_cast<const typename TR::frag_t*>(
-                &al[row * BK + kk + (lane >> 4) * (TR::KSTEP / 4)]);
+                &al[row * BK + glds_col<T>(row, col)]);
           }
 #pragma unroll
           for (int f = 0; f < FN; ++f) {
             int row = wn + f * 16 + (lane & 15);
+            int col = kk + (lane >> 4) * (TR::KSTEP / 4);
             b_frag[f] = *reinterpret_cast<const typename TR::frag_t*>(
-                &bl[row * BK + kk + (lane >> 4) * (TR::KSTEP / 4)]);
+                &bl[row * BK + glds_col<T>(row, col)]);
           }
 #pragma unroll
           for (int fm = 0; fm < FM; ++fm)
