@@ -84,9 +84,12 @@ __device__ inline void stage_glds(T* lds, const T* __restrict__ src,
     const int row = ci * RPC + r_in;
     // rule 21: the LDS image is XOR-swizzled by swizzling the SOURCE
     // address per lane (glds writes lane-linear); reads apply the same XOR.
-    // Spreads the 16-lane ds_read_b128 groups from an 8-way bank conflict
-    // (128 B linear rows) to <= 2-way.
-    const int kc = (slot ^ (row & (LPR - 1))) * EPB;
+    // ONE-bit (32 B-pair) swizzle, the 8-phase template's st_16x32 form:
+    // full-slot permutations also kill bank conflicts but destroy the
+    // request coalescer's lane-order contiguity (13x slower at L3-resident
+    // sizes); the pair swap keeps 32 B runs contiguous and still cuts
+    // ds_read_b128 conflicts 8-way -> 4-way.
+    const int kc = (slot ^ (((row >> 2) & 1) << 1)) * EPB;
     const T* g = src + (int64_t)(row0 + row) * lda + k0 + kc;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g,
@@ -103,7 +106,7 @@ __device__ inline int glds_col(int row, int col) {
   constexpr int EPB = 16 / (int)sizeof(T);
   constexpr int LPR = TR::BK / EPB;
   int slot = col / EPB;
-  return (slot ^ (row & (LPR - 1))) * EPB + (col % EPB);
+  return (slot ^ (((row >> 2) & 1) << 1)) * EPB + (col % EPB);
 }
 
 // Column offset inside an LDS tile row. K-major-staged tiles XOR the
