@@ -92,14 +92,21 @@ class ConcatLayer(Layer):
         top[0].reshape(shape)
 
     def forward(self, bottom, top) -> None:
-        top[0].data = torch.cat([b.data for b in bottom], dim=self.dim)
+        if self.dim == 1 and len(bottom[0].shape) == 4:
+            top[0].data = ops.concat_channels([b.data for b in bottom])
+        else:
+            top[0].data = torch.cat([b.data for b in bottom], dim=self.dim)
 
     def backward(self, top, propagate_down, bottom) -> None:
         offset = 0
+        chan4 = self.dim == 1 and len(bottom[0].shape) == 4
         for i, b in enumerate(bottom):
             n = b.shape[self.dim]
             if propagate_down[i]:
-                b.diff = top[0].diff.narrow(self.dim, offset, n).contiguous()
+                if chan4:
+                    b.diff = ops.slice_channels(top[0].diff, offset, n)
+                else:
+                    b.diff = top[0].diff.narrow(self.dim, offset, n).contiguous()
             offset += n
 
 
@@ -129,12 +136,20 @@ class SliceLayer(Layer):
 
     def forward(self, bottom, top) -> None:
         offset = 0
+        chan4 = self.dim == 1 and len(bottom[0].shape) == 4
         for t, sz in zip(top, self.sizes):
-            t.data = bottom[0].data.narrow(self.dim, offset, sz).contiguous()
+            if chan4:
+                t.data = ops.slice_channels(bottom[0].data, offset, sz)
+            else:
+                t.data = bottom[0].data.narrow(self.dim, offset, sz).contiguous()
             offset += sz
 
     def backward(self, top, propagate_down, bottom) -> None:
-        if propagate_down[0]:
+        if not propagate_down[0]:
+            return
+        if self.dim == 1 and len(bottom[0].shape) == 4:
+            bottom[0].diff = ops.concat_channels([t.diff for t in top])
+        else:
             bottom[0].diff = torch.cat([t.diff for t in top], dim=self.dim)
 
 

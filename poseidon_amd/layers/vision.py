@@ -86,12 +86,15 @@ class ConvolutionLayer(Layer):
         for i, (bo, t) in enumerate(zip(bottom, top)):
             dy = t.diff
             db = self.blobs[1].diff.view(-1) if self.bias_term else None
+            cache = self._colT[i]
+            colT, wkT = cache if isinstance(cache, tuple) else (cache, None)
             ops.conv2d_backward_weight_acc(
-                bo.data, self._colT[i], dy, self.blobs[0].diff, db,
+                bo.data, colT, dy, self.blobs[0].diff, db,
                 self.stride, self.pad, self.group)
             if propagate_down[i]:
                 bo.diff = ops.conv2d_backward_input(
-                    w, dy, bo.shape, self.stride, self.pad, self.group)
+                    w, dy, bo.shape, self.stride, self.pad, self.group,
+                    wkT_cache=wkT)
         self._colT = []
 
 

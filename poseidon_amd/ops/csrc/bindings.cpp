@@ -221,6 +221,21 @@ Tensor weight_khwc(const Tensor& w, bool bf16) {
   return wk;
 }
 
+Tensor weight_khwc_tr(const Tensor& w, int G, bool bf16) {
+  auto wc = w.contiguous();
+  int Co = wc.size(0), Cig = wc.size(1), kh = wc.size(2), kw = wc.size(3);
+  int Kg = kh * kw * Cig;
+  Tensor wkT = at::empty({(int64_t)G * Kg, (int64_t)(Co / G)},
+                         w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
+  if (bf16)
+    ps_weight_to_khwc_tr_f32_bf16(wc.data_ptr<float>(), wkT.data_ptr(), Co,
+                                  Cig, kh, kw, G, stream());
+  else
+    ps_weight_to_khwc_tr_f32(wc.data_ptr<float>(), wkT.data_ptr<float>(), Co,
+                             Cig, kh, kw, G, stream());
+  return wkT;
+}
+
 std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
                                       const c10::optional<Tensor>& bias,
                                       int sh, int sw, int ph, int pw, int G) {
@@ -267,12 +282,16 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
              /*c_off=*/(int64_t)grp * Cog,
              true, true, 1.0f, 0.0f);
   }
-  return {y, colT};
+  // transposed repack for the dgrad NT GEMM, computed once per iteration
+  // and cached by the layer alongside colT
+  Tensor wkT = weight_khwc_tr(w, G, bf16);
+  return {y, colT, wkT};
 }
 
 Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
                              std::vector<int64_t> x_shape, int sh, int sw,
-                             int ph, int pw, int G) {
+                             int ph, int pw, int G,
+                             const c10::optional<Tensor>& wkT_cache) {
   check_float_like(dy, "dy");
   const bool bf16 = is_bf16(dy);
   auto dy_cl = cl4(dy);
@@ -290,15 +309,10 @@ Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
 
   // transposed khwc repack [G][Kg][Cog]: the dgrad GEMM becomes pure NT
   // (both operands K-last) instead of a K-major-staged NN
-  auto wc2 = w.contiguous();
-  Tensor wkT = at::empty({(int64_t)G * Kg, (int64_t)Cog},
-                         w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
-  if (bf16)
-    ps_weight_to_khwc_tr_f32_bf16(wc2.data_ptr<float>(), wkT.data_ptr(),
-                                  Co, w.size(1), kh, kw, G, stream());
-  else
-    ps_weight_to_khwc_tr_f32(wc2.data_ptr<float>(), wkT.data_ptr<float>(),
-                             Co, w.size(1), kh, kw, G, stream());
+  Tensor wkT = wkT_cache.has_value() && wkT_cache->scalar_type() ==
+                       (bf16 ? at::kBFloat16 : at::kFloat)
+                   ? *wkT_cache
+                   : weight_khwc_tr(w, G, bf16);
 
   bool is_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0);
   Tensor dcolT = is_1x1 ? rows2d(dx)
@@ -359,6 +373,49 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
       ps_colsum_f32(dy_cl.data_ptr<float>(), db_out->data_ptr<float>(), NP,
                     Co, stream());
   }
+}
+
+// ---------------------------------------------------------------------------
+// NHWC channel concat / slice (CONCAT + SLICE layers, GoogLeNet inception)
+// ---------------------------------------------------------------------------
+
+Tensor concat_channels(std::vector<Tensor> inputs) {
+  TORCH_CHECK(!inputs.empty());
+  auto x0 = cl4(inputs[0]);
+  int64_t N = x0.size(0), H = x0.size(2), W = x0.size(3);
+  int64_t rows = N * H * W;
+  int C_out = 0;
+  for (auto& t : inputs) C_out += t.size(1);
+  Tensor y = at::empty({N, (int64_t)C_out, H, W},
+                       x0.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int off = 0;
+  for (auto& t : inputs) {
+    auto tc = cl4(t);
+    int Ci = tc.size(1);
+    if (is_bf16(tc))
+      ps_chan_copy_bf16(tc.data_ptr(), y.data_ptr(), rows, Ci, C_out, off,
+                        stream());
+    else
+      ps_chan_copy_f32(tc.data_ptr<float>(), y.data_ptr<float>(), rows, Ci,
+                       C_out, off, stream());
+    off += Ci;
+  }
+  return y;
+}
+
+Tensor slice_channels(const Tensor& x, int64_t c_off, int64_t c_len) {
+  auto xc = cl4(x);
+  int64_t N = xc.size(0), H = xc.size(2), W = xc.size(3);
+  int64_t rows = N * H * W;
+  Tensor y = at::empty({N, c_len, H, W},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  if (is_bf16(xc))
+    ps_chan_slice_bf16(xc.data_ptr(), y.data_ptr(), rows, xc.size(1),
+                       (int)c_len, (int)c_off, stream());
+  else
+    ps_chan_slice_f32(xc.data_ptr<float>(), y.data_ptr<float>(), rows,
+                      xc.size(1), (int)c_len, (int)c_off, stream());
+  return y;
 }
 
 // ---------------------------------------------------------------------------
@@ -779,6 +836,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_forward_ex", &conv2d_forward_ex);
   m.def("conv2d_backward_input", &conv2d_backward_input);
   m.def("conv2d_backward_weight_acc", &conv2d_backward_weight_acc);
+  m.def("concat_channels", &concat_channels);
+  m.def("slice_channels", &slice_channels);
   m.def("pool_max_forward", &pool_max_forward);
   m.def("pool_max_backward", &pool_max_backward);
   m.def("pool_ave_forward", &pool_ave_forward);

@@ -222,7 +222,77 @@ __global__ void nchw_to_nhwc_k(const TI* src, TO* dst, int N, int C, int H, int 
   }
 }
 
+// NHWC channel-block copy: out[:, c_off:c_off+C_in] = in (rows = N*H*W).
+// Serves CONCAT forward (one call per bottom) and SLICE/CONCAT backward.
+template <typename T, int V>
+__global__ void chan_copy_k(const T* __restrict__ in, T* __restrict__ out,
+                            int64_t rows, int C_in, int C_out, int c_off) {
+  typedef T vec_t __attribute__((ext_vector_type(V)));
+  const int CV = C_in / V;
+  int64_t total = rows * CV;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int cv = i % CV;
+    int64_t r = i / CV;
+    *reinterpret_cast<vec_t*>(&out[r * C_out + c_off + cv * V]) =
+        *reinterpret_cast<const vec_t*>(&in[r * C_in + cv * V]);
+  }
+}
+
+// gather variant: out = in[:, c_off:c_off+C_out] (slice forward / concat bwd)
+template <typename T, int V>
+__global__ void chan_slice_k(const T* __restrict__ in, T* __restrict__ out,
+                             int64_t rows, int C_in, int C_out, int c_off) {
+  typedef T vec_t __attribute__((ext_vector_type(V)));
+  const int CV = C_out / V;
+  int64_t total = rows * CV;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int cv = i % CV;
+    int64_t r = i / CV;
+    *reinterpret_cast<vec_t*>(&out[r * C_out + cv * V]) =
+        *reinterpret_cast<const vec_t*>(&in[r * C_in + c_off + cv * V]);
+  }
+}
+
 extern "C" {
+
+void ps_chan_copy_f32(const float* in, float* out, int64_t rows, int C_in,
+                      int C_out, int c_off, hipStream_t s) {
+  if (C_in % 4 == 0 && c_off % 4 == 0)
+    chan_copy_k<float, 4><<<ew_grid(rows * (C_in / 4)), 256, 0, s>>>(
+        in, out, rows, C_in, C_out, c_off);
+  else
+    chan_copy_k<float, 1><<<ew_grid(rows * C_in), 256, 0, s>>>(
+        in, out, rows, C_in, C_out, c_off);
+}
+void ps_chan_copy_bf16(const void* in, void* out, int64_t rows, int C_in,
+                       int C_out, int c_off, hipStream_t s) {
+  if (C_in % 8 == 0 && c_off % 8 == 0)
+    chan_copy_k<__bf16, 8><<<ew_grid(rows * (C_in / 8)), 256, 0, s>>>(
+        (const __bf16*)in, (__bf16*)out, rows, C_in, C_out, c_off);
+  else
+    chan_copy_k<__bf16, 1><<<ew_grid(rows * C_in), 256, 0, s>>>(
+        (const __bf16*)in, (__bf16*)out, rows, C_in, C_out, c_off);
+}
+void ps_chan_slice_f32(const float* in, float* out, int64_t rows, int C_in,
+                       int C_out, int c_off, hipStream_t s) {
+  if (C_out % 4 == 0 && c_off % 4 == 0)
+    chan_slice_k<float, 4><<<ew_grid(rows * (C_out / 4)), 256, 0, s>>>(
+        in, out, rows, C_in, C_out, c_off);
+  else
+    chan_slice_k<float, 1><<<ew_grid(rows * C_out), 256, 0, s>>>(
+        in, out, rows, C_in, C_out, c_off);
+}
+void ps_chan_slice_bf16(const void* in, void* out, int64_t rows, int C_in,
+                        int C_out, int c_off, hipStream_t s) {
+  if (C_out % 8 == 0 && c_off % 8 == 0)
+    chan_slice_k<__bf16, 8><<<ew_grid(rows * (C_out / 8)), 256, 0, s>>>(
+        (const __bf16*)in, (__bf16*)out, rows, C_in, C_out, c_off);
+  else
+    chan_slice_k<__bf16, 1><<<ew_grid(rows * C_out), 256, 0, s>>>(
+        (const __bf16*)in, (__bf16*)out, rows, C_in, C_out, c_off);
+}
 
 void ps_im2col_nhwc_f32(const float* x, float* colT, const ConvGeom* g, hipStream_t s) {
   int Cg = g->C / g->G;

@@ -145,6 +145,10 @@ void ps_dropout_fwd_bf16_offdev(const void*, void*, uint8_t*, int64_t, float,
                                 uint64_t, const void*, hipStream_t);
 
 // im2col.hip
+void ps_chan_copy_f32(const float*, float*, int64_t, int, int, int, hipStream_t);
+void ps_chan_copy_bf16(const void*, void*, int64_t, int, int, int, hipStream_t);
+void ps_chan_slice_f32(const float*, float*, int64_t, int, int, int, hipStream_t);
+void ps_chan_slice_bf16(const void*, void*, int64_t, int, int, int, hipStream_t);
 void ps_im2col_nhwc_f32(const float*, float*, const ConvGeom*, hipStream_t);
 void ps_col2im_nhwc_f32(const float*, float*, const ConvGeom*, hipStream_t);
 void ps_im2col_nhwc_bf16(const void*, void*, const ConvGeom*, hipStream_t);

@@ -77,17 +77,19 @@ def conv2d_forward_ex(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor
     """Returns (y, colT_cache). colT is the im2col matrix on GPU (reused by
     the backward GEMMs); None on CPU."""
     if x.is_cuda:
-        y, colT = _ext().conv2d_forward_ex(x, w, b, stride[0], stride[1],
-                                           pad[0], pad[1], groups)
-        return y, colT
+        y, colT, wkT = _ext().conv2d_forward_ex(x, w, b, stride[0], stride[1],
+                                                pad[0], pad[1], groups)
+        return y, (colT, wkT)
     return F.conv2d(x, w, b, stride=stride, padding=pad, groups=groups), None
 
 
 def conv2d_backward_input(w: torch.Tensor, dy: torch.Tensor,
-                          x_shape, stride, pad, groups: int) -> torch.Tensor:
+                          x_shape, stride, pad, groups: int,
+                          wkT_cache=None) -> torch.Tensor:
     if dy.is_cuda:
         return _ext().conv2d_backward_input(w, dy, list(x_shape), stride[0],
-                                            stride[1], pad[0], pad[1], groups)
+                                            stride[1], pad[0], pad[1], groups,
+                                            wkT_cache)
     return torch.nn.grad.conv2d_input(list(x_shape), w, dy, stride=stride,
                                       padding=pad, groups=groups)
 
@@ -448,6 +450,22 @@ def adagrad_update(w, grad, hist, local_rate: float, delta: float,
     gw = grad if decay == 0.0 else grad + decay * w
     hist.add_(gw * gw)
     w.sub_(local_rate * gw / (hist.sqrt() + delta))
+
+
+# ---------------------------------------------------------------------------
+# NHWC channel concat / slice (CONCAT + SLICE layers)
+# ---------------------------------------------------------------------------
+
+def concat_channels(tensors):
+    if tensors[0].is_cuda and tensors[0].dim() == 4:
+        return _ext().concat_channels(list(tensors))
+    return torch.cat(tensors, dim=1)
+
+
+def slice_channels(x, c_off: int, c_len: int):
+    if x.is_cuda and x.dim() == 4:
+        return _ext().slice_channels(x, c_off, c_len)
+    return x.narrow(1, c_off, c_len).contiguous()
 
 
 # ---------------------------------------------------------------------------

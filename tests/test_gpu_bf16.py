@@ -52,7 +52,8 @@ def test_conv_bf16():
     stride, pad, g = (1, 1), (2, 2), 2
     y_ref, _ = ops.conv2d_forward_ex(x, w, b, stride, pad, g)
     xg = x.to(DEV, torch.bfloat16)
-    y, colT = ops.conv2d_forward_ex(xg, w.to(DEV), b.to(DEV), stride, pad, g)
+    y, cache = ops.conv2d_forward_ex(xg, w.to(DEV), b.to(DEV), stride, pad, g)
+    colT, _wkT = cache
     assert y.dtype == torch.bfloat16
     close_bf16(y, y_ref, what="conv bf16 fwd")
 

@@ -95,7 +95,8 @@ def test_conv_forward_backward(cfg):
     stride, pad = (s, s), (p, p)
 
     y_ref, _ = ops.conv2d_forward_ex(x, w, b, stride, pad, g)
-    y, colT = ops.conv2d_forward_ex(x.to(DEV), w.to(DEV), b.to(DEV), stride, pad, g)
+    y, cache = ops.conv2d_forward_ex(x.to(DEV), w.to(DEV), b.to(DEV), stride, pad, g)
+    colT, _wkT = cache
     close(y, y_ref, what="conv fwd")
 
     dy = rnd(*y_ref.shape, seed=14)
