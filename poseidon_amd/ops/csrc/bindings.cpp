@@ -58,12 +58,22 @@ Tensor weight_shadow(const Tensor& w_f32, bool bf16) {
 // GEMM plumbing
 // ---------------------------------------------------------------------------
 
+// 64B zero page for implicit-GEMM padding loads (per device, persistent)
+const void* zero_page(const Tensor& like) {
+  static Tensor z;
+  if (!z.defined())
+    z = at::zeros({64}, like.options().dtype(at::kByte));
+  return z.data_ptr();
+}
+
 // in_bf16: A/B are bf16; out_f32: C is fp32 (else C matches input dtype)
 void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
               const float* bias, int M, int N, int K,
               int64_t lda, int64_t ldb, int64_t ldc,
               int64_t a_off, int64_t b_off, int64_t c_off,
-              bool a_klast, bool b_klast, float alpha, float beta) {
+              bool a_klast, bool b_klast, float alpha, float beta,
+              const GatherDesc* gather_a = nullptr,
+              const GatherDesc* gather_b = nullptr) {
   const bool in_bf16 = is_bf16(A);
   TORCH_CHECK(is_bf16(B) == in_bf16, "gemm: A/B dtype mismatch");
   const bool out_f32 = !is_bf16(C);
@@ -83,6 +93,8 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   g.a_klast = a_klast; g.b_klast = b_klast;
   g.ws = nullptr;
   g.splitk = 1;
+  g.gather_a = gather_a;
+  g.gather_b = gather_b;
   // Split-K when the output tile grid cannot fill 256 CUs but K is deep
   // (conv wgrad: M=Cout, N=Kcol, K=N*OH*OW up to ~800k): target ~512
   // workgroups, cap the f32 workspace at 256 MB.
@@ -254,10 +266,16 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   Tensor wk = weight_khwc(w, bf16);
 
   bool is_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0);
+  const int VEC = bf16 ? 8 : 4;
+  // implicit GEMM: gather im2col rows inside the GEMM staging whenever the
+  // group channel count keeps 16B runs contiguous -- no column matrix at
+  // all (the wgrad GEMM gathers too); conv1-style small-C layers still
+  // materialize (and 1x1 convs alias x directly)
+  bool implicit = !is_1x1 && (Cg % VEC == 0);
   Tensor colT;
   if (is_1x1) {
     colT = rows2d(x_cl);  // alias: x rows ARE the col rows
-  } else {
+  } else if (!implicit) {
     colT = at::empty({NP, (int64_t)Kcol}, x.options());
     if (bf16)
       ps_im2col_nhwc_bf16(x_cl.data_ptr(), colT.data_ptr(), &g, stream());
@@ -275,16 +293,33 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
     bp = bc.data_ptr<float>();
   }
   for (int grp = 0; grp < G; ++grp) {
-    run_gemm(colT, wk, y, bp ? bp + grp * Cog : nullptr,
-             (int)NP, Cog, Kg,
-             /*lda=*/Kcol, /*ldb=*/Kg, /*ldc=*/Co,
-             /*a_off=*/(int64_t)grp * Kg, /*b_off=*/(int64_t)grp * Cog * Kg,
-             /*c_off=*/(int64_t)grp * Cog,
-             true, true, 1.0f, 0.0f);
+    if (implicit) {
+      GatherDesc ga;
+      ga.x = x_cl.data_ptr();
+      ga.zero = zero_page(x_cl);
+      ga.C = g.C; ga.H = g.H; ga.W = g.W; ga.Ho = g.Ho; ga.Wo = g.Wo;
+      ga.kh = kh; ga.kw = kw; ga.sh = sh; ga.sw = sw; ga.ph = ph; ga.pw = pw;
+      ga.Cg = Cg; ga.c0 = grp * Cg;
+      run_gemm(x_cl, wk, y, bp ? bp + grp * Cog : nullptr,
+               (int)NP, Cog, Kg,
+               /*lda=*/Kg, /*ldb=*/Kg, /*ldc=*/Co,
+               /*a_off=*/0, /*b_off=*/(int64_t)grp * Cog * Kg,
+               /*c_off=*/(int64_t)grp * Cog,
+               true, true, 1.0f, 0.0f, &ga, nullptr);
+    } else {
+      run_gemm(colT, wk, y, bp ? bp + grp * Cog : nullptr,
+               (int)NP, Cog, Kg,
+               /*lda=*/Kcol, /*ldb=*/Kg, /*ldc=*/Co,
+               /*a_off=*/(int64_t)grp * Kg, /*b_off=*/(int64_t)grp * Cog * Kg,
+               /*c_off=*/(int64_t)grp * Cog,
+               true, true, 1.0f, 0.0f);
+    }
   }
   // transposed repack for the dgrad NT GEMM, computed once per iteration
   // and cached by the layer alongside colT
   Tensor wkT = weight_khwc_tr(w, G, bf16);
+  if (!colT.defined())
+    colT = at::empty({0}, x.options());  // implicit mode: wgrad gathers too
   return {y, colT, wkT};
 }
 
@@ -344,24 +379,42 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
                                 int sh, int sw, int ph, int pw, int G) {
   check_float_like(dy, "dy");
   auto dy_cl = cl4(dy);
+  auto x_cl = cl4(x);
   int Co = dw_out.size(0), Cig = dw_out.size(1);
   int kh = dw_out.size(2), kw = dw_out.size(3);
   int Cog = Co / G;
   int Kg = kh * kw * Cig;
   int64_t NP = (int64_t)dy_cl.size(0) * dy_cl.size(2) * dy_cl.size(3);
-  int64_t Kcol = colT.size(1);
+  const bool implicit = colT.numel() == 0 && colT.dim() == 1;
+  int64_t Kcol = implicit ? 0 : colT.size(1);
 
   // fp32 gradient accumulation regardless of activation dtype
   Tensor dwk = at::empty({Co, Kg}, dy.options().dtype(at::kFloat));
   Tensor dy2 = rows2d(dy_cl);
   for (int grp = 0; grp < G; ++grp) {
     // dwk_g[Cog, Kg] = dy_g^T[Cog, NP] @ colT_g[NP, Kg]: contraction NP
-    run_gemm(dy2, colT, dwk, nullptr,
-             Cog, Kg, (int)NP,
-             /*lda=*/Co, /*ldb=*/Kcol, /*ldc=*/Kg,
-             /*a_off=*/(int64_t)grp * Cog, /*b_off=*/(int64_t)grp * Kg,
-             /*c_off=*/(int64_t)grp * Cog * Kg,
-             false, false, 1.0f, 0.0f);
+    if (implicit) {
+      GatherDesc gb;
+      gb.x = x_cl.data_ptr();
+      gb.zero = zero_page(x_cl);
+      gb.C = x_cl.size(1); gb.H = x_cl.size(2); gb.W = x_cl.size(3);
+      gb.Ho = dy_cl.size(2); gb.Wo = dy_cl.size(3);
+      gb.kh = kh; gb.kw = kw; gb.sh = sh; gb.sw = sw; gb.ph = ph; gb.pw = pw;
+      gb.Cg = Cig; gb.c0 = grp * Cig;
+      run_gemm(dy2, x_cl, dwk, nullptr,
+               Cog, Kg, (int)NP,
+               /*lda=*/Co, /*ldb=*/Kg, /*ldc=*/Kg,
+               /*a_off=*/(int64_t)grp * Cog, /*b_off=*/0,
+               /*c_off=*/(int64_t)grp * Cog * Kg,
+               false, false, 1.0f, 0.0f, nullptr, &gb);
+    } else {
+      run_gemm(dy2, colT, dwk, nullptr,
+               Cog, Kg, (int)NP,
+               /*lda=*/Co, /*ldb=*/Kcol, /*ldc=*/Kg,
+               /*a_off=*/(int64_t)grp * Cog, /*b_off=*/(int64_t)grp * Kg,
+               /*c_off=*/(int64_t)grp * Cog * Kg,
+               false, false, 1.0f, 0.0f);
+    }
   }
   ps_weight_from_khwc_f32(dwk.data_ptr<float>(), dw_out.data_ptr<float>(),
                           Co, Cig, kh, kw, /*beta=*/1.0f, stream());

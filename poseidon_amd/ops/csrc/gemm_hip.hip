@@ -62,6 +62,27 @@ template <> struct GemmTraits<__bf16> {
   }
 };
 
+// Implicit-GEMM address: row = im2col row (n,oh,ow), k = column within the
+// group slice (khw*Cg + cg). Returns the NHWC x address of k's element run
+// (contiguous along cg), or the zero page for padding. Valid for vector
+// loads that stay inside one cg-run (callers guarantee k%VEC==0, Cg%VEC==0).
+template <typename T>
+__device__ inline const T* gather_addr(const GatherDesc& ga, int64_t row, int k) {
+  int khw = k / ga.Cg;
+  int cg = k - khw * ga.Cg;
+  int kkh = khw / ga.kw, kkw = khw - kkh * ga.kw;
+  int ow = row % ga.Wo;
+  int64_t t = row / ga.Wo;
+  int oh = t % ga.Ho;
+  int n = t / ga.Ho;
+  int ih = oh * ga.sh - ga.ph + kkh;
+  int iw = ow * ga.sw - ga.pw + kkw;
+  if (ih < 0 || ih >= ga.H || iw < 0 || iw >= ga.W)
+    return (const T*)ga.zero;
+  return (const T*)ga.x + (((int64_t)n * ga.H + ih) * ga.W + iw) * ga.C
+         + ga.c0 + cg;
+}
+
 // Direct global->LDS staging (glds): each wave-instruction moves 1 KiB
 // (64 lanes x 16 B) HBM -> LDS without a VGPR round trip
 // (__builtin_amdgcn_global_load_lds, width 16 -- guide §5 step 3: the
@@ -69,10 +90,10 @@ template <> struct GemmTraits<__bf16> {
 // (the builtin writes wave-uniform-base + lane*16), so this path is used
 // for interior tiles of K-last operands only; fragment reads use the
 // unpadded BK stride.
-template <typename T, int ROWS>
+template <typename T, int ROWS, bool GATHER = false>
 __device__ inline void stage_glds(T* lds, const T* __restrict__ src,
                                   int64_t lda, int row0, int k0, int wid,
-                                  int lane) {
+                                  int lane, const GatherDesc* ga = nullptr) {
   using TR = GemmTraits<T>;
   constexpr int EPB = 16 / sizeof(T);            // elems per 16B lane-load
   constexpr int LPR = TR::BK / EPB;              // lanes per row
@@ -91,7 +112,8 @@ __device__ inline void stage_glds(T* lds, const T* __restrict__ src,
     // sizes); the pair swap keeps 32 B runs contiguous and still cuts
     // ds_read_b128 conflicts 8-way -> 4-way.
     const int kc = (slot ^ (((row >> 2) & 1) << 1)) * EPB;
-    const T* g = src + (int64_t)(row0 + row) * lda + k0 + kc;
+    const T* g = GATHER ? gather_addr<T>(*ga, row0 + row, k0 + kc)
+                        : src + (int64_t)(row0 + row) * lda + k0 + kc;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g,
         (__attribute__((address_space(3))) void*)(lds + ci * (1024 / (int)sizeof(T))),
@@ -128,7 +150,7 @@ __device__ inline int lds_col(int row, int kk) {
 // phase (guide T14 / Guideline 15: write AFTER the barrier, re-issue at
 // once). A [ROWS x BK] tile is ROWS*BK/(256*VEC) vectors per thread.
 // Guarded, zero-filled out of range.
-template <typename T, int ROWS, bool KLAST>
+template <typename T, int ROWS, bool KLAST, bool GATHER = false>
 struct Stager {
   using TR = GemmTraits<T>;
   using vec_t = typename TR::vec_t;
@@ -140,7 +162,8 @@ struct Stager {
   vec_t v[KLAST ? NV : NVB * TR::VEC];
 
   __device__ inline void load(const T* src, int64_t lda, int row0,
-                              int rows_max, int k0, int K, int tid) {
+                              int rows_max, int k0, int K, int tid,
+                              const GatherDesc* ga = nullptr) {
     if (KLAST) {
       constexpr int CK = TR::BK / TR::VEC;
 #pragma unroll
@@ -151,7 +174,12 @@ struct Stager {
         int gr = row0 + r, gk = k0 + kc * TR::VEC;
         vec_t val = {};
         if (gr < rows_max && gk < K) {
-          if (gk + TR::VEC <= K) {
+          if (GATHER) {
+            // implicit im2col: one VEC run per (row, k-chunk); bindings
+            // guarantee Cg % VEC == 0 so the run is contiguous
+            val = *reinterpret_cast<const vec_t*>(
+                gather_addr<T>(*ga, gr, gk));
+          } else if (gk + TR::VEC <= K) {
             val = *reinterpret_cast<const vec_t*>(&src[(int64_t)gr * lda + gk]);
           } else {
             for (int j = 0; j < TR::VEC; ++j)
@@ -159,6 +187,31 @@ struct Stager {
           }
         }
         v[i] = val;
+      }
+    } else if (GATHER) {
+      // K-major gather (conv wgrad B = im2col): rows dim is Kg (contiguous
+      // within cg runs), K dim is NP (im2col rows)
+      constexpr int BLK_M = ROWS / TR::VEC;
+#pragma unroll
+      for (int i = 0; i < NVB; ++i) {
+        int c = tid + i * 256;
+        if (c >= TB) break;
+        int kb = c / BLK_M, mb = c % BLK_M;
+        int gm = row0 + mb * TR::VEC;  // kg (column of im2col)
+#pragma unroll
+        for (int j = 0; j < TR::VEC; ++j) {
+          int gk = k0 + kb * TR::VEC + j;  // np (im2col row)
+          vec_t val = {};
+          if (gk < K && gm + TR::VEC <= rows_max)
+            val = *reinterpret_cast<const vec_t*>(
+                gather_addr<T>(*ga, gk, gm));
+          else if (gk < K && gm < rows_max) {
+            for (int e = 0; e < TR::VEC; ++e)
+              if (gm + e < rows_max)
+                val[e] = *gather_addr<T>(*ga, gk, gm + e);
+          }
+          v[i * TR::VEC + j] = val;
+        }
       }
     } else {
       // K-major: each thread owns a VEC x VEC block (k-block kb, m-block mb)
@@ -225,7 +278,8 @@ struct Stager {
 // Tile geometry: BM x BN block tile, 4 waves arranged WGM x WGN, each wave
 // owns a (BM/WGM) x (BN/WGN) sub-tile as FM x FN fragments of 16x16.
 template <typename T, typename OUT, int BM, int BN, int WGM, int WGN,
-          bool A_KLAST, bool B_KLAST, bool HAS_BIAS, bool SPLITK>
+          bool A_KLAST, bool B_KLAST, bool HAS_BIAS, bool SPLITK,
+          bool GA = false, bool GB = false>
 __global__ __launch_bounds__(256)
 void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
                  OUT* __restrict__ Cbase, const float* __restrict__ bias,
@@ -233,7 +287,8 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
                  int64_t lda, int64_t ldb, int64_t ldc,
                  int64_t strideA, int64_t strideB, int64_t strideC,
                  float alpha, float beta,
-                 float* __restrict__ ws, int kchunk) {
+                 float* __restrict__ ws, int kchunk,
+                 GatherDesc ga_a = {}, GatherDesc ga_b = {}) {
   using TR = GemmTraits<T>;
   constexpr int BK = TR::BK, RS = TR::RS;
   constexpr int FM = BM / WGM / 16, FN = BN / WGN / 16;
@@ -268,24 +323,33 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
   {
     constexpr int EPB = 16 / (int)sizeof(T);
     const int k_span = k_end - k_begin;
-    const bool glds_ok = A_KLAST && B_KLAST && (m0 + BM <= M) &&
-                         (n0 + BN <= N) && (k_span % BK) == 0 && k_span > 0 &&
-                         (lda % EPB) == 0 && (ldb % EPB) == 0 &&
-                         (k_begin % EPB) == 0 &&
-                         (((uintptr_t)A & 15) == 0) && (((uintptr_t)B & 15) == 0);
+    bool glds_ok = A_KLAST && B_KLAST && (m0 + BM <= M) &&
+                   (n0 + BN <= N) && (k_span % BK) == 0 && k_span > 0 &&
+                   (ldb % EPB) == 0 && (k_begin % EPB) == 0 &&
+                   (((uintptr_t)B & 15) == 0);
+    if (GA) {
+      // gathered A: every BK-wide row must stay inside one cg run and the
+      // run base must be 16B-aligned
+      glds_ok = glds_ok && (ga_a.Cg % BK) == 0 && (ga_a.C % EPB) == 0 &&
+                (ga_a.c0 % EPB) == 0;
+    } else {
+      glds_ok = glds_ok && (lda % EPB) == 0 && (((uintptr_t)A & 15) == 0);
+    }
     if (glds_ok) {
       T* a_lin0 = a_lds[0];
       T* b_lin0 = b_lds[0];
       T* a_lin1 = a_lds[1];
       T* b_lin1 = b_lds[1];
-      stage_glds<T, BM>(a_lin0, A, lda, m0, k_begin, wid, lane);
+      stage_glds<T, BM, GA>(a_lin0, A, lda, m0, k_begin, wid, lane, &ga_a);
       stage_glds<T, BN>(b_lin0, B, ldb, n0, k_begin, wid, lane);
       __syncthreads();  // drains the in-flight glds (vmcnt 0) + barrier
       int cur2 = 0;
       for (int k0 = k_begin; k0 < k_end; k0 += BK) {
         if (k0 + BK < k_end) {
-          stage_glds<T, BM>(cur2 ? a_lin0 : a_lin1, A, lda, m0, k0 + BK, wid, lane);
-          stage_glds<T, BN>(cur2 ? b_lin0 : b_lin1, B, ldb, n0, k0 + BK, wid, lane);
+          stage_glds<T, BM, GA>(cur2 ? a_lin0 : a_lin1, A, lda, m0, k0 + BK,
+                                wid, lane, &ga_a);
+          stage_glds<T, BN>(cur2 ? b_lin0 : b_lin1, B, ldb, n0, k0 + BK, wid,
+                            lane);
         }
         const T* al = cur2 ? a_lin1 : a_lin0;
         const T* bl = cur2 ? b_lin1 : b_lin0;
@@ -319,16 +383,16 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
     }
   }
   {
-  Stager<T, BM, A_KLAST> sa;
-  Stager<T, BN, B_KLAST> sb;
+  Stager<T, BM, A_KLAST, GA> sa;
+  Stager<T, BN, B_KLAST, GB> sb;
   // prologue: tile 0 -> LDS[0]; issue tile 1 loads
-  sa.load(A, lda, m0, M, k_begin, k_end, tid);
-  sb.load(B, ldb, n0, N, k_begin, k_end, tid);
+  sa.load(A, lda, m0, M, k_begin, k_end, tid, &ga_a);
+  sb.load(B, ldb, n0, N, k_begin, k_end, tid, &ga_b);
   sa.write(a_lds[0], tid);
   sb.write(b_lds[0], tid);
   if (k_begin + BK < k_end) {
-    sa.load(A, lda, m0, M, k_begin + BK, k_end, tid);
-    sb.load(B, ldb, n0, N, k_begin + BK, k_end, tid);
+    sa.load(A, lda, m0, M, k_begin + BK, k_end, tid, &ga_a);
+    sb.load(B, ldb, n0, N, k_begin + BK, k_end, tid, &ga_b);
   }
   __syncthreads();
 
@@ -340,8 +404,8 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
       sa.write(a_lds[cur ^ 1], tid);
       sb.write(b_lds[cur ^ 1], tid);
       if (k0 + 2 * BK < k_end) {
-        sa.load(A, lda, m0, M, k0 + 2 * BK, k_end, tid);
-        sb.load(B, ldb, n0, N, k0 + 2 * BK, k_end, tid);
+        sa.load(A, lda, m0, M, k0 + 2 * BK, k_end, tid, &ga_a);
+        sb.load(B, ldb, n0, N, k0 + 2 * BK, k_end, tid, &ga_b);
       }
     }
 
@@ -427,8 +491,10 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
 // ---------------------------------------------------------------------------
 
 template <typename T, typename OUT, int BM, int BN, int WGM, int WGN,
-          bool AK, bool BK_, bool HB>
+          bool AK, bool BK_, bool HB, bool GA = false, bool GB = false>
 static void launch_tile(const GemmArgs& g, hipStream_t s) {
+  GatherDesc da = GA ? *g.gather_a : GatherDesc{};
+  GatherDesc db = GB ? *g.gather_b : GatherDesc{};
   const bool sk = g.splitk > 1;
   dim3 grid(cdiv(g.N, BN), cdiv(g.M, BM), sk ? g.splitk : g.batch);
   dim3 block(256);
@@ -438,11 +504,11 @@ static void launch_tile(const GemmArgs& g, hipStream_t s) {
     kchunk = cdiv(cdiv(g.K, g.splitk), TBK) * TBK;
   }
   if (sk) {
-   hipLaunchKernelGGL(( gemm_kernel<T, OUT, BM, BN, WGM, WGN, AK, BK_, HB, true>)
+   hipLaunchKernelGGL(( gemm_kernel<T, OUT, BM, BN, WGM, WGN, AK, BK_, HB, true, GA, GB>)
         , dim3(grid), dim3(block), 0, s, 
             (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,
             g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha,
-            g.beta, (float*)g.ws, kchunk);
+            g.beta, (float*)g.ws, kchunk, da, db);
     int64_t MN = (int64_t)g.M * g.N;
     int64_t rb = cdiv64(MN, 256);
     if (rb > 2048) rb = 2048;
@@ -455,41 +521,52 @@ static void launch_tile(const GemmArgs& g, hipStream_t s) {
           (const float*)g.ws, (OUT*)g.C, g.bias, g.M, g.N, g.ldc, g.splitk,
           g.alpha, g.beta);
   } else {
-   hipLaunchKernelGGL(( gemm_kernel<T, OUT, BM, BN, WGM, WGN, AK, BK_, HB, false>)
+   hipLaunchKernelGGL(( gemm_kernel<T, OUT, BM, BN, WGM, WGN, AK, BK_, HB, false, GA, GB>)
         , dim3(grid), dim3(block), 0, s, 
             (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,
             g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha,
-            g.beta, nullptr, 0);
+            g.beta, nullptr, 0, da, db);
   }
 }
 
-template <typename T, typename OUT, bool AK, bool BK_, bool HB>
+template <typename T, typename OUT, bool AK, bool BK_, bool HB,
+          bool GA = false, bool GB = false>
 static void dispatch_tiles(const GemmArgs& g, hipStream_t s) {
   int bm, bn;
   ps_pick_gemm_tile(g.M, g.N, &bm, &bn);
   if (bm == 128 && bn == 32)
-    launch_tile<T, OUT, 128, 32, 4, 1, AK, BK_, HB>(g, s);
+    launch_tile<T, OUT, 128, 32, 4, 1, AK, BK_, HB, GA, GB>(g, s);
   else if (bm == 32 && bn == 128)
-    launch_tile<T, OUT, 32, 128, 1, 4, AK, BK_, HB>(g, s);
+    launch_tile<T, OUT, 32, 128, 1, 4, AK, BK_, HB, GA, GB>(g, s);
   else if (bm == 64 && bn == 64)
-    launch_tile<T, OUT, 64, 64, 2, 2, AK, BK_, HB>(g, s);
+    launch_tile<T, OUT, 64, 64, 2, 2, AK, BK_, HB, GA, GB>(g, s);
   else
-    launch_tile<T, OUT, 128, 128, 2, 2, AK, BK_, HB>(g, s);
+    launch_tile<T, OUT, 128, 128, 2, 2, AK, BK_, HB, GA, GB>(g, s);
 }
 
 template <typename T, typename OUT>
 static void gemm_dispatch(const GemmArgs& g, hipStream_t s) {
   const bool hb = g.bias != nullptr;
   if (g.a_klast && g.b_klast) {
-    if (hb) dispatch_tiles<T, OUT, true, true, true>(g, s);
-    else dispatch_tiles<T, OUT, true, true, false>(g, s);
+    if (g.gather_a) {
+      // implicit-GEMM conv forward: A rows gathered from NHWC x
+      if (hb) dispatch_tiles<T, OUT, true, true, true, true, false>(g, s);
+      else dispatch_tiles<T, OUT, true, true, false, true, false>(g, s);
+    } else if (hb) {
+      dispatch_tiles<T, OUT, true, true, true>(g, s);
+    } else {
+      dispatch_tiles<T, OUT, true, true, false>(g, s);
+    }
   } else if (g.a_klast && !g.b_klast) {
     dispatch_tiles<T, OUT, true, false, false>(g, s);
   } else if (!g.a_klast && g.b_klast) {
     // unused in the framework (kept for the generic test entry): 128x128 only
     launch_tile<T, OUT, 128, 128, 2, 2, false, true, false>(g, s);
   } else {
-    dispatch_tiles<T, OUT, false, false, false>(g, s);
+    if (g.gather_b)  // implicit-GEMM conv wgrad: K-major B gathered from x
+      dispatch_tiles<T, OUT, false, false, false, false, true>(g, s);
+    else
+      dispatch_tiles<T, OUT, false, false, false>(g, s);
   }
 }
 

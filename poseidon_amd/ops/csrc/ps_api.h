@@ -3,6 +3,18 @@
 #include <hip/hip_runtime.h>
 #include <cstdint>
 
+// Implicit-GEMM gather descriptor: the GEMM's A operand (conv fwd) or
+// K-major B operand (conv wgrad) is the im2col matrix, gathered on the fly
+// from NHWC x instead of materialized. Out-of-bounds (padding) lanes load
+// from a 16-byte zero page.
+struct GatherDesc {
+  const void* x;      // NHWC activations
+  const void* zero;   // >=16B of zeros
+  int C, H, W, Ho, Wo;
+  int kh, kw, sh, sw, ph, pw;
+  int Cg, c0;         // group channel count and channel offset
+};
+
 struct GemmArgs {
   const void* A;
   const void* B;
@@ -18,6 +30,9 @@ struct GemmArgs {
   // ws must hold splitk * M * N floats and batch must be 1.
   void* ws;
   int splitk;
+  // implicit-GEMM: gather A (K-last rows = im2col rows) / gather B (K-major)
+  const GatherDesc* gather_a;
+  const GatherDesc* gather_b;
 };
 
 struct PoolGeom {

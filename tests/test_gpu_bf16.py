@@ -45,11 +45,17 @@ def test_gemm_bf16_layouts(ak, bk):
     close_bf16(out, ref_q, rtol=1e-2, what=f"gemm bf16 ak={ak} bk={bk}")
 
 
-def test_conv_bf16():
-    x = rnd(2, 8, 13, 13, seed=11)
-    w = rnd(12, 4, 5, 5, seed=12, scale=0.2)
-    b = rnd(12, seed=13)
-    stride, pad, g = (1, 1), (2, 2), 2
+@pytest.mark.parametrize("cfg", [
+    dict(C=8, Co=12, k=5, s=1, p=2, g=2, hw=13),    # materialized (Cg=4)
+    dict(C=16, Co=12, k=3, s=1, p=1, g=1, hw=11),   # implicit (Cg=16)
+    dict(C=64, Co=32, k=3, s=2, p=1, g=1, hw=14),   # implicit + glds (Cg=64)
+    dict(C=128, Co=24, k=3, s=1, p=1, g=2, hw=9),   # implicit grouped Cg=64
+])
+def test_conv_bf16(cfg):
+    x = rnd(2, cfg["C"], cfg["hw"], cfg["hw"], seed=11)
+    w = rnd(cfg["Co"], cfg["C"] // cfg["g"], cfg["k"], cfg["k"], seed=12, scale=0.2)
+    b = rnd(cfg["Co"], seed=13)
+    stride, pad, g = (cfg["s"], cfg["s"]), (cfg["p"], cfg["p"]), cfg["g"]
     y_ref, _ = ops.conv2d_forward_ex(x, w, b, stride, pad, g)
     xg = x.to(DEV, torch.bfloat16)
     y, cache = ops.conv2d_forward_ex(xg, w.to(DEV), b.to(DEV), stride, pad, g)
@@ -68,10 +74,10 @@ def test_conv_bf16():
     xq = x.to(torch.bfloat16).float()
     dyq = dy.to(torch.bfloat16).float()
     dw_ref = torch.zeros_like(w)
-    db_ref = torch.zeros(12)
+    db_ref = torch.zeros(cfg["Co"])
     ops.conv2d_backward_weight_acc(xq, None, dyq, dw_ref, db_ref, stride, pad, g)
     dw = torch.zeros_like(w).to(DEV)
-    db = torch.zeros(12).to(DEV)
+    db = torch.zeros(cfg["Co"]).to(DEV)
     ops.conv2d_backward_weight_acc(xg, colT, dy.to(DEV, torch.bfloat16),
                                    dw, db, stride, pad, g)
     assert dw.dtype == torch.float32

@@ -85,6 +85,9 @@ def test_gemm_at_b():
     dict(N=2, C=8, H=13, W=13, Co=12, k=5, s=1, p=2, g=2),    # grouped
     dict(N=3, C=16, H=9, W=9, Co=24, k=1, s=1, p=0, g=1),     # 1x1 fast path
     dict(N=1, C=4, H=8, W=8, Co=6, k=3, s=2, p=1, g=1),
+    dict(N=2, C=8, H=12, W=12, Co=10, k=3, s=1, p=1, g=1),   # implicit (f32 V=4)
+    dict(N=2, C=64, H=10, W=10, Co=32, k=3, s=1, p=1, g=1),  # implicit + glds
+    dict(N=2, C=16, H=9, W=9, Co=8, k=5, s=2, p=2, g=2),     # implicit grouped
 ])
 def test_conv_forward_backward(cfg):
     N, C, H, W = cfg["N"], cfg["C"], cfg["H"], cfg["W"]
