@@ -300,6 +300,7 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
       ga.C = g.C; ga.H = g.H; ga.W = g.W; ga.Ho = g.Ho; ga.Wo = g.Wo;
       ga.kh = kh; ga.kw = kw; ga.sh = sh; ga.sw = sw; ga.ph = ph; ga.pw = pw;
       ga.Cg = Cg; ga.c0 = grp * Cg;
+      ps_fill_gather_inv(&ga);
       run_gemm(x_cl, wk, y, bp ? bp + grp * Cog : nullptr,
                (int)NP, Cog, Kg,
                /*lda=*/Kg, /*ldb=*/Kg, /*ldc=*/Co,
@@ -401,6 +402,7 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
       gb.Ho = dy_cl.size(2); gb.Wo = dy_cl.size(3);
       gb.kh = kh; gb.kw = kw; gb.sh = sh; gb.sw = sw; gb.ph = ph; gb.pw = pw;
       gb.Cg = Cig; gb.c0 = grp * Cig;
+      ps_fill_gather_inv(&gb);
       run_gemm(dy2, x_cl, dwk, nullptr,
                Cog, Kg, (int)NP,
                /*lda=*/Co, /*ldb=*/Kg, /*ldc=*/Kg,

@@ -66,15 +66,24 @@ template <> struct GemmTraits<__bf16> {
 // group slice (khw*Cg + cg). Returns the NHWC x address of k's element run
 // (contiguous along cg), or the zero page for padding. Valid for vector
 // loads that stay inside one cg-run (callers guarantee k%VEC==0, Cg%VEC==0).
+// exact unsigned divide by a small runtime constant via f32 reciprocal +
+// fixup (x < 2^24; integer division is ~40 cycles each and this decode runs
+// per staged vector)
+__device__ inline int fdiv_fix(int x, int d, float inv, int& rem) {
+  int q = (int)((float)x * inv);
+  rem = x - q * d;
+  if (rem < 0) { --q; rem += d; }
+  else if (rem >= d) { ++q; rem -= d; }
+  return q;
+}
+
 template <typename T>
 __device__ inline const T* gather_addr(const GatherDesc& ga, int64_t row, int k) {
-  int khw = k / ga.Cg;
-  int cg = k - khw * ga.Cg;
-  int kkh = khw / ga.kw, kkw = khw - kkh * ga.kw;
-  int ow = row % ga.Wo;
-  int64_t t = row / ga.Wo;
-  int oh = t % ga.Ho;
-  int n = t / ga.Ho;
+  int cg, kkw, ow, oh;
+  int khw = fdiv_fix(k, ga.Cg, ga.inv_Cg, cg);
+  int kkh = fdiv_fix(khw, ga.kw, ga.inv_kw, kkw);
+  int t = fdiv_fix((int)row, ga.Wo, ga.inv_Wo, ow);
+  int n = fdiv_fix(t, ga.Ho, ga.inv_Ho, oh);
   int ih = oh * ga.sh - ga.ph + kkh;
   int iw = ow * ga.sw - ga.pw + kkw;
   if (ih < 0 || ih >= ga.H || iw < 0 || iw >= ga.W)

@@ -13,7 +13,17 @@ struct GatherDesc {
   int C, H, W, Ho, Wo;
   int kh, kw, sh, sw, ph, pw;
   int Cg, c0;         // group channel count and channel offset
+  // f32 reciprocals for division-free index decode (exact with the +-1
+  // fixup in the kernel; operands < 2^24)
+  float inv_Cg, inv_kw, inv_Wo, inv_Ho;
 };
+
+inline void ps_fill_gather_inv(GatherDesc* g) {
+  g->inv_Cg = 1.0f / g->Cg;
+  g->inv_kw = 1.0f / g->kw;
+  g->inv_Wo = 1.0f / g->Wo;
+  g->inv_Ho = 1.0f / g->Ho;
+}
 
 struct GemmArgs {
   const void* A;

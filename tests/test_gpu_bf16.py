@@ -56,18 +56,24 @@ def test_conv_bf16(cfg):
     w = rnd(cfg["Co"], cfg["C"] // cfg["g"], cfg["k"], cfg["k"], seed=12, scale=0.2)
     b = rnd(cfg["Co"], seed=13)
     stride, pad, g = (cfg["s"], cfg["s"]), (cfg["p"], cfg["p"]), cfg["g"]
-    y_ref, _ = ops.conv2d_forward_ex(x, w, b, stride, pad, g)
+    # reference from bf16-QUANTIZED inputs: the GPU kernel sees bf16 x/w,
+    # so comparing against unquantized fp32 mixes input-quantization noise
+    # (which grows with K) into the kernel check
+    xq = x.to(torch.bfloat16).float()
+    wq = w.to(torch.bfloat16).float()
+    y_ref, _ = ops.conv2d_forward_ex(xq, wq, b, stride, pad, g)
     xg = x.to(DEV, torch.bfloat16)
     y, cache = ops.conv2d_forward_ex(xg, w.to(DEV), b.to(DEV), stride, pad, g)
     colT, _wkT = cache
     assert y.dtype == torch.bfloat16
-    close_bf16(y, y_ref, what="conv bf16 fwd")
+    close_bf16(y, y_ref, rtol=1e-2, atol=1e-2, what="conv bf16 fwd")
 
     dy = rnd(*y_ref.shape, seed=14)
-    dx_ref = ops.conv2d_backward_input(w, dy, x.shape, stride, pad, g)
+    dyq = dy.to(torch.bfloat16).float()
+    dx_ref = ops.conv2d_backward_input(wq, dyq, x.shape, stride, pad, g)
     dx = ops.conv2d_backward_input(w.to(DEV), dy.to(DEV, torch.bfloat16),
                                    x.shape, stride, pad, g)
-    close_bf16(dx, dx_ref, what="conv bf16 dgrad")
+    close_bf16(dx, dx_ref, rtol=1e-2, atol=1e-2, what="conv bf16 dgrad")
 
     # wgrad reference from bf16-QUANTIZED inputs (isolates kernel bugs from
     # input quantization; the GPU kernel sees bf16 x/dy)
