@@ -16,6 +16,7 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <atomic>
 #include <cmath>
 
 #include "ps_api.h"
@@ -57,6 +58,16 @@ Tensor weight_shadow(const Tensor& w_f32, bool bf16) {
 // ---------------------------------------------------------------------------
 // GEMM plumbing
 // ---------------------------------------------------------------------------
+
+// Implicit-GEMM mode: gather im2col inside the GEMM staging instead of
+// materializing the column matrix. Correct for every Cg%VEC==0 conv and
+// covered by tests in both modes; measured slightly behind the
+// materialized+vectorized-im2col pipeline on VGG/GoogLeNet shapes (gather
+// address ALU in the staging inner loop), so default off -- toggle with
+// set_implicit_gemm for experiments.
+std::atomic<bool> g_implicit_gemm{false};
+
+void set_implicit_gemm(bool on) { g_implicit_gemm.store(on); }
 
 // 64B zero page for implicit-GEMM padding loads (per device, persistent)
 const void* zero_page(const Tensor& like) {
@@ -271,7 +282,7 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   // group channel count keeps 16B runs contiguous -- no column matrix at
   // all (the wgrad GEMM gathers too); conv1-style small-C layers still
   // materialize (and 1x1 convs alias x directly)
-  bool implicit = !is_1x1 && (Cg % VEC == 0);
+  bool implicit = g_implicit_gemm.load() && !is_1x1 && (Cg % VEC == 0);
   Tensor colT;
   if (is_1x1) {
     colT = rows2d(x_cl);  // alias: x rows ARE the col rows
@@ -891,6 +902,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_forward_ex", &conv2d_forward_ex);
   m.def("conv2d_backward_input", &conv2d_backward_input);
   m.def("conv2d_backward_weight_acc", &conv2d_backward_weight_acc);
+  m.def("set_implicit_gemm", &set_implicit_gemm);
   m.def("concat_channels", &concat_channels);
   m.def("slice_channels", &slice_channels);
   m.def("pool_max_forward", &pool_max_forward);

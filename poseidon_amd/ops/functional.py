@@ -42,6 +42,13 @@ def _ext():
     return _EXT
 
 
+def set_implicit_gemm(on: bool) -> None:
+    """Toggle implicit-GEMM convolution (im2col gathered inside the MFMA
+    GEMM staging; no column matrix). Default off: the materialized pipeline
+    measures faster on current shapes."""
+    _ext().set_implicit_gemm(bool(on))
+
+
 def ext_available() -> bool:
     try:
         _ext()

@@ -45,13 +45,22 @@ def test_gemm_bf16_layouts(ak, bk):
     close_bf16(out, ref_q, rtol=1e-2, what=f"gemm bf16 ak={ak} bk={bk}")
 
 
+@pytest.mark.parametrize("implicit", [False, True])
 @pytest.mark.parametrize("cfg", [
     dict(C=8, Co=12, k=5, s=1, p=2, g=2, hw=13),    # materialized (Cg=4)
     dict(C=16, Co=12, k=3, s=1, p=1, g=1, hw=11),   # implicit (Cg=16)
     dict(C=64, Co=32, k=3, s=2, p=1, g=1, hw=14),   # implicit + glds (Cg=64)
     dict(C=128, Co=24, k=3, s=1, p=1, g=2, hw=9),   # implicit grouped Cg=64
 ])
-def test_conv_bf16(cfg):
+def test_conv_bf16(cfg, implicit):
+    ops.set_implicit_gemm(implicit)
+    try:
+        _run_conv_bf16(cfg)
+    finally:
+        ops.set_implicit_gemm(False)
+
+
+def _run_conv_bf16(cfg):
     x = rnd(2, cfg["C"], cfg["hw"], cfg["hw"], seed=11)
     w = rnd(cfg["Co"], cfg["C"] // cfg["g"], cfg["k"], cfg["k"], seed=12, scale=0.2)
     b = rnd(cfg["Co"], seed=13)

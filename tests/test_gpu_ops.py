@@ -89,7 +89,16 @@ def test_gemm_at_b():
     dict(N=2, C=64, H=10, W=10, Co=32, k=3, s=1, p=1, g=1),  # implicit + glds
     dict(N=2, C=16, H=9, W=9, Co=8, k=5, s=2, p=2, g=2),     # implicit grouped
 ])
-def test_conv_forward_backward(cfg):
+@pytest.mark.parametrize("implicit", [False, True])
+def test_conv_forward_backward(cfg, implicit):
+    ops.set_implicit_gemm(implicit)
+    try:
+        _run_conv_case(cfg)
+    finally:
+        ops.set_implicit_gemm(False)
+
+
+def _run_conv_case(cfg):
     N, C, H, W = cfg["N"], cfg["C"], cfg["H"], cfg["W"]
     Co, k, s, p, g = cfg["Co"], cfg["k"], cfg["s"], cfg["p"], cfg["g"]
     x = rnd(N, C, H, W, seed=11)
