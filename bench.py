@@ -82,11 +82,10 @@ def main():
     sync()
     elapsed = time.perf_counter() - t0
 
-    # MAX elapsed over ranks
+    # MAX elapsed over ranks (staged through the device for nccl/RCCL)
     t = torch.tensor([elapsed], dtype=torch.float64)
     if solver.distributed:
-        import torch.distributed as dist
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        comm.allreduce_max(t)
     elapsed = float(t[0])
 
     if rank == 0:

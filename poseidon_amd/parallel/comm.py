@@ -57,8 +57,28 @@ def broadcast_params(params: List) -> None:
 
 
 def allreduce_metrics(t: torch.Tensor) -> torch.Tensor:
-    if dist.is_initialized():
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    """Small host-side metric reduction. The nccl/RCCL backend only moves
+    device tensors, so stage through the GPU when that is the backend."""
+    if not dist.is_initialized():
+        return t
+    if dist.get_backend() == "nccl" and not t.is_cuda:
+        d = t.to(ctx().torch_device)
+        dist.all_reduce(d, op=dist.ReduceOp.SUM)
+        t.copy_(d.cpu())
+        return t
+    dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return t
+
+
+def allreduce_max(t: torch.Tensor) -> torch.Tensor:
+    if not dist.is_initialized():
+        return t
+    if dist.get_backend() == "nccl" and not t.is_cuda:
+        d = t.to(ctx().torch_device)
+        dist.all_reduce(d, op=dist.ReduceOp.MAX)
+        t.copy_(d.cpu())
+        return t
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
     return t
 
 
