@@ -300,10 +300,117 @@ class HDF5OutputLayer(Layer):
 
 
 @register_layer("WINDOW_DATA")
-class WindowDataLayer(Layer):
+class WindowDataLayer(_PrefetchingDataLayer):
+    """R-CNN-style window sampling (window_data_layer.cpp): a window file
+    lists images and class/overlap/box windows; each batch draws
+    fg_fraction foreground windows (overlap >= fg_threshold) and fills the
+    rest with background (overlap < bg_threshold, class 0), crops with
+    context_pad and warps to crop_size.
+
+    Window-file format (the reference's):
+        # <image_index>
+        <image_path>
+        <channels> <height> <width>
+        <num_windows>
+        <class> <overlap> <x1> <y1> <x2> <y2>
+    """
+
     def layer_setup(self, bottom, top) -> None:
-        raise NotImplementedError(
-            "WINDOW_DATA (R-CNN window sampling) is not wired yet")
+        wp = self.param.ensure("window_data_param")
+        self.batch = int(wp.batch_size)
+        self.crop = int(wp.crop_size)
+        assert self.crop > 0, "WINDOW_DATA needs crop_size"
+        self.fg_thresh = float(wp.fg_threshold)
+        self.bg_thresh = float(wp.bg_threshold)
+        self.fg_frac = float(wp.fg_fraction)
+        self.ctx_pad = int(wp.context_pad)
+        self.scale = float(wp.scale)
+        self.mirror = bool(wp.mirror)
+        self.mean = None
+        if wp.has("mean_file"):
+            from ..proto import read_proto_binary
+            proto = read_proto_binary(wp.mean_file, "BlobProto")
+            arr = np.asarray(proto.data, dtype=np.float32)
+            self.mean = arr.reshape(proto.channels, proto.height, proto.width)
+
+        self.images: List[str] = []
+        self.fg: List[tuple] = []   # (img_idx, cls, x1, y1, x2, y2)
+        self.bg: List[tuple] = []
+        with open(wp.source) as f:
+            lines = [l.rstrip("\n") for l in f]
+        i = 0
+        while i < len(lines):
+            if not lines[i].startswith("#"):
+                i += 1
+                continue
+            path = lines[i + 1].strip()
+            nwin = int(lines[i + 3])
+            img_idx = len(self.images)
+            self.images.append(path)
+            for w in range(nwin):
+                parts = lines[i + 4 + w].split()
+                cls, ov = int(parts[0]), float(parts[1])
+                box = tuple(int(v) for v in parts[2:6])
+                if ov >= self.fg_thresh:
+                    self.fg.append((img_idx, cls) + box)
+                elif ov < self.bg_thresh:
+                    self.bg.append((img_idx, 0) + box)
+            i += 4 + nwin
+        if not self.fg or not self.bg:
+            raise ValueError("window file needs both fg and bg windows")
+        c = ctx()
+        self.rng = np.random.default_rng(c.seed + 389 * c.rank)
+        self._img_cache: dict = {}
+        self._start_prefetch()
+
+    def _read_img(self, idx):
+        if idx not in self._img_cache:
+            from PIL import Image
+            img = Image.open(self.images[idx]).convert("RGB")
+            arr = np.asarray(img, dtype=np.float32)[:, :, ::-1]  # BGR
+            self._img_cache[idx] = np.ascontiguousarray(arr)
+            if len(self._img_cache) > 64:
+                self._img_cache.pop(next(iter(self._img_cache)))
+        return self._img_cache[idx]
+
+    def _crop_window(self, win):
+        from PIL import Image
+        img_idx, cls, x1, y1, x2, y2 = win
+        arr = self._read_img(img_idx)
+        H, W = arr.shape[:2]
+        if self.ctx_pad:
+            pw = int(round((x2 - x1 + 1) * self.ctx_pad / self.crop))
+            ph = int(round((y2 - y1 + 1) * self.ctx_pad / self.crop))
+            x1, x2 = x1 - pw, x2 + pw
+            y1, y2 = y1 - ph, y2 + ph
+        x1c, y1c = max(0, x1), max(0, y1)
+        x2c, y2c = min(W - 1, x2), min(H - 1, y2)
+        patch = arr[y1c:y2c + 1, x1c:x2c + 1]
+        im = Image.fromarray(patch.astype(np.uint8))
+        im = im.resize((self.crop, self.crop))  # warp mode
+        out = np.asarray(im, dtype=np.float32).transpose(2, 0, 1)
+        if self.mean is not None:
+            out = out - self.mean[:, :self.crop, :self.crop]
+        if self.mirror and self.rng.integers(0, 2):
+            out = out[:, :, ::-1]
+        return np.ascontiguousarray(out * self.scale), cls
+
+    def _load_batch(self):
+        n_fg = int(round(self.batch * self.fg_frac))
+        data = np.empty((self.batch, 3, self.crop, self.crop), dtype=np.float32)
+        labels = np.empty((self.batch,), dtype=np.float32)
+        for i in range(self.batch):
+            pool = self.fg if i < n_fg else self.bg
+            win = pool[int(self.rng.integers(0, len(pool)))]
+            data[i], labels[i] = self._crop_window(win)
+        return data, labels
 
     def reshape(self, bottom, top) -> None:
-        pass
+        top[0].reshape(self.batch, 3, self.crop, self.crop)
+        top[1].reshape(self.batch)
+
+    def forward(self, bottom, top) -> None:
+        data, labels = self._next_batch()
+        c = ctx()
+        top[0].data = torch.from_numpy(data).to(c.torch_device, c.compute_dtype)
+        top[1].data = torch.from_numpy(labels).to(c.torch_device)
