@@ -90,6 +90,7 @@ class SGDSolver:
                           f"{[l.name for l in marked]}", flush=True)
 
         self._net_outputs_rows: List[List[float]] = []
+        self._net_outputs_cols: List[str] = []
         self._t0 = time.time()
 
         # hipGraph capture of the steady-state iteration (HIP streams and
@@ -288,15 +289,24 @@ class SGDSolver:
     # ------------------------------------------------------------------
     def _display(self, loss: float, rate: float) -> None:
         c = ctx()
-        vals = torch.tensor([loss], dtype=torch.float64)
+        outs = []
+        for name in sorted(self.net.output_blob_names):
+            b = self.net.blobs[name]
+            if b.count == 1:
+                outs.append((name, float(b.data.reshape(()).item())))
+        vals = torch.tensor([loss] + [v for _, v in outs], dtype=torch.float64)
         if self.distributed:
             comm.allreduce_metrics(vals)
             vals /= c.world_size
-        row = [float(self.iter), time.time() - self._t0, float(vals[0])]
+        if not self._net_outputs_cols:
+            self._net_outputs_cols = ["iter", "time", "loss"] + [n for n, _ in outs]
+        row = [float(self.iter), time.time() - self._t0] +             [float(v) for v in vals]
         self._net_outputs_rows.append(row)
         if self.verbose:
+            extra = " ".join(f"{n}={float(v):.4f}"
+                             for (n, _), v in zip(outs, vals[1:]))
             print(f"[poseidon] iter {self.iter} loss {vals[0]:.6f} "
-                  f"lr {rate:.6g}", flush=True)
+                  f"lr {rate:.6g} {extra}", flush=True)
 
     def test_all(self) -> List[Dict[str, float]]:
         results = []
@@ -395,8 +405,9 @@ class SGDSolver:
         """CSV of display-time metrics (PrintNetOutputs, solver.cpp:699-756)."""
         if not ctx().is_root():
             return
+        cols = self._net_outputs_cols or ["iter", "time", "loss"]
         with open(path + ".netoutputs", "w") as f:
-            f.write("iter,time,loss\n")
+            f.write(",".join(cols) + "\n")
             for row in self._net_outputs_rows:
                 f.write(",".join(f"{v:.6f}" for v in row) + "\n")
 

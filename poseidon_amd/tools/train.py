@@ -33,6 +33,9 @@ def main(argv=None):
                     help="sufficient-factor broadcast for FC layers (default on)")
     ap.add_argument("--no-svb", dest="svb", action="store_false")
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--stats", default="",
+                    help="dump per-layer fwd/bwd timing YAML here (the "
+                         "Bösen caffe_stats.yaml analogue)")
     ap.add_argument("--cpu", action="store_true", help="force CPU mode")
     args = ap.parse_args(argv)
 
@@ -52,7 +55,18 @@ def main(argv=None):
     solver = get_solver(solver_param, use_sfb=args.svb)
     if args.weights:
         solver.load_weights(args.weights)
+    stats = None
+    if args.stats:
+        from poseidon_amd.utils.stats import LayerStats
+        stats = LayerStats(solver.net, roctx=False)
+        ctx_mgr = stats.timed()
+        ctx_mgr.__enter__()
     solver.solve(resume_file=args.snapshot or None)
+    if stats is not None:
+        ctx_mgr.__exit__(None, None, None)
+        if pa.ctx().is_root():
+            stats.dump(args.stats)
+            print(stats.report())
     if args.net_outputs:
         solver.write_net_outputs(args.net_outputs)
     if pa.ctx().is_root():
