@@ -113,8 +113,12 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   ps_pick_gemm_tile(M, N, &tbm, &tbn);
   int64_t tiles = (int64_t)((M + tbm - 1) / tbm) * ((N + tbn - 1) / tbn);
   Tensor ws;  // keep alive until launch returns
-  if (tiles < 96 && K >= 4096) {
-    int sk = (int)std::min<int64_t>(512 / tiles, (K + 2047) / 2048);
+  // 2 blocks/CU resident -> ~512 workgroups fill the chip; split K until
+  // the grid gets there (wgrad at 512x4608 is 144 tiles = 28% occupancy
+  // without this)
+  if (tiles < 384 && K >= 4096) {
+    int sk = (int)std::min<int64_t>((512 + tiles - 1) / tiles,
+                                    (K + 2047) / 2048);
     int64_t ws_elems = (int64_t)sk * M * N;
     if (sk > 1 && ws_elems * 4 <= (256LL << 20)) {
       g.splitk = sk;
