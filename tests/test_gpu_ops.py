@@ -309,3 +309,46 @@ def test_alexnet_converges_gpu():
                    for p in solver.net.learnable_params)
     finally:
         pa.init(device="cpu")
+
+
+def test_solver_checkpoint_gpu(tmp_path):
+    """Snapshot/restore with device tensors + test-net eval on GPU."""
+    from poseidon_amd.solver.solver import SGDSolver
+    from poseidon_amd.proto import Message
+    from poseidon_amd.models import zoo
+    pa.init(device="cuda", seed=11)
+    try:
+        sp = Message("SolverParameter", base_lr=0.01, lr_policy="fixed",
+                     momentum=0.9, max_iter=100,
+                     snapshot_prefix=str(tmp_path / "gpuck"))
+        sp.net_param = zoo.cifar10_quick(batch=16, num_classes=4)
+        solver = SGDSolver(sp, verbose=False)
+        solver.step(3)
+        path = solver.snapshot()
+        w0 = next(l for l in solver.net.layers if l.name == "conv1").blobs[0]
+        ref = w0.data.clone()
+
+        solver2 = SGDSolver(sp, verbose=False)
+        solver2.restore(str(tmp_path / "gpuck") + "_iter_3.solverstate.0.0")
+        w1 = next(l for l in solver2.net.layers if l.name == "conv1").blobs[0]
+        assert solver2.iter == 3
+        assert torch.allclose(w1.data, ref)
+    finally:
+        pa.init(device="cpu")
+
+
+def test_googlenet_step_gpu():
+    from poseidon_amd.core.net import Net, TRAIN
+    from poseidon_amd.models import zoo
+    pa.init(device="cuda", seed=2, compute_dtype=torch.bfloat16)
+    try:
+        net = Net(zoo.googlenet(batch=4, num_classes=100), phase=TRAIN)
+        loss = net.forward()
+        assert torch.isfinite(torch.tensor(loss))
+        net.zero_param_diffs()
+        net.backward()
+        bad = [p.blob.name for p in net.learnable_params
+               if not torch.isfinite(p.blob.diff).all()]
+        assert not bad, bad
+    finally:
+        pa.init(device="cpu", compute_dtype=torch.float32)
