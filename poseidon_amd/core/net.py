@@ -226,6 +226,56 @@ class Net:
             else:
                 self.layer_need_bwd[li] = False
 
+        self._fuse_relu_epilogues()
+
+    def _fuse_relu_epilogues(self) -> None:
+        """Fold an in-place slope-0 ReLU into the producing conv/IP GEMM
+        epilogue. insert_splits guarantees each produced blob instance has
+        a single consumer, so `top of conv == bottom of relu (in-place)`
+        means the clamp commutes with nothing else. The ReLU layer stays
+        in the net (its backward masks by the clamped activations, which
+        is unchanged); its forward becomes a no-op. Mirrors cuDNN-style
+        activation fusion; reference applied ReLU as a separate kernel
+        (relu_layer.cu)."""
+        p = self.param
+        producer_of: Dict[str, int] = {}
+        consumers: Dict[str, List[int]] = {}
+        for li, lp in enumerate(p.layers):
+            for t in lp.top:
+                producer_of[t] = li  # in-place chains: latest producer wins
+            for b in lp.bottom:
+                consumers.setdefault(b, []).append(li)
+        for li, lp in enumerate(p.layers):
+            if lp.enum_name("type") != "RELU" or len(lp.bottom) != 1:
+                continue
+            if lp.top[0] != lp.bottom[0]:
+                continue  # only in-place ReLU keeps backward semantics
+            relu = self.layers[li]
+            if relu.slope != 0.0 or relu.loss_weights[0] != 0.0:
+                continue
+            name = lp.bottom[0]
+            # direct producer must be this blob's conv/IP (not another
+            # in-place layer stacked in between)
+            prod_li = None
+            for lj in range(li - 1, -1, -1):
+                if name in p.layers[lj].top:
+                    prod_li = lj
+                    break
+            if prod_li is None:
+                continue
+            prod_lp = p.layers[prod_li]
+            if prod_lp.enum_name("type") not in ("CONVOLUTION", "INNER_PRODUCT"):
+                continue
+            if len(prod_lp.top) != 1 or self.layers[prod_li].loss_weights[0] != 0.0:
+                continue
+            # the pre-ReLU value must not feed anything else: consumers
+            # after the in-place ReLU read the post-ReLU instance (fine),
+            # one strictly between producer and ReLU reads pre-ReLU (bad)
+            if any(prod_li < lj < li for lj in consumers.get(name, [])):
+                continue
+            self.layers[prod_li].fuse_relu = True
+            relu.fused = True
+
     # ------------------------------------------------------------------
     @property
     def learnable_params(self) -> List[ParamSpec]:

@@ -42,6 +42,7 @@ class InnerProductLayer(Layer):
             b = Blob((1, 1, 1, self.N), dtype=dtype, name=f"{self.name}.bias")
             filler.fill(b, ip.bias_filler if ip.has("bias_filler") else None)
             self.blobs.append(b)
+        self.fuse_relu = False  # set by Net's IP+ReLU fusion pass
         # SFB: when set by the distributed solver, backward defers the dW GEMM
         self.sfb_active = False
         self.sfb_factors: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
@@ -55,7 +56,9 @@ class InnerProductLayer(Layer):
     def forward(self, bottom, top) -> None:
         x = bottom[0].data.reshape(self.M, self.K)
         b = self.blobs[1].data.view(-1) if self.bias_term else None
-        top[0].data = ops.linear_forward(x, self.blobs[0].data.view(self.N, self.K), b)
+        top[0].data = ops.linear_forward(
+            x, self.blobs[0].data.view(self.N, self.K), b,
+            fuse_relu=self.fuse_relu)
 
     def backward(self, top, propagate_down, bottom) -> None:
         x = bottom[0].data.reshape(self.M, self.K)

@@ -29,8 +29,12 @@ class ReLULayer(NeuronLayer):
     def layer_setup(self, bottom, top) -> None:
         rp = self.param.relu_param
         self.slope = float(rp.negative_slope) if rp is not None else 0.0
+        self.fused = False  # producer GEMM applied the clamp in its epilogue
 
     def forward(self, bottom, top) -> None:
+        if self.fused:  # in-place: bottom[0] already holds relu(y)
+            top[0].data = bottom[0].data
+            return
         top[0].data = ops.relu_forward(bottom[0].data, self.slope)
 
     def backward(self, top, propagate_down, bottom) -> None:

@@ -45,6 +45,7 @@ class ConvolutionLayer(Layer):
         self.group = int(cp.group)
         self.num_output = int(cp.num_output)
         self.bias_term = bool(cp.bias_term)
+        self.fuse_relu = False  # set by Net's conv+ReLU fusion pass
         channels = bottom[0].channels
         assert channels % self.group == 0 and self.num_output % self.group == 0
 
@@ -76,7 +77,8 @@ class ConvolutionLayer(Layer):
         self._colT = []
         for bo, t in zip(bottom, top):
             y, colT = ops.conv2d_forward_ex(bo.data, w, b, self.stride,
-                                            self.pad, self.group)
+                                            self.pad, self.group,
+                                            fuse_relu=self.fuse_relu)
             t.data = y
             self._colT.append(colT)
 

@@ -84,7 +84,7 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
               int64_t a_off, int64_t b_off, int64_t c_off,
               bool a_klast, bool b_klast, float alpha, float beta,
               const GatherDesc* gather_a = nullptr,
-              const GatherDesc* gather_b = nullptr) {
+              const GatherDesc* gather_b = nullptr, bool relu = false) {
   const bool in_bf16 = is_bf16(A);
   TORCH_CHECK(is_bf16(B) == in_bf16, "gemm: A/B dtype mismatch");
   const bool out_f32 = !is_bf16(C);
@@ -106,6 +106,7 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   g.splitk = 1;
   g.gather_a = gather_a;
   g.gather_b = gather_b;
+  g.relu = relu;
   // Split-K when the output tile grid cannot fill 256 CUs but K is deep
   // (conv wgrad: M=Cout, N=Kcol, K=N*OH*OW up to ~800k): target ~512
   // workgroups, cap the f32 workspace at 256 MB.
@@ -155,7 +156,7 @@ Tensor gemm(const Tensor& A, const Tensor& B, int M, int N, int K,
 // ---------------------------------------------------------------------------
 
 Tensor linear_forward(const Tensor& x, const Tensor& w,
-                      const c10::optional<Tensor>& bias) {
+                      const c10::optional<Tensor>& bias, bool fuse_relu) {
   check_float_like(x, "x");
   auto xc = x.contiguous();
   auto wc = weight_shadow(w, is_bf16(x));
@@ -168,7 +169,8 @@ Tensor linear_forward(const Tensor& x, const Tensor& w,
     bc = bias->contiguous();
     bp = bc.data_ptr<float>();
   }
-  run_gemm(xc, wc, y, bp, M, N, K, K, K, N, 0, 0, 0, true, true, 1.0f, 0.0f);
+  run_gemm(xc, wc, y, bp, M, N, K, K, K, N, 0, 0, 0, true, true, 1.0f, 0.0f,
+           nullptr, nullptr, fuse_relu);
   return y;
 }
 
@@ -265,7 +267,8 @@ Tensor weight_khwc_tr(const Tensor& w, int G, bool bf16) {
 
 std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
                                       const c10::optional<Tensor>& bias,
-                                      int sh, int sw, int ph, int pw, int G) {
+                                      int sh, int sw, int ph, int pw, int G,
+                                      bool fuse_relu) {
   check_float_like(x, "x");
   const bool bf16 = is_bf16(x);
   auto x_cl = cl4(x);
@@ -321,14 +324,14 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
                /*lda=*/Kg, /*ldb=*/Kg, /*ldc=*/Co,
                /*a_off=*/0, /*b_off=*/(int64_t)grp * Cog * Kg,
                /*c_off=*/(int64_t)grp * Cog,
-               true, true, 1.0f, 0.0f, &ga, nullptr);
+               true, true, 1.0f, 0.0f, &ga, nullptr, fuse_relu);
     } else {
       run_gemm(colT, wk, y, bp ? bp + grp * Cog : nullptr,
                (int)NP, Cog, Kg,
                /*lda=*/Kcol, /*ldb=*/Kg, /*ldc=*/Co,
                /*a_off=*/(int64_t)grp * Kg, /*b_off=*/(int64_t)grp * Cog * Kg,
                /*c_off=*/(int64_t)grp * Cog,
-               true, true, 1.0f, 0.0f);
+               true, true, 1.0f, 0.0f, nullptr, nullptr, fuse_relu);
     }
   }
   // transposed repack for the dgrad NT GEMM, computed once per iteration

@@ -296,7 +296,8 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
                  int64_t strideA, int64_t strideB, int64_t strideC,
                  float alpha, float beta,
                  float* __restrict__ ws, int kchunk,
-                 GatherDesc ga_a = {}, GatherDesc ga_b = {}) {
+                 GatherDesc ga_a = {}, GatherDesc ga_b = {},
+                 bool relu = false) {
   using TR = GemmTraits<T>;
   constexpr int BK = TR::BK, RS = TR::RS;
   constexpr int FM = BM / WGM / 16, FN = BN / WGN / 16;
@@ -466,6 +467,7 @@ epilogue:
           int64_t idx = (int64_t)row * ldc + col;
           float v = alpha * acc[fm][fn][r] + bv;
           if (beta != 0.0f) v += beta * to_f32(C[idx]);
+          if (relu && v < 0.0f) v = 0.0f;  // fused in-place ReLU epilogue
           from_f32(v, C[idx]);
         }
       }
@@ -479,7 +481,7 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
                                 OUT* __restrict__ C,
                                 const float* __restrict__ bias,
                                 int M, int N, int64_t ldc, int splitk,
-                                float alpha, float beta) {
+                                float alpha, float beta, bool relu) {
   int64_t MN = (int64_t)M * N;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < MN;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -490,6 +492,7 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
     if (HAS_BIAS) v += bias[col];
     int64_t idx = (int64_t)row * ldc + col;
     if (beta != 0.0f) v += beta * to_f32(C[idx]);
+    if (relu && v < 0.0f) v = 0.0f;
     from_f32(v, C[idx]);
   }
 }
@@ -516,24 +519,24 @@ static void launch_tile(const GemmArgs& g, hipStream_t s) {
         <<<grid, block, 0, s>>>(
             (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,
             g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha,
-            g.beta, (float*)g.ws, kchunk, da, db);
+            g.beta, (float*)g.ws, kchunk, da, db, g.relu);
     int64_t MN = (int64_t)g.M * g.N;
     int64_t rb = cdiv64(MN, 256);
     if (rb > 2048) rb = 2048;
     if (HB)
       splitk_reduce_k<OUT, true><<<dim3((unsigned)rb), 256, 0, s>>>(
           (const float*)g.ws, (OUT*)g.C, g.bias, g.M, g.N, g.ldc, g.splitk,
-          g.alpha, g.beta);
+          g.alpha, g.beta, g.relu);
     else
       splitk_reduce_k<OUT, false><<<dim3((unsigned)rb), 256, 0, s>>>(
           (const float*)g.ws, (OUT*)g.C, g.bias, g.M, g.N, g.ldc, g.splitk,
-          g.alpha, g.beta);
+          g.alpha, g.beta, g.relu);
   } else {
     gemm_kernel<T, OUT, BM, BN, WGM, WGN, AK, BK_, HB, false, GA, GB>
         <<<grid, block, 0, s>>>(
             (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,
             g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha,
-            g.beta, nullptr, 0, da, db);
+            g.beta, nullptr, 0, da, db, g.relu);
   }
 }
 

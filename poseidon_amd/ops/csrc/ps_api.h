@@ -43,6 +43,8 @@ struct GemmArgs {
   // implicit-GEMM: gather A (K-last rows = im2col rows) / gather B (K-major)
   const GatherDesc* gather_a;
   const GatherDesc* gather_b;
+  // fused epilogue: y = max(y, 0) (conv/IP + in-place ReLU pairs)
+  bool relu;
 };
 
 struct PoolGeom {

@@ -80,14 +80,19 @@ def pool_out_size(h: int, k: int, p: int, s: int) -> Tuple[int, bool]:
 
 def conv2d_forward_ex(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor],
                       stride: Tuple[int, int], pad: Tuple[int, int],
-                      groups: int):
+                      groups: int, fuse_relu: bool = False):
     """Returns (y, colT_cache). colT is the im2col matrix on GPU (reused by
-    the backward GEMMs); None on CPU."""
+    the backward GEMMs); None on CPU. fuse_relu clamps the output in the
+    GEMM epilogue (used by the Net-level conv+ReLU fusion pass)."""
     if x.is_cuda:
         y, colT, wkT = _ext().conv2d_forward_ex(x, w, b, stride[0], stride[1],
-                                                pad[0], pad[1], groups)
+                                                pad[0], pad[1], groups,
+                                                fuse_relu)
         return y, (colT, wkT)
-    return F.conv2d(x, w, b, stride=stride, padding=pad, groups=groups), None
+    y = F.conv2d(x, w, b, stride=stride, padding=pad, groups=groups)
+    if fuse_relu:
+        y = F.relu(y)
+    return y, None
 
 
 def conv2d_backward_input(w: torch.Tensor, dy: torch.Tensor,
@@ -121,12 +126,15 @@ def conv2d_backward_weight_acc(x: torch.Tensor, colT, dy: torch.Tensor,
 # ---------------------------------------------------------------------------
 
 def linear_forward(x: torch.Tensor, w: torch.Tensor,
-                   b: Optional[torch.Tensor]) -> torch.Tensor:
+                   b: Optional[torch.Tensor],
+                   fuse_relu: bool = False) -> torch.Tensor:
     if x.is_cuda:
-        return _ext().linear_forward(x, w, b)
+        return _ext().linear_forward(x, w, b, fuse_relu)
     y = x.matmul(w.t())
     if b is not None:
         y = y + b
+    if fuse_relu:
+        y = F.relu(y)
     return y
 
 

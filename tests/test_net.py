@@ -128,3 +128,59 @@ def test_checkpoint_roundtrip(tmp_path):
     after = net2.layers[1].blobs[0].data
     assert torch.allclose(after, net.layers[1].blobs[0].data)
     assert not torch.allclose(before, after)
+
+
+def test_relu_fusion_matches_unfused():
+    """conv/IP + in-place ReLU folds into the GEMM epilogue (net.py
+    _fuse_relu_epilogues); forward/backward must match the unfused net."""
+    import numpy as np
+    import torch
+    import poseidon_amd as pa
+    from poseidon_amd.core.net import Net, TRAIN
+    from poseidon_amd.proto import parse_text
+
+    txt = """
+        name: "fuse"
+        layers { name: "data" type: DUMMY_DATA top: "data" top: "label"
+                 dummy_data_param { num: 4 channels: 3 height: 8 width: 8
+                     num: 4 channels: 1 height: 1 width: 1
+                     data_filler { type: "gaussian" std: 1.0 } } }
+        layers { name: "conv" type: CONVOLUTION bottom: "data" top: "conv"
+                 convolution_param { num_output: 6 kernel_size: 3
+                     weight_filler { type: "xavier" }
+                     bias_filler { type: "constant" value: 0.1 } } }
+        layers { name: "relu1" type: RELU bottom: "conv" top: "conv" }
+        layers { name: "ip" type: INNER_PRODUCT bottom: "conv" top: "ip"
+                 inner_product_param { num_output: 10
+                     weight_filler { type: "xavier" } } }
+        layers { name: "relu2" type: RELU bottom: "ip" top: "ip" }
+        layers { name: "fc" type: INNER_PRODUCT bottom: "ip" top: "fc"
+                 inner_product_param { num_output: 5
+                     weight_filler { type: "xavier" } } }
+        layers { name: "loss" type: SOFTMAX_LOSS bottom: "fc" bottom: "label"
+                 top: "loss" }
+    """
+
+    def run(disable_fusion):
+        pa.init(device="cpu", seed=11)
+        net = Net(parse_text("NetParameter", txt), phase=TRAIN)
+        fused = [l for l in net.layers if getattr(l, "fuse_relu", False)]
+        if disable_fusion:
+            for l in net.layers:
+                if hasattr(l, "fuse_relu"):
+                    l.fuse_relu = False
+                if hasattr(l, "fused"):
+                    l.fused = False
+        else:
+            assert len(fused) == 2  # conv+relu1, ip+relu2 (fc feeds loss)
+        loss = net.forward()
+        net.zero_param_diffs()
+        net.backward()
+        grads = [p.blob.diff.clone() for p in net.params]
+        return loss, grads
+
+    l1, g1 = run(disable_fusion=False)
+    l0, g0 = run(disable_fusion=True)
+    assert np.isclose(l1, l0)
+    for a, b in zip(g1, g0):
+        assert torch.allclose(a, b), "fused/unfused gradients differ"
