@@ -352,3 +352,38 @@ def test_googlenet_step_gpu():
         assert not bad, bad
     finally:
         pa.init(device="cpu", compute_dtype=torch.float32)
+
+
+def test_full_solver_protocol_gpu(tmp_path):
+    """The complete Solve() protocol on GPU in one run (solver.cpp:246-402):
+    test_initialization eval, periodic TestAll + display, snapshot at
+    snapshot interval and at max_iter, then restore + continue."""
+    from poseidon_amd.solver.solver import SGDSolver
+    from poseidon_amd.proto import Message
+    from poseidon_amd.models import zoo
+    pa.init(device="cuda", seed=23, compute_dtype=torch.bfloat16)
+    try:
+        sp = Message("SolverParameter", base_lr=0.005, lr_policy="step",
+                     gamma=0.5, stepsize=8, momentum=0.9,
+                     weight_decay=0.0005, max_iter=20, display=5,
+                     test_interval=10, test_initialization=True,
+                     snapshot=10, snapshot_prefix=str(tmp_path / "proto"))
+        sp.test_iter.append(2)
+        sp.net_param = zoo.cifar10_quick(batch=32, num_classes=4)
+        solver = SGDSolver(sp, verbose=False)
+        solver.solve()
+        assert solver.iter == 20
+        # snapshot at iter 10 (interval) and 20 (snapshot_after_train)
+        for it in (10, 20):
+            assert (tmp_path / f"proto_iter_{it}.caffemodel").exists()
+            assert (tmp_path / f"proto_iter_{it}.solverstate.0.0").exists()
+        # display path recorded scalar net outputs
+        assert solver._net_outputs_rows, "display recorded no outputs"
+        # resume from the mid snapshot and continue to max_iter
+        solver2 = SGDSolver(sp, verbose=False)
+        solver2.solve(resume_file=str(tmp_path / "proto_iter_10.solverstate.0.0"))
+        assert solver2.iter == 20
+        w0 = solver.net.learnable_params[0].blob.data
+        assert torch.isfinite(w0).all()
+    finally:
+        pa.init(device="cpu")
