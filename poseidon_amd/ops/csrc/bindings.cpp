@@ -643,7 +643,10 @@ Tensor lrn_backward(const Tensor& x, const Tensor& y, const Tensor& scale,
   int64_t rows = (int64_t)x_cl.size(0) * x_cl.size(2) * x_cl.size(3);
   int C = x_cl.size(1);
   Tensor dx = at::empty_like(x_cl, x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  Tensor ratio = at::empty({rows * C}, x.options().dtype(at::kFloat));
+  // ratio workspace only feeds the C>256 fallback path; the row-block
+  // kernels keep the ratio in LDS
+  Tensor ratio = at::empty({C > 256 ? rows * C : 0},
+                           x.options().dtype(at::kFloat));
   if (is_bf16(x))
     ps_lrn_bwd_bf16(x_cl.data_ptr(), y_cl.data_ptr(), sc_cl.data_ptr<float>(),
                     dy_cl.data_ptr(), dx.data_ptr(), ratio.data_ptr<float>(),
@@ -899,6 +902,99 @@ void adagrad_update(Tensor w, const Tensor& g, Tensor h, double lr,
                     (float)wd, stream());
 }
 
+// ---------------------------------------------------------------------------
+// Multi-tensor apply tables (sgd.hip MTDesc/MTZeroDesc/MTChunk). Host-side
+// mirror structs -- layouts must match sgd.hip exactly (same compiler, both
+// trivially copyable, so they do).
+// ---------------------------------------------------------------------------
+struct MTDescHost {
+  float* w;
+  const float* g;
+  float* h;
+  int64_t n;
+  float lr_mult;
+  float wd;
+};
+struct MTZeroDescHost {
+  float* p;
+  int64_t n;
+};
+struct MTChunkHost {
+  int t;
+  int64_t off;
+};
+
+Tensor blob_to_dev(const void* src, int64_t bytes, const Tensor& like) {
+  Tensor host = at::from_blob(const_cast<void*>(src), {bytes},
+                              at::TensorOptions().dtype(at::kByte));
+  return host.to(like.device());
+}
+
+// returns {desc_dev(u8), chunk_dev(u8), nchunks} -- build once, reuse forever
+std::vector<Tensor> sgd_mt_prepare(std::vector<Tensor> ws,
+                                   std::vector<Tensor> gs,
+                                   std::vector<Tensor> hs,
+                                   std::vector<double> lr_mults,
+                                   std::vector<double> wds) {
+  const int CHUNK = ps_mt_chunk_elts();
+  const size_t nt = ws.size();
+  TORCH_CHECK(gs.size() == nt && hs.size() == nt && lr_mults.size() == nt &&
+              wds.size() == nt, "sgd_mt_prepare: length mismatch");
+  std::vector<MTDescHost> descs(nt);
+  std::vector<MTChunkHost> chunks;
+  for (size_t t = 0; t < nt; ++t) {
+    TORCH_CHECK(ws[t].is_cuda() && ws[t].is_contiguous() &&
+                ws[t].scalar_type() == at::kFloat &&
+                gs[t].is_contiguous() && hs[t].is_contiguous(),
+                "sgd_mt_prepare: params must be contiguous f32 CUDA");
+    int64_t n = ws[t].numel();
+    TORCH_CHECK(gs[t].numel() == n && hs[t].numel() == n);
+    descs[t] = {ws[t].data_ptr<float>(), gs[t].data_ptr<float>(),
+                hs[t].data_ptr<float>(), n, (float)lr_mults[t],
+                (float)wds[t]};
+    for (int64_t off = 0; off < n; off += CHUNK)
+      chunks.push_back({(int)t, off});
+  }
+  return {blob_to_dev(descs.data(), nt * sizeof(MTDescHost), ws[0]),
+          blob_to_dev(chunks.data(), chunks.size() * sizeof(MTChunkHost),
+                      ws[0]),
+          at::scalar_tensor((int64_t)chunks.size())};
+}
+
+void sgd_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
+                int64_t nchunks, double lr, double mom,
+                const c10::optional<Tensor>& lr_dev) {
+  ps_sgd_mt(desc_dev.data_ptr(), chunk_dev.data_ptr(), (int)nchunks,
+            (float)lr, (float)mom,
+            lr_dev.has_value() ? lr_dev->data_ptr<float>() : nullptr,
+            stream());
+}
+
+std::vector<Tensor> zero_mt_prepare(std::vector<Tensor> ts) {
+  const int CHUNK = ps_mt_chunk_elts();
+  std::vector<MTZeroDescHost> descs(ts.size());
+  std::vector<MTChunkHost> chunks;
+  for (size_t t = 0; t < ts.size(); ++t) {
+    TORCH_CHECK(ts[t].is_cuda() && ts[t].is_contiguous() &&
+                ts[t].scalar_type() == at::kFloat,
+                "zero_mt_prepare: tensors must be contiguous f32 CUDA");
+    descs[t] = {ts[t].data_ptr<float>(), ts[t].numel()};
+    for (int64_t off = 0; off < ts[t].numel(); off += CHUNK)
+      chunks.push_back({(int)t, off});
+  }
+  return {blob_to_dev(descs.data(), descs.size() * sizeof(MTZeroDescHost),
+                      ts[0]),
+          blob_to_dev(chunks.data(), chunks.size() * sizeof(MTChunkHost),
+                      ts[0]),
+          at::scalar_tensor((int64_t)chunks.size())};
+}
+
+void zero_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
+                 int64_t nchunks) {
+  ps_zero_mt(desc_dev.data_ptr(), chunk_dev.data_ptr(), (int)nchunks,
+             stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -936,6 +1032,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_backward", &dropout_backward);
   m.def("sgd_update", &sgd_update);
   m.def("sgd_update_lrdev", &sgd_update_lrdev);
+  m.def("sgd_mt_prepare", &sgd_mt_prepare);
+  m.def("sgd_mt_run", &sgd_mt_run);
+  m.def("zero_mt_prepare", &zero_mt_prepare);
+  m.def("zero_mt_run", &zero_mt_run);
   m.def("dropout_forward_offdev", &dropout_forward_offdev);
   m.def("nesterov_update", &nesterov_update);
   m.def("adagrad_update", &adagrad_update);

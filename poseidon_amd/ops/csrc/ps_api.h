@@ -166,6 +166,13 @@ void ps_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
 void ps_sgd_update_lrdev(float*, const float*, float*, int64_t, float, float,
                          float, const float*, hipStream_t);
 void ps_u64_inc(void*, hipStream_t);
+// multi-tensor apply (descriptor/chunk tables built by bindings; layouts in
+// sgd.hip: MTDesc {w,g,h,n,lr_mult,wd}, MTZeroDesc {p,n}, MTChunk {t,off})
+int ps_mt_chunk_elts(void);
+void ps_sgd_mt(const void* descs, const void* chunks, int nchunks, float lr,
+               float mom, const float* lr_dev, hipStream_t);
+void ps_zero_mt(const void* descs, const void* chunks, int nchunks,
+                hipStream_t);
 void ps_dropout_fwd_f32_offdev(const float*, float*, uint8_t*, int64_t, float,
                                uint64_t, const void*, hipStream_t);
 void ps_dropout_fwd_bf16_offdev(const void*, void*, uint8_t*, int64_t, float,

@@ -444,6 +444,34 @@ def sgd_update(w: torch.Tensor, grad: torch.Tensor, hist: torch.Tensor,
     w.sub_(hist)
 
 
+# -- multi-tensor apply: one launch for ALL params (GPU only) ---------------
+
+def sgd_mt_prepare(ws, gs, hs, lr_mults, wds):
+    """Build device-resident descriptor/chunk tables for sgd_mt_run. Shapes
+    and tensor identities must stay fixed afterwards (they do: param blobs
+    are allocated once at net build)."""
+    d, c, n = _ext().sgd_mt_prepare(list(ws), list(gs), list(hs),
+                                    [float(v) for v in lr_mults],
+                                    [float(v) for v in wds])
+    return d, c, int(n.item())
+
+
+def sgd_mt_run(mt, lr: float, momentum: float, lr_dev=None) -> None:
+    desc, chunk, nchunks = mt
+    _ext().sgd_mt_run(desc, chunk, nchunks, float(lr), float(momentum),
+                      lr_dev)
+
+
+def zero_mt_prepare(tensors):
+    d, c, n = _ext().zero_mt_prepare(list(tensors))
+    return d, c, int(n.item())
+
+
+def zero_mt_run(mt) -> None:
+    desc, chunk, nchunks = mt
+    _ext().zero_mt_run(desc, chunk, nchunks)
+
+
 def nesterov_update(w, grad, hist, local_rate: float, momentum: float,
                     decay: float) -> None:
     """update = (1+mu)*h_new - mu*h_old (solver.cpp:1013-1120)."""

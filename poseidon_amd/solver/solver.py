@@ -99,6 +99,9 @@ class SGDSolver:
         self._graph = None
         self._graph_loss = None
         self._lr_dev: Optional[torch.Tensor] = None
+        # multi-tensor update table (one kernel for ALL params); keyed on
+        # tensor identities so restore()/load_weights() invalidate it
+        self._mt = None
 
     # ------------------------------------------------------------------
     # hipGraph iteration capture
@@ -126,15 +129,48 @@ class SGDSolver:
 
     def _graph_body(self) -> torch.Tensor:
         loss = self.forward_backward()
-        for i, ps in enumerate(self.net.params):
-            if ps.owner == i and ps.lr_mult != 0.0:
-                wd = float(self.param.weight_decay or 0.0) * ps.decay_mult
-                if self.distributed:
-                    wd *= ctx().world_size
-                ops.sgd_update(ps.blob.data, ps.blob.diff, self.history[i],
-                               ps.lr_mult, float(self.param.momentum or 0.0),
-                               wd, lr_dev=self._lr_dev)
+        if not self._mt_update(0.0, lr_dev=self._lr_dev):
+            for i, ps in enumerate(self.net.params):
+                if ps.owner == i and ps.lr_mult != 0.0:
+                    wd = float(self.param.weight_decay or 0.0) * ps.decay_mult
+                    if self.distributed:
+                        wd *= ctx().world_size
+                    ops.sgd_update(ps.blob.data, ps.blob.diff,
+                                   self.history[i], ps.lr_mult,
+                                   float(self.param.momentum or 0.0),
+                                   wd, lr_dev=self._lr_dev)
         return loss
+
+    # -- multi-tensor update: one sgd_mt kernel covers every param ------
+    def _mt_update(self, rate: float, lr_dev=None) -> bool:
+        """Returns True if the whole update ran as one multi-tensor kernel
+        (SGD + L2 + GPU only; Nesterov/AdaGrad subclasses and the L1 path
+        keep per-param kernels). The descriptor table is rebuilt whenever
+        any tensor identity changes (restore, first backward, ...)."""
+        if type(self) is not SGDSolver or ctx().device != "cuda":
+            return False
+        if self.param.regularization_type == "L1":
+            return False
+        items = [(i, ps) for i, ps in enumerate(self.net.params)
+                 if ps.owner == i and ps.lr_mult != 0.0]
+        if not items or not all(ps.blob.has_diff() for _, ps in items):
+            return False
+        key = [(id(ps.blob.data), id(ps.blob.diff), id(self.history[i]))
+               for i, ps in items]
+        if self._mt is None or self._mt[1] != key:
+            wd0 = float(self.param.weight_decay or 0.0)
+            if self.distributed:
+                wd0 *= ctx().world_size
+            mt = ops.sgd_mt_prepare(
+                [ps.blob.data for _, ps in items],
+                [ps.blob.diff for _, ps in items],
+                [self.history[i] for i, _ in items],
+                [ps.lr_mult for _, ps in items],
+                [wd0 * ps.decay_mult for _, ps in items])
+            self._mt = (mt, key)
+        ops.sgd_mt_run(self._mt[0], rate,
+                       float(self.param.momentum or 0.0), lr_dev)
+        return True
 
     def _capture_graph(self) -> None:
         dev = ctx().torch_device
@@ -264,9 +300,10 @@ class SGDSolver:
                 self.test_all()
             loss_t = self.forward_backward()
             rate = self.get_learning_rate()
-            for i, ps in enumerate(self.net.params):
-                if ps.owner == i and ps.lr_mult != 0.0:
-                    self._apply_update(i, ps, rate)
+            if not self._mt_update(rate):
+                for i, ps in enumerate(self.net.params):
+                    if ps.owner == i and ps.lr_mult != 0.0:
+                        self._apply_update(i, ps, rate)
             if p.display and self.iter % p.display == 0:
                 last_loss = float(loss_t.item())
                 self._display(last_loss, rate)
