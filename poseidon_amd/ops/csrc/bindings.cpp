@@ -117,7 +117,19 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   // 2 blocks/CU resident -> ~512 workgroups fill the chip; split K until
   // the grid gets there (wgrad at 512x4608 is 144 tiles = 28% occupancy
   // without this)
-  if (tiles < 384 && K >= 4096) {
+  const bool atomic_ok = out_f32 && bias == nullptr && beta == 0.0f &&
+                         !relu && ldc == N;
+  if (atomic_ok && tiles < 384 && K >= 1024) {
+    // atomic split-K: no workspace, so split much deeper (chunk >= 512)
+    // -- the un-split grid leaves most of the 256 CUs idle
+    int sk = (int)std::min<int64_t>((512 + tiles - 1) / tiles,
+                                    (K + 511) / 512);
+    if (sk > 1) {
+      g.splitk = sk;
+      g.ws = nullptr;
+      hipMemsetAsync(g.C, 0, (size_t)M * N * sizeof(float), stream());
+    }
+  } else if (tiles < 384 && K >= 4096) {
     int sk = (int)std::min<int64_t>((512 + tiles - 1) / tiles,
                                     (K + 2047) / 2048);
     int64_t ws_elems = (int64_t)sk * M * N;

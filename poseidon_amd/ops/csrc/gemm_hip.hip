@@ -23,6 +23,8 @@
 // Replaces the reference's cuBLAS call sites (math_functions.cu:16-65,
 // conv_layer.cu:25-121, inner_product_layer.cu:18-63).
 
+#include <type_traits>
+
 #include "ps_common_hip.h"
 #include "ps_api.h"
 
@@ -462,8 +464,15 @@ epilogue:
         int row = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
         if (row >= M) continue;
         if (SPLITK) {
-          // raw partial; alpha/beta/bias applied by the reduce kernel
-          ws[((int64_t)blockIdx.z * M + row) * N + col] = acc[fm][fn][r];
+          if (ws) {
+            // raw partial; alpha/beta/bias applied by the reduce kernel
+            ws[((int64_t)blockIdx.z * M + row) * N + col] = acc[fm][fn][r];
+          } else if constexpr (std::is_same<OUT, float>::value) {
+            // atomic split-K: accumulate straight into zeroed f32 C --
+            // no workspace round-trip, no reduce kernel; lets the launcher
+            // split K much deeper (wgrad tiles are tiny: M=Cout)
+            atomicAdd(&C[(int64_t)row * ldc + col], alpha * acc[fm][fn][r]);
+          }
         } else {
           int64_t idx = (int64_t)row * ldc + col;
           float v = alpha * acc[fm][fn][r] + bv;
@@ -521,6 +530,7 @@ static void launch_tile(const GemmArgs& g, hipStream_t s) {
             (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,
             g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha,
             g.beta, (float*)g.ws, kchunk, da, db, g.relu);
+    if (!g.ws) return;  // atomic split-K accumulated into C directly
     int64_t MN = (int64_t)g.M * g.N;
     int64_t rb = cdiv64(MN, 256);
     if (rb > 2048) rb = 2048;

@@ -39,7 +39,8 @@ def close(a, b, rtol=2e-4, atol=2e-4, what=""):
 
 @pytest.mark.parametrize("ak,bk", [(True, True), (True, False),
                                    (False, True), (False, False)])
-@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (190, 70, 85), (64, 257, 300)])
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (190, 70, 85), (64, 257, 300),
+                                   (64, 576, 25088)])  # wgrad shape: atomic split-K
 def test_gemm_layouts(ak, bk, M, N, K):
     from poseidon_amd.ops._backend import load
     ext = load()
