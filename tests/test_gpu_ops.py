@@ -50,7 +50,10 @@ def test_gemm_layouts(ak, bk, M, N, K):
     A = opA.contiguous() if ak else opA.t().contiguous()
     B = opB.t().contiguous() if bk else opB.contiguous()
     out = ext.gemm(A.to(DEV), B.to(DEV), M, N, K, ak, bk)
-    close(out, ref, what=f"gemm ak={ak} bk={bk} {M}x{N}x{K}")
+    # f32 rounding grows ~sqrt(K) with accumulation depth (split-K partial
+    # order differs from torch's); loosen for the deep-K wgrad shape
+    tol = 2e-4 if K < 4096 else 1e-3
+    close(out, ref, rtol=tol, atol=tol, what=f"gemm ak={ak} bk={bk} {M}x{N}x{K}")
 
 
 def test_linear_fwd_bwd():
