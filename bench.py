@@ -35,6 +35,9 @@ def main():
                     choices=list(DEFAULT_BATCH))
     ap.add_argument("--batch", type=int, default=0, help="per-GPU batch")
     ap.add_argument("--no-sfb", action="store_true")
+    ap.add_argument("--implicit", action="store_true",
+                    help="implicit-GEMM convolutions (gather in GEMM staging"
+                         " instead of materialized im2col)")
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph capture of the iteration")
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"],
@@ -58,6 +61,9 @@ def main():
     cd = torch.bfloat16 if (args.dtype == "bf16" and device == "cuda") \
         else torch.float32
     pa.init(device=device, seed=1234, compute_dtype=cd)
+    if args.implicit and device == "cuda":
+        from poseidon_amd.ops import functional as F
+        F.set_implicit_gemm(True)
     batch = args.batch or DEFAULT_BATCH[args.model]
 
     sp = Message("SolverParameter", base_lr=0.01, lr_policy="fixed",

@@ -634,6 +634,16 @@ std::vector<Tensor> lrn_forward(const Tensor& x, int size, double alpha,
   int C = x_cl.size(1);
   auto opts_cl = x.options().memory_format(at::MemoryFormat::ChannelsLast);
   Tensor y = at::empty_like(x_cl, opts_cl);
+  if (ps_lrn_v8_ok(C, size)) {
+    // halo-register path: no scale tensor at all (backward recomputes it)
+    if (is_bf16(x))
+      ps_lrn_fwd_v8_bf16(x_cl.data_ptr(), y.data_ptr(), rows, C, size,
+                         (float)alpha, (float)beta, stream());
+    else
+      ps_lrn_fwd_v8_f32(x_cl.data_ptr<float>(), y.data_ptr<float>(), rows, C,
+                        size, (float)alpha, (float)beta, stream());
+    return {y, at::empty({0}, x.options().dtype(at::kFloat))};
+  }
   Tensor scale = at::empty_like(x_cl,
       x.options().dtype(at::kFloat).memory_format(at::MemoryFormat::ChannelsLast));
   if (is_bf16(x))
@@ -650,11 +660,23 @@ Tensor lrn_backward(const Tensor& x, const Tensor& y, const Tensor& scale,
                     const Tensor& dy, int size, double alpha, double beta) {
   auto x_cl = cl4(x);
   auto y_cl = cl4(y);
-  auto sc_cl = cl4(scale);
   auto dy_cl = cl4(dy);
   int64_t rows = (int64_t)x_cl.size(0) * x_cl.size(2) * x_cl.size(3);
   int C = x_cl.size(1);
   Tensor dx = at::empty_like(x_cl, x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  if (scale.numel() == 0) {  // forward used the halo-register path
+    TORCH_CHECK(ps_lrn_v8_ok(C, size), "lrn: empty scale but v8 ineligible");
+    if (is_bf16(x))
+      ps_lrn_bwd_v8_bf16(x_cl.data_ptr(), y_cl.data_ptr(), dy_cl.data_ptr(),
+                         dx.data_ptr(), rows, C, size, (float)alpha,
+                         (float)beta, stream());
+    else
+      ps_lrn_bwd_v8_f32(x_cl.data_ptr<float>(), y_cl.data_ptr<float>(),
+                        dy_cl.data_ptr<float>(), dx.data_ptr<float>(), rows,
+                        C, size, (float)alpha, (float)beta, stream());
+    return dx;
+  }
+  auto sc_cl = cl4(scale);
   // ratio workspace only feeds the C>256 fallback path; the row-block
   // kernels keep the ratio in LDS
   Tensor ratio = at::empty({C > 256 ? rows * C : 0},

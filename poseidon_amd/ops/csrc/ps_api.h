@@ -166,6 +166,15 @@ void ps_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
 void ps_sgd_update_lrdev(float*, const float*, float*, int64_t, float, float,
                          float, const float*, hipStream_t);
 void ps_u64_inc(void*, hipStream_t);
+int ps_lrn_v8_ok(int C, int size);
+void ps_lrn_fwd_v8_f32(const float*, float*, int64_t, int, int, float, float,
+                       hipStream_t);
+void ps_lrn_fwd_v8_bf16(const void*, void*, int64_t, int, int, float, float,
+                        hipStream_t);
+void ps_lrn_bwd_v8_f32(const float*, const float*, const float*, float*,
+                       int64_t, int, int, float, float, hipStream_t);
+void ps_lrn_bwd_v8_bf16(const void*, const void*, const void*, void*, int64_t,
+                        int, int, float, float, hipStream_t);
 // multi-tensor apply (descriptor/chunk tables built by bindings; layouts in
 // sgd.hip: MTDesc {w,g,h,n,lr_mult,wd}, MTZeroDesc {p,n}, MTChunk {t,off})
 int ps_mt_chunk_elts(void);
