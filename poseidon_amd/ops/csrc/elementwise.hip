@@ -165,6 +165,44 @@ __global__ void colsum_k(const T* __restrict__ in, float* __restrict__ out,
   }
 }
 
+// Flat-vector colsum: whole-wave 16 B loads regardless of C (the banded
+// kernel above degrades to C*sizeof(T) bytes per wave transaction when
+// C < 2*64 lanes, e.g. conv bias grads with C<=128). Each thread walks the
+// matrix as a flat vec stream, tracking its column phase incrementally
+// (c0 advances by grid_stride*VEC mod C -- no per-iteration division), and
+// accumulates into LDS f32 column partials; one global atomic per column
+// per block at the end. Requires C % VEC == 0.
+template <typename T, int VEC>
+__global__ void colsum_flat_k(const T* __restrict__ in,
+                              float* __restrict__ out, int64_t R, int C) {
+  extern __shared__ float part[];  // C floats
+  for (int c = threadIdx.x; c < C; c += blockDim.x) part[c] = 0.f;
+  __syncthreads();
+  typedef T vecT __attribute__((ext_vector_type(VEC)));
+  const int64_t nvec = R * C / VEC;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int c0 = (int)((i * VEC) % C);  // multiple of VEC since C % VEC == 0
+  const int rot = (int)((stride * VEC) % C);
+  for (; i < nvec; i += stride) {
+    vecT v = *((const vecT*)in + i);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) atomicAdd(&part[c0 + j], to_f32(v[j]));
+    c0 += rot;
+    if (c0 >= C) c0 -= C;
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x)
+    atomicAdd(&out[c], part[c]);
+}
+
+static inline dim3 colsum_flat_grid(int64_t nvec) {
+  int64_t blocks = cdiv64(nvec, 256);
+  if (blocks > 2048) blocks = 2048;
+  return dim3((unsigned)blocks);
+}
+
+
 // ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
@@ -249,43 +287,6 @@ void ps_dropout_bwd_bf16(const void* dy, const uint8_t* mask, void* dx,
                          int64_t n, float ratio, hipStream_t s) {
   PS_EW_LAUNCH(dropout_bwd_k<__bf16>, (const __bf16*)dy, mask, (__bf16*)dx, n,
                1.0f / (1.0f - ratio));
-}
-
-// Flat-vector colsum: whole-wave 16 B loads regardless of C (the banded
-// kernel above degrades to C*sizeof(T) bytes per wave transaction when
-// C < 2*64 lanes, e.g. conv bias grads with C<=128). Each thread walks the
-// matrix as a flat vec stream, tracking its column phase incrementally
-// (c0 advances by grid_stride*VEC mod C -- no per-iteration division), and
-// accumulates into LDS f32 column partials; one global atomic per column
-// per block at the end. Requires C % VEC == 0.
-template <typename T, int VEC>
-__global__ void colsum_flat_k(const T* __restrict__ in,
-                              float* __restrict__ out, int64_t R, int C) {
-  extern __shared__ float part[];  // C floats
-  for (int c = threadIdx.x; c < C; c += blockDim.x) part[c] = 0.f;
-  __syncthreads();
-  typedef T vecT __attribute__((ext_vector_type(VEC)));
-  const int64_t nvec = R * C / VEC;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int c0 = (int)((i * VEC) % C);  // multiple of VEC since C % VEC == 0
-  const int rot = (int)((stride * VEC) % C);
-  for (; i < nvec; i += stride) {
-    vecT v = *((const vecT*)in + i);
-#pragma unroll
-    for (int j = 0; j < VEC; ++j) atomicAdd(&part[c0 + j], to_f32(v[j]));
-    c0 += rot;
-    if (c0 >= C) c0 -= C;
-  }
-  __syncthreads();
-  for (int c = threadIdx.x; c < C; c += blockDim.x)
-    atomicAdd(&out[c], part[c]);
-}
-
-static inline dim3 colsum_flat_grid(int64_t nvec) {
-  int64_t blocks = cdiv64(nvec, 256);
-  if (blocks > 2048) blocks = 2048;
-  return dim3((unsigned)blocks);
 }
 
 static inline dim3 colsum_grid(int64_t R, int C) {

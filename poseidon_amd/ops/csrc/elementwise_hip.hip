@@ -166,6 +166,44 @@ __global__ void colsum_k(const T* __restrict__ in, float* __restrict__ out,
   }
 }
 
+// Flat-vector colsum: whole-wave 16 B loads regardless of C (the banded
+// kernel above degrades to C*sizeof(T) bytes per wave transaction when
+// C < 2*64 lanes, e.g. conv bias grads with C<=128). Each thread walks the
+// matrix as a flat vec stream, tracking its column phase incrementally
+// (c0 advances by grid_stride*VEC mod C -- no per-iteration division), and
+// accumulates into LDS f32 column partials; one global atomic per column
+// per block at the end. Requires C % VEC == 0.
+template <typename T, int VEC>
+__global__ void colsum_flat_k(const T* __restrict__ in,
+                              float* __restrict__ out, int64_t R, int C) {
+  extern __shared__ float part[];  // C floats
+  for (int c = threadIdx.x; c < C; c += blockDim.x) part[c] = 0.f;
+  __syncthreads();
+  typedef T vecT __attribute__((ext_vector_type(VEC)));
+  const int64_t nvec = R * C / VEC;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int c0 = (int)((i * VEC) % C);  // multiple of VEC since C % VEC == 0
+  const int rot = (int)((stride * VEC) % C);
+  for (; i < nvec; i += stride) {
+    vecT v = *((const vecT*)in + i);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) atomicAdd(&part[c0 + j], to_f32(v[j]));
+    c0 += rot;
+    if (c0 >= C) c0 -= C;
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x)
+    atomicAdd(&out[c], part[c]);
+}
+
+static inline dim3 colsum_flat_grid(int64_t nvec) {
+  int64_t blocks = cdiv64(nvec, 256);
+  if (blocks > 2048) blocks = 2048;
+  return dim3((unsigned)blocks);
+}
+
+
 // ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
@@ -260,10 +298,18 @@ static inline dim3 colsum_grid(int64_t R, int C) {
   return dim3((unsigned)blocks);
 }
 void ps_colsum_f32(const float* in, float* out, int64_t R, int C, hipStream_t s) {
- hipLaunchKernelGGL(( colsum_k<float>), dim3(colsum_grid(R, C)), dim3(256), 0, s, in, out, R, C);
+  if (C % 4 == 0 && C <= 8192)
+   hipLaunchKernelGGL(( colsum_flat_k<float, 4>), dim3(colsum_flat_grid(R * C / 4)), dim3(256), C * 4, s, 
+        in, out, R, C);
+  else
+   hipLaunchKernelGGL(( colsum_k<float>), dim3(colsum_grid(R, C)), dim3(256), 0, s, in, out, R, C);
 }
 void ps_colsum_bf16(const void* in, float* out, int64_t R, int C, hipStream_t s) {
- hipLaunchKernelGGL(( colsum_k<__bf16>), dim3(colsum_grid(R, C)), dim3(256), 0, s, (const __bf16*)in, out, R, C);
+  if (C % 8 == 0 && C <= 8192)
+   hipLaunchKernelGGL(( colsum_flat_k<__bf16, 8>), dim3(colsum_flat_grid(R * C / 8)), dim3(256), C * 4, s, 
+        (const __bf16*)in, out, R, C);
+  else
+   hipLaunchKernelGGL(( colsum_k<__bf16>), dim3(colsum_grid(R, C)), dim3(256), 0, s, (const __bf16*)in, out, R, C);
 }
 
 }  // extern "C"

@@ -151,7 +151,186 @@ static inline dim3 lrn_grid(int64_t rows, int rpb) {
   return dim3((unsigned)blocks);
 }
 
+// ---------------------------------------------------------------------------
+// Vec8 halo-register variants (C % 8 == 0, odd size, pre <= 4): each thread
+// owns 8 contiguous channels of one pixel, loads them as ONE 16 B vector
+// plus a few scalar halo loads (L1 hits -- neighbours just fetched them),
+// and keeps the whole window arithmetic in registers. No LDS, no barriers,
+// no stored scale tensor (backward recomputes it from x, trading a little
+// ALU for 4 B/elt of HBM writes+reads). The row-block kernels above remain
+// the fallback (they store scale) for odd channel counts.
+// ---------------------------------------------------------------------------
+
+template <typename T> struct LrnV8;
+template <> struct LrnV8<float> {
+  typedef f32x4 half_t;
+  __device__ static inline void load8(const float* p, float* out) {
+    f32x4 a = *(const f32x4*)p, b = *(const f32x4*)(p + 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) { out[j] = a[j]; out[4 + j] = b[j]; }
+  }
+  __device__ static inline void store8(float* p, const float* v) {
+    f32x4 a, b;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) { a[j] = v[j]; b[j] = v[4 + j]; }
+    *(f32x4*)p = a;
+    *(f32x4*)(p + 4) = b;
+  }
+};
+template <> struct LrnV8<__bf16> {
+  __device__ static inline void load8(const __bf16* p, float* out) {
+    bf16x8 v = *(const bf16x8*)p;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = (float)v[j];
+  }
+  __device__ static inline void store8(__bf16* p, const float* v) {
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = (__bf16)v[j];
+    *(bf16x8*)p = o;
+  }
+};
+
+template <typename T, int PRE>
+__global__ void lrn_fwd_v8_k(const T* __restrict__ x, T* __restrict__ y,
+                             int64_t rows, int C, float alpha_over_n,
+                             float beta, int gpr, int rpb) {
+  const int t = threadIdx.x;
+  const int lr = t / gpr, lg = t - lr * gpr;  // once per thread
+  const int cb = lg * 8;
+  const bool active = lr < rpb;
+  for (int64_t r0 = (int64_t)blockIdx.x * rpb; r0 < rows;
+       r0 += (int64_t)gridDim.x * rpb) {
+    const int64_t row = r0 + lr;
+    if (!active || row >= rows) continue;  // barrier-free: safe to skip
+    const T* xr = x + row * C;
+    float xv[8], xs[8 + 2 * PRE];
+    LrnV8<T>::load8(xr + cb, xv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) xs[PRE + j] = xv[j] * xv[j];
+#pragma unroll
+    for (int j = 0; j < PRE; ++j) {
+      const int cl = cb - PRE + j, cr2 = cb + 8 + j;
+      float a = (cl >= 0) ? to_f32(xr[cl]) : 0.f;   // zeros implement the
+      float b = (cr2 < C) ? to_f32(xr[cr2]) : 0.f;  // clipped edge window
+      xs[j] = a * a;
+      xs[PRE + 8 + j] = b * b;
+    }
+    float out[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float ss = 0.f;
+#pragma unroll
+      for (int k = 0; k <= 2 * PRE; ++k) ss += xs[j + k];
+      out[j] = xv[j] * __powf(1.0f + alpha_over_n * ss, -beta);
+    }
+    LrnV8<T>::store8(y + row * C + cb, out);
+  }
+}
+
+template <typename T, int PRE>
+__global__ void lrn_bwd_v8_k(const T* __restrict__ x, const T* __restrict__ y,
+                             const T* __restrict__ dy, T* __restrict__ dx,
+                             int64_t rows, int C, float alpha_over_n,
+                             float cache_ratio, float beta, int gpr, int rpb) {
+  const int t = threadIdx.x;
+  const int lr = t / gpr, lg = t - lr * gpr;
+  const int cb = lg * 8;
+  const bool active = lr < rpb;
+  for (int64_t r0 = (int64_t)blockIdx.x * rpb; r0 < rows;
+       r0 += (int64_t)gridDim.x * rpb) {
+    const int64_t row = r0 + lr;
+    if (!active || row >= rows) continue;
+    const T* xr = x + row * C;
+    const T* yr = y + row * C;
+    const T* dyr = dy + row * C;
+    // x^2 over [cb-2P, cb+8+2P): recomputes the scale of every window
+    // position touching this thread's 8 outputs
+    float xs[8 + 4 * PRE];
+#pragma unroll
+    for (int j = 0; j < 8 + 4 * PRE; ++j) {
+      const int c = cb - 2 * PRE + j;
+      const float v = (c >= 0 && c < C) ? to_f32(xr[c]) : 0.f;
+      xs[j] = v * v;
+    }
+    // ratio = dy*y/scale over [cb-P, cb+8+P)
+    float ratio[8 + 2 * PRE], scc[8];
+#pragma unroll
+    for (int m = 0; m < 8 + 2 * PRE; ++m) {
+      const int c = cb - PRE + m;
+      float ss = 0.f;
+#pragma unroll
+      for (int k = 0; k <= 2 * PRE; ++k) ss += xs[m + k];
+      const float sc = 1.0f + alpha_over_n * ss;
+      if (m >= PRE && m < PRE + 8) scc[m - PRE] = sc;
+      ratio[m] = (c >= 0 && c < C)
+                     ? to_f32(dyr[c]) * to_f32(yr[c]) / sc
+                     : 0.f;
+    }
+    float out[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float acc = 0.f;
+#pragma unroll
+      for (int k = 0; k <= 2 * PRE; ++k) acc += ratio[j + k];
+      out[j] = to_f32(dyr[cb + j]) * __powf(scc[j], -beta)
+               - cache_ratio * to_f32(xr[cb + j]) * acc;
+    }
+    LrnV8<T>::store8(dx + row * C + cb, out);
+  }
+}
+
+template <typename T>
+static bool lrn_v8_ok(int C, int size) {
+  const int pre = (size - 1) / 2;
+  return (C % 8 == 0) && C >= 8 && (C / 8) <= 256 && (size % 2 == 1) &&
+         pre >= 1 && pre <= 4;
+}
+
+#define LRN_V8_DISPATCH(kernel, T, ...)                                   \
+  do {                                                                    \
+    const int pre_ = (size - 1) / 2;                                      \
+    const int gpr_ = C / 8;                                               \
+    const int rpb_ = 256 / gpr_;                                          \
+    dim3 grid_ = lrn_grid(rows, rpb_);                                    \
+    switch (pre_) {                                                       \
+      case 1:hipLaunchKernelGGL(( kernel<T, 1>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
+      case 2:hipLaunchKernelGGL(( kernel<T, 2>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
+      case 3:hipLaunchKernelGGL(( kernel<T, 3>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
+      default:hipLaunchKernelGGL(( kernel<T, 4>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
+    }                                                                     \
+  } while (0)
+
 extern "C" {
+
+int ps_lrn_v8_ok(int C, int size) { return lrn_v8_ok<float>(C, size) ? 1 : 0; }
+
+void ps_lrn_fwd_v8_f32(const float* x, float* y, int64_t rows, int C,
+                       int size, float alpha, float beta, hipStream_t s) {
+  const float an = alpha / size;
+  LRN_V8_DISPATCH(lrn_fwd_v8_k, float, x, y, rows, C, an, beta);
+}
+void ps_lrn_fwd_v8_bf16(const void* x, void* y, int64_t rows, int C,
+                        int size, float alpha, float beta, hipStream_t s) {
+  const float an = alpha / size;
+  LRN_V8_DISPATCH(lrn_fwd_v8_k, __bf16, (const __bf16*)x, (__bf16*)y, rows, C,
+                  an, beta);
+}
+void ps_lrn_bwd_v8_f32(const float* x, const float* y, const float* dy,
+                       float* dx, int64_t rows, int C, int size, float alpha,
+                       float beta, hipStream_t s) {
+  const float an = alpha / size;
+  const float cr = 2.0f * alpha * beta / size;
+  LRN_V8_DISPATCH(lrn_bwd_v8_k, float, x, y, dy, dx, rows, C, an, cr, beta);
+}
+void ps_lrn_bwd_v8_bf16(const void* x, const void* y, const void* dy,
+                        void* dx, int64_t rows, int C, int size, float alpha,
+                        float beta, hipStream_t s) {
+  const float an = alpha / size;
+  const float cr = 2.0f * alpha * beta / size;
+  LRN_V8_DISPATCH(lrn_bwd_v8_k, __bf16, (const __bf16*)x, (const __bf16*)y,
+                  (const __bf16*)dy, (__bf16*)dx, rows, C, an, cr, beta);
+}
 
 void ps_lrn_fwd_f32(const float* x, float* y, float* scale, int64_t rows,
                     int C, int size, float alpha, float beta, hipStream_t s) {
