@@ -168,13 +168,15 @@ __global__ void colsum_k(const T* __restrict__ in, float* __restrict__ out,
 // Flat-vector colsum: whole-wave 16 B loads regardless of C (the banded
 // kernel above degrades to C*sizeof(T) bytes per wave transaction when
 // C < 2*64 lanes, e.g. conv bias grads with C<=128). Each thread walks the
-// matrix as a flat vec stream, tracking its column phase incrementally
-// (c0 advances by grid_stride*VEC mod C -- no per-iteration division), and
-// accumulates into LDS f32 column partials; one global atomic per column
-// per block at the end. Requires C % VEC == 0.
-template <typename T, int VEC>
+// matrix as a flat vec stream; its column phase cycles with period
+// PH = C/gcd(grid_stride*VEC, C), so it accumulates into PH*VEC REGISTERS
+// and touches LDS only once at flush (a per-load LDS-atomic variant
+// serialized on slot contention and measured slower than the banded
+// kernel). PH > 4 falls back to the banded kernel.
+template <typename T, int VEC, int PH>
 __global__ void colsum_flat_k(const T* __restrict__ in,
-                              float* __restrict__ out, int64_t R, int C) {
+                              float* __restrict__ out, int64_t R, int C,
+                              int rot) {
   extern __shared__ float part[];  // C floats
   for (int c = threadIdx.x; c < C; c += blockDim.x) part[c] = 0.f;
   __syncthreads();
@@ -182,18 +184,34 @@ __global__ void colsum_flat_k(const T* __restrict__ in,
   const int64_t nvec = R * C / VEC;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int c0 = (int)((i * VEC) % C);  // multiple of VEC since C % VEC == 0
-  const int rot = (int)((stride * VEC) % C);
+  const int c0 = (int)((i * VEC) % C);  // multiple of VEC (C % VEC == 0)
+  float acc[PH][VEC];
+#pragma unroll
+  for (int q = 0; q < PH; ++q)
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) acc[q][j] = 0.f;
+  int p = 0;
   for (; i < nvec; i += stride) {
     vecT v = *((const vecT*)in + i);
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) atomicAdd(&part[c0 + j], to_f32(v[j]));
-    c0 += rot;
-    if (c0 >= C) c0 -= C;
+    for (int j = 0; j < VEC; ++j) acc[p][j] += to_f32(v[j]);
+    if (PH > 1 && ++p == PH) p = 0;
+  }
+#pragma unroll
+  for (int q = 0; q < PH; ++q) {
+    int c = c0 + q * rot;
+    while (c >= C) c -= C;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) atomicAdd(&part[c + j], acc[q][j]);
   }
   __syncthreads();
   for (int c = threadIdx.x; c < C; c += blockDim.x)
     atomicAdd(&out[c], part[c]);
+}
+
+static inline int colsum_gcd(int a, int b) {
+  while (b) { int t = a % b; a = b; b = t; }
+  return a;
 }
 
 static inline dim3 colsum_flat_grid(int64_t nvec) {
@@ -296,17 +314,36 @@ static inline dim3 colsum_grid(int64_t R, int C) {
   if (blocks > 1024) blocks = 1024;
   return dim3((unsigned)blocks);
 }
+#define PS_COLSUM_FLAT(T, VEC, inexpr)                                        \
+  do {                                                                        \
+    dim3 grid_ = colsum_flat_grid(R * C / VEC);                               \
+    int rot_ = (int)(((int64_t)grid_.x * 256 * VEC) % C);                     \
+    int ph_ = rot_ ? C / colsum_gcd(rot_, C) : 1;                             \
+    if (ph_ == 1)                                                             \
+      colsum_flat_k<T, VEC, 1><<<grid_, 256, C * 4, s>>>(inexpr, out, R, C,   \
+                                                         rot_);              \
+    else if (ph_ == 2)                                                        \
+      colsum_flat_k<T, VEC, 2><<<grid_, 256, C * 4, s>>>(inexpr, out, R, C,   \
+                                                         rot_);              \
+    else if (ph_ == 3)                                                        \
+      colsum_flat_k<T, VEC, 3><<<grid_, 256, C * 4, s>>>(inexpr, out, R, C,   \
+                                                         rot_);              \
+    else if (ph_ == 4)                                                        \
+      colsum_flat_k<T, VEC, 4><<<grid_, 256, C * 4, s>>>(inexpr, out, R, C,   \
+                                                         rot_);              \
+    else                                                                      \
+      colsum_k<T><<<colsum_grid(R, C), 256, 0, s>>>(inexpr, out, R, C);       \
+  } while (0)
+
 void ps_colsum_f32(const float* in, float* out, int64_t R, int C, hipStream_t s) {
-  if (C % 4 == 0 && C <= 8192)
-    colsum_flat_k<float, 4><<<colsum_flat_grid(R * C / 4), 256, C * 4, s>>>(
-        in, out, R, C);
+  if (C % 4 == 0 && C <= 8192 && R * C >= (64 << 10))
+    PS_COLSUM_FLAT(float, 4, in);
   else
     colsum_k<float><<<colsum_grid(R, C), 256, 0, s>>>(in, out, R, C);
 }
 void ps_colsum_bf16(const void* in, float* out, int64_t R, int C, hipStream_t s) {
-  if (C % 8 == 0 && C <= 8192)
-    colsum_flat_k<__bf16, 8><<<colsum_flat_grid(R * C / 8), 256, C * 4, s>>>(
-        (const __bf16*)in, out, R, C);
+  if (C % 8 == 0 && C <= 8192 && R * C >= (64 << 10))
+    PS_COLSUM_FLAT(__bf16, 8, (const __bf16*)in);
   else
     colsum_k<__bf16><<<colsum_grid(R, C), 256, 0, s>>>((const __bf16*)in, out, R, C);
 }
