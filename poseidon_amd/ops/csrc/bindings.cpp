@@ -66,8 +66,15 @@ Tensor weight_shadow(const Tensor& w_f32, bool bf16) {
 // address ALU in the staging inner loop), so default off -- toggle with
 // set_implicit_gemm for experiments.
 std::atomic<bool> g_implicit_gemm{false};
+// Per-layer auto-policy: a conv whose materialized colT would exceed this
+// goes implicit regardless of the global flag (VGG's 224/112-resolution
+// 3x3 convs build 1-2 GB column matrices; gathering from x reads KHW x
+// less HBM). Measured: global implicit loses ~4-8% on AlexNet/GoogLeNet
+// (gather decode overhead on small colT), wins on the giant-colT layers.
+std::atomic<int64_t> g_implicit_thresh{512LL << 20};
 
 void set_implicit_gemm(bool on) { g_implicit_gemm.store(on); }
+void set_implicit_threshold(int64_t bytes) { g_implicit_thresh.store(bytes); }
 
 // 64B zero page for implicit-GEMM padding loads (per device, persistent)
 const void* zero_page(const Tensor& like) {
@@ -301,7 +308,9 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   // group channel count keeps 16B runs contiguous -- no column matrix at
   // all (the wgrad GEMM gathers too); conv1-style small-C layers still
   // materialize (and 1x1 convs alias x directly)
-  bool implicit = g_implicit_gemm.load() && !is_1x1 && (Cg % VEC == 0);
+  int64_t colT_bytes = NP * Kcol * (bf16 ? 2 : 4);
+  bool implicit = (g_implicit_gemm.load() || colT_bytes > g_implicit_thresh.load())
+                  && !is_1x1 && (Cg % VEC == 0);
   Tensor colT;
   if (is_1x1) {
     colT = rows2d(x_cl);  // alias: x rows ARE the col rows
@@ -1040,6 +1049,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_backward_input", &conv2d_backward_input);
   m.def("conv2d_backward_weight_acc", &conv2d_backward_weight_acc);
   m.def("set_implicit_gemm", &set_implicit_gemm);
+  m.def("set_implicit_threshold", &set_implicit_threshold);
   m.def("concat_channels", &concat_channels);
   m.def("slice_channels", &slice_channels);
   m.def("pool_max_forward", &pool_max_forward);

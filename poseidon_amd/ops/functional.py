@@ -49,6 +49,12 @@ def set_implicit_gemm(on: bool) -> None:
     _ext().set_implicit_gemm(bool(on))
 
 
+def set_implicit_threshold(bytes_: int) -> None:
+    """colT size above which a conv auto-selects implicit GEMM per layer."""
+    if ext_available():
+        _ext().set_implicit_threshold(int(bytes_))
+
+
 def ext_available() -> bool:
     try:
         _ext()

@@ -36,10 +36,18 @@ def main(argv=None):
     A = (opA if ak else opA.t()).contiguous().to("cuda", dt)
     B = (opB.t() if bk else opB).contiguous().to("cuda", dt)
 
-    out = ext.gemm(A, B, M, N, K, ak, bk)  # warmup + correctness spot
+    # correctness spot-check FIRST: the CPU reference matmul takes seconds,
+    # and any GPU-idle gap right before the timed region lets DVFS drop the
+    # clocks for the whole measurement (guide rule 24; measured 50x once)
+    out = ext.gemm(A, B, M, N, K, ak, bk)
     ref = (opA.to(dt).float() @ opB.to(dt).float())
     err = (out.cpu() - ref).abs().max().item()
+    # clock ramp: keep the GPU busy ~100 ms right before timing
     torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < 0.1:
+        ext.gemm(A, B, M, N, K, ak, bk)
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.reps):
         ext.gemm(A, B, M, N, K, ak, bk)
