@@ -306,13 +306,33 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
   constexpr int FM = BM / WGM / 16, FN = BN / WGN / 16;
   static_assert(WGM * WGN == 4, "4 waves per block");
 
+  // T1 XCD-aware swizzle: the dispatcher places linear block b on XCD b%8
+  // (each with a private 4 MiB L2); consecutive tiles share A/B panels, so
+  // give each XCD a CONTIGUOUS run of tiles instead of a round-robin comb.
+  // Bijective only when the flattened grid is a multiple of 8 -- identity
+  // otherwise. z (split-K slice / batch) folds into the flatten so the
+  // remap stays a permutation of the whole grid.
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  {
+    const int nbx = gridDim.x, nby = gridDim.y;
+    const int64_t nwg = (int64_t)nbx * nby * gridDim.z;
+    if ((nwg & 7) == 0 && nwg > 8) {
+      int64_t id = ((int64_t)bz * nby + by) * nbx + bx;
+      const int64_t cpx = nwg >> 3;
+      id = (id & 7) * cpx + (id >> 3);
+      bx = (int)(id % nbx);
+      by = (int)((id / nbx) % nby);
+      bz = (int)(id / ((int64_t)nbx * nby));
+    }
+  }
+
   // split-K: grid.z indexes the K-slice (batch must be 1); otherwise batch
-  const T* A = Abase + (SPLITK ? 0 : (int64_t)blockIdx.z * strideA);
-  const T* B = Bbase + (SPLITK ? 0 : (int64_t)blockIdx.z * strideB);
-  OUT* C = Cbase + (SPLITK ? 0 : (int64_t)blockIdx.z * strideC);
+  const T* A = Abase + (SPLITK ? 0 : (int64_t)bz * strideA);
+  const T* B = Bbase + (SPLITK ? 0 : (int64_t)bz * strideB);
+  OUT* C = Cbase + (SPLITK ? 0 : (int64_t)bz * strideC);
   int k_begin = 0, k_end = K;
   if (SPLITK) {
-    k_begin = blockIdx.z * kchunk;
+    k_begin = bz * kchunk;
     k_end = min(K, k_begin + kchunk);
   }
 
@@ -324,8 +344,8 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
   const int wid = tid >> 6;
   const int wm = (wid / WGN) * (BM / WGM);  // wave row offset in block tile
   const int wn = (wid % WGN) * (BN / WGN);
-  const int m0 = blockIdx.y * BM;
-  const int n0 = blockIdx.x * BN;
+  const int m0 = by * BM;
+  const int n0 = bx * BN;
 
   f32x4 acc[FM][FN] = {};
 
@@ -466,7 +486,7 @@ epilogue:
         if (SPLITK) {
           if (ws) {
             // raw partial; alpha/beta/bias applied by the reduce kernel
-            ws[((int64_t)blockIdx.z * M + row) * N + col] = acc[fm][fn][r];
+            ws[((int64_t)bz * M + row) * N + col] = acc[fm][fn][r];
           } else if constexpr (std::is_same<OUT, float>::value) {
             // atomic split-K: accumulate straight into zeroed f32 C --
             // no workspace round-trip, no reduce kernel; lets the launcher
