@@ -174,7 +174,23 @@ struct Stager {
   __device__ inline void load(const T* src, int64_t lda, int row0,
                               int rows_max, int k0, int K, int tid,
                               const GatherDesc* ga = nullptr) {
-    if (KLAST) {
+    // Interior tiles (the overwhelmingly common case) take a branch with NO
+    // per-vector guards: the guarded forms below make the compiler wrap
+    // EVERY 16 B load in s_and_saveexec exec-mask juggling and keep a
+    // scalarized (global_load_ushort) edge clone -- measured as the
+    // dominant cost of the K-major (TN/NN) staging path.
+    const bool interior = (row0 + ROWS <= rows_max) && (k0 + TR::BK <= K);
+    if (KLAST && !GATHER && interior) {
+      constexpr int CK = TR::BK / TR::VEC;
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        int c = tid + i * 256;
+        if (c >= TV) break;
+        int r = c / CK, kc = c % CK;
+        v[i] = *reinterpret_cast<const vec_t*>(
+            &src[(int64_t)(row0 + r) * lda + k0 + kc * TR::VEC]);
+      }
+    } else if (KLAST) {
       constexpr int CK = TR::BK / TR::VEC;
 #pragma unroll
       for (int i = 0; i < NV; ++i) {
@@ -197,6 +213,19 @@ struct Stager {
           }
         }
         v[i] = val;
+      }
+    } else if (GATHER && interior) {
+      constexpr int BLK_M = ROWS / TR::VEC;
+#pragma unroll
+      for (int i = 0; i < NVB; ++i) {
+        int c = tid + i * 256;
+        if (c >= TB) break;
+        int kb = c / BLK_M, mb = c % BLK_M;
+        int gm = row0 + mb * TR::VEC;
+#pragma unroll
+        for (int j = 0; j < TR::VEC; ++j)
+          v[i * TR::VEC + j] = *reinterpret_cast<const vec_t*>(
+              gather_addr<T>(*ga, k0 + kb * TR::VEC + j, gm));
       }
     } else if (GATHER) {
       // K-major gather (conv wgrad B = im2col): rows dim is Kg (contiguous
@@ -222,6 +251,21 @@ struct Stager {
           }
           v[i * TR::VEC + j] = val;
         }
+      }
+    } else if (interior) {
+      // K-major interior: VEC unguarded row-vector loads per block
+      constexpr int BLK_M = ROWS / TR::VEC;
+#pragma unroll
+      for (int i = 0; i < NVB; ++i) {
+        int c = tid + i * 256;
+        if (c >= TB) break;
+        int kb = c / BLK_M, mb = c % BLK_M;
+        const T* base = &src[(int64_t)(k0 + kb * TR::VEC) * lda
+                             + row0 + mb * TR::VEC];
+#pragma unroll
+        for (int j = 0; j < TR::VEC; ++j)
+          v[i * TR::VEC + j] =
+              *reinterpret_cast<const vec_t*>(base + (int64_t)j * lda);
       }
     } else {
       // K-major: each thread owns a VEC x VEC block (k-block kb, m-block mb)
