@@ -161,7 +161,8 @@ __device__ inline int lds_col(int row, int kk) {
 // phase (guide T14 / Guideline 15: write AFTER the barrier, re-issue at
 // once). A [ROWS x BK] tile is ROWS*BK/(256*VEC) vectors per thread.
 // Guarded, zero-filled out of range.
-template <typename T, int ROWS, bool KLAST, bool GATHER = false>
+template <typename T, int ROWS, bool KLAST, bool GATHER = false,
+          bool FASTI = false>
 struct Stager {
   using TR = GemmTraits<T>;
   using vec_t = typename TR::vec_t;
@@ -175,23 +176,17 @@ struct Stager {
   __device__ inline void load(const T* src, int64_t lda, int row0,
                               int rows_max, int k0, int K, int tid,
                               const GatherDesc* ga = nullptr) {
-    // Interior tiles (the overwhelmingly common case) take a branch with NO
-    // per-vector guards: the guarded forms below make the compiler wrap
-    // EVERY 16 B load in s_and_saveexec exec-mask juggling and keep a
-    // scalarized (global_load_ushort) edge clone -- measured as the
-    // dominant cost of the K-major (TN/NN) staging path.
-    const bool interior = (row0 + ROWS <= rows_max) && (k0 + TR::BK <= K);
-    if (KLAST && !GATHER && interior) {
-      constexpr int CK = TR::BK / TR::VEC;
-#pragma unroll
-      for (int i = 0; i < NV; ++i) {
-        int c = tid + i * 256;
-        if (c >= TV) break;
-        int r = c / CK, kc = c % CK;
-        v[i] = *reinterpret_cast<const vec_t*>(
-            &src[(int64_t)(row0 + r) * lda + k0 + kc * TR::VEC]);
-      }
-    } else if (KLAST) {
+    // Interior tiles of K-MAJOR operands take a branch with NO per-vector
+    // guards: the guarded form makes the compiler wrap EVERY 16 B load in
+    // s_and_saveexec exec-mask juggling plus a scalarized edge clone --
+    // measured as the dominant cost of the TN/NN staging path. The code
+    // duplication costs ~36 VGPRs, which pushes the NON-split-K 128x128
+    // bf16 instantiation over the 256-VGPR occupancy cliff (224 -> 260,
+    // -30% at 4096^3), so FASTI (= SPLITK at the launch site, 200 VGPR)
+    // gates it to the split-K kernels where it measures +9%.
+    const bool interior = FASTI &&
+        (row0 + ROWS <= rows_max) && (k0 + TR::BK <= K);
+    if (KLAST) {
       constexpr int CK = TR::BK / TR::VEC;
 #pragma unroll
       for (int i = 0; i < NV; ++i) {
@@ -214,19 +209,6 @@ struct Stager {
           }
         }
         v[i] = val;
-      }
-    } else if (GATHER && interior) {
-      constexpr int BLK_M = ROWS / TR::VEC;
-#pragma unroll
-      for (int i = 0; i < NVB; ++i) {
-        int c = tid + i * 256;
-        if (c >= TB) break;
-        int kb = c / BLK_M, mb = c % BLK_M;
-        int gm = row0 + mb * TR::VEC;
-#pragma unroll
-        for (int j = 0; j < TR::VEC; ++j)
-          v[i * TR::VEC + j] = *reinterpret_cast<const vec_t*>(
-              gather_addr<T>(*ga, k0 + kb * TR::VEC + j, gm));
       }
     } else if (GATHER) {
       // K-major gather (conv wgrad B = im2col): rows dim is Kg (contiguous
@@ -253,7 +235,7 @@ struct Stager {
           v[i * TR::VEC + j] = val;
         }
       }
-    } else if (interior) {
+    } else if (FASTI && interior) {
       // K-major interior: VEC unguarded row-vector loads per block
       constexpr int BLK_M = ROWS / TR::VEC;
 #pragma unroll
@@ -459,8 +441,8 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
     }
   }
   {
-  Stager<T, BM, A_KLAST, GA> sa;
-  Stager<T, BN, B_KLAST, GB> sb;
+  Stager<T, BM, A_KLAST, GA, SPLITK> sa;
+  Stager<T, BN, B_KLAST, GB, SPLITK> sb;
   // prologue: tile 0 -> LDS[0]; issue tile 1 loads
   sa.load(A, lda, m0, M, k_begin, k_end, tid, &ga_a);
   sb.load(B, ldb, n0, N, k_begin, k_end, tid, &ga_b);
