@@ -300,7 +300,21 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
   int Cog = Co / G;
 
-  Tensor wk = weight_khwc(w, bf16);
+  // fused repack: one kernel writes both the khwc fwd operand and the
+  // per-group transpose the dgrad GEMM wants (cached by the layer)
+  auto wc = w.contiguous();
+  Tensor wk = at::empty({Co, (int64_t)kh * kw * Cig},
+                        w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
+  Tensor wkT = at::empty({(int64_t)G * kh * kw * Cig, (int64_t)(Co / G)},
+                         w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
+  if (bf16)
+    ps_weight_to_khwc_both_f32_bf16(wc.data_ptr<float>(), wk.data_ptr(),
+                                    wkT.data_ptr(), Co, Cig, kh, kw, G,
+                                    stream());
+  else
+    ps_weight_to_khwc_both_f32(wc.data_ptr<float>(), wk.data_ptr<float>(),
+                               wkT.data_ptr<float>(), Co, Cig, kh, kw, G,
+                               stream());
 
   bool is_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0);
   const int VEC = bf16 ? 8 : 4;
@@ -355,9 +369,6 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
                true, true, 1.0f, 0.0f, nullptr, nullptr, fuse_relu);
     }
   }
-  // transposed repack for the dgrad NT GEMM, computed once per iteration
-  // and cached by the layer alongside colT
-  Tensor wkT = weight_khwc_tr(w, G, bf16);
   if (!colT.defined())
     colT = at::empty({0}, x.options());  // implicit mode: wgrad gathers too
   return {y, colT, wkT};

@@ -188,6 +188,30 @@ __global__ void weight_to_khwc_tr_k(const TI* src, TO* dst, int Co, int Cig,
   }
 }
 
+// Fused repack: one read of the fp32 master emits BOTH the khwc forward
+// operand and its per-group transpose for dgrad (they were two kernels =
+// two master reads + two launches per conv per iteration; GoogLeNet has 59
+// convs).
+template <typename TI, typename TO>
+__global__ void weight_to_khwc_both_k(const TI* src, TO* dst, TO* dst_tr,
+                                      int Co, int Cig, int KH, int KW, int G) {
+  const int Cog = Co / G;
+  const int Kg = KH * KW * Cig;
+  int64_t total = (int64_t)Co * Kg;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int kkw = i % KW;
+    int64_t t = i / KW;
+    int kkh = t % KH; t /= KH;
+    int ci = t % Cig;
+    int co = t / Cig;
+    const float v = to_f32(src[i]);
+    const int kg = (kkh * KW + kkw) * Cig + ci;
+    from_f32(v, dst[(int64_t)co * Kg + kg]);
+    from_f32(v, dst_tr[((int64_t)(co / Cog) * Kg + kg) * Cog + co % Cog]);
+  }
+}
+
 template <typename TI, typename TO>
 __global__ void weight_from_khwc_k(const TI* src, TO* dst, int Co, int Cig,
                                    int KH, int KW, float beta) {
@@ -360,6 +384,20 @@ void ps_weight_to_khwc_tr_f32_bf16(const float* src, void* dst, int Co, int Cig,
   weight_to_khwc_tr_k<float, __bf16>
       <<<ew_grid((int64_t)Co * Cig * KH * KW), 256, 0, s>>>(
           src, (__bf16*)dst, Co, Cig, KH, KW, G);
+}
+void ps_weight_to_khwc_both_f32(const float* src, float* dst, float* dst_tr,
+                                int Co, int Cig, int KH, int KW, int G,
+                                hipStream_t s) {
+  weight_to_khwc_both_k<float, float>
+      <<<ew_grid((int64_t)Co * Cig * KH * KW), 256, 0, s>>>(
+          src, dst, dst_tr, Co, Cig, KH, KW, G);
+}
+void ps_weight_to_khwc_both_f32_bf16(const float* src, void* dst,
+                                     void* dst_tr, int Co, int Cig, int KH,
+                                     int KW, int G, hipStream_t s) {
+  weight_to_khwc_both_k<float, __bf16>
+      <<<ew_grid((int64_t)Co * Cig * KH * KW), 256, 0, s>>>(
+          src, (__bf16*)dst, (__bf16*)dst_tr, Co, Cig, KH, KW, G);
 }
 void ps_weight_from_khwc_f32(const float* src, float* dst, int Co, int Cig,
                              int KH, int KW, float beta, hipStream_t s) {

@@ -204,4 +204,8 @@ void ps_weight_to_khwc_tr_f32(const float*, float*, int, int, int, int, int,
                               hipStream_t);
 void ps_weight_to_khwc_tr_f32_bf16(const float*, void*, int, int, int, int,
                                    int, hipStream_t);
+void ps_weight_to_khwc_both_f32(const float*, float*, float*, int, int, int,
+                                int, int, hipStream_t);
+void ps_weight_to_khwc_both_f32_bf16(const float*, void*, void*, int, int,
+                                     int, int, int, hipStream_t);
 }
