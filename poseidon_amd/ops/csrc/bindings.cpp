@@ -126,11 +126,11 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   // without this)
   const bool atomic_ok = out_f32 && bias == nullptr && beta == 0.0f &&
                          !relu && ldc == N;
-  if (atomic_ok && tiles < 384 && K >= 1024) {
-    // atomic split-K: no workspace, so split much deeper (chunk >= 512)
+  if (atomic_ok && tiles < 384 && K >= 512) {
+    // atomic split-K: no workspace, so split much deeper (chunk >= 256)
     // -- the un-split grid leaves most of the 256 CUs idle
     int sk = (int)std::min<int64_t>((512 + tiles - 1) / tiles,
-                                    (K + 511) / 512);
+                                    (K + 255) / 256);
     if (sk > 1) {
       g.splitk = sk;
       g.ws = nullptr;

@@ -605,6 +605,10 @@ static void dispatch_tiles(const GemmArgs& g, hipStream_t s) {
     launch_tile<T, OUT, 128, 32, 4, 1, AK, BK_, HB, GA, GB>(g, s);
   else if (bm == 32 && bn == 128)
     launch_tile<T, OUT, 32, 128, 1, 4, AK, BK_, HB, GA, GB>(g, s);
+  else if (bm == 128 && bn == 16)
+    launch_tile<T, OUT, 128, 16, 4, 1, AK, BK_, HB, GA, GB>(g, s);
+  else if (bm == 16 && bn == 128)
+    launch_tile<T, OUT, 16, 128, 1, 4, AK, BK_, HB, GA, GB>(g, s);
   else if (bm == 64 && bn == 64)
     launch_tile<T, OUT, 64, 64, 2, 2, AK, BK_, HB, GA, GB>(g, s);
   else

@@ -65,10 +65,11 @@ struct ConvGeom {
 // versa). ALPHA ~= MACs the matrix cores retire per element the memory
 // system delivers (bf16: ~1.25e15 MAC/s vs ~3.2e12 elem/s -> ~400).
 inline void ps_pick_gemm_tile(int M, int N, int* bm_out, int* bn_out) {
-  const int cand[4][2] = {{128, 128}, {128, 32}, {32, 128}, {64, 64}};
+  const int cand[6][2] = {{128, 128}, {128, 32}, {32, 128}, {64, 64},
+                          {128, 16},  {16, 128}};
   const double ALPHA = 400.0;
   double best = 0;
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < 6; ++i) {
     int bm = cand[i][0], bn = cand[i][1];
     int64_t tm = (M + bm - 1) / bm, tn = (N + bn - 1) / bn;
     double flops = (double)(tm * bm) * (tn * bn) / ALPHA;  // (x K, common)
