@@ -62,6 +62,9 @@ def init(device: Optional[str] = None, rank: Optional[int] = None,
         _ctx.device = device
     if _ctx.device == "cuda":
         local = int(os.environ.get("LOCAL_RANK", str(rank)))
+        # oversubscription (more ranks than GPUs, e.g. debug runs on a
+        # 1-GPU box) wraps onto the available devices
+        local %= max(1, torch.cuda.device_count())
         _ctx.device_index = local
         torch.cuda.set_device(local)
     if seed is not None:
