@@ -40,7 +40,8 @@ def init_distributed(backend: Optional[str] = None) -> bool:
     if dist.is_initialized():
         return True
     if backend is None:
-        backend = "nccl" if c.device == "cuda" else "gloo"
+        backend = os.environ.get("PS_BACKEND") or (
+            "nccl" if c.device == "cuda" else "gloo")
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
     dist.init_process_group(backend=backend, rank=c.rank, world_size=c.world_size)
