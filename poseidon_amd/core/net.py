@@ -109,8 +109,18 @@ class Net:
         self.output_blob_names: List[str] = []
         self._loss_tops: List = []  # (layer_idx, top_idx, weight)
         self._zero_mt = None  # multi-tensor zero table (GPU, built lazily)
+        # inter-branch stream parallelism (inception-style nets): built by
+        # _build_stream_schedule, activated lazily on GPU
+        self._lstream: List[int] = []
+        self._producers: List[List[int]] = []
+        self._consumers: List[List[int]] = []
+        self._ms_nstreams = 0
+        self._ms_pool = None
+        self._fwd_ev = None
+        self._bwd_ev = None
 
         self._build()
+        self._build_stream_schedule()
 
     # ------------------------------------------------------------------
     def _build(self) -> None:
@@ -230,6 +240,62 @@ class Net:
 
         self._fuse_relu_epilogues()
 
+    def _build_stream_schedule(self, n_streams: int = 4) -> None:
+        """Dataflow schedule for inter-branch stream parallelism.
+
+        GoogLeNet's inception branches are independent chains whose GEMMs
+        individually fill <20% of the 256 CUs; running sibling branches on
+        separate HIP streams overlaps them. Assignment: a layer inherits
+        its most recent producer's stream, except when that producer fans
+        out to several consumers (a SPLIT) -- then siblings spread across
+        the pool round-robin. Joins (CONCAT) wait on the other branches'
+        events. insert_splits guarantees single-writer blob instances, so
+        cross-stream hazards reduce to the recorded events plus
+        record_stream() for the torch caching allocator."""
+        L = len(self.layers)
+        prods: List[List[int]] = [[] for _ in range(L)]
+        last_writer: Dict[int, int] = {}
+        for i in range(L):
+            for b in self.bottoms[i]:
+                j = last_writer.get(id(b))
+                if j is not None and j not in prods[i]:
+                    prods[i].append(j)
+            for t in self.tops[i]:
+                last_writer[id(t)] = i
+        cons: List[List[int]] = [[] for _ in range(L)]
+        for i in range(L):
+            for pj in prods[i]:
+                cons[pj].append(i)
+        ls = [0] * L
+        for i in range(L):
+            if prods[i]:
+                p0 = max(prods[i])
+                sibs = cons[p0]
+                ls[i] = sibs.index(i) % n_streams if len(sibs) > 1 else ls[p0]
+        self._producers, self._consumers, self._lstream = prods, cons, ls
+        # layers sharing a param blob must not race on its diff
+        groups: Dict[int, List[int]] = {}
+        for psp in self.params:
+            groups.setdefault(psp.owner, []).append(psp.layer_idx)
+        hazard = any(len({ls[li] for li in g}) > 1
+                     for g in groups.values() if len(g) > 1)
+        distinct = len(set(ls))
+        self._ms_nstreams = n_streams if (distinct > 1 and not hazard) else 0
+
+    def _ms_active(self) -> bool:
+        import os
+        return (self._ms_nstreams > 1 and ctx().device == "cuda"
+                and torch.cuda.is_available()
+                and os.environ.get("PS_MULTI_STREAM", "1") != "0")
+
+    def _ms_streams(self):
+        if self._ms_pool is None:
+            self._ms_pool = [torch.cuda.Stream()
+                             for _ in range(self._ms_nstreams - 1)]
+            self._fwd_ev = [torch.cuda.Event() for _ in self.layers]
+            self._bwd_ev = [torch.cuda.Event() for _ in self.layers]
+        return [torch.cuda.current_stream()] + self._ms_pool
+
     def _fuse_relu_epilogues(self) -> None:
         """Fold an in-place slope-0 ReLU into the producing conv/IP GEMM
         epilogue. insert_splits guarantees each produced blob instance has
@@ -318,6 +384,12 @@ class Net:
         """Forward pass without host synchronization: returns the loss as a
         0-d device tensor (sum of weighted loss tops)."""
         dev = ctx().torch_device
+        if self._ms_active():
+            self._forward_ms()
+            loss = torch.zeros((), dtype=torch.float32, device=dev)
+            for (li, ti, w) in self._loss_tops:
+                loss = loss + w * self.tops[li][ti].data.sum().to(torch.float32)
+            return loss
         loss = torch.zeros((), dtype=torch.float32, device=dev)
         marks = {}
         for (li, ti, w) in self._loss_tops:
@@ -327,6 +399,30 @@ class Net:
             for (ti, w) in marks.get(i, []):
                 loss = loss + w * self.tops[i][ti].data.sum().to(torch.float32)
         return loss
+
+    def _forward_ms(self) -> None:
+        """Forward with sibling branches on separate HIP streams. The main
+        (current) stream is stream 0; every cross-stream read waits the
+        producer's event and record_stream()s the tensor for the caching
+        allocator; the main stream joins every used stream at the end."""
+        streams = self._ms_streams()
+        ls = self._lstream
+        last_on_stream: Dict[int, int] = {}
+        for i, layer in enumerate(self.layers):
+            s = streams[ls[i]]
+            for pj in self._producers[i]:
+                if ls[pj] != ls[i]:
+                    s.wait_event(self._fwd_ev[pj])
+            with torch.cuda.stream(s):
+                for b in self.bottoms[i]:
+                    b.data.record_stream(s)
+                layer.forward(self.bottoms[i], self.tops[i])
+            self._fwd_ev[i].record(s)
+            last_on_stream[ls[i]] = i
+        cur = streams[0]
+        for k, li in last_on_stream.items():
+            if k != 0:
+                cur.wait_event(self._fwd_ev[li])
 
     def backward(self, post_layer_cb: Optional[Callable[[int, Layer], None]] = None
                  ) -> None:
@@ -340,6 +436,9 @@ class Net:
         for (li, ti, w) in self._loss_tops:
             t = self.tops[li][ti]
             t.diff.fill_(w)
+        if self._ms_active():
+            self._backward_ms(post_layer_cb)
+            return
         for i in range(len(self.layers) - 1, -1, -1):
             if not self.layer_need_bwd[i]:
                 continue
@@ -347,6 +446,45 @@ class Net:
                                     self.bottoms[i])
             if post_layer_cb is not None and self.layers[i].blobs:
                 post_layer_cb(i, self.layers[i])
+
+    def _backward_ms(self, post_layer_cb) -> None:
+        """Backward mirror of _forward_ms: layer i runs on its forward
+        stream, waiting the backward events of every consumer on another
+        stream (their backward wrote this layer's top diffs)."""
+        streams = self._ms_streams()
+        ls = self._lstream
+        # loss-top diff seeding above ran on the main stream; branch streams
+        # with a loss layer (aux heads) have no consumer event to wait on,
+        # so fork them explicitly from the seed point
+        seed_ev = torch.cuda.Event()
+        seed_ev.record(streams[0])
+        for st in streams[1:]:
+            st.wait_event(seed_ev)
+        ran = [False] * len(self.layers)
+        last_on_stream: Dict[int, int] = {}
+        for i in range(len(self.layers) - 1, -1, -1):
+            if not self.layer_need_bwd[i]:
+                continue
+            s = streams[ls[i]]
+            for cj in self._consumers[i]:
+                if ran[cj] and ls[cj] != ls[i]:
+                    s.wait_event(self._bwd_ev[cj])
+            with torch.cuda.stream(s):
+                for t in self.tops[i]:
+                    if t.has_diff():
+                        t.diff.record_stream(s)
+                self.layers[i].backward(self.tops[i],
+                                        self.bottom_need_bwd[i],
+                                        self.bottoms[i])
+                if post_layer_cb is not None and self.layers[i].blobs:
+                    post_layer_cb(i, self.layers[i])
+            self._bwd_ev[i].record(s)
+            ran[i] = True
+            last_on_stream[ls[i]] = i
+        cur = streams[0]
+        for k, li in last_on_stream.items():
+            if k != 0:
+                cur.wait_event(self._bwd_ev[li])
 
     def clear_activation_diffs(self) -> None:
         """Zero intermediate blob diffs between iterations (layers accumulate
