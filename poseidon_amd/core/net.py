@@ -407,6 +407,10 @@ class Net:
         allocator; the main stream joins every used stream at the end."""
         streams = self._ms_streams()
         ls = self._lstream
+        # record_stream is allocator bookkeeping for cross-stream tensor
+        # lifetime; under hipGraph capture the graph pool freezes addresses,
+        # and calling it there aborts the capture
+        rec = not torch.cuda.is_current_stream_capturing()
         last_on_stream: Dict[int, int] = {}
         for i, layer in enumerate(self.layers):
             s = streams[ls[i]]
@@ -414,8 +418,9 @@ class Net:
                 if ls[pj] != ls[i]:
                     s.wait_event(self._fwd_ev[pj])
             with torch.cuda.stream(s):
-                for b in self.bottoms[i]:
-                    b.data.record_stream(s)
+                if rec:
+                    for b in self.bottoms[i]:
+                        b.data.record_stream(s)
                 layer.forward(self.bottoms[i], self.tops[i])
             self._fwd_ev[i].record(s)
             last_on_stream[ls[i]] = i
@@ -460,6 +465,7 @@ class Net:
         seed_ev.record(streams[0])
         for st in streams[1:]:
             st.wait_event(seed_ev)
+        rec = not torch.cuda.is_current_stream_capturing()
         ran = [False] * len(self.layers)
         last_on_stream: Dict[int, int] = {}
         for i in range(len(self.layers) - 1, -1, -1):
@@ -470,9 +476,10 @@ class Net:
                 if ran[cj] and ls[cj] != ls[i]:
                     s.wait_event(self._bwd_ev[cj])
             with torch.cuda.stream(s):
-                for t in self.tops[i]:
-                    if t.has_diff():
-                        t.diff.record_stream(s)
+                if rec:
+                    for t in self.tops[i]:
+                        if t.has_diff():
+                            t.diff.record_stream(s)
                 self.layers[i].backward(self.tops[i],
                                         self.bottom_need_bwd[i],
                                         self.bottoms[i])
