@@ -283,10 +283,16 @@ class Net:
         self._ms_nstreams = n_streams if (distinct > 1 and not hazard) else 0
 
     def _ms_active(self) -> bool:
+        """Opt-in (PS_MULTI_STREAM=1): branch overlap measured, but eager-
+        mode host-side stream/event churn outweighs the GPU overlap on
+        GoogLeNet (3.36k vs 4.00k img/s serial), and hipGraph capture of
+        cross-stream event graphs segfaults inside ROCm 7.2 -- hipGraph
+        (which already removes the launch gaps these streams target) is
+        the winning configuration today."""
         import os
         return (self._ms_nstreams > 1 and ctx().device == "cuda"
                 and torch.cuda.is_available()
-                and os.environ.get("PS_MULTI_STREAM", "1") != "0")
+                and os.environ.get("PS_MULTI_STREAM", "0") == "1")
 
     def _ms_streams(self):
         if self._ms_pool is None:
