@@ -202,26 +202,40 @@ __global__ void lrn_fwd_v8_k(const T* __restrict__ x, T* __restrict__ y,
   for (int64_t r0 = (int64_t)blockIdx.x * rpb; r0 < rows;
        r0 += (int64_t)gridDim.x * rpb) {
     const int64_t row = r0 + lr;
-    if (!active || row >= rows) continue;  // barrier-free: safe to skip
-    const T* xr = x + row * C;
-    float xv[8], xs[8 + 2 * PRE];
-    LrnV8<T>::load8(xr + cb, xv);
+    const bool ok = active && row < rows;
+    float xv[8] = {};
+    if (ok) LrnV8<T>::load8(x + row * C + cb, xv);
+    float sq[8];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) xs[PRE + j] = xv[j] * xv[j];
+    for (int j = 0; j < 8; ++j) sq[j] = xv[j] * xv[j];
+    // halo x^2 from the NEIGHBOR LANES' registers (same row: lg +- 1),
+    // not from memory -- the scalar halo loads of the previous version
+    // quadrupled the memory request rate (4 extra 2 B requests per 16 B
+    // vector) and capped the kernel at ~1.1 TB/s. Window clipping at the
+    // channel edges = zeroed halo (lg boundaries).
+    float lh[PRE], rh[PRE];
 #pragma unroll
     for (int j = 0; j < PRE; ++j) {
-      const int cl = cb - PRE + j, cr2 = cb + 8 + j;
-      float a = (cl >= 0) ? to_f32(xr[cl]) : 0.f;   // zeros implement the
-      float b = (cr2 < C) ? to_f32(xr[cr2]) : 0.f;  // clipped edge window
-      xs[j] = a * a;
-      xs[PRE + 8 + j] = b * b;
+      lh[j] = __shfl_up(sq[8 - PRE + j], 1);
+      rh[j] = __shfl_down(sq[j], 1);
     }
+    if (lg == 0)
+#pragma unroll
+      for (int j = 0; j < PRE; ++j) lh[j] = 0.f;
+    if (lg == gpr - 1)
+#pragma unroll
+      for (int j = 0; j < PRE; ++j) rh[j] = 0.f;
+    if (!ok) continue;
     float out[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float ss = 0.f;
 #pragma unroll
-      for (int k = 0; k <= 2 * PRE; ++k) ss += xs[j + k];
+      for (int k = 0; k <= 2 * PRE; ++k) {
+        const int m = j + k;  // xs-space [0, 8+2P): [halo][sq][halo]
+        ss += (m < PRE) ? lh[m] : (m < PRE + 8) ? sq[m - PRE]
+                                                : rh[m - PRE - 8];
+      }
       out[j] = xv[j] * __powf(1.0f + alpha_over_n * ss, -beta);
     }
     LrnV8<T>::store8(y + row * C + cb, out);
@@ -240,32 +254,62 @@ __global__ void lrn_bwd_v8_k(const T* __restrict__ x, const T* __restrict__ y,
   for (int64_t r0 = (int64_t)blockIdx.x * rpb; r0 < rows;
        r0 += (int64_t)gridDim.x * rpb) {
     const int64_t row = r0 + lr;
-    if (!active || row >= rows) continue;
-    const T* xr = x + row * C;
-    const T* yr = y + row * C;
-    const T* dyr = dy + row * C;
-    // x^2 over [cb-2P, cb+8+2P): recomputes the scale of every window
-    // position touching this thread's 8 outputs
-    float xs[8 + 4 * PRE];
-#pragma unroll
-    for (int j = 0; j < 8 + 4 * PRE; ++j) {
-      const int c = cb - 2 * PRE + j;
-      const float v = (c >= 0 && c < C) ? to_f32(xr[c]) : 0.f;
-      xs[j] = v * v;
+    const bool ok = active && row < rows;
+    const int64_t base = row * C + cb;
+    float xv[8] = {}, yv[8] = {}, dyv[8] = {};
+    if (ok) {
+      LrnV8<T>::load8(x + base, xv);
+      LrnV8<T>::load8(y + base, yv);
+      LrnV8<T>::load8(dy + base, dyv);
     }
-    // ratio = dy*y/scale over [cb-P, cb+8+P)
+    float sq[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sq[j] = xv[j] * xv[j];
+    // x^2 halo reaches 2P (scale of the halo ratio positions), y/dy halo P
+    float lhx[2 * PRE], rhx[2 * PRE], lhy[PRE], rhy[PRE], lhd[PRE], rhd[PRE];
+#pragma unroll
+    for (int j = 0; j < 2 * PRE; ++j) {
+      lhx[j] = __shfl_up(sq[8 - 2 * PRE + j], 1);
+      rhx[j] = __shfl_down(sq[j], 1);
+    }
+#pragma unroll
+    for (int j = 0; j < PRE; ++j) {
+      lhy[j] = __shfl_up(yv[8 - PRE + j], 1);
+      rhy[j] = __shfl_down(yv[j], 1);
+      lhd[j] = __shfl_up(dyv[8 - PRE + j], 1);
+      rhd[j] = __shfl_down(dyv[j], 1);
+    }
+    if (lg == 0) {
+#pragma unroll
+      for (int j = 0; j < 2 * PRE; ++j) lhx[j] = 0.f;
+#pragma unroll
+      for (int j = 0; j < PRE; ++j) lhy[j] = lhd[j] = 0.f;
+    }
+    if (lg == gpr - 1) {
+#pragma unroll
+      for (int j = 0; j < 2 * PRE; ++j) rhx[j] = 0.f;
+#pragma unroll
+      for (int j = 0; j < PRE; ++j) rhy[j] = rhd[j] = 0.f;
+    }
+    if (!ok) continue;
+    // xs-space index m in [-2P, 8+2P) -> value
+    auto XS = [&](int m) {
+      return (m < 0) ? lhx[m + 2 * PRE]
+                     : (m < 8) ? sq[m] : rhx[m - 8];
+    };
+    // scale at positions [-P, 8+P), ratio = dy*y/scale there
     float ratio[8 + 2 * PRE], scc[8];
 #pragma unroll
-    for (int m = 0; m < 8 + 2 * PRE; ++m) {
-      const int c = cb - PRE + m;
+    for (int i2 = 0; i2 < 8 + 2 * PRE; ++i2) {
+      const int m = i2 - PRE;
       float ss = 0.f;
 #pragma unroll
-      for (int k = 0; k <= 2 * PRE; ++k) ss += xs[m + k];
+      for (int k = -PRE; k <= PRE; ++k) ss += XS(m + k);
       const float sc = 1.0f + alpha_over_n * ss;
-      if (m >= PRE && m < PRE + 8) scc[m - PRE] = sc;
-      ratio[m] = (c >= 0 && c < C)
-                     ? to_f32(dyr[c]) * to_f32(yr[c]) / sc
-                     : 0.f;
+      if (m >= 0 && m < 8) scc[m] = sc;
+      const float ym = (m < 0) ? lhy[m + PRE] : (m < 8) ? yv[m] : rhy[m - 8];
+      const float dm = (m < 0) ? lhd[m + PRE] : (m < 8) ? dyv[m] : rhd[m - 8];
+      ratio[i2] = dm * ym / sc;
     }
     float out[8];
 #pragma unroll
@@ -273,8 +317,7 @@ __global__ void lrn_bwd_v8_k(const T* __restrict__ x, const T* __restrict__ y,
       float acc = 0.f;
 #pragma unroll
       for (int k = 0; k <= 2 * PRE; ++k) acc += ratio[j + k];
-      out[j] = to_f32(dyr[cb + j]) * __powf(scc[j], -beta)
-               - cache_ratio * to_f32(xr[cb + j]) * acc;
+      out[j] = dyv[j] * __powf(scc[j], -beta) - cache_ratio * xv[j] * acc;
     }
     LrnV8<T>::store8(dx + row * C + cb, out);
   }
@@ -283,8 +326,10 @@ __global__ void lrn_bwd_v8_k(const T* __restrict__ x, const T* __restrict__ y,
 template <typename T>
 static bool lrn_v8_ok(int C, int size) {
   const int pre = (size - 1) / 2;
+  // pre <= 2: the shfl halo reaches one neighbor lane (8 channels); wider
+  // windows fall back to the LDS row-block kernels
   return (C % 8 == 0) && C >= 8 && (C / 8) <= 256 && (size % 2 == 1) &&
-         pre >= 1 && pre <= 4;
+         pre >= 1 && pre <= 2;
 }
 
 #define LRN_V8_DISPATCH(kernel, T, ...)                                   \
@@ -295,9 +340,7 @@ static bool lrn_v8_ok(int C, int size) {
     dim3 grid_ = lrn_grid(rows, rpb_);                                    \
     switch (pre_) {                                                       \
       case 1:hipLaunchKernelGGL(( kernel<T, 1>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
-      case 2:hipLaunchKernelGGL(( kernel<T, 2>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
-      case 3:hipLaunchKernelGGL(( kernel<T, 3>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
-      default:hipLaunchKernelGGL(( kernel<T, 4>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
+      default:hipLaunchKernelGGL(( kernel<T, 2>), dim3(grid_), dim3(256), 0, s, __VA_ARGS__, gpr_, rpb_); break; \
     }                                                                     \
   } while (0)
 
