@@ -95,6 +95,106 @@ __global__ void im2col_nhwc_rowrun_k(const T* __restrict__ x,
   }
 }
 
+
+// exact small-divisor division via f32 reciprocal (operands < 2^24)
+__device__ inline int fdiv_fix2(int x2, int d, float inv, int& rem) {
+  int q = (int)((float)x2 * inv);
+  rem = x2 - q * d;
+  if (rem < 0) { --q; rem += d; }
+  else if (rem >= d) { ++q; rem -= d; }
+  return q;
+}
+
+// LDS row-staged variant for small-C first layers (G==1, C % 8 != 0 --
+// conv1-style). The rowrun kernel above issues unaligned 16 B ops on 66 B
+// runs (every op splits into two requests) and re-reads overlapping
+// windows from HBM; here a block stages the kh input rows of one (n, oh)
+// in LDS once (coalesced), then emits the (Wo x Kcol) output span as
+// GLOBALLY 16 B-ALIGNED vector stores, decoding (ow, kh, r) incrementally
+// (one f32-reciprocal divide per 8 elements, no per-element IDIV).
+template <typename T>
+__global__ void im2col_rowstage_k(const T* __restrict__ x,
+                                  T* __restrict__ colT, ConvGeom g,
+                                  int rsw) {
+  extern __shared__ char smem[];
+  T* xs = (T*)smem;  // [kh][rsw]; staged elem s <-> iw_elem = s - pw*C
+  constexpr int V = 16 / (int)sizeof(T);
+  typedef T uvec_t __attribute__((ext_vector_type(V), aligned(sizeof(T))));
+  typedef T avec_t __attribute__((ext_vector_type(V)));
+  const int RUN = g.kw * g.C;
+  const int Kcol = g.kh * RUN;
+  const int WC = g.W * g.C;
+  const int row_elems = g.Wo * Kcol;  // contiguous colT span per (n, oh)
+  const float inv_Kcol = 1.0f / Kcol, inv_RUN = 1.0f / RUN;
+  for (int64_t bo = blockIdx.x; bo < (int64_t)g.N * g.Ho; bo += gridDim.x) {
+    const int oh = (int)(bo % g.Ho);
+    const int n = (int)(bo / g.Ho);
+    // stage kh rows (zero rows for padded ih; left/right iw pad as zeros)
+    for (int kh = 0; kh < g.kh; ++kh) {
+      const int ih = oh * g.sh - g.ph + kh;
+      T* row = xs + kh * rsw;
+      if (ih < 0 || ih >= g.H) {
+        for (int e = threadIdx.x; e < rsw; e += blockDim.x) row[e] = (T)0.0f;
+      } else {
+        const T* src = x + (((int64_t)n * g.H + ih) * g.W) * g.C - g.pw * g.C;
+        for (int e = threadIdx.x * V; e < rsw; e += blockDim.x * V) {
+          uvec_t v;
+#pragma unroll
+          for (int j2 = 0; j2 < V; ++j2) {
+            const int ie = e + j2 - g.pw * g.C;
+            v[j2] = (ie >= 0 && ie < WC && e + j2 < rsw) ? src[e + j2 + 0]
+                                                         : (T)0.0f;
+          }
+          if (e + V <= rsw)
+            *reinterpret_cast<uvec_t*>(&row[e]) = v;
+          else
+            for (int j2 = 0; e + j2 < rsw; ++j2) row[e + j2] = v[j2];
+        }
+      }
+    }
+    __syncthreads();
+    // emit: flat span [np0*Kcol, np0*Kcol + row_elems), aligned V pieces
+    T* out = colT + bo * (int64_t)g.Wo * Kcol;
+    const int64_t gbase = bo * (int64_t)g.Wo * Kcol;  // global elem index
+    // first aligned element within the span
+    const int head = (int)((V - (gbase & (V - 1))) & (V - 1));
+    for (int e = head + (int)threadIdx.x * V; e < row_elems;
+         e += blockDim.x * V) {
+      int rem;
+      int ow = fdiv_fix2(e, Kcol, inv_Kcol, rem);
+      int r;
+      int kh = fdiv_fix2(rem, RUN, inv_RUN, r);
+      int sbase = ow * g.sw * g.C;
+      avec_t v;
+#pragma unroll
+      for (int j2 = 0; j2 < V; ++j2) {
+        v[j2] = xs[kh * rsw + sbase + r];
+        if (++r == RUN) {
+          r = 0;
+          if (++kh == g.kh) {
+            kh = 0;
+            ++ow;
+            sbase = ow * g.sw * g.C;
+          }
+        }
+      }
+      if (e + V <= row_elems)
+        *reinterpret_cast<avec_t*>(&out[e]) = v;
+      else
+        for (int j2 = 0; e + j2 < row_elems; ++j2) out[e + j2] = v[j2];
+    }
+    // head elements (before the first aligned piece)
+    for (int e = threadIdx.x; e < head && e < row_elems; e += blockDim.x) {
+      int rem;
+      int ow = fdiv_fix2(e, Kcol, inv_Kcol, rem);
+      int r;
+      int kh = fdiv_fix2(rem, RUN, inv_RUN, r);
+      out[e] = xs[kh * rsw + ow * g.sw * g.C + r];
+    }
+    __syncthreads();
+  }
+}
+
 // gather: one thread per (n, ih, iw, V-chunk of channels) -- channel chunks
 // are contiguous in BOTH dcolT rows and NHWC dx, so loads/stores vectorize
 template <typename T, int V>
@@ -324,10 +424,20 @@ void ps_im2col_nhwc_f32(const float* x, float* colT, const ConvGeom* g, hipStrea
   int64_t base = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
   if (Cg % 4 == 0)
    hipLaunchKernelGGL(( im2col_nhwc_k<float, 4>), dim3(ew_grid(base * (Cg / 4))), dim3(256), 0, s, x, colT, *g);
-  else if (g->G == 1)
-   hipLaunchKernelGGL(( im2col_nhwc_rowrun_k<float, 16>)
-        , dim3(ew_grid((int64_t)g->N * g->Ho * g->Wo * g->kh)), dim3(256), 0, s, 
-            x, colT, *g);
+  else if (g->G == 1) {
+    int rsw = ((g->Wo - 1) * g->sw + g->kw) * g->C;
+    int64_t lds = (int64_t)g->kh * rsw * sizeof(float);
+    if (lds <= (48 << 10)) {
+      int64_t blocks = (int64_t)g->N * g->Ho;
+      if (blocks > (64 << 10)) blocks = 64 << 10;
+     hipLaunchKernelGGL(( im2col_rowstage_k<float>)
+          , dim3(dim3((unsigned)blocks)), dim3(256), lds, s, x, colT, *g, rsw);
+    } else {
+     hipLaunchKernelGGL(( im2col_nhwc_rowrun_k<float, 16>)
+          , dim3(ew_grid((int64_t)g->N * g->Ho * g->Wo * g->kh)), dim3(256), 0, s, 
+              x, colT, *g);
+    }
+  }
   else
    hipLaunchKernelGGL(( im2col_nhwc_k<float, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, x, colT, *g);
 }
@@ -337,10 +447,21 @@ void ps_im2col_nhwc_bf16(const void* x, void* colT, const ConvGeom* g, hipStream
   if (Cg % 8 == 0)
    hipLaunchKernelGGL(( im2col_nhwc_k<__bf16, 8>), dim3(ew_grid(base * (Cg / 8))), dim3(256), 0, s, 
         (const __bf16*)x, (__bf16*)colT, *g);
-  else if (g->G == 1)
-   hipLaunchKernelGGL(( im2col_nhwc_rowrun_k<__bf16, 16>)
-        , dim3(ew_grid((int64_t)g->N * g->Ho * g->Wo * g->kh)), dim3(256), 0, s, 
-            (const __bf16*)x, (__bf16*)colT, *g);
+  else if (g->G == 1) {
+    int rsw = ((g->Wo - 1) * g->sw + g->kw) * g->C;
+    int64_t lds = (int64_t)g->kh * rsw * sizeof(__bf16);
+    if (lds <= (48 << 10)) {
+      int64_t blocks = (int64_t)g->N * g->Ho;
+      if (blocks > (64 << 10)) blocks = 64 << 10;
+     hipLaunchKernelGGL(( im2col_rowstage_k<__bf16>)
+          , dim3(dim3((unsigned)blocks)), dim3(256), lds, s, (const __bf16*)x,
+                                                    (__bf16*)colT, *g, rsw);
+    } else {
+     hipLaunchKernelGGL(( im2col_nhwc_rowrun_k<__bf16, 16>)
+          , dim3(ew_grid((int64_t)g->N * g->Ho * g->Wo * g->kh)), dim3(256), 0, s, 
+              (const __bf16*)x, (__bf16*)colT, *g);
+    }
+  }
   else
    hipLaunchKernelGGL(( im2col_nhwc_k<__bf16, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, 
         (const __bf16*)x, (__bf16*)colT, *g);
