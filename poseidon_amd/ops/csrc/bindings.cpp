@@ -255,20 +255,6 @@ ConvGeom conv_geom(const Tensor& x_cl, int kh, int kw, int sh, int sw,
 }
 
 // weights [Co,Cig,kh,kw] fp32 -> khwc [Co, kh*kw*Cig] in compute dtype
-Tensor weight_khwc(const Tensor& w, bool bf16) {
-  auto wc = w.contiguous();
-  int Co = wc.size(0), Cig = wc.size(1), kh = wc.size(2), kw = wc.size(3);
-  Tensor wk = at::empty({Co, (int64_t)kh * kw * Cig},
-                        wc.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
-  if (bf16)
-    ps_weight_to_khwc_f32_bf16(wc.data_ptr<float>(), wk.data_ptr(), Co, Cig,
-                               kh, kw, stream());
-  else
-    ps_weight_to_khwc_f32(wc.data_ptr<float>(), wk.data_ptr<float>(), Co, Cig,
-                          kh, kw, stream());
-  return wk;
-}
-
 Tensor weight_khwc_tr(const Tensor& w, int G, bool bf16) {
   auto wc = w.contiguous();
   int Co = wc.size(0), Cig = wc.size(1), kh = wc.size(2), kw = wc.size(3);
