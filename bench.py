@@ -113,7 +113,9 @@ def main():
             "config": {
                 "model": args.model,
                 "global_batch": batch * n_gpus,
-                "input": "3x227x227" if args.model == "alexnet" else "3x224x224",
+                "input": {"alexnet": "3x227x227", "googlenet": "3x224x224",
+                          "vgg16": "3x224x224", "cifar10_quick": "3x32x32",
+                          "lenet": "1x28x28"}[args.model],
                 "parallelism": f"dp{n_gpus}",
                 "sfb": not args.no_sfb and n_gpus > 1,
                 "hipgraph": graphed,

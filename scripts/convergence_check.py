@@ -9,8 +9,12 @@ profiles/r01_convergence.md.
 """
 
 import argparse
+import os
+import sys
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def main():
