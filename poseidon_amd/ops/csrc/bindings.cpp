@@ -285,25 +285,30 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   int Kcol = G * Kg;
   int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
   int Cog = Co / G;
+  const int VEC = bf16 ? 8 : 4;
+  // small-K pad (ps_api.h ps_colT_ld): zero columns up to one full BK so
+  // the fwd GEMM is a single interior k-tile (glds path)
+  const int ldc_col = ps_colT_ld(G, g.C, kh, kw, VEC);
+  const bool kpad = ldc_col != Kcol;
 
   // fused repack: one kernel writes both the khwc fwd operand and the
   // per-group transpose the dgrad GEMM wants (cached by the layer)
   auto wc = w.contiguous();
-  Tensor wk = at::empty({Co, (int64_t)kh * kw * Cig},
-                        w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
+  auto wkopts = w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat);
+  Tensor wk = kpad ? at::zeros({Co, (int64_t)ldc_col}, wkopts)
+                   : at::empty({Co, (int64_t)Kg}, wkopts);
   Tensor wkT = at::empty({(int64_t)G * kh * kw * Cig, (int64_t)(Co / G)},
-                         w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat));
+                         wkopts);
   if (bf16)
     ps_weight_to_khwc_both_f32_bf16(wc.data_ptr<float>(), wk.data_ptr(),
                                     wkT.data_ptr(), Co, Cig, kh, kw, G,
-                                    stream());
+                                    ldc_col / G, stream());
   else
     ps_weight_to_khwc_both_f32(wc.data_ptr<float>(), wk.data_ptr<float>(),
                                wkT.data_ptr<float>(), Co, Cig, kh, kw, G,
-                               stream());
+                               ldc_col / G, stream());
 
   bool is_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0);
-  const int VEC = bf16 ? 8 : 4;
   // implicit GEMM: gather im2col rows inside the GEMM staging whenever the
   // group channel count keeps 16B runs contiguous -- no column matrix at
   // all (the wgrad GEMM gathers too); conv1-style small-C layers still
@@ -315,12 +320,14 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   if (is_1x1) {
     colT = rows2d(x_cl);  // alias: x rows ARE the col rows
   } else if (!implicit) {
-    colT = at::empty({NP, (int64_t)Kcol}, x.options());
+    colT = kpad ? at::zeros({NP, (int64_t)ldc_col}, x.options())
+                : at::empty({NP, (int64_t)Kcol}, x.options());
     if (bf16)
-      ps_im2col_nhwc_bf16(x_cl.data_ptr(), colT.data_ptr(), &g, stream());
+      ps_im2col_nhwc_bf16(x_cl.data_ptr(), colT.data_ptr(), &g, ldc_col,
+                          stream());
     else
       ps_im2col_nhwc_f32(x_cl.data_ptr<float>(), colT.data_ptr<float>(), &g,
-                         stream());
+                         ldc_col, stream());
   }
 
   Tensor y = at::empty({g.N, Co, g.Ho, g.Wo},
@@ -347,10 +354,12 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
                /*c_off=*/(int64_t)grp * Cog,
                true, true, 1.0f, 0.0f, &ga, nullptr, fuse_relu);
     } else {
+      const int Kgl = ldc_col / G;  // per-group K incl. pad
       run_gemm(colT, wk, y, bp ? bp + grp * Cog : nullptr,
-               (int)NP, Cog, Kg,
-               /*lda=*/Kcol, /*ldb=*/Kg, /*ldc=*/Co,
-               /*a_off=*/(int64_t)grp * Kg, /*b_off=*/(int64_t)grp * Cog * Kg,
+               (int)NP, Cog, Kgl,
+               /*lda=*/ldc_col, /*ldb=*/Kgl, /*ldc=*/Co,
+               /*a_off=*/(int64_t)grp * Kgl,
+               /*b_off=*/(int64_t)grp * Cog * Kgl,
                /*c_off=*/(int64_t)grp * Cog,
                true, true, 1.0f, 0.0f, nullptr, nullptr, fuse_relu);
     }
@@ -424,9 +433,12 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
   int64_t NP = (int64_t)dy_cl.size(0) * dy_cl.size(2) * dy_cl.size(3);
   const bool implicit = colT.numel() == 0 && colT.dim() == 1;
   int64_t Kcol = implicit ? 0 : colT.size(1);
+  // colT may carry zero pad columns (ps_colT_ld): run the GEMM over the
+  // padded width too (zero B columns -> zero dwk columns, skipped below)
+  const int Kgw = implicit ? Kg : (int)(Kcol / G);
 
   // fp32 gradient accumulation regardless of activation dtype
-  Tensor dwk = at::empty({Co, Kg}, dy.options().dtype(at::kFloat));
+  Tensor dwk = at::empty({Co, (int64_t)Kgw}, dy.options().dtype(at::kFloat));
   Tensor dy2 = rows2d(dy_cl);
   for (int grp = 0; grp < G; ++grp) {
     // dwk_g[Cog, Kg] = dy_g^T[Cog, NP] @ colT_g[NP, Kg]: contraction NP
@@ -447,15 +459,16 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
                false, false, 1.0f, 0.0f, nullptr, &gb);
     } else {
       run_gemm(dy2, colT, dwk, nullptr,
-               Cog, Kg, (int)NP,
-               /*lda=*/Co, /*ldb=*/Kcol, /*ldc=*/Kg,
-               /*a_off=*/(int64_t)grp * Cog, /*b_off=*/(int64_t)grp * Kg,
-               /*c_off=*/(int64_t)grp * Cog * Kg,
+               Cog, Kgw, (int)NP,
+               /*lda=*/Co, /*ldb=*/Kcol, /*ldc=*/Kgw,
+               /*a_off=*/(int64_t)grp * Cog, /*b_off=*/(int64_t)grp * Kgw,
+               /*c_off=*/(int64_t)grp * Cog * Kgw,
                false, false, 1.0f, 0.0f);
     }
   }
   ps_weight_from_khwc_f32(dwk.data_ptr<float>(), dw_out.data_ptr<float>(),
-                          Co, Cig, kh, kw, /*beta=*/1.0f, stream());
+                          Co, Cig, kh, kw, /*ld=*/Kgw, /*beta=*/1.0f,
+                          stream());
   if (db_out.has_value()) {
     if (is_bf16(dy_cl))
       ps_colsum_bf16(dy_cl.data_ptr(), db_out->data_ptr<float>(), NP, Co,

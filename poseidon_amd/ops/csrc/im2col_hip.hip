@@ -17,7 +17,7 @@ namespace ps {
 // group channel count allows (guide Guideline 13: always vectorize)
 template <typename T, int V>
 __global__ void im2col_nhwc_k(const T* __restrict__ x, T* __restrict__ colT,
-                              ConvGeom g) {
+                              ConvGeom g, int ldcol) {
   typedef T vec_t __attribute__((ext_vector_type(V)));
   const int Cg = g.C / g.G;
   const int CV = Cg / V;
@@ -39,7 +39,8 @@ __global__ void im2col_nhwc_k(const T* __restrict__ x, T* __restrict__ colT,
     int kkh = khw / g.kw, kkw = khw % g.kw;
     int ih = oh * g.sh - g.ph + kkh;
     int iw = ow * g.sw - g.pw + kkw;
-    T* dst = colT + np * Kcol + ((int64_t)grp * KHW + khw) * Cg + cv * V;
+    T* dst = colT + np * (int64_t)ldcol
+             + ((int64_t)grp * KHW + khw) * Cg + cv * V;
     vec_t v = {};
     if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
       const T* src = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C
@@ -57,7 +58,8 @@ __global__ void im2col_nhwc_k(const T* __restrict__ x, T* __restrict__ colT,
 // cannot vectorize.
 template <typename T, int VBYTES>
 __global__ void im2col_nhwc_rowrun_k(const T* __restrict__ x,
-                                     T* __restrict__ colT, ConvGeom g) {
+                                     T* __restrict__ colT, ConvGeom g,
+                                     int ldcol) {
   constexpr int V = VBYTES / sizeof(T);
   // element-aligned only: run starts (np*Kcol, kh*RUN) are not 16B-aligned
   // for odd K (conv1 Kcol=363); clang emits the widest legal loads
@@ -76,7 +78,7 @@ __global__ void im2col_nhwc_rowrun_k(const T* __restrict__ x,
     int n = t2 / g.Ho;
     int ih = oh * g.sh - g.ph + kkh;
     int iw0 = ow * g.sw - g.pw;
-    T* dst = colT + np * Kcol + (int64_t)kkh * RUN;
+    T* dst = colT + np * (int64_t)ldcol + (int64_t)kkh * RUN;
     if (ih < 0 || ih >= g.H) {
       for (int e = 0; e < RUN; ++e) dst[e] = (T)0.0f;
       continue;
@@ -115,7 +117,7 @@ __device__ inline int fdiv_fix2(int x2, int d, float inv, int& rem) {
 template <typename T>
 __global__ void im2col_rowstage_k(const T* __restrict__ x,
                                   T* __restrict__ colT, ConvGeom g,
-                                  int rsw) {
+                                  int rsw, int ldcol) {
   extern __shared__ char smem[];
   T* xs = (T*)smem;  // [kh][rsw]; staged elem s <-> iw_elem = s - pw*C
   constexpr int V = 16 / (int)sizeof(T);
@@ -153,6 +155,33 @@ __global__ void im2col_rowstage_k(const T* __restrict__ x,
       }
     }
     __syncthreads();
+    if (ldcol != Kcol) {
+      // padded rows (ldcol % V == 0): each colT row is [Kcol valid)[pad]
+      // and V-aligned on its own, so emit row-by-row with no head/tail
+      T* out0 = colT + bo * (int64_t)g.Wo * ldcol;
+      const int pieces = ldcol / V;
+      const int total_p = g.Wo * pieces;
+      for (int pi = threadIdx.x; pi < total_p; pi += blockDim.x) {
+        const int ow = pi / pieces;
+        int e = (pi - ow * pieces) * V;
+        avec_t v;
+        if (e >= Kcol) {
+          v = avec_t{};  // zero pad columns
+        } else {
+          int r;
+          int kh = fdiv_fix2(e, RUN, inv_RUN, r);
+          const int sb0 = ow * g.sw * g.C;
+#pragma unroll
+          for (int j2 = 0; j2 < V; ++j2) {
+            v[j2] = (e + j2 < Kcol) ? xs[kh * rsw + sb0 + r] : (T)0.0f;
+            if (++r == RUN) { r = 0; ++kh; }
+          }
+        }
+        *reinterpret_cast<avec_t*>(&out0[(int64_t)ow * ldcol + e]) = v;
+      }
+      __syncthreads();
+      continue;
+    }
     // emit: flat span [np0*Kcol, np0*Kcol + row_elems), aligned V pieces
     T* out = colT + bo * (int64_t)g.Wo * Kcol;
     const int64_t gbase = bo * (int64_t)g.Wo * Kcol;  // global elem index
@@ -295,7 +324,8 @@ __global__ void weight_to_khwc_tr_k(const TI* src, TO* dst, int Co, int Cig,
 // convs).
 template <typename TI, typename TO>
 __global__ void weight_to_khwc_both_k(const TI* src, TO* dst, TO* dst_tr,
-                                      int Co, int Cig, int KH, int KW, int G) {
+                                      int Co, int Cig, int KH, int KW, int G,
+                                      int ldk) {
   const int Cog = Co / G;
   const int Kg = KH * KW * Cig;
   int64_t total = (int64_t)Co * Kg;
@@ -308,14 +338,14 @@ __global__ void weight_to_khwc_both_k(const TI* src, TO* dst, TO* dst_tr,
     int co = t / Cig;
     const float v = to_f32(src[i]);
     const int kg = (kkh * KW + kkw) * Cig + ci;
-    from_f32(v, dst[(int64_t)co * Kg + kg]);
+    from_f32(v, dst[(int64_t)co * ldk + kg]);
     from_f32(v, dst_tr[((int64_t)(co / Cog) * Kg + kg) * Cog + co % Cog]);
   }
 }
 
 template <typename TI, typename TO>
 __global__ void weight_from_khwc_k(const TI* src, TO* dst, int Co, int Cig,
-                                   int KH, int KW, float beta) {
+                                   int KH, int KW, int ld, float beta) {
   int64_t total = (int64_t)Co * Cig * KH * KW;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -324,7 +354,7 @@ __global__ void weight_from_khwc_k(const TI* src, TO* dst, int Co, int Cig,
     int kkh = t % KH; t /= KH;
     int ci = t % Cig;
     int co = t / Cig;
-    int64_t src_i = (((int64_t)co * KH + kkh) * KW + kkw) * Cig + ci;
+    int64_t src_i = (int64_t)co * ld + ((int64_t)kkh * KW + kkw) * Cig + ci;
     float v = to_f32(src[src_i]);
     if (beta != 0.f) v += beta * to_f32(dst[i]);
     from_f32(v, dst[i]);
@@ -419,11 +449,12 @@ void ps_chan_slice_bf16(const void* in, void* out, int64_t rows, int C_in,
         (const __bf16*)in, (__bf16*)out, rows, C_in, C_out, c_off);
 }
 
-void ps_im2col_nhwc_f32(const float* x, float* colT, const ConvGeom* g, hipStream_t s) {
+void ps_im2col_nhwc_f32(const float* x, float* colT, const ConvGeom* g,
+                        int ldcol, hipStream_t s) {
   int Cg = g->C / g->G;
   int64_t base = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
   if (Cg % 4 == 0)
-   hipLaunchKernelGGL(( im2col_nhwc_k<float, 4>), dim3(ew_grid(base * (Cg / 4))), dim3(256), 0, s, x, colT, *g);
+   hipLaunchKernelGGL(( im2col_nhwc_k<float, 4>), dim3(ew_grid(base * (Cg / 4))), dim3(256), 0, s, x, colT, *g, ldcol);
   else if (g->G == 1) {
     int rsw = ((g->Wo - 1) * g->sw + g->kw) * g->C;
     int64_t lds = (int64_t)g->kh * rsw * sizeof(float);
@@ -431,22 +462,23 @@ void ps_im2col_nhwc_f32(const float* x, float* colT, const ConvGeom* g, hipStrea
       int64_t blocks = (int64_t)g->N * g->Ho;
       if (blocks > (64 << 10)) blocks = 64 << 10;
      hipLaunchKernelGGL(( im2col_rowstage_k<float>)
-          , dim3(dim3((unsigned)blocks)), dim3(256), lds, s, x, colT, *g, rsw);
+          , dim3(dim3((unsigned)blocks)), dim3(256), lds, s, x, colT, *g, rsw, ldcol);
     } else {
      hipLaunchKernelGGL(( im2col_nhwc_rowrun_k<float, 16>)
           , dim3(ew_grid((int64_t)g->N * g->Ho * g->Wo * g->kh)), dim3(256), 0, s, 
-              x, colT, *g);
+              x, colT, *g, ldcol);
     }
   }
   else
-   hipLaunchKernelGGL(( im2col_nhwc_k<float, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, x, colT, *g);
+   hipLaunchKernelGGL(( im2col_nhwc_k<float, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, x, colT, *g, ldcol);
 }
-void ps_im2col_nhwc_bf16(const void* x, void* colT, const ConvGeom* g, hipStream_t s) {
+void ps_im2col_nhwc_bf16(const void* x, void* colT, const ConvGeom* g,
+                         int ldcol, hipStream_t s) {
   int Cg = g->C / g->G;
   int64_t base = (int64_t)g->N * g->Ho * g->Wo * g->G * g->kh * g->kw;
   if (Cg % 8 == 0)
    hipLaunchKernelGGL(( im2col_nhwc_k<__bf16, 8>), dim3(ew_grid(base * (Cg / 8))), dim3(256), 0, s, 
-        (const __bf16*)x, (__bf16*)colT, *g);
+        (const __bf16*)x, (__bf16*)colT, *g, ldcol);
   else if (g->G == 1) {
     int rsw = ((g->Wo - 1) * g->sw + g->kw) * g->C;
     int64_t lds = (int64_t)g->kh * rsw * sizeof(__bf16);
@@ -455,16 +487,17 @@ void ps_im2col_nhwc_bf16(const void* x, void* colT, const ConvGeom* g, hipStream
       if (blocks > (64 << 10)) blocks = 64 << 10;
      hipLaunchKernelGGL(( im2col_rowstage_k<__bf16>)
           , dim3(dim3((unsigned)blocks)), dim3(256), lds, s, (const __bf16*)x,
-                                                    (__bf16*)colT, *g, rsw);
+                                                    (__bf16*)colT, *g, rsw,
+                                                    ldcol);
     } else {
      hipLaunchKernelGGL(( im2col_nhwc_rowrun_k<__bf16, 16>)
           , dim3(ew_grid((int64_t)g->N * g->Ho * g->Wo * g->kh)), dim3(256), 0, s, 
-              (const __bf16*)x, (__bf16*)colT, *g);
+              (const __bf16*)x, (__bf16*)colT, *g, ldcol);
     }
   }
   else
    hipLaunchKernelGGL(( im2col_nhwc_k<__bf16, 1>), dim3(ew_grid(base * Cg)), dim3(256), 0, s, 
-        (const __bf16*)x, (__bf16*)colT, *g);
+        (const __bf16*)x, (__bf16*)colT, *g, ldcol);
 }
 void ps_col2im_nhwc_f32(const float* colT, float* dx, const ConvGeom* g, hipStream_t s) {
   int Cg = g->C / g->G;
@@ -509,23 +542,24 @@ void ps_weight_to_khwc_tr_f32_bf16(const float* src, void* dst, int Co, int Cig,
 }
 void ps_weight_to_khwc_both_f32(const float* src, float* dst, float* dst_tr,
                                 int Co, int Cig, int KH, int KW, int G,
-                                hipStream_t s) {
+                                int ldk, hipStream_t s) {
  hipLaunchKernelGGL(( weight_to_khwc_both_k<float, float>)
       , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, 
-          src, dst, dst_tr, Co, Cig, KH, KW, G);
+          src, dst, dst_tr, Co, Cig, KH, KW, G, ldk);
 }
 void ps_weight_to_khwc_both_f32_bf16(const float* src, void* dst,
                                      void* dst_tr, int Co, int Cig, int KH,
-                                     int KW, int G, hipStream_t s) {
+                                     int KW, int G, int ldk, hipStream_t s) {
  hipLaunchKernelGGL(( weight_to_khwc_both_k<float, __bf16>)
       , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, 
-          src, (__bf16*)dst, (__bf16*)dst_tr, Co, Cig, KH, KW, G);
+          src, (__bf16*)dst, (__bf16*)dst_tr, Co, Cig, KH, KW, G, ldk);
 }
 void ps_weight_from_khwc_f32(const float* src, float* dst, int Co, int Cig,
-                             int KH, int KW, float beta, hipStream_t s) {
+                             int KH, int KW, int ld, float beta,
+                             hipStream_t s) {
  hipLaunchKernelGGL(( weight_from_khwc_k<float, float>)
       , dim3(ew_grid((int64_t)Co * Cig * KH * KW)), dim3(256), 0, s, src, dst, Co, Cig,
-                                                            KH, KW, beta);
+                                                            KH, KW, ld, beta);
 }
 
 }  // extern "C"

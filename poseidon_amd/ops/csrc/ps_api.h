@@ -83,6 +83,18 @@ inline void ps_pick_gemm_tile(int M, int N, int* bm_out, int* bn_out) {
   }
 }
 
+// Materialized colT column stride: small-K first layers (G==1, channel
+// count not vector-aligned, Kcol < 64) pad to 64 ZERO-FILLED columns so
+// the forward GEMM's K-span is one full BK tile -- the K-edge otherwise
+// forces the guarded (scalarized) staging path for every tile (VGG
+// conv1_1 fwd, K=27: 2.57 ms -> glds path). Callers allocate colT/wk
+// zero-initialized when the returned stride != Kcol.
+inline int ps_colT_ld(int G, int C, int kh, int kw, int vec) {
+  int Kcol = kh * kw * C;  // G*kh*kw*(C/G)
+  if (G == 1 && (C % vec) != 0 && Kcol < 64) return 64;
+  return Kcol;
+}
+
 extern "C" {
 // gemm.hip
 void ps_gemm_f32(const GemmArgs* g, hipStream_t s);
@@ -193,20 +205,22 @@ void ps_chan_copy_f32(const float*, float*, int64_t, int, int, int, hipStream_t)
 void ps_chan_copy_bf16(const void*, void*, int64_t, int, int, int, hipStream_t);
 void ps_chan_slice_f32(const float*, float*, int64_t, int, int, int, hipStream_t);
 void ps_chan_slice_bf16(const void*, void*, int64_t, int, int, int, hipStream_t);
-void ps_im2col_nhwc_f32(const float*, float*, const ConvGeom*, hipStream_t);
+void ps_im2col_nhwc_f32(const float*, float*, const ConvGeom*, int ldcol,
+                        hipStream_t);
 void ps_col2im_nhwc_f32(const float*, float*, const ConvGeom*, hipStream_t);
-void ps_im2col_nhwc_bf16(const void*, void*, const ConvGeom*, hipStream_t);
+void ps_im2col_nhwc_bf16(const void*, void*, const ConvGeom*, int ldcol,
+                         hipStream_t);
 void ps_col2im_nhwc_bf16(const void*, void*, const ConvGeom*, hipStream_t);
 void ps_weight_to_khwc_f32(const float*, float*, int, int, int, int, hipStream_t);
 void ps_weight_to_khwc_f32_bf16(const float*, void*, int, int, int, int, hipStream_t);
-void ps_weight_from_khwc_f32(const float*, float*, int, int, int, int, float,
-                             hipStream_t);
+void ps_weight_from_khwc_f32(const float*, float*, int, int, int, int,
+                             int ld, float beta, hipStream_t);
 void ps_weight_to_khwc_tr_f32(const float*, float*, int, int, int, int, int,
                               hipStream_t);
 void ps_weight_to_khwc_tr_f32_bf16(const float*, void*, int, int, int, int,
                                    int, hipStream_t);
 void ps_weight_to_khwc_both_f32(const float*, float*, float*, int, int, int,
-                                int, int, hipStream_t);
+                                int, int, int ldk, hipStream_t);
 void ps_weight_to_khwc_both_f32_bf16(const float*, void*, void*, int, int,
-                                     int, int, int, hipStream_t);
+                                     int, int, int, int ldk, hipStream_t);
 }
