@@ -49,10 +49,12 @@ __device__ inline bf16x8 tr_frag(unsigned lds_base, int kk, int cb, int l,
       lds_base + (unsigned)(((kk + (l >> 2)) * ldt + cb + 4 * (l & 3)) * 2);
   const unsigned a2 = a1 + 4u * ldt * 2u;
   bf16x4 v1, v2;
+  // "=&v": early-clobber -- insn 1 writes v1 BEFORE insn 2 consumes a2,
+  // so the allocator must not overlap them (silent corruption otherwise)
   asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
                "ds_read_b64_tr_b16 %1, %3\n\t"
                "s_waitcnt lgkmcnt(0)"
-               : "=v"(v1), "=v"(v2) : "v"(a1), "v"(a2));
+               : "=&v"(v1), "=&v"(v2) : "v"(a1), "v"(a2));
   bf16x8 f;
 #pragma unroll
   for (int i = 0; i < 4; ++i) { f[i] = v1[i]; f[4 + i] = v2[i]; }

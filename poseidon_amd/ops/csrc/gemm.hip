@@ -552,6 +552,184 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// TN fast path: both operands K-major, staged by glds in their NATURAL
+// [BK][cols] row-major-in-k images and consumed through gfx950's
+// ds_read_b64_tr_b16 hardware transpose-read (guide T10). Replaces the
+// register block-transpose staging (59.5% wave-parked, 5.5 VALU per MFMA
+// -- profiles/r01_tn_gemm_pmc.md) for eligible tiles: measured 46 -> 246
+// TF/s on the VGG conv1_2 wgrad shape (64x576x1.6M), +14..40% on other
+// wgrad shapes. Semantics probe: experiments/tr16_probe.hip -- within a
+// 16-lane group, lane l loads 4 contiguous bf16 at its own 8B-aligned
+// address and lane j receives element j of each 16-element chunk of the
+// group's concatenated loads; pointing lane l at row kk+l/4, col
+// cb+4*(l%4) of the [k][col] image hands lane j the k-run of col cb+j.
+// Eligibility (checked by the launcher): M,N,K % 64 == 0, lda/ldb % 8 == 0,
+// 16B-aligned bases, bf16 in / f32 out, no gathers, no bias, beta == 0.
+// ---------------------------------------------------------------------------
+
+#define TRBM 64
+#define TRBN 64
+
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+__device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
+                                     int64_t ld, int k0, int c0, int wid,
+                                     int lane) {
+  // [BK=64 rows][64 cols]: 128 B rows = 8 lanes x 16 B, 8 rows/1KB chunk
+  const int r_in = lane >> 3;
+  const int slot = lane & 7;
+#pragma unroll
+  for (int ci = wid; ci < 8; ci += 4) {
+    const int row = ci * 8 + r_in;
+    const __bf16* g2 = src + (int64_t)(k0 + row) * ld + c0 + slot * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g2,
+        (__attribute__((address_space(3))) void*)(lds + ci * 512), 16, 0, 0);
+  }
+}
+
+__device__ inline bf16x8 tr_frag(unsigned lds_base, int kk, int cb, int l,
+                                 int ldt) {
+  const unsigned a1 =
+      lds_base + (unsigned)(((kk + (l >> 2)) * ldt + cb + 4 * (l & 3)) * 2);
+  const unsigned a2 = a1 + 4u * ldt * 2u;
+  bf16x4 v1, v2;
+  // "=&v" early-clobber: insn 1 writes v1 before insn 2 consumes a2
+  asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+               "ds_read_b64_tr_b16 %1, %3\n\t"
+               "s_waitcnt lgkmcnt(0)"
+               : "=&v"(v1), "=&v"(v2) : "v"(a1), "v"(a2));
+  bf16x8 f;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) { f[i] = v1[i]; f[4 + i] = v2[i]; }
+  return f;
+}
+
+template <bool SPLITK>
+__global__ __launch_bounds__(256)
+void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
+                       const __bf16* __restrict__ B, float* __restrict__ C,
+                       int M, int N, int K, int64_t ldA, int64_t ldB,
+                       int64_t ldC, float alpha, int kchunk) {
+  __shared__ __attribute__((aligned(16))) __bf16 a_lds[2][64 * TRBM];
+  __shared__ __attribute__((aligned(16))) __bf16 b_lds[2][64 * TRBN];
+  // T1 XCD swizzle (same as gemm_kernel)
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  {
+    const int nbx = gridDim.x, nby = gridDim.y;
+    const int64_t nwg = (int64_t)nbx * nby * gridDim.z;
+    if ((nwg & 7) == 0 && nwg > 8) {
+      int64_t id = ((int64_t)bz * nby + by) * nbx + bx;
+      const int64_t cpx = nwg >> 3;
+      id = (id & 7) * cpx + (id >> 3);
+      bx = (int)(id % nbx);
+      by = (int)((id / nbx) % nby);
+      bz = (int)(id / ((int64_t)nbx * nby));
+    }
+  }
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 32;
+  const int wn = (wid & 1) * 32;
+  const int m0 = by * TRBM;
+  const int n0 = bx * TRBN;
+  int k_begin = 0, k_end = K;
+  if (SPLITK) {
+    k_begin = bz * kchunk;
+    k_end = min(K, k_begin + kchunk);
+    if (k_begin >= k_end) return;
+  }
+  const int l = lane & 15, q = lane >> 4;
+
+  f32x4 acc[2][2] = {};
+  stage_kmaj_tr(a_lds[0], A, ldA, k_begin, m0, wid, lane);
+  stage_kmaj_tr(b_lds[0], B, ldB, k_begin, n0, wid, lane);
+  __syncthreads();
+  unsigned ab[2], bb[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    ab[i] = (unsigned)(unsigned long long)(
+        __attribute__((address_space(3))) __bf16*)a_lds[i];
+    bb[i] = (unsigned)(unsigned long long)(
+        __attribute__((address_space(3))) __bf16*)b_lds[i];
+  }
+  int cur = 0;
+  for (int k0 = k_begin; k0 < k_end; k0 += 64) {
+    if (k0 + 64 < k_end) {
+      stage_kmaj_tr(a_lds[cur ^ 1], A, ldA, k0 + 64, m0, wid, lane);
+      stage_kmaj_tr(b_lds[cur ^ 1], B, ldB, k0 + 64, n0, wid, lane);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 64; kk += 32) {
+      bf16x8 af[2], bfr[2];
+#pragma unroll
+      for (int f = 0; f < 2; ++f)
+        af[f] = tr_frag(ab[cur], kk + q * 8, wm + f * 16, l, TRBM);
+#pragma unroll
+      for (int f = 0; f < 2; ++f)
+        bfr[f] = tr_frag(bb[cur], kk + q * 8, wn + f * 16, l, TRBN);
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+      const int col = n0 + wn + fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        if (SPLITK)
+          atomicAdd(&C[(int64_t)row * ldC + col], alpha * acc[fm][fn][r]);
+        else
+          C[(int64_t)row * ldC + col] = alpha * acc[fm][fn][r];
+      }
+    }
+}
+
+// eligibility + launch; returns false if the caller must use the generic
+// path. C must be zeroed by the caller when split-K fires (atomic adds).
+static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
+  if (g.a_klast || g.b_klast || g.gather_a || g.gather_b) return false;
+  if (g.bias || g.relu || g.beta != 0.0f || g.batch > 1) return false;
+  if (g.M % TRBM || g.N % TRBN || g.K % 64) return false;
+  if (g.lda % 8 || g.ldb % 8 || g.ldc % 4) return false;
+  if (((uintptr_t)g.A & 15) || ((uintptr_t)g.B & 15)) return false;
+  int64_t tiles = (int64_t)(g.M / TRBM) * (g.N / TRBN);
+  if (tiles >= 1024) return false;  // huge grids: generic path measures better
+  int sk = 1, kchunk = g.K;
+  if (tiles < 384 && g.K >= 512) {
+    int want = (int)((512 + tiles - 1) / tiles);
+    int maxsk = (g.K + 255) / 256;
+    sk = want < maxsk ? want : maxsk;
+    if (sk < 1) sk = 1;
+    kchunk = ((g.K / sk + 63) / 64) * 64;
+    sk = (g.K + kchunk - 1) / kchunk;
+  }
+  dim3 grid(g.N / TRBN, g.M / TRBM, sk);
+  if (sk > 1) {
+    (void)hipMemsetAsync(g.C, 0, (size_t)g.M * g.N * sizeof(float), s);
+    gemm_tn_tr_kernel<true><<<grid, 256, 0, s>>>(
+        (const __bf16*)g.A, (const __bf16*)g.B, (float*)g.C, g.M, g.N, g.K,
+        g.lda, g.ldb, g.ldc, g.alpha, kchunk);
+  } else {
+    gemm_tn_tr_kernel<false><<<grid, 256, 0, s>>>(
+        (const __bf16*)g.A, (const __bf16*)g.B, (float*)g.C, g.M, g.N, g.K,
+        g.lda, g.ldb, g.ldc, g.alpha, kchunk);
+  }
+  return true;
+}
+
 // ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
@@ -647,6 +825,7 @@ void ps_gemm_f32(const GemmArgs* g, hipStream_t s) {
   gemm_dispatch<float, float>(*g, s);
 }
 void ps_gemm_bf16_f32out(const GemmArgs* g, hipStream_t s) {
+  if (try_gemm_tn_tr(*g, s)) return;
   gemm_dispatch<__bf16, float>(*g, s);
 }
 void ps_gemm_bf16(const GemmArgs* g, hipStream_t s) {
