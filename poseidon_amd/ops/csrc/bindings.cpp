@@ -314,7 +314,13 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   // all (the wgrad GEMM gathers too); conv1-style small-C layers still
   // materialize (and 1x1 convs alias x directly)
   int64_t colT_bytes = NP * Kcol * (bf16 ? 2 : 4);
-  bool implicit = (g_implicit_gemm.load() || colT_bytes > g_implicit_thresh.load())
+  // giant colT layers go implicit -- UNLESS their wgrad qualifies for the
+  // tr16 TN fast path (gemm.hip try_gemm_tn_tr), which reads the
+  // materialized colT ~5x faster than the gather-staged implicit wgrad
+  bool wgrad_tr_ok = bf16 && G == 1 && (Co % 64) == 0 &&
+                     (ldc_col % 64) == 0 && (NP % 64) == 0;
+  bool implicit = (g_implicit_gemm.load() ||
+                   (colT_bytes > g_implicit_thresh.load() && !wgrad_tr_ok))
                   && !is_1x1 && (Cg % VEC == 0);
   Tensor colT;
   if (is_1x1) {
