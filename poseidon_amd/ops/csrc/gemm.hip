@@ -699,12 +699,40 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
 
 // eligibility + launch; returns false if the caller must use the generic
 // path. C must be zeroed by the caller when split-K fires (atomic adds).
+template <typename T, typename OUT>
+static void gemm_dispatch(const GemmArgs& g, hipStream_t s);
+
 static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
   if (g.a_klast || g.b_klast || g.gather_a || g.gather_b) return false;
   if (g.bias || g.relu || g.beta != 0.0f || g.batch > 1) return false;
-  if (g.M % TRBM || g.N % TRBN || g.K % 64) return false;
-  if (g.lda % 8 || g.ldb % 8 || g.ldc % 4) return false;
+  if (g.ws) return false;  // caller committed to workspace split-K layout
+  if (g.K % 64) return false;
+  if (g.lda % 8 || g.ldb % 8) return false;
   if (((uintptr_t)g.A & 15) || ((uintptr_t)g.B & 15)) return false;
+  const int M0 = g.M & ~63, N0 = g.N & ~63;
+  if (M0 == 0 || N0 == 0) return false;
+  if (M0 != g.M || N0 != g.N) {
+    // edge strips go through the generic kernel; the [M0 x N0] interior
+    // through the tr path. Strips: [0,M) x [N0,N) and [M0,M) x [0,N0).
+    GemmArgs gi = g;
+    gi.M = M0; gi.N = N0;
+    if (!try_gemm_tn_tr(gi, s)) return false;
+    if (N0 != g.N) {
+      GemmArgs gt = g;
+      gt.N = g.N - N0;
+      gt.B = (const void*)((const __bf16*)g.B + N0);
+      gt.C = (void*)((float*)g.C + N0);
+      gemm_dispatch<__bf16, float>(gt, s);
+    }
+    if (M0 != g.M) {
+      GemmArgs gt = g;
+      gt.M = g.M - M0; gt.N = N0;
+      gt.A = (const void*)((const __bf16*)g.A + M0);
+      gt.C = (void*)((float*)g.C + (int64_t)M0 * g.ldc);
+      gemm_dispatch<__bf16, float>(gt, s);
+    }
+    return true;
+  }
   int64_t tiles = (int64_t)(g.M / TRBM) * (g.N / TRBN);
   if (tiles >= 1024) return false;  // huge grids: generic path measures better
   int sk = 1, kchunk = g.K;
