@@ -712,6 +712,10 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
   const int M0 = g.M & ~63, N0 = g.N & ~63;
   if (M0 == 0 || N0 == 0) return false;
   if (M0 != g.M || N0 != g.N) {
+    // edge strips cost 1-2 extra launches: only worth it when the
+    // interior carries real work (GoogLeNet's small inception wgrads
+    // measured slower split)
+    if ((int64_t)M0 * N0 * g.K < (1LL << 33)) return false;
     // edge strips go through the generic kernel; the [M0 x N0] interior
     // through the tr path. Strips: [0,M) x [N0,N) and [M0,M) x [0,N0).
     GemmArgs gi = g;
