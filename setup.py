@@ -28,6 +28,17 @@ sources = [
     ]
 ]
 
+# Sanitizer preset (host side): PS_SANITIZE=address adds ASan to the C++
+# binding layer -- a deliberate improvement over the reference, which had
+# no sanitizer story (concurrency safety there was
+# by-construction mutexes). Device code relies on numerics tests +
+# compute-sanitizer-style tools where available.
+_cxx = ["-O3"]
+_ld = []
+if os.environ.get("PS_SANITIZE") == "address":
+    _cxx += ["-fsanitize=address", "-fno-omit-frame-pointer", "-g"]
+    _ld += ["-fsanitize=address"]
+
 setup(
     name="poseidon_amd_hip",
     ext_modules=[
@@ -35,9 +46,10 @@ setup(
             name="poseidon_amd.ops._hip",
             sources=sources,
             extra_compile_args={
-                "cxx": ["-O3"],
+                "cxx": _cxx,
                 "nvcc": ["-O3", "-std=c++17"],
             },
+            extra_link_args=_ld,
         )
     ],
     cmdclass={"build_ext": BuildExtension},
