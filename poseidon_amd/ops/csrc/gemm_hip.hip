@@ -591,6 +591,9 @@ __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
   }
 }
 
+// Issues the two transpose-reads WITHOUT waiting -- the caller batches all
+// fragment reads of a k-step behind one s_waitcnt (tr_wait) so the LDS
+// latency of 16 reads overlaps instead of serializing.
 __device__ inline bf16x8 tr_frag(unsigned lds_base, int kk, int cb, int l,
                                  int ldt) {
   const unsigned a1 =
@@ -599,13 +602,21 @@ __device__ inline bf16x8 tr_frag(unsigned lds_base, int kk, int cb, int l,
   bf16x4 v1, v2;
   // "=&v" early-clobber: insn 1 writes v1 before insn 2 consumes a2
   asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
-               "ds_read_b64_tr_b16 %1, %3\n\t"
-               "s_waitcnt lgkmcnt(0)"
+               "ds_read_b64_tr_b16 %1, %3"
                : "=&v"(v1), "=&v"(v2) : "v"(a1), "v"(a2));
   bf16x8 f;
 #pragma unroll
   for (int i = 0; i < 4; ++i) { f[i] = v1[i]; f[4 + i] = v2[i]; }
   return f;
+}
+
+// drain the outstanding tr reads; the "+v" operands make every consumer of
+// the fragments order AFTER this wait (a bare clobber would let the
+// compiler hoist register uses above it)
+__device__ inline void tr_wait(bf16x8& a0, bf16x8& a1, bf16x8& b0,
+                               bf16x8& b1) {
+  asm volatile("s_waitcnt lgkmcnt(0)"
+               : "+v"(a0), "+v"(a1), "+v"(b0), "+v"(b1)::"memory");
 }
 
 template <bool SPLITK>
@@ -672,6 +683,7 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
 #pragma unroll
       for (int f = 0; f < 2; ++f)
         bfr[f] = tr_frag(bb[cur], kk + q * 8, wn + f * 16, l, TRBN);
+      tr_wait(af[0], af[1], bfr[0], bfr[1]);
 #pragma unroll
       for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
