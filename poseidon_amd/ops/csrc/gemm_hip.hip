@@ -575,15 +575,19 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
 
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
+template <int COLS>
 __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
                                      int64_t ld, int k0, int c0, int wid,
                                      int lane) {
-  // [BK=64 rows][64 cols]: 128 B rows = 8 lanes x 16 B, 8 rows/1KB chunk
-  const int r_in = lane >> 3;
-  const int slot = lane & 7;
+  // [BK=64 rows][COLS cols]: COLS*2 B rows, 16 B lane chunks
+  constexpr int LPR = COLS / 8;       // lanes per row
+  constexpr int RPC = 64 / LPR;       // rows per 1 KB chunk
+  constexpr int CHUNKS = 64 / RPC;
+  const int r_in = lane / LPR;
+  const int slot = lane % LPR;
 #pragma unroll
-  for (int ci = wid; ci < 8; ci += 4) {
-    const int row = ci * 8 + r_in;
+  for (int ci = wid; ci < CHUNKS; ci += 4) {
+    const int row = ci * RPC + r_in;
     const __bf16* g2 = src + (int64_t)(k0 + row) * ld + c0 + slot * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g2,
@@ -610,23 +614,28 @@ __device__ inline bf16x8 tr_frag(unsigned lds_base, int kk, int cb, int l,
   return f;
 }
 
-// drain the outstanding tr reads; the "+v" operands make every consumer of
-// the fragments order AFTER this wait (a bare clobber would let the
-// compiler hoist register uses above it)
-__device__ inline void tr_wait(bf16x8& a0, bf16x8& a1, bf16x8& b0,
-                               bf16x8& b1) {
-  asm volatile("s_waitcnt lgkmcnt(0)"
-               : "+v"(a0), "+v"(a1), "+v"(b0), "+v"(b1)::"memory");
+// drain the outstanding tr reads, then pin EVERY fragment behind the wait
+// with an empty volatile asm (volatile asms are ordered against each
+// other; a bare clobber would let the compiler hoist register uses)
+template <int NA, int NB>
+__device__ inline void tr_wait(bf16x8 (&a)[NA], bf16x8 (&b)[NB]) {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+  for (int i = 0; i < NA; ++i) asm volatile("" : "+v"(a[i]));
+#pragma unroll
+  for (int i = 0; i < NB; ++i) asm volatile("" : "+v"(b[i]));
 }
 
-template <bool SPLITK>
+template <int BM2, int BN2, int WGM2, int WGN2, bool SPLITK>
 __global__ __launch_bounds__(256)
 void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
                        const __bf16* __restrict__ B, float* __restrict__ C,
                        int M, int N, int K, int64_t ldA, int64_t ldB,
                        int64_t ldC, float alpha, int kchunk) {
-  __shared__ __attribute__((aligned(16))) __bf16 a_lds[2][64 * TRBM];
-  __shared__ __attribute__((aligned(16))) __bf16 b_lds[2][64 * TRBN];
+  constexpr int FM2 = BM2 / WGM2 / 16, FN2 = BN2 / WGN2 / 16;
+  static_assert(WGM2 * WGN2 == 4, "4 waves");
+  __shared__ __attribute__((aligned(16))) __bf16 a_lds[2][64 * BM2];
+  __shared__ __attribute__((aligned(16))) __bf16 b_lds[2][64 * BN2];
   // T1 XCD swizzle (same as gemm_kernel)
   int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
   {
@@ -644,10 +653,10 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int wm = (wid >> 1) * 32;
-  const int wn = (wid & 1) * 32;
-  const int m0 = by * TRBM;
-  const int n0 = bx * TRBN;
+  const int wm = (wid / WGN2) * (BM2 / WGM2);
+  const int wn = (wid % WGN2) * (BN2 / WGN2);
+  const int m0 = by * BM2;
+  const int n0 = bx * BN2;
   int k_begin = 0, k_end = K;
   if (SPLITK) {
     k_begin = bz * kchunk;
@@ -656,9 +665,9 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
   }
   const int l = lane & 15, q = lane >> 4;
 
-  f32x4 acc[2][2] = {};
-  stage_kmaj_tr(a_lds[0], A, ldA, k_begin, m0, wid, lane);
-  stage_kmaj_tr(b_lds[0], B, ldB, k_begin, n0, wid, lane);
+  f32x4 acc[FM2][FN2] = {};
+  stage_kmaj_tr<BM2>(a_lds[0], A, ldA, k_begin, m0, wid, lane);
+  stage_kmaj_tr<BN2>(b_lds[0], B, ldB, k_begin, n0, wid, lane);
   __syncthreads();
   unsigned ab[2], bb[2];
 #pragma unroll
@@ -671,23 +680,23 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
   int cur = 0;
   for (int k0 = k_begin; k0 < k_end; k0 += 64) {
     if (k0 + 64 < k_end) {
-      stage_kmaj_tr(a_lds[cur ^ 1], A, ldA, k0 + 64, m0, wid, lane);
-      stage_kmaj_tr(b_lds[cur ^ 1], B, ldB, k0 + 64, n0, wid, lane);
+      stage_kmaj_tr<BM2>(a_lds[cur ^ 1], A, ldA, k0 + 64, m0, wid, lane);
+      stage_kmaj_tr<BN2>(b_lds[cur ^ 1], B, ldB, k0 + 64, n0, wid, lane);
     }
 #pragma unroll
     for (int kk = 0; kk < 64; kk += 32) {
-      bf16x8 af[2], bfr[2];
+      bf16x8 af[FM2], bfr[FN2];
 #pragma unroll
-      for (int f = 0; f < 2; ++f)
-        af[f] = tr_frag(ab[cur], kk + q * 8, wm + f * 16, l, TRBM);
+      for (int f = 0; f < FM2; ++f)
+        af[f] = tr_frag(ab[cur], kk + q * 8, wm + f * 16, l, BM2);
 #pragma unroll
-      for (int f = 0; f < 2; ++f)
-        bfr[f] = tr_frag(bb[cur], kk + q * 8, wn + f * 16, l, TRBN);
-      tr_wait(af[0], af[1], bfr[0], bfr[1]);
+      for (int f = 0; f < FN2; ++f)
+        bfr[f] = tr_frag(bb[cur], kk + q * 8, wn + f * 16, l, BN2);
+      tr_wait(af, bfr);
 #pragma unroll
-      for (int fm = 0; fm < 2; ++fm)
+      for (int fm = 0; fm < FM2; ++fm)
 #pragma unroll
-        for (int fn = 0; fn < 2; ++fn)
+        for (int fn = 0; fn < FN2; ++fn)
           acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
     }
@@ -695,9 +704,9 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
     cur ^= 1;
   }
 #pragma unroll
-  for (int fm = 0; fm < 2; ++fm)
+  for (int fm = 0; fm < FM2; ++fm)
 #pragma unroll
-    for (int fn = 0; fn < 2; ++fn) {
+    for (int fn = 0; fn < FN2; ++fn) {
       const int col = n0 + wn + fn * 16 + (lane & 15);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -722,7 +731,7 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
   if (g.K % 64) return false;
   if (g.lda % 8 || g.ldb % 8) return false;
   if (((uintptr_t)g.A & 15) || ((uintptr_t)g.B & 15)) return false;
-  const int M0 = g.M & ~63, N0 = g.N & ~63;
+  const int M0 = g.M & ~15, N0 = g.N & ~63;
   if (M0 == 0 || N0 == 0) return false;
   if (M0 != g.M || N0 != g.N) {
     // edge strips cost 1-2 extra launches: only worth it when the
@@ -750,7 +759,12 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
     }
     return true;
   }
-  int64_t tiles = (int64_t)(g.M / TRBM) * (g.N / TRBN);
+  // tile config by shape: thin-M wgrads (inception Cog 16-48) use 16/32-row
+  // tiles; wide-N with N % 128 == 0 halves the barrier count per output
+  int bm = g.M % 64 ? (g.M % 32 ? 16 : 32) : 64;
+  int bn = (g.N % 128 == 0) ? 128 : 64;
+  if (g.N % bn) return false;  // caller's N0 strip guarantees N % 64 == 0
+  int64_t tiles = (int64_t)(g.M / bm) * (g.N / bn);
   if (tiles >= 1024) return false;  // huge grids: generic path measures better
   int sk = 1, kchunk = g.K;
   if (tiles < 384 && g.K >= 512) {
@@ -761,17 +775,27 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
     kchunk = ((g.K / sk + 63) / 64) * 64;
     sk = (g.K + kchunk - 1) / kchunk;
   }
-  dim3 grid(g.N / TRBN, g.M / TRBM, sk);
-  if (sk > 1) {
+  dim3 grid(g.N / bn, g.M / bm, sk);
+  if (sk > 1)
     (void)hipMemsetAsync(g.C, 0, (size_t)g.M * g.N * sizeof(float), s);
-   hipLaunchKernelGGL(( gemm_tn_tr_kernel<true>), dim3(grid), dim3(256), 0, s, 
-        (const __bf16*)g.A, (const __bf16*)g.B, (float*)g.C, g.M, g.N, g.K,
-        g.lda, g.ldb, g.ldc, g.alpha, kchunk);
-  } else {
-   hipLaunchKernelGGL(( gemm_tn_tr_kernel<false>), dim3(grid), dim3(256), 0, s, 
-        (const __bf16*)g.A, (const __bf16*)g.B, (float*)g.C, g.M, g.N, g.K,
-        g.lda, g.ldb, g.ldc, g.alpha, kchunk);
-  }
+#define PS_TR_LAUNCH(BM_, BN_, WGM_, WGN_)                                  \
+  do {                                                                      \
+    if (sk > 1)                                                             \
+     hipLaunchKernelGGL(( gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, true>), dim3(grid), dim3(256), 0, s,    \
+          (const __bf16*)g.A, (const __bf16*)g.B, (float*)g.C, g.M, g.N,    \
+          g.K, g.lda, g.ldb, g.ldc, g.alpha, kchunk);                       \
+    else                                                                    \
+     hipLaunchKernelGGL(( gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, false>), dim3(grid), dim3(256), 0, s,   \
+          (const __bf16*)g.A, (const __bf16*)g.B, (float*)g.C, g.M, g.N,    \
+          g.K, g.lda, g.ldb, g.ldc, g.alpha, kchunk);                       \
+  } while (0)
+  if (bm == 16 && bn == 128) PS_TR_LAUNCH(16, 128, 1, 4);
+  else if (bm == 16) PS_TR_LAUNCH(16, 64, 1, 4);
+  else if (bm == 32 && bn == 128) PS_TR_LAUNCH(32, 128, 1, 4);
+  else if (bm == 32) PS_TR_LAUNCH(32, 64, 1, 4);
+  else if (bn == 128) PS_TR_LAUNCH(64, 128, 2, 2);
+  else PS_TR_LAUNCH(64, 64, 2, 2);
+#undef PS_TR_LAUNCH
   return true;
 }
 
