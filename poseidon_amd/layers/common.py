@@ -11,7 +11,7 @@ exposed for the solver's all-gather + MFMA outer-product reconstruction
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 

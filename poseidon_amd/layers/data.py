@@ -19,7 +19,6 @@ from typing import List, Optional
 import numpy as np
 import torch
 
-from ..core.blob import Blob
 from ..core.context import ctx
 from ..core.layer import Layer, register_layer
 from ..core import filler as fillers

@@ -13,7 +13,6 @@ from __future__ import annotations
 import numpy as np
 import torch
 
-from ..core.blob import Blob
 from ..core.layer import Layer, register_layer
 from ..ops import functional as ops
 from ..proto import read_proto_binary

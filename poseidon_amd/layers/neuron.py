@@ -10,7 +10,6 @@ from __future__ import annotations
 import torch
 import zlib
 
-from ..core.blob import Blob
 from ..core.layer import Layer, register_layer
 from ..core.context import ctx
 from ..ops import functional as ops

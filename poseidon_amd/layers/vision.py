@@ -16,7 +16,6 @@ from ..core.blob import Blob
 from ..core.layer import Layer, register_layer
 from ..core import filler
 from ..ops import functional as ops
-from ..proto import Message
 
 
 def _pair(param, base: str, generic: str):

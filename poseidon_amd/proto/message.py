@@ -10,7 +10,7 @@ order), so .caffemodel files interoperate.
 from __future__ import annotations
 
 import struct
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 import numpy as np
 

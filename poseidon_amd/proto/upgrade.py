@@ -12,7 +12,6 @@ convolution, as the reference does.
 from __future__ import annotations
 
 from .message import Message
-from . import spec
 
 _V0_TYPE_MAP = {
     "accuracy": "ACCURACY", "bnll": "BNLL", "concat": "CONCAT",

@@ -1,8 +1,6 @@
 """Aux subsystems: V0 proto upgrade, per-layer stats, net-output CSV."""
 
-import os
 
-import torch
 
 import poseidon_amd as pa
 from poseidon_amd.core.net import Net, TRAIN

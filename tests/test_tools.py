@@ -3,8 +3,6 @@
 import os
 
 import numpy as np
-import pytest
-import torch
 
 import poseidon_amd as pa
 from poseidon_amd.data.pdb import PDBReader, PDBWriter, array_to_datum, datum_to_array

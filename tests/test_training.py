@@ -3,7 +3,6 @@ convergence smoke test (the reference's acceptance style, SURVEY.md §4:
 example configs must reach expected accuracy; here synthetic separable data
 replaces MNIST since the image has no datasets)."""
 
-import math
 import os
 
 import numpy as np
@@ -15,7 +14,6 @@ import poseidon_amd as pa
 
 def _ip1(net):
     return next(l for l in net.layers if l.name == "ip1")
-from poseidon_amd.core.net import Net, TRAIN
 from poseidon_amd.proto import Message, parse_text
 from poseidon_amd.solver.solver import SGDSolver, get_solver
 
