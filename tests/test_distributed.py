@@ -4,7 +4,6 @@ must reproduce the single-process summed-gradient result exactly."""
 
 import os
 
-import numpy as np
 import pytest
 import torch
 import torch.multiprocessing as mp

@@ -1,7 +1,6 @@
 """Finite-difference gradient checks for every differentiable layer (CPU
 fp64 where possible; fp32 with loose tolerance elsewhere)."""
 
-import numpy as np
 import pytest
 import torch
 

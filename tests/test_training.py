@@ -5,7 +5,6 @@ replaces MNIST since the image has no datasets)."""
 
 import os
 
-import numpy as np
 import pytest
 import torch
 
