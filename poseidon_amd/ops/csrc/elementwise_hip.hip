@@ -378,13 +378,16 @@ static inline dim3 colsum_grid(int64_t R, int C) {
   } while (0)
 
 void ps_colsum_f32(const float* in, float* out, int64_t R, int C, hipStream_t s) {
-  if (C % 4 == 0 && C <= 8192 && R * C >= (8 << 20))
+  // flat path: big matrices (amortizes the LDS flush) OR small-R wide-C
+  // (the banded kernel's row-slab grid collapses to R/64 blocks: an fc
+  // bias grad at R=256 ran 4 workgroups)
+  if (C % 4 == 0 && C <= 8192 && (R * C >= (8 << 20) || (R < 1024 && C >= 512)))
     PS_COLSUM_FLAT(float, 4, in);
   else
    hipLaunchKernelGGL(( colsum_k<float>), dim3(colsum_grid(R, C)), dim3(256), 0, s, in, out, R, C);
 }
 void ps_colsum_bf16(const void* in, float* out, int64_t R, int C, hipStream_t s) {
-  if (C % 8 == 0 && C <= 8192 && R * C >= (8 << 20))
+  if (C % 8 == 0 && C <= 8192 && (R * C >= (8 << 20) || (R < 1024 && C >= 512)))
     PS_COLSUM_FLAT(__bf16, 8, (const __bf16*)in);
   else
    hipLaunchKernelGGL(( colsum_k<__bf16>), dim3(colsum_grid(R, C)), dim3(256), 0, s, (const __bf16*)in, out, R, C);
