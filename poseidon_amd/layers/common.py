@@ -220,10 +220,8 @@ class EltwiseLayer(Layer):
             y = bottom[0].data.clone()
             for b in bottom[1:]:
                 y.mul_(b.data)
-        else:  # MAX with argmax mask
-            stacked = torch.stack([b.data for b in bottom])
-            y, idx = stacked.max(dim=0)
-            self._argmax = idx
+        else:  # MAX with argmax mask (MaxForward, eltwise_layer.cu:11)
+            y, self._argmax = ops.eltwise_max([b.data for b in bottom])
         top[0].data = y
 
     def backward(self, top, propagate_down, bottom) -> None:
@@ -244,7 +242,7 @@ class EltwiseLayer(Layer):
                 else:
                     b.diff = dy * top[0].data / b.data
             else:  # MAX
-                b.diff = dy * (self._argmax == i).to(dy.dtype)
+                b.diff = ops.eltwise_max_backward(dy, self._argmax, i)
 
 
 @register_layer("MVN")

@@ -221,7 +221,10 @@ class ContrastiveLossLayer(LossLayer):
         n = a.shape[0]
         self._cache = (diff, d2, y)
         if top:
-            loss = (y * d2 + (1 - y) * (self.margin - d2).clamp(min=0)).sum() / (2 * n)
+            # per-pair terms via the CLLForward kernel on GPU
+            # (contrastive_loss_layer.cu:49, legacy margin-d^2 form)
+            terms = ops.contrastive_terms(d2, y, self.margin, legacy=True)
+            loss = (terms.to(d2.dtype)).sum() / (2 * n)
             top[0].data = loss.reshape(())
 
     def backward(self, top, propagate_down, bottom) -> None:

@@ -162,7 +162,7 @@ class ThresholdLayer(NeuronLayer):
         self.threshold = float(tp.threshold) if tp is not None else 0.0
 
     def forward(self, bottom, top) -> None:
-        top[0].data = (bottom[0].data > self.threshold).to(bottom[0].data.dtype)
+        top[0].data = ops.threshold_forward(bottom[0].data, self.threshold)
 
     def backward(self, top, propagate_down, bottom) -> None:
         raise NotImplementedError("THRESHOLD has no backward")

@@ -935,6 +935,57 @@ Tensor dropout_backward(const Tensor& dy, const Tensor& mask, double ratio) {
 // optimizer updates (in-place on fp32 master params)
 // ---------------------------------------------------------------------------
 
+Tensor threshold_forward(const Tensor& x, double thr) {
+  auto xc = any_contig(x);
+  Tensor y = at::empty_like(xc);
+  if (is_bf16(x))
+    ps_threshold_fwd_bf16(xc.data_ptr(), y.data_ptr(), xc.numel(),
+                          (float)thr, stream());
+  else
+    ps_threshold_fwd_f32(xc.data_ptr<float>(), y.data_ptr<float>(),
+                         xc.numel(), (float)thr, stream());
+  return y;
+}
+
+// pairwise running max: y/mask updated in place; mask[i] = blob index of
+// the current max (start by calling with mask filled 0 and y = blob 0)
+void eltwise_max_step(Tensor y, const Tensor& b, Tensor mask, int64_t idx_b) {
+  auto bc = any_contig(b);
+  if (is_bf16(y))
+    ps_eltwise_max_fwd_bf16(y.data_ptr(), bc.data_ptr(), y.data_ptr(),
+                            mask.data_ptr<uint8_t>(), y.numel(), (int)idx_b,
+                            stream());
+  else
+    ps_eltwise_max_fwd_f32(y.data_ptr<float>(), bc.data_ptr<float>(),
+                           y.data_ptr<float>(), mask.data_ptr<uint8_t>(),
+                           y.numel(), (int)idx_b, stream());
+}
+
+Tensor eltwise_max_backward(const Tensor& dy, const Tensor& mask,
+                            int64_t idx) {
+  auto dyc = any_contig(dy);
+  Tensor dx = at::empty_like(dyc);
+  if (is_bf16(dy))
+    ps_eltwise_max_bwd_bf16(dyc.data_ptr(), mask.data_ptr<uint8_t>(),
+                            dx.data_ptr(), dyc.numel(), (int)idx, stream());
+  else
+    ps_eltwise_max_bwd_f32(dyc.data_ptr<float>(), mask.data_ptr<uint8_t>(),
+                           dx.data_ptr<float>(), dyc.numel(), (int)idx,
+                           stream());
+  return dx;
+}
+
+Tensor contrastive_forward(const Tensor& dist_sq, const Tensor& sim,
+                           double margin, bool legacy) {
+  auto d = dist_sq.contiguous().to(at::kFloat);
+  auto sm = sim.contiguous().to(at::kFloat);
+  Tensor loss = at::empty_like(d);
+  ps_contrastive_fwd_f32(d.data_ptr<float>(), sm.data_ptr<float>(),
+                         loss.data_ptr<float>(), d.numel(), (float)margin,
+                         legacy ? 1 : 0, stream());
+  return loss;
+}
+
 void sgd_update(Tensor w, const Tensor& g, Tensor h, double lr, double mom,
                 double wd) {
   ps_sgd_update(w.data_ptr<float>(), g.data_ptr<float>(), h.data_ptr<float>(),
@@ -1092,6 +1143,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_backward", &dropout_backward);
   m.def("sgd_update", &sgd_update);
   m.def("sgd_update_lrdev", &sgd_update_lrdev);
+  m.def("threshold_forward", &threshold_forward);
+  m.def("eltwise_max_step", &eltwise_max_step);
+  m.def("eltwise_max_backward", &eltwise_max_backward);
+  m.def("contrastive_forward", &contrastive_forward);
   m.def("sgd_mt_prepare", &sgd_mt_prepare);
   m.def("sgd_mt_run", &sgd_mt_run);
   m.def("zero_mt_prepare", &zero_mt_prepare);

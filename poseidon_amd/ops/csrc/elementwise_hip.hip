@@ -145,6 +145,58 @@ __global__ void dropout_bwd_k(const T* dy, const uint8_t* mask, T* dx,
     from_f32(mask[i] ? to_f32(dy[i]) * scale : 0.0f, dx[i]);
 }
 
+template <typename T>
+__global__ void threshold_fwd_k(const T* x, T* y, int64_t n, float thr) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    from_f32(to_f32(x[i]) > thr ? 1.0f : 0.0f, y[i]);
+}
+
+// pairwise eltwise max with argmax mask (ELTWISE MAX over >=2 blobs runs
+// this iteratively, like the reference's MaxForward, eltwise_layer.cu:11)
+template <typename T>
+__global__ void eltwise_max_fwd_k(const T* a, const T* b, T* y,
+                                  uint8_t* mask, int64_t n, int idx_b) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float av = to_f32(a[i]), bv = to_f32(b[i]);
+    if (bv > av) {
+      mask[i] = (uint8_t)idx_b;
+      from_f32(bv, y[i]);
+    } else {
+      from_f32(av, y[i]);
+    }
+  }
+}
+
+template <typename T>
+__global__ void eltwise_max_bwd_k(const T* dy, const uint8_t* mask, T* dx,
+                                  int64_t n, int idx) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    from_f32(mask[i] == idx ? to_f32(dy[i]) : 0.0f, dx[i]);
+}
+
+// contrastive-loss per-pair terms (contrastive_loss_layer.cu:49 CLLForward):
+// legacy=false variant: sim pairs contribute d^2, dissim max(margin-d, 0)^2
+__global__ void contrastive_fwd_k(const float* dist_sq, const float* sim,
+                                  float* loss, int64_t n, float margin,
+                                  int legacy) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (sim[i] != 0.0f) {
+      loss[i] = dist_sq[i];
+    } else if (legacy) {
+      float m = margin - dist_sq[i];
+      loss[i] = m > 0.f ? m : 0.f;
+    } else {
+      float m = margin - sqrtf(dist_sq[i]);
+      m = m > 0.f ? m : 0.f;
+      loss[i] = m * m;
+    }
+  }
+}
+
 // column-sum: out[c] += sum_r in[r][c] for in[R][C] row-major -- the bias
 // gradient for both linear (dy[M][N]) and NHWC conv (dy[(N*OH*OW)][C]).
 // Each block owns a row slab; within the block, 64-lane groups sweep 64
@@ -271,6 +323,40 @@ extern "C" {
   void ps_##opname##_bf16(const void* a, void* b, int64_t n, hipStream_t s) { \
     PS_EW_LAUNCH(kern<__bf16>, (const __bf16*)a, (__bf16*)b, n);              \
   }
+
+void ps_threshold_fwd_f32(const float* x, float* y, int64_t n, float thr,
+                          hipStream_t s) {
+  PS_EW_LAUNCH(threshold_fwd_k<float>, x, y, n, thr);
+}
+void ps_threshold_fwd_bf16(const void* x, void* y, int64_t n, float thr,
+                           hipStream_t s) {
+  PS_EW_LAUNCH(threshold_fwd_k<__bf16>, (const __bf16*)x, (__bf16*)y, n, thr);
+}
+void ps_eltwise_max_fwd_f32(const float* a, const float* b, float* y,
+                            uint8_t* mask, int64_t n, int idx_b,
+                            hipStream_t s) {
+  PS_EW_LAUNCH(eltwise_max_fwd_k<float>, a, b, y, mask, n, idx_b);
+}
+void ps_eltwise_max_fwd_bf16(const void* a, const void* b, void* y,
+                             uint8_t* mask, int64_t n, int idx_b,
+                             hipStream_t s) {
+  PS_EW_LAUNCH(eltwise_max_fwd_k<__bf16>, (const __bf16*)a, (const __bf16*)b,
+               (__bf16*)y, mask, n, idx_b);
+}
+void ps_eltwise_max_bwd_f32(const float* dy, const uint8_t* mask, float* dx,
+                            int64_t n, int idx, hipStream_t s) {
+  PS_EW_LAUNCH(eltwise_max_bwd_k<float>, dy, mask, dx, n, idx);
+}
+void ps_eltwise_max_bwd_bf16(const void* dy, const uint8_t* mask, void* dx,
+                             int64_t n, int idx, hipStream_t s) {
+  PS_EW_LAUNCH(eltwise_max_bwd_k<__bf16>, (const __bf16*)dy, mask,
+               (__bf16*)dx, n, idx);
+}
+void ps_contrastive_fwd_f32(const float* dist_sq, const float* sim,
+                            float* loss, int64_t n, float margin, int legacy,
+                            hipStream_t s) {
+  PS_EW_LAUNCH(contrastive_fwd_k, dist_sq, sim, loss, n, margin, legacy);
+}
 
 PS_DEF_UNARY(sigmoid_fwd, sigmoid_fwd_k)
 PS_DEF_UNARY(tanh_fwd, tanh_fwd_k)

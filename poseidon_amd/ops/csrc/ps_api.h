@@ -128,6 +128,18 @@ void ps_dropout_fwd_bf16(const void*, void*, uint8_t*, int64_t, float,
 void ps_dropout_bwd_bf16(const void*, const uint8_t*, void*, int64_t, float,
                          hipStream_t);
 void ps_colsum_bf16(const void*, float*, int64_t, int, hipStream_t);
+void ps_threshold_fwd_f32(const float*, float*, int64_t, float, hipStream_t);
+void ps_threshold_fwd_bf16(const void*, void*, int64_t, float, hipStream_t);
+void ps_eltwise_max_fwd_f32(const float*, const float*, float*, uint8_t*,
+                            int64_t, int, hipStream_t);
+void ps_eltwise_max_fwd_bf16(const void*, const void*, void*, uint8_t*,
+                             int64_t, int, hipStream_t);
+void ps_eltwise_max_bwd_f32(const float*, const uint8_t*, float*, int64_t,
+                            int, hipStream_t);
+void ps_eltwise_max_bwd_bf16(const void*, const uint8_t*, void*, int64_t,
+                             int, hipStream_t);
+void ps_contrastive_fwd_f32(const float*, const float*, float*, int64_t,
+                            float, int, hipStream_t);
 
 // pool.hip
 void ps_maxpool_fwd_f32(const float*, float*, uint8_t*, const PoolGeom*, hipStream_t);

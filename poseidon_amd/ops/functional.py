@@ -435,6 +435,45 @@ def dropout_backward(dy, mask, ratio: float):
 #   w   -= hist
 # ---------------------------------------------------------------------------
 
+def threshold_forward(x: torch.Tensor, thr: float) -> torch.Tensor:
+    if x.is_cuda:
+        return _ext().threshold_forward(x, float(thr))
+    return (x > thr).to(x.dtype)
+
+
+def eltwise_max(blobs) -> tuple:
+    """Running pairwise max with a u8 argmax mask (ELTWISE MAX,
+    eltwise_layer.cu:11 MaxForward semantics)."""
+    if blobs[0].is_cuda:
+        y = blobs[0].contiguous().clone()
+        mask = torch.zeros(y.numel(), dtype=torch.uint8, device=y.device)
+        for i, b in enumerate(blobs[1:], start=1):
+            _ext().eltwise_max_step(y.view(-1), b.reshape(-1), mask, i)
+        return y, mask.view(y.shape)
+    stacked = torch.stack(list(blobs))
+    y, idx = stacked.max(dim=0)
+    return y, idx.to(torch.uint8)
+
+
+def eltwise_max_backward(dy: torch.Tensor, mask: torch.Tensor,
+                         idx: int) -> torch.Tensor:
+    if dy.is_cuda:
+        return _ext().eltwise_max_backward(dy.reshape(-1),
+                                           mask.reshape(-1),
+                                           idx).view(dy.shape)
+    return dy * (mask == idx).to(dy.dtype)
+
+
+def contrastive_terms(dist_sq: torch.Tensor, sim: torch.Tensor,
+                      margin: float, legacy: bool) -> torch.Tensor:
+    """Per-pair contrastive loss terms (contrastive_loss_layer.cu:49)."""
+    if dist_sq.is_cuda:
+        return _ext().contrastive_forward(dist_sq, sim, float(margin),
+                                          bool(legacy))
+    m = torch.clamp(margin - (dist_sq if legacy else dist_sq.sqrt()), min=0)
+    return torch.where(sim != 0, dist_sq, m if legacy else m * m)
+
+
 def sgd_update(w: torch.Tensor, grad: torch.Tensor, hist: torch.Tensor,
                local_rate: float, momentum: float, decay: float,
                lr_dev=None) -> None:

@@ -395,3 +395,27 @@ def test_full_solver_protocol_gpu(tmp_path):
         assert torch.isfinite(w0).all()
     finally:
         pa.init(device="cpu")
+
+
+def test_threshold_eltwise_contrastive_kernels():
+    """Cold-layer CDNA4 kernels (reference eltwise_layer.cu:11,72,
+    threshold_layer.cu:10, contrastive_loss_layer.cu:49)."""
+    x = rnd(4, 8, 5, 5, seed=31)
+    y = ops.threshold_forward(x.to(DEV), 0.1)
+    close(y, (x > 0.1).float(), what="threshold")
+
+    blobs = [rnd(3, 6, 4, 4, seed=s) for s in (32, 33, 34)]
+    yg, mg = ops.eltwise_max([b.to(DEV) for b in blobs])
+    yr, mr = ops.eltwise_max(blobs)
+    close(yg, yr, what="eltwise max fwd")
+    dy = rnd(3, 6, 4, 4, seed=35)
+    for i in range(3):
+        dg = ops.eltwise_max_backward(dy.to(DEV), mg, i)
+        dr = ops.eltwise_max_backward(dy, mr, i)
+        close(dg, dr, what=f"eltwise max bwd {i}")
+
+    d2 = rnd(16, seed=36).abs()
+    sim = (rnd(16, seed=37) > 0).float()
+    tg = ops.contrastive_terms(d2.to(DEV), sim.to(DEV), 1.0, True)
+    tr2 = ops.contrastive_terms(d2, sim, 1.0, True)
+    close(tg, tr2, what="contrastive terms")
