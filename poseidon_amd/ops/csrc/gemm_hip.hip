@@ -575,11 +575,14 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
 
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
-template <int COLS>
+template <int COLS, bool GATHER = false>
 __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
                                      int64_t ld, int k0, int c0, int wid,
-                                     int lane) {
-  // [BK=64 rows][COLS cols]: COLS*2 B rows, 16 B lane chunks
+                                     int lane,
+                                     const GatherDesc* ga = nullptr) {
+  // [BK=64 rows][COLS cols]: COLS*2 B rows, 16 B lane chunks. GATHER:
+  // the operand is the im2col matrix gathered on the fly from NHWC x
+  // (row = im2col row np, col = kg) -- implicit wgrad without a colT.
   constexpr int LPR = COLS / 8;       // lanes per row
   constexpr int RPC = 64 / LPR;       // rows per 1 KB chunk
   constexpr int CHUNKS = 64 / RPC;
@@ -588,7 +591,9 @@ __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
 #pragma unroll
   for (int ci = wid; ci < CHUNKS; ci += 4) {
     const int row = ci * RPC + r_in;
-    const __bf16* g2 = src + (int64_t)(k0 + row) * ld + c0 + slot * 8;
+    const __bf16* g2 =
+        GATHER ? gather_addr<__bf16>(*ga, k0 + row, c0 + slot * 8)
+               : src + (int64_t)(k0 + row) * ld + c0 + slot * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g2,
         (__attribute__((address_space(3))) void*)(lds + ci * 512), 16, 0, 0);
@@ -626,12 +631,14 @@ __device__ inline void tr_wait(bf16x8 (&a)[NA], bf16x8 (&b)[NB]) {
   for (int i = 0; i < NB; ++i) asm volatile("" : "+v"(b[i]));
 }
 
-template <int BM2, int BN2, int WGM2, int WGN2, bool SPLITK>
+template <int BM2, int BN2, int WGM2, int WGN2, bool SPLITK,
+          bool GB2 = false>
 __global__ __launch_bounds__(256)
 void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
                        const __bf16* __restrict__ B, float* __restrict__ C,
                        int M, int N, int K, int64_t ldA, int64_t ldB,
-                       int64_t ldC, float alpha, int kchunk) {
+                       int64_t ldC, float alpha, int kchunk,
+                       GatherDesc ga_b = {}) {
   constexpr int FM2 = BM2 / WGM2 / 16, FN2 = BN2 / WGN2 / 16;
   static_assert(WGM2 * WGN2 == 4, "4 waves");
   __shared__ __attribute__((aligned(16))) __bf16 a_lds[2][64 * BM2];
@@ -667,7 +674,7 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
 
   f32x4 acc[FM2][FN2] = {};
   stage_kmaj_tr<BM2>(a_lds[0], A, ldA, k_begin, m0, wid, lane);
-  stage_kmaj_tr<BN2>(b_lds[0], B, ldB, k_begin, n0, wid, lane);
+  stage_kmaj_tr<BN2, GB2>(b_lds[0], B, ldB, k_begin, n0, wid, lane, &ga_b);
   __syncthreads();
   unsigned ab[2], bb[2];
 #pragma unroll
@@ -681,7 +688,8 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
   for (int k0 = k_begin; k0 < k_end; k0 += 64) {
     if (k0 + 64 < k_end) {
       stage_kmaj_tr<BM2>(a_lds[cur ^ 1], A, ldA, k0 + 64, m0, wid, lane);
-      stage_kmaj_tr<BN2>(b_lds[cur ^ 1], B, ldB, k0 + 64, n0, wid, lane);
+      stage_kmaj_tr<BN2, GB2>(b_lds[cur ^ 1], B, ldB, k0 + 64, n0, wid,
+                              lane, &ga_b);
     }
 #pragma unroll
     for (int kk = 0; kk < 64; kk += 32) {
@@ -725,14 +733,25 @@ template <typename T, typename OUT>
 static void gemm_dispatch(const GemmArgs& g, hipStream_t s);
 
 static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
-  if (g.a_klast || g.b_klast || g.gather_a || g.gather_b) return false;
+  if (g.a_klast || g.b_klast || g.gather_a) return false;
   if (g.bias || g.relu || g.beta != 0.0f || g.batch > 1) return false;
   if (g.ws) return false;  // caller committed to workspace split-K layout
   if (g.K % 64) return false;
-  if (g.lda % 8 || g.ldb % 8) return false;
-  if (((uintptr_t)g.A & 15) || ((uintptr_t)g.B & 15)) return false;
+  if (g.lda % 8) return false;
+  if ((uintptr_t)g.A & 15) return false;
+  if (g.gather_b) {
+    // gathered B (implicit wgrad): 16 B lane chunks must stay inside one
+    // contiguous cg run of NHWC x, aligned for glds
+    const GatherDesc& gb = *g.gather_b;
+    if (gb.Cg % 8 || gb.C % 8 || gb.c0 % 8) return false;
+  } else {
+    if (g.ldb % 8 || ((uintptr_t)g.B & 15)) return false;
+  }
   const int M0 = g.M & ~15, N0 = g.N & ~63;
   if (M0 == 0 || N0 == 0) return false;
+  if (g.gather_b && (M0 != g.M || N0 != g.N))
+    return false;  // a strip's column offset is not expressible in the
+                   // gather descriptor (kg shift crosses khw boundaries)
   if (M0 != g.M || N0 != g.N) {
     // edge strips cost 1-2 extra launches: only worth it when the
     // interior carries real work (GoogLeNet's small inception wgrads
@@ -778,9 +797,21 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
   dim3 grid(g.N / bn, g.M / bm, sk);
   if (sk > 1)
     (void)hipMemsetAsync(g.C, 0, (size_t)g.M * g.N * sizeof(float), s);
+  GatherDesc gb = g.gather_b ? *g.gather_b : GatherDesc{};
 #define PS_TR_LAUNCH(BM_, BN_, WGM_, WGN_)                                  \
   do {                                                                      \
-    if (sk > 1)                                                             \
+    if (g.gather_b) {                                                       \
+      if (sk > 1)                                                           \
+       hipLaunchKernelGGL(( gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, true, true>)                 \
+            , dim3(grid), dim3(256), 0, s, (const __bf16*)g.A, (const __bf16*)g.B,   \
+                                  (float*)g.C, g.M, g.N, g.K, g.lda,        \
+                                  g.ldb, g.ldc, g.alpha, kchunk, gb);       \
+      else                                                                  \
+       hipLaunchKernelGGL(( gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, false, true>)                \
+            , dim3(grid), dim3(256), 0, s, (const __bf16*)g.A, (const __bf16*)g.B,   \
+                                  (float*)g.C, g.M, g.N, g.K, g.lda,        \
+                                  g.ldb, g.ldc, g.alpha, kchunk, gb);       \
+    } else if (sk > 1)                                                      \
      hipLaunchKernelGGL(( gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, true>), dim3(grid), dim3(256), 0, s,    \
           (const __bf16*)g.A, (const __bf16*)g.B, (float*)g.C, g.M, g.N,    \
           g.K, g.lda, g.ldb, g.ldc, g.alpha, kchunk);                       \
