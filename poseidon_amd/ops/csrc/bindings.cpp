@@ -71,7 +71,7 @@ std::atomic<bool> g_implicit_gemm{false};
 // 3x3 convs build 1-2 GB column matrices; gathering from x reads KHW x
 // less HBM). Measured: global implicit loses ~4-8% on AlexNet/GoogLeNet
 // (gather decode overhead on small colT), wins on the giant-colT layers.
-std::atomic<int64_t> g_implicit_thresh{512LL << 20};
+std::atomic<int64_t> g_implicit_thresh{192LL << 20};
 
 void set_implicit_gemm(bool on) { g_implicit_gemm.store(on); }
 void set_implicit_threshold(int64_t bytes) { g_implicit_thresh.store(bytes); }
@@ -314,13 +314,13 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   // all (the wgrad GEMM gathers too); conv1-style small-C layers still
   // materialize (and 1x1 convs alias x directly)
   int64_t colT_bytes = NP * Kcol * (bf16 ? 2 : 4);
-  // giant colT layers go implicit -- UNLESS their wgrad qualifies for the
-  // tr16 TN fast path (gemm.hip try_gemm_tn_tr), which reads the
-  // materialized colT ~5x faster than the gather-staged implicit wgrad
-  bool wgrad_tr_ok = bf16 && G == 1 && (Co % 64) == 0 &&
-                     (ldc_col % 64) == 0 && (NP % 64) == 0;
+  // big-colT layers go implicit: with the gather-staged tr16 wgrad
+  // (gemm.hip stage_kmaj_tr<GATHER>) the implicit path now beats
+  // materialization wherever the column matrix is large (VGG/AlexNet
+  // early-mid layers: no im2col write, wgrad gathers x directly);
+  // small-colT inception-style layers still measure faster materialized
   bool implicit = (g_implicit_gemm.load() ||
-                   (colT_bytes > g_implicit_thresh.load() && !wgrad_tr_ok))
+                   colT_bytes > g_implicit_thresh.load())
                   && !is_1x1 && (Cg % VEC == 0);
   Tensor colT;
   if (is_1x1) {
