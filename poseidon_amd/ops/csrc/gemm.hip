@@ -385,9 +385,11 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
                    (ldb % EPB) == 0 && (k_begin % EPB) == 0 &&
                    (((uintptr_t)B & 15) == 0);
     if (GA) {
-      // gathered A: every BK-wide row must stay inside one cg run and the
-      // run base must be 16B-aligned
-      glds_ok = glds_ok && (ga_a.Cg % BK) == 0 && (ga_a.C % EPB) == 0 &&
+      // gathered A: every 16 B lane chunk must stay inside one cg run
+      // (chunk starts are EPB-aligned, so Cg % EPB suffices -- the old
+      // Cg % BK gate was overly strict and pushed grouped convs like
+      // AlexNet conv2 (Cg=48) onto the guarded register-gather path)
+      glds_ok = glds_ok && (ga_a.Cg % EPB) == 0 && (ga_a.C % EPB) == 0 &&
                 (ga_a.c0 % EPB) == 0;
     } else {
       glds_ok = glds_ok && (lda % EPB) == 0 && (((uintptr_t)A & 15) == 0);
