@@ -589,12 +589,32 @@ __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
   constexpr int CHUNKS = 64 / RPC;
   const int r_in = lane / LPR;
   const int slot = lane % LPR;
+  // gather: the column decode (kg -> kh,kw,cg) is loop-invariant per lane
+  // (slot is fixed); only the row decode (np -> n,oh,ow) runs per chunk.
+  // PMC showed the naive per-chunk gather_addr at 18-20 VALU per MFMA.
+  int kkh = 0, kkw = 0, cg = 0;
+  if (GATHER) {
+    int khw = fdiv_fix(c0 + slot * 8, ga->Cg, ga->inv_Cg, cg);
+    kkh = fdiv_fix(khw, ga->kw, ga->inv_kw, kkw);
+  }
 #pragma unroll
   for (int ci = wid; ci < CHUNKS; ci += 4) {
     const int row = ci * RPC + r_in;
-    const __bf16* g2 =
-        GATHER ? gather_addr<__bf16>(*ga, k0 + row, c0 + slot * 8)
-               : src + (int64_t)(k0 + row) * ld + c0 + slot * 8;
+    const __bf16* g2;
+    if (GATHER) {
+      int ow, oh;
+      const int t2 = fdiv_fix(k0 + row, ga->Wo, ga->inv_Wo, ow);
+      const int n2 = fdiv_fix(t2, ga->Ho, ga->inv_Ho, oh);
+      const int ih = oh * ga->sh - ga->ph + kkh;
+      const int iw = ow * ga->sw - ga->pw + kkw;
+      g2 = (ih < 0 || ih >= ga->H || iw < 0 || iw >= ga->W)
+               ? (const __bf16*)ga->zero
+               : (const __bf16*)ga->x +
+                     (((int64_t)n2 * ga->H + ih) * ga->W + iw) * ga->C +
+                     ga->c0 + cg;
+    } else {
+      g2 = src + (int64_t)(k0 + row) * ld + c0 + slot * 8;
+    }
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g2,
         (__attribute__((address_space(3))) void*)(lds + ci * 512), 16, 0, 0);
