@@ -394,6 +394,44 @@ Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
   int Kcol = G * Kg;
   int64_t NP = (int64_t)g.N * g.Ho * g.Wo;
 
+  // stride-1 dgrad runs as a FORWARD conv of dy with rotated weights
+  // (dx = conv(dy, rot180 W, pad = K-1-p)) through the implicit-gather
+  // GEMM: no dcolT round-trip (up to 2 GB for VGG conv1_2) and no col2im.
+  const int VEC = bf16 ? 8 : 4;
+  bool is_1x1e = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 &&
+                  pw == 0);
+  if (sh == 1 && sw == 1 && !is_1x1e && (Cog % VEC) == 0 && ph <= kh - 1 &&
+      pw <= kw - 1) {
+    const int K2 = kh * kw * Cog;
+    auto wc = w.contiguous();
+    Tensor wr = at::empty({(int64_t)G * Cg, (int64_t)K2}, dy.options());
+    if (bf16)
+      ps_weight_to_dgrad_f32_bf16(wc.data_ptr<float>(), wr.data_ptr(), Co,
+                                  Cg, kh, kw, G, stream());
+    else
+      ps_weight_to_dgrad_f32(wc.data_ptr<float>(), wr.data_ptr<float>(), Co,
+                             Cg, kh, kw, G, stream());
+    int64_t NP2 = (int64_t)g.N * g.H * g.W;
+    for (int grp = 0; grp < G; ++grp) {
+      GatherDesc ga;
+      ga.x = dy_cl.data_ptr();
+      ga.zero = zero_page(dy_cl);
+      ga.C = Co; ga.H = g.Ho; ga.W = g.Wo;
+      ga.Ho = g.H; ga.Wo = g.W;
+      ga.kh = kh; ga.kw = kw; ga.sh = 1; ga.sw = 1;
+      ga.ph = kh - 1 - ph; ga.pw = kw - 1 - pw;
+      ga.Cg = Cog; ga.c0 = grp * Cog;
+      ps_fill_gather_inv(&ga);
+      run_gemm(dy_cl, wr, dx, nullptr,
+               (int)NP2, Cg, K2,
+               /*lda=*/K2, /*ldb=*/K2, /*ldc=*/g.C,
+               /*a_off=*/0, /*b_off=*/(int64_t)grp * Cg * K2,
+               /*c_off=*/(int64_t)grp * Cg,
+               true, true, 1.0f, 0.0f, &ga);
+    }
+    return dx;
+  }
+
   // transposed khwc repack [G][Kg][Cog]: the dgrad GEMM becomes pure NT
   // (both operands K-last) instead of a K-major-staged NN
   Tensor wkT = wkT_cache.has_value() && wkT_cache->scalar_type() ==

@@ -342,6 +342,30 @@ __global__ void weight_to_khwc_both_k(const TI* src, TO* dst, TO* dst_tr,
   }
 }
 
+// dgrad-as-forward weight repack: W[co][ci][kh][kw] (NCHW, per-group) ->
+// wrot[(grp*Cg + ci)][ ((KH-1-kh)*KW + (KW-1-kw)) * Cog + cog ] -- the
+// K-last B operand of the stride-1 dgrad-as-conv GEMM
+// (dx = conv(dy, rot180(W), pad = K-1-p)).
+template <typename TI, typename TO>
+__global__ void weight_to_dgrad_k(const TI* src, TO* dst, int Co, int Cig,
+                                  int KH, int KW, int G) {
+  const int Cog = Co / G;
+  const int K2 = KH * KW * Cog;
+  int64_t total = (int64_t)Co * Cig * KH * KW;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int kkw = i % KW;
+    int64_t t = i / KW;
+    int kkh = t % KH; t /= KH;
+    int ci = t % Cig;
+    int co = t / Cig;
+    const int grp = co / Cog, cog = co % Cog;
+    const int k2 = ((KH - 1 - kkh) * KW + (KW - 1 - kkw)) * Cog + cog;
+    from_f32(to_f32(src[i]),
+             dst[((int64_t)grp * Cig + ci) * K2 + k2]);
+  }
+}
+
 template <typename TI, typename TO>
 __global__ void weight_from_khwc_k(const TI* src, TO* dst, int Co, int Cig,
                                    int KH, int KW, int ld, float beta) {
@@ -552,6 +576,19 @@ void ps_weight_to_khwc_both_f32_bf16(const float* src, void* dst,
   weight_to_khwc_both_k<float, __bf16>
       <<<ew_grid((int64_t)Co * Cig * KH * KW), 256, 0, s>>>(
           src, (__bf16*)dst, (__bf16*)dst_tr, Co, Cig, KH, KW, G, ldk);
+}
+void ps_weight_to_dgrad_f32(const float* src, float* dst, int Co, int Cig,
+                            int KH, int KW, int G, hipStream_t s) {
+  weight_to_dgrad_k<float, float>
+      <<<ew_grid((int64_t)Co * Cig * KH * KW), 256, 0, s>>>(src, dst, Co,
+                                                            Cig, KH, KW, G);
+}
+void ps_weight_to_dgrad_f32_bf16(const float* src, void* dst, int Co,
+                                 int Cig, int KH, int KW, int G,
+                                 hipStream_t s) {
+  weight_to_dgrad_k<float, __bf16>
+      <<<ew_grid((int64_t)Co * Cig * KH * KW), 256, 0, s>>>(
+          src, (__bf16*)dst, Co, Cig, KH, KW, G);
 }
 void ps_weight_from_khwc_f32(const float* src, float* dst, int Co, int Cig,
                              int KH, int KW, int ld, float beta,

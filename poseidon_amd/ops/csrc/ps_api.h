@@ -225,6 +225,10 @@ void ps_im2col_nhwc_bf16(const void*, void*, const ConvGeom*, int ldcol,
 void ps_col2im_nhwc_bf16(const void*, void*, const ConvGeom*, hipStream_t);
 void ps_weight_to_khwc_f32(const float*, float*, int, int, int, int, hipStream_t);
 void ps_weight_to_khwc_f32_bf16(const float*, void*, int, int, int, int, hipStream_t);
+void ps_weight_to_dgrad_f32(const float*, float*, int, int, int, int, int,
+                            hipStream_t);
+void ps_weight_to_dgrad_f32_bf16(const float*, void*, int, int, int, int,
+                                 int, hipStream_t);
 void ps_weight_from_khwc_f32(const float*, float*, int, int, int, int,
                              int ld, float beta, hipStream_t);
 void ps_weight_to_khwc_tr_f32(const float*, float*, int, int, int, int, int,
