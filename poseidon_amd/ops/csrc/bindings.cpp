@@ -400,8 +400,11 @@ Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
   const int VEC = bf16 ? 8 : 4;
   bool is_1x1e = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 &&
                   pw == 0);
+  // worth it only when the dcolT it eliminates is substantial (GoogLeNet's
+  // small inception dgrads measured faster on the materialized NT+col2im)
+  int64_t dcolT_bytes = NP * Kcol * (bf16 ? 2 : 4);
   if (sh == 1 && sw == 1 && !is_1x1e && (Cog % VEC) == 0 && ph <= kh - 1 &&
-      pw <= kw - 1) {
+      pw <= kw - 1 && dcolT_bytes > (128LL << 20)) {
     const int K2 = kh * kw * Cog;
     auto wc = w.contiguous();
     Tensor wr = at::empty({(int64_t)G * Cg, (int64_t)K2}, dy.options());
