@@ -160,6 +160,18 @@ static inline dim3 lrn_grid(int64_t rows, int rpb) {
 // the fallback (they store scale) for odd channel counts.
 // ---------------------------------------------------------------------------
 
+// sc^(-beta) with a fast path for the universal beta = 0.75:
+// sc^(-3/4) = rsqrt(sc) * sqrt(rsqrt(sc)) -- two hardware transcendentals
+// instead of __powf's ~20-op expansion (PMC: the v8 kernels were 66-78%
+// issue-stalled on the powf dependency chain, ~300 VALU/iteration).
+__device__ inline float lrn_pow_negbeta(float sc, float beta) {
+  if (beta == 0.75f) {
+    const float r = rsqrtf(sc);
+    return r * sqrtf(r);
+  }
+  return __powf(sc, -beta);
+}
+
 template <typename T> struct LrnV8;
 template <> struct LrnV8<float> {
   typedef f32x4 half_t;
@@ -235,7 +247,7 @@ __global__ void lrn_fwd_v8_k(const T* __restrict__ x, T* __restrict__ y,
         ss += (m < PRE) ? lh[m] : (m < PRE + 8) ? sq[m - PRE]
                                                 : rh[m - PRE - 8];
       }
-      out[j] = xv[j] * __powf(1.0f + alpha_over_n * ss, -beta);
+      out[j] = xv[j] * lrn_pow_negbeta(1.0f + alpha_over_n * ss, beta);
     }
     LrnV8<T>::store8(y + row * C + cb, out);
   }
@@ -305,10 +317,11 @@ __global__ void lrn_bwd_v8_k(const T* __restrict__ x, const T* __restrict__ y,
 #pragma unroll
       for (int k = -PRE; k <= PRE; ++k) ss += XS(m + k);
       const float sc = 1.0f + alpha_over_n * ss;
+      const float r = rsqrtf(sc);
       if (m >= 0 && m < 8) scc[m] = sc;
       const float ym = (m < 0) ? lhy[m + PRE] : (m < 8) ? yv[m] : rhy[m - 8];
       const float dm = (m < 0) ? lhd[m + PRE] : (m < 8) ? dyv[m] : rhd[m - 8];
-      ratio[i2] = dm * ym / sc;
+      ratio[i2] = dm * ym * (r * r);  // 1/sc via rsqrt^2 (no v_div chain)
     }
     float out[8];
 #pragma unroll
@@ -316,7 +329,8 @@ __global__ void lrn_bwd_v8_k(const T* __restrict__ x, const T* __restrict__ y,
       float acc = 0.f;
 #pragma unroll
       for (int k = 0; k <= 2 * PRE; ++k) acc += ratio[j + k];
-      out[j] = dyv[j] * __powf(scc[j], -beta) - cache_ratio * xv[j] * acc;
+      out[j] = dyv[j] * lrn_pow_negbeta(scc[j], beta)
+               - cache_ratio * xv[j] * acc;
     }
     LrnV8<T>::store8(dx + row * C + cb, out);
   }
