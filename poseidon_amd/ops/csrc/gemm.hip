@@ -131,6 +131,17 @@ __device__ inline void stage_glds(T* lds, const T* __restrict__ src,
   }
 }
 
+// Guarded scalar staging of a PARTIAL final K-tile into the glds-layout
+// image (zero-filled beyond k_end): lets a GEMM whose K is not a BK
+// multiple still take the glds path for the full tiles (AlexNet conv2's
+// implicit fwd at K=1200 was otherwise stuck on the guarded register
+// pipeline for every tile: 674 us -> glds + this tail).
+template <typename T, int ROWS, bool GATHER = false>
+__device__ inline void stage_tail_linear(T* lds, const T* __restrict__ src,
+                                         int64_t lda, int row0, int k0,
+                                         int k_end, int tid,
+                                         const GatherDesc* ga = nullptr);
+
 // matching read-side XOR for glds-staged tiles: element offset within a
 // linear [row][BK] image whose 16 B slots are swizzled by row
 template <typename T>
@@ -140,6 +151,28 @@ __device__ inline int glds_col(int row, int col) {
   constexpr int LPR = TR::BK / EPB;
   int slot = col / EPB;
   return (slot ^ (((row >> 2) & 1) << 1)) * EPB + (col % EPB);
+}
+
+template <typename T, int ROWS, bool GATHER>
+__device__ inline void stage_tail_linear(T* lds, const T* __restrict__ src,
+                                         int64_t lda, int row0, int k0,
+                                         int k_end, int tid,
+                                         const GatherDesc* ga) {
+  using TR = GemmTraits<T>;
+  constexpr int BK = TR::BK;
+  const int rem = k_end - k0;
+  for (int i = tid; i < ROWS * BK; i += 256) {
+    const int row = i / BK, col = i - row * BK;
+    T v = (T)0.0f;
+    if (col < rem) {
+      if (GATHER)
+        v = gather_addr<T>(*ga, row0 + row, k0 + (col & ~(TR::VEC - 1)))
+                [col & (TR::VEC - 1)];
+      else
+        v = src[(int64_t)(row0 + row) * lda + k0 + col];
+    }
+    lds[row * BK + glds_col<T>(row, col)] = v;
+  }
 }
 
 // Column offset inside an LDS tile row. K-major-staged tiles XOR the
@@ -381,7 +414,7 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
     constexpr int EPB = 16 / (int)sizeof(T);
     const int k_span = k_end - k_begin;
     bool glds_ok = A_KLAST && B_KLAST && (m0 + BM <= M) &&
-                   (n0 + BN <= N) && (k_span % BK) == 0 && k_span > 0 &&
+                   (n0 + BN <= N) && k_span >= BK &&
                    (ldb % EPB) == 0 && (k_begin % EPB) == 0 &&
                    (((uintptr_t)B & 15) == 0);
     if (GA) {
@@ -395,6 +428,9 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
       glds_ok = glds_ok && (lda % EPB) == 0 && (((uintptr_t)A & 15) == 0);
     }
     if (glds_ok) {
+      // full BK tiles via glds; a partial final tile (K % BK != 0) is
+      // staged by guarded scalar writes into the same swizzled image
+      const int k_full = k_begin + (k_span / BK) * BK;
       T* a_lin0 = a_lds[0];
       T* b_lin0 = b_lds[0];
       T* a_lin1 = a_lds[1];
@@ -404,11 +440,16 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
       __syncthreads();  // drains the in-flight glds (vmcnt 0) + barrier
       int cur2 = 0;
       for (int k0 = k_begin; k0 < k_end; k0 += BK) {
-        if (k0 + BK < k_end) {
+        if (k0 + BK < k_full) {
           stage_glds<T, BM, GA>(cur2 ? a_lin0 : a_lin1, A, lda, m0, k0 + BK,
                                 wid, lane, &ga_a);
           stage_glds<T, BN>(cur2 ? b_lin0 : b_lin1, B, ldb, n0, k0 + BK, wid,
                             lane);
+        } else if (k0 + BK < k_end) {
+          stage_tail_linear<T, BM, GA>(cur2 ? a_lin0 : a_lin1, A, lda, m0,
+                                       k0 + BK, k_end, tid, &ga_a);
+          stage_tail_linear<T, BN, false>(cur2 ? b_lin0 : b_lin1, B, ldb, n0,
+                                          k0 + BK, k_end, tid, nullptr);
         }
         const T* al = cur2 ? a_lin1 : a_lin0;
         const T* bl = cur2 ? b_lin1 : b_lin0;
