@@ -288,7 +288,10 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   const int VEC = bf16 ? 8 : 4;
   // small-K pad (ps_api.h ps_colT_ld): zero columns up to one full BK so
   // the fwd GEMM is a single interior k-tile (glds path)
-  const int ldc_col = ps_colT_ld(G, g.C, kh, kw, VEC);
+  // 1x1 convs alias x rows as colT (physical stride = C): never pad those
+  const bool pre_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 &&
+                        ph == 0 && pw == 0);
+  const int ldc_col = pre_1x1 ? Kcol : ps_colT_ld(G, g.C, kh, kw, VEC);
   const bool kpad = ldc_col != Kcol;
 
   // fused repack: one kernel writes both the khwc fwd operand and the
@@ -326,14 +329,23 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   if (is_1x1) {
     colT = rows2d(x_cl);  // alias: x rows ARE the col rows
   } else if (!implicit) {
-    colT = kpad ? at::zeros({NP, (int64_t)ldc_col}, x.options())
-                : at::empty({NP, (int64_t)Kcol}, x.options());
+    colT = at::empty({NP, (int64_t)ldc_col}, x.options());
     if (bf16)
       ps_im2col_nhwc_bf16(x_cl.data_ptr(), colT.data_ptr(), &g, ldc_col,
                           stream());
     else
       ps_im2col_nhwc_f32(x_cl.data_ptr<float>(), colT.data_ptr<float>(), &g,
                          ldc_col, stream());
+    // pad columns (ldc_col > Kcol) must be zero for the GEMMs; the
+    // rowstage/rowrun kernels already write them, the vectorized
+    // im2col_nhwc_k does not
+    if (kpad && Cg % VEC == 0) {
+      if (bf16)
+        ps_zero_cols_bf16(colT.data_ptr(), NP, ldc_col, Kcol, stream());
+      else
+        ps_zero_cols_f32(colT.data_ptr<float>(), NP, ldc_col, Kcol,
+                         stream());
+    }
   }
 
   Tensor y = at::empty({g.N, Co, g.Ho, g.Wo},
@@ -352,6 +364,7 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
       ga.C = g.C; ga.H = g.H; ga.W = g.W; ga.Ho = g.Ho; ga.Wo = g.Wo;
       ga.kh = kh; ga.kw = kw; ga.sh = sh; ga.sw = sw; ga.ph = ph; ga.pw = pw;
       ga.Cg = Cg; ga.c0 = grp * Cg;
+      ga.kg_max = Kg;
       ps_fill_gather_inv(&ga);
       run_gemm(x_cl, wk, y, bp ? bp + grp * Cog : nullptr,
                (int)NP, Cog, Kg,
@@ -424,6 +437,7 @@ Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
       ga.kh = kh; ga.kw = kw; ga.sh = 1; ga.sw = 1;
       ga.ph = kh - 1 - ph; ga.pw = kw - 1 - pw;
       ga.Cg = Cog; ga.c0 = grp * Cog;
+      ga.kg_max = K2;
       ps_fill_gather_inv(&ga);
       run_gemm(dy_cl, wr, dx, nullptr,
                (int)NP2, Cg, K2,
@@ -481,8 +495,11 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
   const bool implicit = colT.numel() == 0 && colT.dim() == 1;
   int64_t Kcol = implicit ? 0 : colT.size(1);
   // colT may carry zero pad columns (ps_colT_ld): run the GEMM over the
-  // padded width too (zero B columns -> zero dwk columns, skipped below)
-  const int Kgw = implicit ? Kg : (int)(Kcol / G);
+  // padded width too (zero B columns -> zero dwk columns, skipped below).
+  // Implicit wgrads pad the same way via the gather's kg_max bound, which
+  // makes them tr16-eligible (N % 64 == 0, no edge strips).
+  const int Kgw = implicit ? (G == 1 ? (Kg + 63) & ~63 : Kg)
+                           : (int)(Kcol / G);
 
   // fp32 gradient accumulation regardless of activation dtype
   Tensor dwk = at::empty({Co, (int64_t)Kgw}, dy.options().dtype(at::kFloat));
@@ -497,12 +514,13 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
       gb.Ho = dy_cl.size(2); gb.Wo = dy_cl.size(3);
       gb.kh = kh; gb.kw = kw; gb.sh = sh; gb.sw = sw; gb.ph = ph; gb.pw = pw;
       gb.Cg = Cig; gb.c0 = grp * Cig;
+      gb.kg_max = Kg;
       ps_fill_gather_inv(&gb);
       run_gemm(dy2, x_cl, dwk, nullptr,
-               Cog, Kg, (int)NP,
-               /*lda=*/Co, /*ldb=*/Kg, /*ldc=*/Kg,
+               Cog, Kgw, (int)NP,
+               /*lda=*/Co, /*ldb=*/Kgw, /*ldc=*/Kgw,
                /*a_off=*/(int64_t)grp * Cog, /*b_off=*/0,
-               /*c_off=*/(int64_t)grp * Cog * Kg,
+               /*c_off=*/(int64_t)grp * Cog * Kgw,
                false, false, 1.0f, 0.0f, nullptr, &gb);
     } else {
       run_gemm(dy2, colT, dwk, nullptr,

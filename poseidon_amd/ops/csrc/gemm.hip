@@ -80,6 +80,7 @@ __device__ inline int fdiv_fix(int x, int d, float inv, int& rem) {
 
 template <typename T>
 __device__ inline const T* gather_addr(const GatherDesc& ga, int64_t row, int k) {
+  if (k >= ga.kg_max) return (const T*)ga.zero;  // padded columns
   int cg, kkw, ow, oh;
   int khw = fdiv_fix(k, ga.Cg, ga.inv_Cg, cg);
   int kkh = fdiv_fix(khw, ga.kw, ga.inv_kw, kkw);

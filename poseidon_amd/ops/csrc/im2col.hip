@@ -342,6 +342,16 @@ __global__ void weight_to_khwc_both_k(const TI* src, TO* dst, TO* dst_tr,
   }
 }
 
+// zero the pad columns [c_lo, ld) of a row-major matrix (colT pad strips)
+template <typename T>
+__global__ void zero_cols_k(T* m, int64_t rows, int ld, int c_lo) {
+  const int pw2 = ld - c_lo;
+  int64_t total = rows * pw2;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x)
+    m[(i / pw2) * (int64_t)ld + c_lo + i % pw2] = (T)0.0f;
+}
+
 // dgrad-as-forward weight repack: W[co][ci][kh][kw] (NCHW, per-group) ->
 // wrot[(grp*Cg + ci)][ ((KH-1-kh)*KW + (KW-1-kw)) * Cog + cog ] -- the
 // K-last B operand of the stride-1 dgrad-as-conv GEMM
@@ -576,6 +586,17 @@ void ps_weight_to_khwc_both_f32_bf16(const float* src, void* dst,
   weight_to_khwc_both_k<float, __bf16>
       <<<ew_grid((int64_t)Co * Cig * KH * KW), 256, 0, s>>>(
           src, (__bf16*)dst, (__bf16*)dst_tr, Co, Cig, KH, KW, G, ldk);
+}
+void ps_zero_cols_f32(float* m, int64_t rows, int ld, int c_lo,
+                      hipStream_t s) {
+  zero_cols_k<float>
+      <<<ew_grid(rows * (ld - c_lo)), 256, 0, s>>>(m, rows, ld, c_lo);
+}
+void ps_zero_cols_bf16(void* m, int64_t rows, int ld, int c_lo,
+                       hipStream_t s) {
+  zero_cols_k<__bf16>
+      <<<ew_grid(rows * (ld - c_lo)), 256, 0, s>>>((__bf16*)m, rows, ld,
+                                                   c_lo);
 }
 void ps_weight_to_dgrad_f32(const float* src, float* dst, int Co, int Cig,
                             int KH, int KW, int G, hipStream_t s) {

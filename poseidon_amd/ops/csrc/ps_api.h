@@ -13,6 +13,7 @@ struct GatherDesc {
   int C, H, W, Ho, Wo;
   int kh, kw, sh, sw, ph, pw;
   int Cg, c0;         // group channel count and channel offset
+  int kg_max;         // k >= kg_max reads the zero page (padded N/K)
   // f32 reciprocals for division-free index decode (exact with the +-1
   // fixup in the kernel; operands < 2^24)
   float inv_Cg, inv_kw, inv_Wo, inv_Ho;
@@ -91,7 +92,11 @@ inline void ps_pick_gemm_tile(int M, int N, int* bm_out, int* bn_out) {
 // zero-initialized when the returned stride != Kcol.
 inline int ps_colT_ld(int G, int C, int kh, int kw, int vec) {
   int Kcol = kh * kw * C;  // G*kh*kw*(C/G)
-  if (G == 1 && (C % vec) != 0 && Kcol < 64) return 64;
+  (void)vec;
+  // G==1: round every colT row up to a BK (64) multiple with zero columns
+  // -- the fwd GEMM's K becomes glds-clean (no k-tail) and the wgrad's N
+  // becomes tr16-clean (no edge strips)
+  if (G == 1) return (Kcol + 63) & ~63;
   return Kcol;
 }
 
@@ -225,6 +230,8 @@ void ps_im2col_nhwc_bf16(const void*, void*, const ConvGeom*, int ldcol,
 void ps_col2im_nhwc_bf16(const void*, void*, const ConvGeom*, hipStream_t);
 void ps_weight_to_khwc_f32(const float*, float*, int, int, int, int, hipStream_t);
 void ps_weight_to_khwc_f32_bf16(const float*, void*, int, int, int, int, hipStream_t);
+void ps_zero_cols_f32(float*, int64_t rows, int ld, int c_lo, hipStream_t);
+void ps_zero_cols_bf16(void*, int64_t rows, int ld, int c_lo, hipStream_t);
 void ps_weight_to_dgrad_f32(const float*, float*, int, int, int, int, int,
                             hipStream_t);
 void ps_weight_to_dgrad_f32_bf16(const float*, void*, int, int, int, int,
