@@ -336,10 +336,10 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
     else
       ps_im2col_nhwc_f32(x_cl.data_ptr<float>(), colT.data_ptr<float>(), &g,
                          ldc_col, stream());
-    // pad columns (ldc_col > Kcol) must be zero for the GEMMs; the
-    // rowstage/rowrun kernels already write them, the vectorized
-    // im2col_nhwc_k does not
-    if (kpad && Cg % VEC == 0) {
+    // pad columns (ldc_col > Kcol) must be zero for the GEMMs; only the
+    // rowstage kernel writes them itself -- zero unconditionally (cheap,
+    // <= 63 columns)
+    if (kpad) {
       if (bf16)
         ps_zero_cols_bf16(colT.data_ptr(), NP, ldc_col, Kcol, stream());
       else
