@@ -366,10 +366,13 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
       ga.Cg = Cg; ga.c0 = grp * Cg;
       ga.kg_max = Kg;
       ps_fill_gather_inv(&ga);
+      // wk rows are PADDED to ldc_col/G; run the GEMM over the padded K
+      // (gather k >= kg_max reads the zero page, wk pad cols are zero)
+      const int Kgi = ldc_col / G;
       run_gemm(x_cl, wk, y, bp ? bp + grp * Cog : nullptr,
-               (int)NP, Cog, Kg,
-               /*lda=*/Kg, /*ldb=*/Kg, /*ldc=*/Co,
-               /*a_off=*/0, /*b_off=*/(int64_t)grp * Cog * Kg,
+               (int)NP, Cog, Kgi,
+               /*lda=*/Kgi, /*ldb=*/Kgi, /*ldc=*/Co,
+               /*a_off=*/0, /*b_off=*/(int64_t)grp * Cog * Kgi,
                /*c_off=*/(int64_t)grp * Cog,
                true, true, 1.0f, 0.0f, &ga, nullptr, fuse_relu);
     } else {
