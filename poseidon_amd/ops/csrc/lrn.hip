@@ -12,6 +12,18 @@
 
 namespace ps {
 
+// sc^(-beta) with a fast path for the universal beta = 0.75:
+// sc^(-3/4) = rsqrt(sc) * sqrt(rsqrt(sc)) -- two hardware transcendentals
+// instead of __powf's ~20-op expansion (PMC: the v8 kernels were 66-78%
+// issue-stalled on the powf dependency chain, ~300 VALU/iteration).
+__device__ inline float lrn_pow_negbeta(float sc, float beta) {
+  if (beta == 0.75f) {
+    const float r = rsqrtf(sc);
+    return r * sqrtf(r);
+  }
+  return __powf(sc, -beta);
+}
+
 template <typename T>
 __global__ void lrn_fwd_k(const T* x, T* y, float* scale, int64_t rows, int C,
                           int size, float alpha_over_n, float beta) {
@@ -30,7 +42,7 @@ __global__ void lrn_fwd_k(const T* x, T* y, float* scale, int64_t rows, int C,
     }
     float sc = 1.0f + alpha_over_n * ss;
     scale[i] = sc;
-    from_f32(to_f32(xr[c]) * __powf(sc, -beta), y[i]);
+    from_f32(to_f32(xr[c]) * lrn_pow_negbeta(sc, beta), y[i]);
   }
 }
 
@@ -60,7 +72,7 @@ __global__ void lrn_bwd_k(const T* x, const float* scale, const T* dy,
     int c0 = max(c - post, 0), c1 = min(c + pre + 1, C);
     float acc = 0.f;
     for (int j = c0; j < c1; ++j) acc += ratio[base + j];
-    from_f32(to_f32(dy[i]) * __powf(scale[i], -beta)
+    from_f32(to_f32(dy[i]) * lrn_pow_negbeta(scale[i], beta)
                  - cache_ratio * to_f32(x[i]) * acc,
              dx[i]);
   }
@@ -101,7 +113,7 @@ __global__ void lrn_fwd_rows_k(const T* __restrict__ x, T* __restrict__ y,
       for (int j = c0; j < c1; ++j) ss += xr[j];
       const float sc = 1.0f + alpha_over_n * ss;
       scale[idx] = sc;
-      from_f32(v * __powf(sc, -beta), y[idx]);
+      from_f32(v * lrn_pow_negbeta(sc, beta), y[idx]);
     }
     __syncthreads();
   }
@@ -138,7 +150,8 @@ __global__ void lrn_bwd_rows_k(const T* __restrict__ x,
       const float* rr = rs + lr * C;
       float acc = 0.f;
       for (int j = c0; j < c1; ++j) acc += rr[j];
-      from_f32(dyv * __powf(scv, -beta) - cache_ratio * xv * acc, dx[idx]);
+      from_f32(dyv * lrn_pow_negbeta(scv, beta) - cache_ratio * xv * acc,
+               dx[idx]);
     }
     __syncthreads();
   }
@@ -159,18 +172,6 @@ static inline dim3 lrn_grid(int64_t rows, int rpb) {
 // ALU for 4 B/elt of HBM writes+reads). The row-block kernels above remain
 // the fallback (they store scale) for odd channel counts.
 // ---------------------------------------------------------------------------
-
-// sc^(-beta) with a fast path for the universal beta = 0.75:
-// sc^(-3/4) = rsqrt(sc) * sqrt(rsqrt(sc)) -- two hardware transcendentals
-// instead of __powf's ~20-op expansion (PMC: the v8 kernels were 66-78%
-// issue-stalled on the powf dependency chain, ~300 VALU/iteration).
-__device__ inline float lrn_pow_negbeta(float sc, float beta) {
-  if (beta == 0.75f) {
-    const float r = rsqrtf(sc);
-    return r * sqrtf(r);
-  }
-  return __powf(sc, -beta);
-}
 
 template <typename T> struct LrnV8;
 template <> struct LrnV8<float> {
