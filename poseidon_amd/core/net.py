@@ -109,6 +109,8 @@ class Net:
         self.output_blob_names: List[str] = []
         self._loss_tops: List = []  # (layer_idx, top_idx, weight)
         self._zero_mt = None  # multi-tensor zero table (GPU, built lazily)
+        self._repack_mt = None  # multi-tensor conv-weight repack table
+        self._repack_key = None
         # inter-branch stream parallelism (inception-style nets): built by
         # _build_stream_schedule, activated lazily on GPU
         self._lstream: List[int] = []
@@ -376,8 +378,45 @@ class Net:
         for b in owned:
             b.zero_diff()
 
+    def _maybe_mt_repack(self) -> None:
+        """One repack_mt kernel refreshes every conv's bf16 khwc shadow +
+        dgrad transpose from the fp32 masters (GoogLeNet: 2 launches
+        instead of 68 per step). Buffers are persistent per layer; the
+        table is keyed on master identities (restore() rebuilds)."""
+        if ctx().device != "cuda" or ctx().compute_dtype != torch.bfloat16:
+            return
+        convs = [l for l in self.layers
+                 if l.type_name == "CONVOLUTION" and l.blobs
+                 and l.blobs[0].data.is_cuda
+                 and l.blobs[0].data.dtype == torch.float32]
+        if not convs:
+            return
+        key = [id(l.blobs[0].data) for l in convs]
+        if self._repack_mt is None or self._repack_key != key:
+            masters, wks, wkTs, Gs = [], [], [], []
+            for l in convs:
+                w = l.blobs[0].data
+                Co, Cig, kh, kw = w.shape
+                G = l.group
+                is_1x1 = (kh == 1 and kw == 1 and l.stride == (1, 1)
+                          and l.pad == (0, 0))
+                ldc = (kh * kw * Cig * G) if is_1x1 else                     ops.conv_colT_ld(G, Cig * G, kh, kw, 8)
+                wk = torch.zeros(Co, ldc // G, dtype=torch.bfloat16,
+                                 device=w.device)
+                wkT = torch.empty(G * kh * kw * Cig, Co // G,
+                                  dtype=torch.bfloat16, device=w.device)
+                l._wk_cache = (wk, wkT)
+                masters.append(w)
+                wks.append(wk)
+                wkTs.append(wkT)
+                Gs.append(G)
+            self._repack_mt = ops.repack_mt_prepare(masters, wks, wkTs, Gs)
+            self._repack_key = key
+        ops.repack_mt_run(self._repack_mt)
+
     def forward(self, start: int = 0, end: Optional[int] = None) -> float:
         end = len(self.layers) if end is None else end
+        self._maybe_mt_repack()
         loss = 0.0
         for i in range(start, end):
             self.layers[i].forward(self.bottoms[i], self.tops[i])
@@ -390,6 +429,7 @@ class Net:
         """Forward pass without host synchronization: returns the loss as a
         0-d device tensor (sum of weighted loss tops)."""
         dev = ctx().torch_device
+        self._maybe_mt_repack()
         if self._ms_active():
             self._forward_ms()
             loss = torch.zeros((), dtype=torch.float32, device=dev)

@@ -45,6 +45,8 @@ class ConvolutionLayer(Layer):
         self.num_output = int(cp.num_output)
         self.bias_term = bool(cp.bias_term)
         self.fuse_relu = False  # set by Net's conv+ReLU fusion pass
+        self._wk_cache = None   # (wk, wkT) persistent buffers filled by the
+                                # net-level multi-tensor repack each step
         channels = bottom[0].channels
         assert channels % self.group == 0 and self.num_output % self.group == 0
 
@@ -77,7 +79,8 @@ class ConvolutionLayer(Layer):
         for bo, t in zip(bottom, top):
             y, colT = ops.conv2d_forward_ex(bo.data, w, b, self.stride,
                                             self.pad, self.group,
-                                            fuse_relu=self.fuse_relu)
+                                            fuse_relu=self.fuse_relu,
+                                            cached=self._wk_cache)
             t.data = y
             self._colT.append(colT)
 

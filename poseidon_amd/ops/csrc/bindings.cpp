@@ -273,7 +273,9 @@ Tensor weight_khwc_tr(const Tensor& w, int G, bool bf16) {
 std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
                                       const c10::optional<Tensor>& bias,
                                       int sh, int sw, int ph, int pw, int G,
-                                      bool fuse_relu) {
+                                      bool fuse_relu,
+                                      const c10::optional<Tensor>& wk_cached,
+                                      const c10::optional<Tensor>& wkT_cached) {
   check_float_like(x, "x");
   const bool bf16 = is_bf16(x);
   auto x_cl = cl4(x);
@@ -295,21 +297,30 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
   const bool kpad = ldc_col != Kcol;
 
   // fused repack: one kernel writes both the khwc fwd operand and the
-  // per-group transpose the dgrad GEMM wants (cached by the layer)
-  auto wc = w.contiguous();
-  auto wkopts = w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat);
-  Tensor wk = kpad ? at::zeros({Co, (int64_t)ldc_col}, wkopts)
-                   : at::empty({Co, (int64_t)Kg}, wkopts);
-  Tensor wkT = at::empty({(int64_t)G * kh * kw * Cig, (int64_t)(Co / G)},
-                         wkopts);
-  if (bf16)
-    ps_weight_to_khwc_both_f32_bf16(wc.data_ptr<float>(), wk.data_ptr(),
-                                    wkT.data_ptr(), Co, Cig, kh, kw, G,
-                                    ldc_col / G, stream());
-  else
-    ps_weight_to_khwc_both_f32(wc.data_ptr<float>(), wk.data_ptr<float>(),
-                               wkT.data_ptr<float>(), Co, Cig, kh, kw, G,
-                               ldc_col / G, stream());
+  // per-group transpose the dgrad GEMM wants (cached by the layer). When
+  // the layer provides PRE-REPACKED buffers (the net-level multi-tensor
+  // repack, repack_mt_run), skip the per-layer kernel entirely.
+  Tensor wk, wkT;
+  if (wk_cached.has_value() && wkT_cached.has_value() &&
+      wk_cached->size(1) == ldc_col / G &&
+      wk_cached->scalar_type() == (bf16 ? at::kBFloat16 : at::kFloat)) {
+    wk = *wk_cached;
+    wkT = *wkT_cached;
+  } else {
+    auto wc = w.contiguous();
+    auto wkopts = w.options().dtype(bf16 ? at::kBFloat16 : at::kFloat);
+    wk = kpad ? at::zeros({Co, (int64_t)ldc_col}, wkopts)
+              : at::empty({Co, (int64_t)Kg}, wkopts);
+    wkT = at::empty({(int64_t)G * kh * kw * Cig, (int64_t)(Co / G)}, wkopts);
+    if (bf16)
+      ps_weight_to_khwc_both_f32_bf16(wc.data_ptr<float>(), wk.data_ptr(),
+                                      wkT.data_ptr(), Co, Cig, kh, kw, G,
+                                      ldc_col / G, stream());
+    else
+      ps_weight_to_khwc_both_f32(wc.data_ptr<float>(), wk.data_ptr<float>(),
+                                 wkT.data_ptr<float>(), Co, Cig, kh, kw, G,
+                                 ldc_col / G, stream());
+  }
 
   bool is_1x1 = (kh == 1 && kw == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0);
   // implicit GEMM: gather im2col rows inside the GEMM staging whenever the
@@ -1167,6 +1178,48 @@ void zero_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
              stream());
 }
 
+struct MTRepackDescHost {
+  const float* src;
+  void* wk;
+  void* wkT;
+  int64_t n;
+  int Co, Cig, kh, kw, G, ldk;
+};
+
+std::vector<Tensor> repack_mt_prepare(std::vector<Tensor> masters,
+                                      std::vector<Tensor> wks,
+                                      std::vector<Tensor> wkTs,
+                                      std::vector<int64_t> Gs) {
+  const int CHUNK = ps_mt_chunk_elts();
+  const size_t nt = masters.size();
+  std::vector<MTRepackDescHost> descs(nt);
+  std::vector<MTChunkHost> chunks;
+  for (size_t t = 0; t < nt; ++t) {
+    const Tensor& m = masters[t];
+    TORCH_CHECK(m.is_cuda() && m.is_contiguous() &&
+                m.scalar_type() == at::kFloat && m.dim() == 4 &&
+                wks[t].scalar_type() == at::kBFloat16 &&
+                wkTs[t].scalar_type() == at::kBFloat16,
+                "repack_mt_prepare: f32 NCHW masters + bf16 outputs");
+    descs[t] = {m.data_ptr<float>(), wks[t].data_ptr(), wkTs[t].data_ptr(),
+                m.numel(), (int)m.size(0), (int)m.size(1), (int)m.size(2),
+                (int)m.size(3), (int)Gs[t], (int)wks[t].size(1)};
+    for (int64_t off = 0; off < m.numel(); off += CHUNK)
+      chunks.push_back({(int)t, off});
+  }
+  return {blob_to_dev(descs.data(), nt * sizeof(MTRepackDescHost),
+                      masters[0]),
+          blob_to_dev(chunks.data(), chunks.size() * sizeof(MTChunkHost),
+                      masters[0]),
+          at::scalar_tensor((int64_t)chunks.size())};
+}
+
+void repack_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
+                   int64_t nchunks) {
+  ps_repack_mt(desc_dev.data_ptr(), chunk_dev.data_ptr(), (int)nchunks,
+               stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1213,6 +1266,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_mt_run", &sgd_mt_run);
   m.def("zero_mt_prepare", &zero_mt_prepare);
   m.def("zero_mt_run", &zero_mt_run);
+  m.def("repack_mt_prepare", &repack_mt_prepare);
+  m.def("repack_mt_run", &repack_mt_run);
+  m.def("colT_ld", [](int64_t G, int64_t C, int64_t kh, int64_t kw,
+                      int64_t vec) {
+    return (int64_t)ps_colT_ld((int)G, (int)C, (int)kh, (int)kw, (int)vec);
+  });
   m.def("dropout_forward_offdev", &dropout_forward_offdev);
   m.def("nesterov_update", &nesterov_update);
   m.def("adagrad_update", &adagrad_update);

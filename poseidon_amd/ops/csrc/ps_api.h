@@ -212,6 +212,8 @@ void ps_sgd_mt(const void* descs, const void* chunks, int nchunks, float lr,
                float mom, const float* lr_dev, hipStream_t);
 void ps_zero_mt(const void* descs, const void* chunks, int nchunks,
                 hipStream_t);
+void ps_repack_mt(const void* descs, const void* chunks, int nchunks,
+                  hipStream_t);
 void ps_dropout_fwd_f32_offdev(const float*, float*, uint8_t*, int64_t, float,
                                uint64_t, const void*, hipStream_t);
 void ps_dropout_fwd_bf16_offdev(const void*, void*, uint8_t*, int64_t, float,

@@ -129,6 +129,39 @@ struct MTZeroDesc {
   int64_t n;
 };
 
+// Multi-tensor conv weight repack: one launch refreshes EVERY conv's bf16
+// khwc shadow + per-group dgrad transpose from the fp32 masters
+// (GoogLeNet: 2 launches instead of 68 x ~5 us repack kernels per step).
+struct MTRepackDesc {
+  const float* src;  // fp32 master, NCHW [Co][Cig][kh][kw]
+  void* wk;          // bf16 [Co][ldk] khwc (pad cols pre-zeroed)
+  void* wkT;         // bf16 [G*Kg][Cog]
+  int64_t n;         // Co*Cig*kh*kw
+  int Co, Cig, kh, kw, G, ldk;
+};
+
+__global__ void repack_mt_k(const MTRepackDesc* __restrict__ descs,
+                            const MTChunk* __restrict__ chunks) {
+  const MTChunk ck = chunks[blockIdx.x];
+  const MTRepackDesc d = descs[ck.t];
+  const int Cog = d.Co / d.G;
+  const int Kg = d.kh * d.kw * d.Cig;
+  const int KW = d.kw, KH = d.kh, Cig = d.Cig;
+  for (int64_t i = ck.off + threadIdx.x;
+       i < ck.off + MT_CHUNK && i < d.n; i += 256) {
+    int kkw = (int)(i % KW);
+    int64_t t = i / KW;
+    int kkh = (int)(t % KH); t /= KH;
+    int ci = (int)(t % Cig);
+    int co = (int)(t / Cig);
+    const float v = d.src[i];
+    const int kg = (kkh * KW + kkw) * Cig + ci;
+    ((__bf16*)d.wk)[(int64_t)co * d.ldk + kg] = (__bf16)v;
+    ((__bf16*)d.wkT)[((int64_t)(co / Cog) * Kg + kg) * Cog + co % Cog] =
+        (__bf16)v;
+  }
+}
+
 __global__ void zero_mt_k(const MTZeroDesc* __restrict__ descs,
                           const MTChunk* __restrict__ chunks) {
   const MTChunk ck = chunks[blockIdx.x];
@@ -192,6 +225,13 @@ void ps_sgd_mt(const void* descs, const void* chunks, int nchunks, float lr,
   else
     sgd_mt_k<false><<<dim3((unsigned)nchunks), 256, 0, s>>>(
         (const MTDesc*)descs, (const MTChunk*)chunks, lr, mom, nullptr);
+}
+
+void ps_repack_mt(const void* descs, const void* chunks, int nchunks,
+                  hipStream_t s) {
+  if (nchunks <= 0) return;
+  repack_mt_k<<<dim3((unsigned)nchunks), 256, 0, s>>>(
+      (const MTRepackDesc*)descs, (const MTChunk*)chunks);
 }
 
 void ps_zero_mt(const void* descs, const void* chunks, int nchunks,

@@ -86,14 +86,15 @@ def pool_out_size(h: int, k: int, p: int, s: int) -> Tuple[int, bool]:
 
 def conv2d_forward_ex(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor],
                       stride: Tuple[int, int], pad: Tuple[int, int],
-                      groups: int, fuse_relu: bool = False):
+                      groups: int, fuse_relu: bool = False, cached=None):
     """Returns (y, colT_cache). colT is the im2col matrix on GPU (reused by
     the backward GEMMs); None on CPU. fuse_relu clamps the output in the
     GEMM epilogue (used by the Net-level conv+ReLU fusion pass)."""
     if x.is_cuda:
+        wk_c, wkT_c = cached if cached is not None else (None, None)
         y, colT, wkT = _ext().conv2d_forward_ex(x, w, b, stride[0], stride[1],
                                                 pad[0], pad[1], groups,
-                                                fuse_relu)
+                                                fuse_relu, wk_c, wkT_c)
         return y, (colT, wkT)
     y = F.conv2d(x, w, b, stride=stride, padding=pad, groups=groups)
     if fuse_relu:
@@ -505,6 +506,21 @@ def sgd_mt_run(mt, lr: float, momentum: float, lr_dev=None) -> None:
     desc, chunk, nchunks = mt
     _ext().sgd_mt_run(desc, chunk, nchunks, float(lr), float(momentum),
                       lr_dev)
+
+
+def conv_colT_ld(G: int, C: int, kh: int, kw: int, vec: int) -> int:
+    return int(_ext().colT_ld(G, C, kh, kw, vec))
+
+
+def repack_mt_prepare(masters, wks, wkTs, Gs):
+    d, c, n = _ext().repack_mt_prepare(list(masters), list(wks), list(wkTs),
+                                       [int(g) for g in Gs])
+    return d, c, int(n.item())
+
+
+def repack_mt_run(mt) -> None:
+    d, c, n = mt
+    _ext().repack_mt_run(d, c, n)
 
 
 def zero_mt_prepare(tensors):
