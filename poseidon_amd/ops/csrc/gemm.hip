@@ -613,9 +613,6 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
 // 16B-aligned bases, bf16 in / f32 out, no gathers, no bias, beta == 0.
 // ---------------------------------------------------------------------------
 
-#define TRBM 64
-#define TRBN 64
-
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
 template <int COLS, bool GATHER = false>
