@@ -164,3 +164,24 @@ def test_bf16_training_converges():
         assert last < first * 0.75, (first, last)
     finally:
         pa.init(device="cpu", compute_dtype=torch.float32)
+
+
+@pytest.mark.parametrize("M,N,K", [
+    (64, 576, 25088),    # tr16 split-K (VGG conv1_2 wgrad class)
+    (128, 832, 6400),    # tr16, 64x128 tile
+    (96, 640, 4096),     # tr16 thin-M 32-row tile
+    (130, 577, 8192),    # tr16 interior + generic edge strips? under floor -> generic
+])
+def test_gemm_bf16_tn_tr16(M, N, K):
+    """TN bf16->f32 shapes routed through the ds_read_b64_tr_b16 kernel
+    (gemm.hip try_gemm_tn_tr) vs the fp32 torch reference."""
+    from poseidon_amd.ops._backend import load
+    ext = load()
+    opA = rnd(M, K, seed=5) * 0.3
+    opB = rnd(K, N, seed=6) * 0.3
+    A = opA.t().contiguous().to(DEV, torch.bfloat16)  # [K, M] K-major
+    B = opB.contiguous().to(DEV, torch.bfloat16)      # [K, N] K-major
+    out = ext.gemm(A, B, M, N, K, False, False)
+    assert out.dtype == torch.float32
+    ref_q = opA.to(torch.bfloat16).float() @ opB.to(torch.bfloat16).float()
+    close_bf16(out, ref_q, rtol=2e-2, what=f"tn tr16 {M}x{N}x{K}")
