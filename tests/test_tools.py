@@ -117,3 +117,24 @@ def test_dump_prototxt_roundtrip(tmp_path):
     from poseidon_amd.core.net import Net, TRAIN
     n = Net(net, phase=TRAIN)
     assert any(l.name == "fc6" for l in n.layers)
+
+
+def test_profile_summary_tool(tmp_path):
+    """profile_summary parses a rocpd-schema sqlite DB."""
+    import sqlite3
+    db = tmp_path / "x_results.db"
+    con = sqlite3.connect(str(db))
+    con.execute("create table rocpd_kernel_dispatch_ab (id int, kernel_id int,"
+                " start int, [end] int)")
+    con.execute("create table rocpd_info_kernel_symbol_ab (id int,"
+                " display_name text)")
+    con.execute("insert into rocpd_info_kernel_symbol_ab values (1, 'k1'),"
+                " (2, 'k2')")
+    for i in range(4):
+        con.execute("insert into rocpd_kernel_dispatch_ab values (?, ?, 0, ?)",
+                    (i, 1 + i % 2, 1000 * (i + 1)))
+    con.commit()
+    from poseidon_amd.tools.profile_summary import summarize
+    rows, total = summarize(str(db))
+    assert len(rows) == 2
+    assert abs(total - 0.01) < 1e-9  # 10,000 ns = 0.01 ms
