@@ -128,3 +128,44 @@ def test_two_rank_training_matches_reference(tmp_path, use_sfb):
         assert torch.allclose(r0[i], ref[i], atol=1e-5), \
             f"param {i}: distributed != reference (max err " \
             f"{(r0[i]-ref[i]).abs().max():.3g})"
+
+
+def _ckpt_worker(rank, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = PORT
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(WORLD)
+    import poseidon_amd as pa
+    from poseidon_amd.solver.solver import SGDSolver
+    pa.init(device="cpu", rank=rank, world_size=WORLD, seed=42)
+    sp = _solver_param()
+    sp.snapshot_prefix = os.path.join(out_dir, "ck")
+    sp.net_param = _net_param(8)
+    solver = SGDSolver(sp, use_sfb=False, verbose=False)
+    data, labels = _dataset()
+    solver.net.layers[0].add_data(data[rank::WORLD], labels[rank::WORLD])
+    solver.step(3)
+    if rank == 0:
+        solver.snapshot()          # rank 0 writes caffemodel + its state
+    import torch.distributed as dist
+    dist.barrier()
+    # BOTH ranks resume; rank 1 has no own solverstate file and must fall
+    # back to rank 0's (solver.cpp:670-696 thread-0 fallback semantics)
+    solver2 = SGDSolver(sp, use_sfb=False, verbose=False)
+    solver2.net.layers[0].add_data(data[rank::WORLD], labels[rank::WORLD])
+    solver2.restore(os.path.join(out_dir, "ck_iter_3.solverstate.0.0"))
+    assert solver2.iter == 3
+    own = [i for i, ps in enumerate(solver2.net.params) if ps.owner == i]
+    ref = {i: solver.net.params[i].blob.data for i in own}
+    for i in own:
+        assert torch.allclose(solver2.net.params[i].blob.data, ref[i],
+                              atol=1e-6), f"restore mismatch p{i} rank{rank}"
+    solver2.step(2)  # resumes training without error
+    dist.destroy_process_group()
+
+
+def test_two_rank_snapshot_restore(tmp_path):
+    global PORT
+    PORT = str(29791)
+    mp.start_processes(_ckpt_worker, args=(str(tmp_path),),
+                       nprocs=WORLD, join=True, start_method="spawn")
