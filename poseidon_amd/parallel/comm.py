@@ -99,9 +99,12 @@ class GradReducer:
     After wait(), every owned param's .diff holds the SUM over ranks.
     """
 
-    def __init__(self, net, bucket_bytes: int = 25 << 20):
+    def __init__(self, net, bucket_bytes: int = 0):
         self.net = net
-        self.bucket_bytes = bucket_bytes
+        # default 25 MB; tune per fabric with PS_BUCKET_MB (xGMI ring
+        # bandwidth is per-link, so bigger buckets amortize latency while
+        # smaller ones start overlapping earlier)
+        self.bucket_bytes = bucket_bytes or             (int(os.environ.get("PS_BUCKET_MB", "25")) << 20)
         c = ctx()
         self.enabled = c.distributed and dist.is_initialized()
         self.use_stream = c.device == "cuda" and self.enabled
