@@ -1,4 +1,10 @@
-"""Protobuf schema + wire/text codecs for the caffe.proto message set."""
+"""Protobuf schema + wire/text codecs for the caffe.proto message set.
+
+Parity: /root/reference/src/caffe/util/io.cpp:38-77
+(ReadProtoFromTextFile / WriteProtoToTextFile / ReadProtoFromBinaryFile /
+WriteProtoToBinaryFile, incl. the reference's large coded-stream limits --
+irrelevant here, the codec is stream-size-agnostic) over the schema in
+spec.py (field numbers byte-compatible with src/caffe/proto/caffe.proto)."""
 
 from . import spec
 from .message import Message
