@@ -19,6 +19,7 @@ import torch
 
 DEFAULT_BATCH = {  # per-GPU batch, matching the reference's training configs
     "alexnet": 256,   # models/bvlc_alexnet/train_val.prototxt:10
+    "caffenet": 256,  # models/bvlc_reference_caffenet/train_val.prototxt
     "googlenet": 32,  # models/bvlc_googlenet/train_test.prototxt:9
     "vgg16": 32,
     "cifar10_quick": 100,
@@ -113,7 +114,8 @@ def main():
             "config": {
                 "model": args.model,
                 "global_batch": batch * n_gpus,
-                "input": {"alexnet": "3x227x227", "googlenet": "3x224x224",
+                "input": {"alexnet": "3x227x227", "caffenet": "3x227x227",
+                          "googlenet": "3x224x224",
                           "vgg16": "3x224x224", "cifar10_quick": "3x32x32",
                           "lenet": "1x28x28"}[args.model],
                 "parallelism": f"dp{n_gpus}",

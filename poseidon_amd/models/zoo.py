@@ -209,6 +209,41 @@ def alexnet(batch: int = 256, num_classes: int = 1000) -> Message:
     return b.build()
 
 
+def caffenet(batch: int = 256, num_classes: int = 1000) -> Message:
+    """CaffeNet (models/bvlc_reference_caffenet/train_val.prototxt): the
+    AlexNet variant the reference also ships -- pooling BEFORE the LRN
+    (pool1 -> norm1) instead of AlexNet's norm -> pool order."""
+    b = NetBuilder("CaffeNet")
+    b.synthetic_images(batch, 3, 227, num_classes)
+    b.conv("conv1", "data", "conv1", 96, 11, stride=4, w_std=0.01, bias=0.0)
+    b.relu("relu1", "conv1")
+    b.pool("pool1", "conv1", "pool1", "MAX", 3, 2)
+    b.lrn("norm1", "pool1", "norm1")
+    b.conv("conv2", "norm1", "conv2", 256, 5, pad=2, group=2, w_std=0.01,
+           bias=0.1)
+    b.relu("relu2", "conv2")
+    b.pool("pool2", "conv2", "pool2", "MAX", 3, 2)
+    b.lrn("norm2", "pool2", "norm2")
+    b.conv("conv3", "norm2", "conv3", 384, 3, pad=1, w_std=0.01)
+    b.relu("relu3", "conv3")
+    b.conv("conv4", "conv3", "conv4", 384, 3, pad=1, group=2, w_std=0.01,
+           bias=0.1)
+    b.relu("relu4", "conv4")
+    b.conv("conv5", "conv4", "conv5", 256, 3, pad=1, group=2, w_std=0.01,
+           bias=0.1)
+    b.relu("relu5", "conv5")
+    b.pool("pool5", "conv5", "pool5", "MAX", 3, 2)
+    b.ip("fc6", "pool5", "fc6", 4096, w_std=0.005, bias=0.1)
+    b.relu("relu6", "fc6")
+    b.dropout("drop6", "fc6", 0.5)
+    b.ip("fc7", "fc6", "fc7", 4096, w_std=0.005, bias=0.1)
+    b.relu("relu7", "fc7")
+    b.dropout("drop7", "fc7", 0.5)
+    b.ip("fc8", "fc7", "fc8", num_classes, w_std=0.01)
+    b.softmax_loss("loss", "fc8", "label")
+    return b.build()
+
+
 def _inception(b: NetBuilder, name: str, bottom: str, c1, c3r, c3, c5r, c5, cp):
     b.conv(f"{name}/1x1", bottom, f"{name}/1x1", c1, 1, w_type="xavier")
     b.relu(f"{name}/relu_1x1", f"{name}/1x1")
@@ -315,6 +350,7 @@ MODEL_ZOO = {
     "lenet": lenet,
     "cifar10_quick": cifar10_quick,
     "alexnet": alexnet,
+    "caffenet": caffenet,
     "googlenet": googlenet,
     "vgg16": vgg16,
 }
