@@ -111,6 +111,7 @@ class Net:
         self._zero_mt = None  # multi-tensor zero table (GPU, built lazily)
         self._repack_mt = None  # multi-tensor conv-weight repack table
         self._repack_key = None
+        self._loss_marks: Dict[int, List] = {}
         # inter-branch stream parallelism (inception-style nets): built by
         # _build_stream_schedule, activated lazily on GPU
         self._lstream: List[int] = []
@@ -241,6 +242,8 @@ class Net:
                 self.layer_need_bwd[li] = False
 
         self._fuse_relu_epilogues()
+        for (li, ti, w) in self._loss_tops:
+            self._loss_marks.setdefault(li, []).append((ti, w))
 
     def _build_stream_schedule(self, n_streams: int = 4) -> None:
         """Dataflow schedule for inter-branch stream parallelism.
@@ -437,13 +440,15 @@ class Net:
                 loss = loss + w * self.tops[li][ti].data.sum().to(torch.float32)
             return loss
         loss = torch.zeros((), dtype=torch.float32, device=dev)
-        marks = {}
-        for (li, ti, w) in self._loss_tops:
-            marks.setdefault(li, []).append((ti, w))
-        for i, layer in enumerate(self.layers):
-            layer.forward(self.bottoms[i], self.tops[i])
-            for (ti, w) in marks.get(i, []):
-                loss = loss + w * self.tops[i][ti].data.sum().to(torch.float32)
+        marks = self._loss_marks
+        layers = self.layers
+        bottoms = self.bottoms
+        tops = self.tops
+        for i in range(len(layers)):
+            layers[i].forward(bottoms[i], tops[i])
+            if i in marks:
+                for (ti, w) in marks[i]:
+                    loss = loss + w * tops[i][ti].data.sum().to(torch.float32)
         return loss
 
     def _forward_ms(self) -> None:
