@@ -10,6 +10,8 @@ Layout (little-endian):
 
 from __future__ import annotations
 
+import os
+import shutil
 import struct
 from typing import Iterator, List
 
@@ -22,28 +24,38 @@ _VERSION = 1
 
 
 class PDBWriter:
+    """Streams records to a sidecar file and assembles header + offset
+    index at close -- O(nrecords) memory for the index only, so
+    ImageNet-scale conversions do not buffer the dataset in RAM (the
+    reference streamed into LevelDB/LMDB batches the same way)."""
+
     def __init__(self, path: str):
         self.path = path
-        self._records: List[bytes] = []
+        self._tmp_path = path + ".tmp"
+        self._tmp = open(self._tmp_path, "wb")
+        self._sizes: List[int] = []
 
     def put(self, datum: Message) -> None:
-        self._records.append(datum.encode())
+        self.put_raw(datum.encode())
 
     def put_raw(self, raw: bytes) -> None:
-        self._records.append(raw)
+        self._tmp.write(raw)
+        self._sizes.append(len(raw))
 
     def close(self) -> None:
-        n = len(self._records)
+        self._tmp.close()
+        n = len(self._sizes)
         header = _MAGIC + struct.pack("<IQ", _VERSION, n)
         base = len(header) + 8 * (n + 1)
         offsets = [base]
-        for r in self._records:
-            offsets.append(offsets[-1] + len(r))
+        for sz in self._sizes:
+            offsets.append(offsets[-1] + sz)
         with open(self.path, "wb") as f:
             f.write(header)
             f.write(struct.pack(f"<{n + 1}Q", *offsets))
-            for r in self._records:
-                f.write(r)
+            with open(self._tmp_path, "rb") as src:
+                shutil.copyfileobj(src, f, 4 << 20)
+        os.remove(self._tmp_path)
 
     def __enter__(self):
         return self
