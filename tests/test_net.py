@@ -184,3 +184,50 @@ def test_relu_fusion_matches_unfused():
     assert np.isclose(l1, l0)
     for a, b in zip(g1, g0):
         assert torch.allclose(a, b), "fused/unfused gradients differ"
+
+
+def test_state_rules_levels_and_stages():
+    """NetStateRule level/stage/not_stage matching (net.cpp:423-468)."""
+    from poseidon_amd.core.net import state_meets_rule
+    from poseidon_amd.proto import Message
+
+    st = Message("NetState", phase=0, level=2)
+    st.stage.append("deploy")
+    st.stage.append("quantized")
+
+    r = Message("NetStateRule", min_level=1, max_level=3)
+    assert state_meets_rule(st, r)
+    r = Message("NetStateRule", min_level=3)
+    assert not state_meets_rule(st, r)
+    r = Message("NetStateRule", max_level=1)
+    assert not state_meets_rule(st, r)
+
+    r = Message("NetStateRule")
+    r.stage.append("deploy")
+    assert state_meets_rule(st, r)
+    r.stage.append("missing")
+    assert not state_meets_rule(st, r)
+
+    r = Message("NetStateRule")
+    r.not_stage.append("quantized")
+    assert not state_meets_rule(st, r)
+    r = Message("NetStateRule")
+    r.not_stage.append("other")
+    assert state_meets_rule(st, r)
+
+    r = Message("NetStateRule", phase=1)
+    assert not state_meets_rule(st, r)
+
+
+def test_filter_net_rejects_include_and_exclude():
+    import pytest as _pt
+    from poseidon_amd.core.net import filter_net
+    from poseidon_amd.proto import parse_text, Message
+    np_ = parse_text("NetParameter", """
+        name: "x"
+        layers { name: "d" type: DUMMY_DATA top: "d"
+                 dummy_data_param { num: 1 channels: 1 height: 1 width: 1 }
+                 include { phase: TRAIN } exclude { phase: TEST } }
+    """)
+    with _pt.raises(ValueError, match="include OR exclude"):
+        filter_net(np_, Message("NetState", phase=0))
