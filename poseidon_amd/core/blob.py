@@ -138,20 +138,34 @@ class Blob:
 
     def from_proto(self, p: Message, reshape: bool = True) -> None:
         shape = (p.num, p.channels, p.height, p.width)
-        if reshape:
-            self._shape = shape
-            self._data = None
-            self._diff = None
+        if reshape and self._shape != shape:
+            if self.count == shape[0] * shape[1] * shape[2] * shape[3]:
+                self.reshape(shape)  # same count: keeps storage (and sharers)
+            else:
+                self._shape = shape
+                self._data = None
+                self._diff = None
         arr = np.asarray(p.data, dtype=np.float32)
         if arr.size != self.count:
             raise ValueError(
                 f"BlobProto data count {arr.size} != blob count {self.count}")
         t = torch.from_numpy(arr.copy()).view(self._shape)
-        self._data = t.to(device=self.device, dtype=self.dtype)
+        # Copy IN PLACE when storage already exists: nets that shared this
+        # blob (test nets via _share_params, name-shared params) must keep
+        # seeing the same tensor, exactly like the reference's
+        # Blob::FromProto memcpy into mutable_cpu_data (blob.cpp:399-426).
+        # Rebinding self._data would silently detach every sharer.
+        if self._data is not None and self._data.numel() == arr.size:
+            self._data.copy_(t.view(self._data.shape))
+        else:
+            self._data = t.to(device=self.device, dtype=self.dtype)
         d = np.asarray(p.diff, dtype=np.float32)
         if d.size == self.count:
-            self._diff = torch.from_numpy(d.copy()).view(self._shape).to(
-                device=self.device, dtype=torch.float32)
+            dt = torch.from_numpy(d.copy()).view(self._shape)
+            if self._diff is not None and self._diff.numel() == d.size:
+                self._diff.copy_(dt.view(self._diff.shape))
+            else:
+                self._diff = dt.to(device=self.device, dtype=torch.float32)
 
     def __repr__(self) -> str:
         return f"Blob({self.name or '?'}, shape={self._shape}, dtype={self.dtype})"

@@ -4,8 +4,11 @@ ContrastiveLoss, Accuracy.
 
 Parity: /root/reference/src/caffe/layers/*_loss_layer.{cpp,cu},
 softmax_layer.cu, accuracy_layer.cpp. Losses are normalized by batch size
-(num), matching Caffe, so distributed data parallelism averages gradients
-across ranks (parallel/comm.py divides the all-reduced sum by world size).
+(num), matching Caffe. Under data parallelism the gradients are SUMMED
+across ranks (parallel/comm.GradReducer all-reduces with SUM -- the same
+semantics as N workers each BatchInc-ing the PS table); only display/test
+metrics are averaged. Effective lr therefore scales with world size exactly
+as it does for the reference's multi-worker runs.
 """
 
 from __future__ import annotations

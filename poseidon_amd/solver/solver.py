@@ -326,11 +326,18 @@ class SGDSolver:
     # ------------------------------------------------------------------
     def _display(self, loss: float, rate: float) -> None:
         c = ctx()
+        # Flatten WHOLE output blobs into the row (reference writes every
+        # element of every net output into the net-output PS table,
+        # solver.cpp:336-366) -- vector outputs get name_<j> columns.
         outs = []
         for name in sorted(self.net.output_blob_names):
             b = self.net.blobs[name]
-            if b.count == 1:
-                outs.append((name, float(b.data.reshape(()).item())))
+            flat = b.data.detach().reshape(-1).to(torch.float64).cpu()
+            if flat.numel() == 1:
+                outs.append((name, float(flat[0])))
+            else:
+                for j in range(flat.numel()):
+                    outs.append((f"{name}_{j}", float(flat[j])))
         vals = torch.tensor([loss] + [v for _, v in outs], dtype=torch.float64)
         if self.distributed:
             comm.allreduce_metrics(vals)
@@ -360,13 +367,17 @@ class SGDSolver:
         my_iters = total_iters // c.world_size
         if c.rank < total_iters % c.world_size:
             my_iters += 1
-        sums: Dict[str, float] = {}
+        # Metric names come from the net topology, NOT from which ranks ran
+        # iterations: a rank with my_iters == 0 (world_size > test_iter)
+        # must still contribute an identically-shaped zero tensor to the
+        # all-reduce or the collective hangs/corrupts.
+        names = sorted(tn.output_blob_names)
+        sums: Dict[str, float] = {n: 0.0 for n in names}
         for _ in range(my_iters):
             tn.forward()
-            for name in tn.output_blob_names:
+            for name in names:
                 b = tn.blobs[name]
                 sums[name] = sums.get(name, 0.0) + float(b.data.sum().item())
-        names = sorted(sums)
         vals = torch.tensor([sums[n] for n in names] or [0.0],
                             dtype=torch.float64)
         if self.distributed:
@@ -410,19 +421,38 @@ class SGDSolver:
     def restore(self, state_file: str) -> None:
         c = ctx()
         if not os.path.exists(state_file):
-            # fall back to rank 0's file (reference falls back to thread 0)
-            base = state_file.rsplit(".", 2)[0]
-            state_file = f"{base}.0.0"
+            # Resolve the conventional suffix-less path first: the snapshot
+            # writer emits '<prefix>_iter_N.solverstate.<rank>.0', so try
+            # appending our rank / rank 0 before assuming state_file already
+            # carries a numeric suffix (reference falls back to thread 0,
+            # solver.cpp:670-696).
+            for cand in (f"{state_file}.{c.rank}.0", f"{state_file}.0.0",
+                         f"{state_file.rsplit('.', 2)[0]}.0.0"):
+                if os.path.exists(cand):
+                    state_file = cand
+                    break
         state = read_proto_binary(state_file, "SolverState")
         self.iter = int(state.iter)
         hist_protos = list(state.history)
         own = [i for i, ps in enumerate(self.net.params)
                if ps.owner == i and i in self.history]
+        # History blobs pair with owned params positionally (the reference
+        # keys by order too) -- but refuse a silent mispairing when the
+        # prototxt changed between save and load (solver.cpp:1004-1008).
+        if len(hist_protos) != len(own):
+            raise ValueError(
+                f"solverstate carries {len(hist_protos)} history blobs but "
+                f"net has {len(own)} learnable params -- prototxt mismatch?")
         for i, hp in zip(own, hist_protos):
-            hb = Blob((), device=self.history[i].device)
+            want = self.history[i]
+            if hp.num * hp.channels * hp.height * hp.width != want.numel():
+                raise ValueError(
+                    f"history blob for param {i} has "
+                    f"{hp.num * hp.channels * hp.height * hp.width} elements, "
+                    f"net param has {want.numel()}")
+            hb = Blob((), device=want.device)
             hb.from_proto(hp)
-            self.history[i] = hb.data.view(self.history[i].shape).to(
-                self.history[i].dtype)
+            self.history[i] = hb.data.view(want.shape).to(want.dtype)
         if state.has("learned_net") and os.path.exists(state.learned_net):
             net_proto = read_proto_binary(state.learned_net, "NetParameter")
             self.net.copy_trained_layers_from(net_proto)

@@ -168,3 +168,80 @@ def test_test_net_evaluation():
     res = solver.test(0)
     assert res["acc"] > 0.9, res
     assert res["loss"] < 0.3, res
+
+
+def test_restore_preserves_test_net_sharing(tmp_path):
+    """restore()/load_weights() must copy weights IN PLACE
+    so test nets sharing storage via _share_params keep seeing the train
+    net's tensors (reference Blob::FromProto memcpy, blob.cpp:399-426)."""
+    pa.init(device="cpu", seed=23)
+    sp = _solver_param(snapshot_prefix=str(tmp_path / "share"))
+    sp.test_iter.append(2)
+    sp.test_interval = 10**9
+    sp.test_initialization = False
+    sp.net_param = _separable_net_param()
+    solver = SGDSolver(sp, verbose=False)
+    data, labels = _toy_dataset(n=32)
+    solver.net.layers[0].add_data(data, labels)
+    solver.test_nets[0].layers[0].add_data(data, labels)
+    solver.step(10)
+    solver.snapshot()
+
+    solver2 = SGDSolver(sp, verbose=False)
+    solver2.net.layers[0].add_data(data, labels)
+    solver2.test_nets[0].layers[0].add_data(data, labels)
+    solver2.restore(str(tmp_path / "share") + "_iter_10.solverstate")
+    assert solver2.iter == 10  # suffix-less path resolved (ADVICE low)
+    # same storage objects after restore
+    tr = _ip1(solver2.net).blobs[0]
+    te = next(l for l in solver2.test_nets[0].layers
+              if l.name == "ip1").blobs[0]
+    assert tr.data.data_ptr() == te.data.data_ptr()
+    # and stays shared through further training
+    solver2.step(5)
+    assert torch.equal(tr.data, te.data)
+
+
+def test_restore_rejects_mismatched_history(tmp_path):
+    pa.init(device="cpu", seed=29)
+    sp = _solver_param(snapshot_prefix=str(tmp_path / "mm"))
+    sp.net_param = _separable_net_param()
+    solver = SGDSolver(sp, verbose=False)
+    data, labels = _toy_dataset(n=32)
+    solver.net.layers[0].add_data(data, labels)
+    solver.step(3)
+    solver.snapshot()
+
+    sp2 = _solver_param(snapshot_prefix=str(tmp_path / "mm"))
+    sp2.net_param = _separable_net_param(classes=7)  # changed prototxt
+    solver2 = SGDSolver(sp2, verbose=False)
+    with pytest.raises(ValueError):
+        solver2.restore(str(tmp_path / "mm") + "_iter_3.solverstate.0.0")
+
+
+def test_netoutputs_flattens_vector_outputs(tmp_path):
+    """.netoutputs must carry every element of every net
+    output (solver.cpp:336-366), not only scalar outputs."""
+    pa.init(device="cpu", seed=31)
+    np_param = parse_text("NetParameter", """
+        name: "vec"
+        layers { name: "data" type: DUMMY_DATA top: "data"
+                 dummy_data_param { num: 4 channels: 3 height: 1 width: 1
+                     data_filler { type: "gaussian" std: 1.0 } } }
+        layers { name: "ip" type: INNER_PRODUCT bottom: "data" top: "ip"
+                 inner_product_param { num_output: 3
+                     weight_filler { type: "xavier" } } }
+        layers { name: "sm" type: SOFTMAX bottom: "ip" top: "sm" }
+    """)
+    sp = _solver_param(max_iter=2, display=1)
+    sp.net_param = np_param
+    solver = SGDSolver(sp, verbose=False)
+    solver.net.forward()
+    solver._display(0.5, 0.1)
+    # 'sm' output blob has 4*3 = 12 elements -> sm_0..sm_11 columns
+    cols = solver._net_outputs_cols
+    assert "sm_0" in cols and "sm_11" in cols, cols
+    out = tmp_path / "vec"
+    solver.write_net_outputs(str(out))
+    header = (tmp_path / "vec.netoutputs").read_text().splitlines()[0]
+    assert header.startswith("iter,time,loss,") and "sm_5" in header
