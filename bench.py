@@ -30,8 +30,8 @@ DEFAULT_BATCH = {  # per-GPU batch, matching the reference's training configs
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--model", default="alexnet",
                     choices=list(DEFAULT_BATCH))
     ap.add_argument("--batch", type=int, default=0, help="per-GPU batch")
@@ -73,9 +73,13 @@ def main():
     sp.net_param = zoo.build_net(args.model, batch=batch)
     solver = SGDSolver(sp, use_sfb=not args.no_sfb, verbose=False)
     graphed = False
-    if not args.no_graph and device == "cuda" and n_gpus == 1:
-        # (multi-rank RCCL graph capture untested on this pool; eager DWBP there)
-        graphed = solver.enable_graph()
+    if not args.no_graph and device == "cuda":
+        # RCCL collectives are hipGraph-capturable: the multi-rank graph
+        # replays DWBP all-reduces + SFB all-gathers in-graph; capture
+        # success is agreed across ranks and falls back to eager everywhere
+        # if any rank fails (solver._capture_graph). PS_GRAPH=0 disables.
+        if os.environ.get("PS_GRAPH", "1") != "0":
+            graphed = solver.enable_graph()
 
     def sync():
         if device == "cuda":
@@ -83,6 +87,8 @@ def main():
         comm.barrier()
 
     solver.step(args.warmup)
+    # report the truth: capture may have fallen back to eager during warmup
+    graphed = bool(solver._use_graph and solver._graph is not None)
     sync()
     t0 = time.perf_counter()
     solver.step(args.steps)

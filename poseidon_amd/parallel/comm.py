@@ -44,7 +44,16 @@ def init_distributed(backend: Optional[str] = None) -> bool:
             "nccl" if c.device == "cuda" else "gloo")
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
-    dist.init_process_group(backend=backend, rank=c.rank, world_size=c.world_size)
+    kw = {}
+    if backend == "nccl":
+        # hipGraph capture of RCCL collectives: the NCCL watchdog's event
+        # polling must not abort captured-but-idle work, and binding the PG
+        # to its device up front lets RCCL init communicators eagerly
+        # (before capture begins) instead of lazily inside it.
+        os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "0")
+        kw["device_id"] = c.torch_device
+    dist.init_process_group(backend=backend, rank=c.rank,
+                            world_size=c.world_size, **kw)
     return True
 
 
@@ -104,7 +113,9 @@ class GradReducer:
         # default 25 MB; tune per fabric with PS_BUCKET_MB (xGMI ring
         # bandwidth is per-link, so bigger buckets amortize latency while
         # smaller ones start overlapping earlier)
-        self.bucket_bytes = bucket_bytes or             (int(os.environ.get("PS_BUCKET_MB", "25")) << 20)
+        self.bucket_bytes = bucket_bytes or int(
+            os.environ.get("PS_BUCKET_BYTES", "0")) or \
+            (int(os.environ.get("PS_BUCKET_MB", "25")) << 20)
         c = ctx()
         self.enabled = c.distributed and dist.is_initialized()
         self.use_stream = c.device == "cuda" and self.enabled

@@ -70,9 +70,13 @@ class SFBReducer:
         if self.comm_stream is not None:
             # flat/out are allocated on the compute stream but used by the
             # comm stream: record_stream stops the caching allocator from
-            # recycling them while the collective is in flight
-            flat.record_stream(self.comm_stream)
-            out.record_stream(self.comm_stream)
+            # recycling them while the collective is in flight. During
+            # hipGraph capture record_stream is disallowed (and unnecessary:
+            # the graph's private memory pool owns the allocation for the
+            # graph's lifetime).
+            if not torch.cuda.is_current_stream_capturing():
+                flat.record_stream(self.comm_stream)
+                out.record_stream(self.comm_stream)
             ev = torch.cuda.Event()
             ev.record(torch.cuda.current_stream())
             with torch.cuda.stream(self.comm_stream):

@@ -110,7 +110,11 @@ class SGDSolver:
         """Capture fwd+bwd+update as one hipGraph and replay per iteration
         (collapses per-kernel launch gaps -- GoogLeNet's hundreds of small
         kernels). Requires: GPU, SGD solver, static data layers (constant
-        DummyData), L2 regularization. Returns True if enabled."""
+        DummyData), L2 regularization. Works under multi-rank DP too: RCCL
+        collectives are hipGraph-capturable, so the DWBP all-reduces and SFB
+        all-gathers replay inside the graph; capture success is agreed
+        across ranks (all-reduce MIN) so no rank replays while another runs
+        eager. Returns True if enabled."""
         c = ctx()
         if c.device != "cuda" or type(self) is not SGDSolver:
             return False
@@ -179,13 +183,30 @@ class SGDSolver:
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
-            for _ in range(2):  # allocation warmup
+            for _ in range(2):  # allocation warmup (inits RCCL comms too)
                 self._graph_body()
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            self._graph_loss = self._graph_body()
+        ok = True
+        err: Optional[Exception] = None
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._graph_loss = self._graph_body()
+        except Exception as e:  # noqa: BLE001
+            ok, err = False, e
+            # capture never executes collectives, only records them, so a
+            # mismatched record count across ranks cannot deadlock here
+            torch.cuda.synchronize()
+        if self.distributed:
+            # all ranks must agree before anyone replays: a rank replaying a
+            # graphed all-reduce against a rank running eager would hang
+            import torch.distributed as dist
+            flag = torch.tensor([1.0 if ok else 0.0], device=dev)
+            dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+            ok = ok and float(flag.item()) > 0.0
+        if not ok:
+            raise RuntimeError(f"graph capture failed on some rank: {err}")
         self._graph = g
 
     def _step_graphed(self, iters: int) -> float:
