@@ -244,6 +244,15 @@ class Net:
         self._fuse_relu_epilogues()
         for (li, ti, w) in self._loss_tops:
             self._loss_marks.setdefault(li, []).append((ti, w))
+        # Loss layers whose loss top feeds NO consumer get the seeded weight
+        # as a host scalar: their backward then never reads top.diff from
+        # the device (a D2H sync per iteration that also aborts hipGraph
+        # capture). Loss tops that ARE consumed downstream keep the device
+        # read (extra gradient may accumulate into the diff).
+        for (li, ti, w) in self._loss_tops:
+            tname = p.layers[li].top[ti] if ti < len(p.layers[li].top) else None
+            if tname is not None and tname not in consumed:
+                self.layers[li].seeded_loss_weight = w
 
     def _build_stream_schedule(self, n_streams: int = 4) -> None:
         """Dataflow schedule for inter-branch stream parallelism.

@@ -30,6 +30,22 @@ class LossLayer(Layer):
         if top:
             top[0].reshape(())  # scalar loss
 
+    def _top_loss_weight(self, top) -> float:
+        """Scalar the backward scales gradients by (the reference reads
+        top cpu_diff()[0], which Net seeded with the loss weight). Reading
+        the device diff would be a D2H sync every iteration -- it
+        serializes the backward launch queue and aborts hipGraph capture --
+        so use the host-known seeded value (Net.backward seeds loss-top
+        diffs from _loss_tops each iteration) and only fall back to the
+        device read for exotic graphs where the loss top feeds consumers
+        that accumulate extra gradient into it."""
+        if not top:
+            return 1.0
+        w = getattr(self, "seeded_loss_weight", None)
+        if w is not None:
+            return float(w)
+        return float(top[0].diff.reshape(-1)[0].item())
+
 
 @register_layer("SOFTMAX")
 class SoftmaxLayer(Layer):
@@ -64,7 +80,7 @@ class SoftmaxWithLossLayer(LossLayer):
         if len(propagate_down) > 1 and propagate_down[1]:
             raise ValueError("cannot backprop to labels")
         if propagate_down[0]:
-            w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
+            w = self._top_loss_weight(top)
             dx = ops.softmax_loss_backward(self._prob, self._labels, w)
             bottom[0].diff = dx.reshape(bottom[0].shape)
 
@@ -84,7 +100,7 @@ class MultinomialLogisticLossLayer(LossLayer):
         if propagate_down[0]:
             prob, labels, picked = self._cache
             n = prob.shape[0]
-            w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
+            w = self._top_loss_weight(top)
             dx = torch.zeros_like(prob)
             dx[torch.arange(n), labels] = -w / (picked * n)
             bottom[0].diff = dx.reshape(bottom[0].shape)
@@ -105,7 +121,7 @@ class EuclideanLossLayer(LossLayer):
 
     def backward(self, top, propagate_down, bottom) -> None:
         n = bottom[0].num
-        w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
+        w = self._top_loss_weight(top)
         for i, sign in ((0, 1.0), (1, -1.0)):
             if propagate_down[i]:
                 bottom[i].diff = (self._diff * (sign * w / n)).reshape(bottom[i].shape)
@@ -138,7 +154,7 @@ class HingeLossLayer(LossLayer):
             return
         margin, labels = self._cache
         n = margin.shape[0]
-        w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
+        w = self._top_loss_weight(top)
         rows = torch.arange(n)
         if self.norm == "L1":
             g = (margin > 0).to(margin.dtype)
@@ -167,7 +183,7 @@ class SigmoidCrossEntropyLossLayer(LossLayer):
         if propagate_down[0]:
             x, t = self._cache
             n = bottom[0].num
-            w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
+            w = self._top_loss_weight(top)
             bottom[0].diff = (torch.sigmoid(x) - t) * (w / n)
 
 
@@ -202,7 +218,7 @@ class InfogainLossLayer(LossLayer):
         if propagate_down[0]:
             prob, labels, H = self._cache
             n = prob.shape[0]
-            w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
+            w = self._top_loss_weight(top)
             bottom[0].diff = (-(H[labels] / prob.clamp(min=1e-20)) * (w / n)) \
                 .view(bottom[0].shape)
 
@@ -233,7 +249,7 @@ class ContrastiveLossLayer(LossLayer):
     def backward(self, top, propagate_down, bottom) -> None:
         diff, d2, y = self._cache
         n = diff.shape[0]
-        w = float(top[0].diff.reshape(-1)[0].item()) if top else 1.0
+        w = self._top_loss_weight(top)
         active = ((self.margin - d2) > 0).to(diff.dtype)
         scale = (y - (1 - y) * active).view(-1, 1) * (w / n)
         for i, sign in ((0, 1.0), (1, -1.0)):

@@ -120,6 +120,10 @@ class SGDSolver:
             return False
         if self.param.regularization_type == "L1":
             return False
+        if self.distributed:
+            import torch.distributed as dist
+            if dist.get_backend() != "nccl":
+                return False  # gloo collectives are host-side: not capturable
         for layer in self.net.layers:
             t = layer.type_name
             if t in ("DATA", "IMAGE_DATA", "MEMORY_DATA", "WINDOW_DATA"):
