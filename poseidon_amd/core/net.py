@@ -375,18 +375,27 @@ class Net:
         if owned and owned[0].data.is_cuda:
             # one zero_mt kernel instead of one zero_() launch per param
             # (GoogLeNet: 116 launches -> 1); diff tensors are identity-
-            # stable after the first backward, so build the table once
+            # stable after the first backward, so build the table once.
+            # Layer-owned scratch that must start each iteration zeroed
+            # (conv dwk buffers -- lets the split-K wgrad skip its memset)
+            # rides in the same launch.
+            extra = [t for l in self.layers
+                     if hasattr(l, "extra_zero_buffers")
+                     for t in l.extra_zero_buffers()]
             if self._zero_mt is None and all(
                     b.has_diff() and b.diff.dtype == torch.float32
                     and b.diff.is_contiguous() for b in owned):
-                self._zero_mt = (ops.zero_mt_prepare([b.diff for b in owned]),
-                                 [id(b.diff) for b in owned])
+                ts = [b.diff for b in owned] + extra
+                self._zero_mt = (ops.zero_mt_prepare(ts), [id(t) for t in ts])
             if self._zero_mt is not None:
                 mt, ids = self._zero_mt
-                if all(id(b.diff) == i for b, i in zip(owned, ids)):
+                cur = [id(b.diff) for b in owned] + [id(t) for t in extra]
+                if cur == ids:
                     ops.zero_mt_run(mt)
                     return
                 self._zero_mt = None  # identity changed; rebuild next time
+            for t in extra:
+                t.zero_()
         for b in owned:
             b.zero_diff()
 

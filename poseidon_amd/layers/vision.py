@@ -47,6 +47,8 @@ class ConvolutionLayer(Layer):
         self.fuse_relu = False  # set by Net's conv+ReLU fusion pass
         self._wk_cache = None   # (wk, wkT) persistent buffers filled by the
                                 # net-level multi-tensor repack each step
+        self._dwk_cache = None  # persistent khwc wgrad scratch (zeroed per
+                                # iter by the net zero table)
         channels = bottom[0].channels
         assert channels % self.group == 0 and self.num_output % self.group == 0
 
@@ -87,19 +89,29 @@ class ConvolutionLayer(Layer):
     def backward(self, top: List[Blob], propagate_down: List[bool],
                  bottom: List[Blob]) -> None:
         w = self.blobs[0].data
+        single = len(bottom) == 1  # dwk reuse assumes one wgrad write/iter
         for i, (bo, t) in enumerate(zip(bottom, top)):
             dy = t.diff
             db = self.blobs[1].diff.view(-1) if self.bias_term else None
             cache = self._colT[i]
             colT, wkT = cache if isinstance(cache, tuple) else (cache, None)
-            ops.conv2d_backward_weight_acc(
+            dwk = ops.conv2d_backward_weight_acc(
                 bo.data, colT, dy, self.blobs[0].diff, db,
-                self.stride, self.pad, self.group)
+                self.stride, self.pad, self.group,
+                dwk_buf=self._dwk_cache if single else None)
+            if single and dwk is not None:
+                # persistent khwc wgrad scratch: the Net's zero table zeroes
+                # it each iteration, letting the atomic split-K GEMM skip
+                # its per-launch memset
+                self._dwk_cache = dwk
             if propagate_down[i]:
                 bo.diff = ops.conv2d_backward_input(
                     w, dy, bo.shape, self.stride, self.pad, self.group,
                     wkT_cache=wkT)
         self._colT = []
+
+    def extra_zero_buffers(self) -> List[torch.Tensor]:
+        return [self._dwk_cache] if self._dwk_cache is not None else []
 
 
 @register_layer("POOLING")

@@ -91,7 +91,8 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
               int64_t a_off, int64_t b_off, int64_t c_off,
               bool a_klast, bool b_klast, float alpha, float beta,
               const GatherDesc* gather_a = nullptr,
-              const GatherDesc* gather_b = nullptr, bool relu = false) {
+              const GatherDesc* gather_b = nullptr, bool relu = false,
+              bool c_prezeroed = false) {
   const bool in_bf16 = is_bf16(A);
   TORCH_CHECK(is_bf16(B) == in_bf16, "gemm: A/B dtype mismatch");
   const bool out_f32 = !is_bf16(C);
@@ -134,7 +135,11 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
     if (sk > 1) {
       g.splitk = sk;
       g.ws = nullptr;
-      hipMemsetAsync(g.C, 0, (size_t)M * N * sizeof(float), stream());
+      // callers owning a buffer that is zeroed once per iteration by the
+      // net-level zero_mt launch skip this per-GEMM memset (GoogLeNet:
+      // ~57 fillBuffer launches/step eliminated)
+      if (!c_prezeroed)
+        hipMemsetAsync(g.C, 0, (size_t)M * N * sizeof(float), stream());
     }
   } else if (tiles < 384 && K >= 4096) {
     int sk = (int)std::min<int64_t>((512 + tiles - 1) / tiles,
@@ -494,10 +499,14 @@ Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
 }
 
 // dW accumulated into dw_out (fp32 NCHW [Co,Cg,kh,kw]); db into db_out.
-void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
-                                const Tensor& dy, Tensor dw_out,
-                                c10::optional<Tensor> db_out,
-                                int sh, int sw, int ph, int pw, int G) {
+// Returns the khwc scratch dwk it used: callers that keep it alive, hand it
+// back as dwk_buf AND guarantee it is zeroed each iteration (net-level
+// zero_mt) save both the allocation and the atomic-split-K memset.
+Tensor conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
+                                  const Tensor& dy, Tensor dw_out,
+                                  c10::optional<Tensor> db_out,
+                                  int sh, int sw, int ph, int pw, int G,
+                                  const c10::optional<Tensor>& dwk_buf) {
   check_float_like(dy, "dy");
   auto dy_cl = cl4(dy);
   auto x_cl = cl4(x);
@@ -516,7 +525,9 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
                            : (int)(Kcol / G);
 
   // fp32 gradient accumulation regardless of activation dtype
-  Tensor dwk = at::empty({Co, (int64_t)Kgw}, dy.options().dtype(at::kFloat));
+  const bool prez = dwk_buf.has_value() && dwk_buf->numel() == (int64_t)Co * Kgw;
+  Tensor dwk = prez ? *dwk_buf
+             : at::empty({Co, (int64_t)Kgw}, dy.options().dtype(at::kFloat));
   Tensor dy2 = rows2d(dy_cl);
   for (int grp = 0; grp < G; ++grp) {
     // dwk_g[Cog, Kg] = dy_g^T[Cog, NP] @ colT_g[NP, Kg]: contraction NP
@@ -535,14 +546,14 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
                /*lda=*/Co, /*ldb=*/Kgw, /*ldc=*/Kgw,
                /*a_off=*/(int64_t)grp * Cog, /*b_off=*/0,
                /*c_off=*/(int64_t)grp * Cog * Kgw,
-               false, false, 1.0f, 0.0f, nullptr, &gb);
+               false, false, 1.0f, 0.0f, nullptr, &gb, false, prez);
     } else {
       run_gemm(dy2, colT, dwk, nullptr,
                Cog, Kgw, (int)NP,
                /*lda=*/Co, /*ldb=*/Kcol, /*ldc=*/Kgw,
                /*a_off=*/(int64_t)grp * Cog, /*b_off=*/(int64_t)grp * Kgw,
                /*c_off=*/(int64_t)grp * Cog * Kgw,
-               false, false, 1.0f, 0.0f);
+               false, false, 1.0f, 0.0f, nullptr, nullptr, false, prez);
     }
   }
   ps_weight_from_khwc_f32(dwk.data_ptr<float>(), dw_out.data_ptr<float>(),
@@ -556,6 +567,7 @@ void conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
       ps_colsum_f32(dy_cl.data_ptr<float>(), db_out->data_ptr<float>(), NP,
                     Co, stream());
   }
+  return dwk;
 }
 
 // ---------------------------------------------------------------------------

@@ -115,17 +115,22 @@ def conv2d_backward_input(w: torch.Tensor, dy: torch.Tensor,
 
 def conv2d_backward_weight_acc(x: torch.Tensor, colT, dy: torch.Tensor,
                                dw: torch.Tensor, db: Optional[torch.Tensor],
-                               stride, pad, groups: int) -> None:
-    """Accumulates into dw (NCHW) and db."""
+                               stride, pad, groups: int,
+                               dwk_buf: Optional[torch.Tensor] = None
+                               ) -> Optional[torch.Tensor]:
+    """Accumulates into dw (NCHW) and db. Returns the khwc dwk scratch used
+    on GPU: hand it back as dwk_buf on later iterations (keeping it zeroed
+    each iteration via the net-level zero table) to skip the per-GEMM
+    split-K memset."""
     if dy.is_cuda:
-        _ext().conv2d_backward_weight_acc(x, colT, dy, dw, db,
-                                          stride[0], stride[1], pad[0], pad[1],
-                                          groups)
-        return
+        return _ext().conv2d_backward_weight_acc(
+            x, colT, dy, dw, db, stride[0], stride[1], pad[0], pad[1],
+            groups, dwk_buf)
     dw.add_(torch.nn.grad.conv2d_weight(x, list(dw.shape), dy, stride=stride,
                                         padding=pad, groups=groups))
     if db is not None:
         db.add_(dy.sum(dim=(0, 2, 3)))
+    return None
 
 
 # ---------------------------------------------------------------------------
