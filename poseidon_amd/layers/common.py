@@ -101,8 +101,14 @@ class ConcatLayer(Layer):
             top[0].data = torch.cat([b.data for b in bottom], dim=self.dim)
 
     def backward(self, top, propagate_down, bottom) -> None:
-        offset = 0
         chan4 = self.dim == 1 and len(bottom[0].shape) == 4
+        if chan4 and all(propagate_down):
+            parts = ops.split_channels(top[0].diff,
+                                       [b.shape[1] for b in bottom])
+            for b, d in zip(bottom, parts):
+                b.diff = d
+            return
+        offset = 0
         for i, b in enumerate(bottom):
             n = b.shape[self.dim]
             if propagate_down[i]:
@@ -138,13 +144,15 @@ class SliceLayer(Layer):
             t.reshape(shape)
 
     def forward(self, bottom, top) -> None:
-        offset = 0
         chan4 = self.dim == 1 and len(bottom[0].shape) == 4
+        if chan4:
+            for t, d in zip(top, ops.split_channels(bottom[0].data,
+                                                    self.sizes)):
+                t.data = d
+            return
+        offset = 0
         for t, sz in zip(top, self.sizes):
-            if chan4:
-                t.data = ops.slice_channels(bottom[0].data, offset, sz)
-            else:
-                t.data = bottom[0].data.narrow(self.dim, offset, sz).contiguous()
+            t.data = bottom[0].data.narrow(self.dim, offset, sz).contiguous()
             offset += sz
 
     def backward(self, top, propagate_down, bottom) -> None:

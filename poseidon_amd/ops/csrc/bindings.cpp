@@ -583,9 +583,37 @@ Tensor concat_channels(std::vector<Tensor> inputs) {
   for (auto& t : inputs) C_out += t.size(1);
   Tensor y = at::empty({N, (int64_t)C_out, H, W},
                        x0.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const bool bf16 = is_bf16(x0);
+  const int V = bf16 ? 8 : 4;
+  bool aligned = true;
+  for (auto& t : inputs) aligned = aligned && (t.size(1) % V == 0);
+  std::vector<Tensor> cls;
+  cls.reserve(inputs.size());
+  for (auto& t : inputs) cls.push_back(cl4(t));
+  if (aligned) {
+    // up to 4 branches per launch (GoogLeNet inception: one launch/join)
+    int off = 0;
+    for (size_t i0 = 0; i0 < cls.size(); i0 += 4) {
+      void* ptrs[4];
+      int c_end[4];
+      int n = (int)std::min<size_t>(4, cls.size() - i0);
+      int c_begin = off;
+      for (int j = 0; j < n; ++j) {
+        ptrs[j] = cls[i0 + j].data_ptr();
+        off += (int)cls[i0 + j].size(1);
+        c_end[j] = off;
+      }
+      if (bf16)
+        ps_chan_concat4_bf16(y.data_ptr(), ptrs, c_end, n, c_begin, rows,
+                             C_out, 0, stream());
+      else
+        ps_chan_concat4_f32(y.data_ptr<float>(), ptrs, c_end, n, c_begin,
+                            rows, C_out, 0, stream());
+    }
+    return y;
+  }
   int off = 0;
-  for (auto& t : inputs) {
-    auto tc = cl4(t);
+  for (auto& tc : cls) {
     int Ci = tc.size(1);
     if (is_bf16(tc))
       ps_chan_copy_bf16(tc.data_ptr(), y.data_ptr(), rows, Ci, C_out, off,
@@ -596,6 +624,58 @@ Tensor concat_channels(std::vector<Tensor> inputs) {
     off += Ci;
   }
   return y;
+}
+
+// split the wide NHWC tensor into per-range narrow tensors (concat backward
+// / slice forward), up to 4 ranges per launch
+std::vector<Tensor> split_channels(const Tensor& x,
+                                   std::vector<int64_t> sizes) {
+  auto xc = cl4(x);
+  int64_t N = xc.size(0), H = xc.size(2), W = xc.size(3);
+  int64_t rows = N * H * W;
+  int C_in = xc.size(1);
+  const bool bf16 = is_bf16(xc);
+  const int V = bf16 ? 8 : 4;
+  std::vector<Tensor> outs;
+  outs.reserve(sizes.size());
+  for (int64_t c : sizes)
+    outs.push_back(at::empty({N, c, H, W},
+        x.options().memory_format(at::MemoryFormat::ChannelsLast)));
+  bool aligned = true;
+  for (int64_t c : sizes) aligned = aligned && (c % V == 0);
+  if (aligned) {
+    int off = 0;
+    for (size_t i0 = 0; i0 < outs.size(); i0 += 4) {
+      void* ptrs[4];
+      int c_end[4];
+      int n = (int)std::min<size_t>(4, outs.size() - i0);
+      int c_begin = off;
+      for (int j = 0; j < n; ++j) {
+        ptrs[j] = outs[i0 + j].data_ptr();
+        off += (int)sizes[i0 + j];
+        c_end[j] = off;
+      }
+      if (bf16)
+        ps_chan_concat4_bf16(xc.data_ptr(), ptrs, c_end, n, c_begin, rows,
+                             C_in, 1, stream());
+      else
+        ps_chan_concat4_f32(xc.data_ptr<float>(), ptrs, c_end, n, c_begin,
+                            rows, C_in, 1, stream());
+    }
+  } else {
+    int off = 0;
+    for (size_t j = 0; j < outs.size(); ++j) {
+      int Ci = (int)sizes[j];
+      if (bf16)
+        ps_chan_slice_bf16(xc.data_ptr(), outs[j].data_ptr(), rows, C_in, Ci,
+                           off, stream());
+      else
+        ps_chan_slice_f32(xc.data_ptr<float>(), outs[j].data_ptr<float>(),
+                          rows, C_in, Ci, off, stream());
+      off += Ci;
+    }
+  }
+  return outs;
 }
 
 Tensor slice_channels(const Tensor& x, int64_t c_off, int64_t c_len) {
@@ -1246,6 +1326,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_implicit_threshold", &set_implicit_threshold);
   m.def("concat_channels", &concat_channels);
   m.def("slice_channels", &slice_channels);
+  m.def("split_channels", &split_channels);
   m.def("pool_max_forward", &pool_max_forward);
   m.def("pool_max_backward", &pool_max_backward);
   m.def("pool_ave_forward", &pool_ave_forward);

@@ -427,6 +427,42 @@ __global__ void chan_copy_k(const T* __restrict__ in, T* __restrict__ out,
   }
 }
 
+// 4-way fused channel concat/split: one launch per inception join instead
+// of one per branch (GoogLeNet: 72 -> 18 concat/slice launches per step).
+// Thread i owns one V-wide chunk of the WIDE tensor; an unrolled compare
+// over the <=4 channel ranges picks the narrow tensor it pairs with.
+struct Chan4 {
+  void* p[4];    // narrow tensors (src for concat, dst for split)
+  int c_end[4];  // exclusive channel end of each range in the wide tensor
+  int c_begin;   // first channel this launch covers
+  int n;         // live entries
+};
+
+template <typename T, int V, bool GATHER>  // GATHER: wide->narrow (split)
+__global__ void chan_concat4_k(T* __restrict__ wide, Chan4 t, int64_t rows,
+                               int C_wide) {
+  typedef T vec_t __attribute__((ext_vector_type(V)));
+  const int CV = (t.c_end[t.n - 1] - t.c_begin) / V;  // slab chunks only
+  int64_t total = rows * CV;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int cv = (int)(i % CV);
+    int64_t r = i / CV;
+    int c = t.c_begin + cv * V;
+    int j = 0, c0 = t.c_begin;
+#pragma unroll
+    for (int q = 0; q < 3; ++q)
+      if (q < t.n - 1 && c >= t.c_end[q]) { j = q + 1; c0 = t.c_end[q]; }
+    const int Cn = t.c_end[j] - c0;
+    T* nar = (T*)t.p[j] + r * Cn + (c - c0);
+    vec_t* wp = reinterpret_cast<vec_t*>(&wide[r * C_wide + c]);
+    if (GATHER)
+      *reinterpret_cast<vec_t*>(nar) = *wp;
+    else
+      *wp = *reinterpret_cast<const vec_t*>(nar);
+  }
+}
+
 // gather variant: out = in[:, c_off:c_off+C_out] (slice forward / concat bwd)
 template <typename T, int V>
 __global__ void chan_slice_k(const T* __restrict__ in, T* __restrict__ out,
@@ -444,6 +480,46 @@ __global__ void chan_slice_k(const T* __restrict__ in, T* __restrict__ out,
 }
 
 extern "C" {
+
+// n <= 4 narrow tensors <-> one wide tensor in a single launch.
+// Requirements checked by caller: every range width and boundary divisible
+// by V. gather=1 splits wide->narrow, 0 concats narrow->wide.
+static inline Chan4 chan4_pack(void* const* ptrs, const int* c_end, int n,
+                               int c_begin) {
+  Chan4 t;
+  t.n = n;
+  t.c_begin = c_begin;
+  for (int i = 0; i < 4; ++i) {
+    t.p[i] = i < n ? const_cast<void*>(ptrs[i]) : nullptr;
+    t.c_end[i] = i < n ? c_end[i] : (n ? c_end[n - 1] : 0);
+  }
+  return t;
+}
+
+void ps_chan_concat4_f32(float* wide, void* const* ptrs, const int* c_end,
+                         int n, int c_begin, int64_t rows, int C_wide,
+                         int gather, hipStream_t s) {
+  Chan4 t = chan4_pack(ptrs, c_end, n, c_begin);
+  int64_t work = rows * ((c_end[n - 1] - c_begin) / 4);
+  if (gather)
+    chan_concat4_k<float, 4, true>
+        <<<ew_grid(work), 256, 0, s>>>(wide, t, rows, C_wide);
+  else
+    chan_concat4_k<float, 4, false>
+        <<<ew_grid(work), 256, 0, s>>>(wide, t, rows, C_wide);
+}
+void ps_chan_concat4_bf16(void* wide, void* const* ptrs, const int* c_end,
+                          int n, int c_begin, int64_t rows, int C_wide,
+                          int gather, hipStream_t s) {
+  Chan4 t = chan4_pack(ptrs, c_end, n, c_begin);
+  int64_t work = rows * ((c_end[n - 1] - c_begin) / 8);
+  if (gather)
+    chan_concat4_k<__bf16, 8, true>
+        <<<ew_grid(work), 256, 0, s>>>((__bf16*)wide, t, rows, C_wide);
+  else
+    chan_concat4_k<__bf16, 8, false>
+        <<<ew_grid(work), 256, 0, s>>>((__bf16*)wide, t, rows, C_wide);
+}
 
 void ps_chan_copy_f32(const float* in, float* out, int64_t rows, int C_in,
                       int C_out, int c_off, hipStream_t s) {

@@ -577,6 +577,18 @@ def slice_channels(x, c_off: int, c_len: int):
     return x.narrow(1, c_off, c_len).contiguous()
 
 
+def split_channels(x, sizes):
+    """All channel ranges in one go: up to 4 ranges per kernel launch
+    (concat backward / slice forward over inception joins)."""
+    if x.is_cuda and x.dim() == 4:
+        return _ext().split_channels(x, [int(s) for s in sizes])
+    out, off = [], 0
+    for s in sizes:
+        out.append(x.narrow(1, off, int(s)).contiguous())
+        off += int(s)
+    return out
+
+
 # ---------------------------------------------------------------------------
 # Metrics
 # ---------------------------------------------------------------------------

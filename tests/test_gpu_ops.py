@@ -419,3 +419,21 @@ def test_threshold_eltwise_contrastive_kernels():
     tg = ops.contrastive_terms(d2.to(DEV), sim.to(DEV), 1.0, True)
     tr2 = ops.contrastive_terms(d2, sim, 1.0, True)
     close(tg, tr2, what="contrastive terms")
+
+
+def test_concat_split_channels_fused():
+    """4-way fused NHWC concat/split kernels (chan_concat4_k) vs torch.cat
+    -- aligned (one launch per <=4 branches), >4 inputs (two slabs), and
+    unaligned widths (per-input fallback)."""
+    for widths in ([64, 128, 32, 32],           # inception-style, 1 launch
+                   [16, 8, 24, 32, 40, 8],      # 6 inputs -> 2 slabs
+                   [10, 6, 12]):                # unaligned -> fallback
+        xs = [rnd(2, c, 5, 7, seed=100 + i) for i, c in enumerate(widths)]
+        yg = ops.concat_channels([x.to(DEV) for x in xs])
+        close(yg, torch.cat(xs, dim=1), what=f"concat {widths}")
+        wide = rnd(2, sum(widths), 5, 7, seed=99)
+        parts = ops.split_channels(wide.to(DEV), widths)
+        off = 0
+        for w, p in zip(widths, parts):
+            close(p, wide.narrow(1, off, w), what=f"split {widths}@{off}")
+            off += w
