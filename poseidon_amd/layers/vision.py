@@ -102,8 +102,15 @@ class ConvolutionLayer(Layer):
             if single and dwk is not None:
                 # persistent khwc wgrad scratch: the Net's zero table zeroes
                 # it each iteration, letting the atomic split-K GEMM skip
-                # its per-launch memset
-                self._dwk_cache = dwk
+                # its per-launch memset. Keep the SAME python object when
+                # the storage is unchanged -- the binding returns a fresh
+                # wrapper around the buffer we passed in, and rebinding it
+                # would churn the zero-table identity key every iteration
+                # (a rebuild + H2D copy per iter, and a capture abort).
+                cur = self._dwk_cache
+                if (cur is None or cur.data_ptr() != dwk.data_ptr()
+                        or cur.numel() != dwk.numel()):
+                    self._dwk_cache = dwk
             if propagate_down[i]:
                 bo.diff = ops.conv2d_backward_input(
                     w, dy, bo.shape, self.stride, self.pad, self.group,

@@ -309,9 +309,9 @@ class SGDSolver:
             try:
                 return self._step_graphed(iters)
             except Exception as e:  # noqa: BLE001 -- capture unsupported
-                if self.verbose:
-                    print(f"[poseidon] graph capture failed ({e}); "
-                          "falling back to eager", flush=True)
+                import sys
+                print(f"[poseidon] graph capture failed ({e}); "
+                      "falling back to eager", file=sys.stderr, flush=True)
                 self._use_graph = False
                 self._graph = None
         last_loss = 0.0
