@@ -187,7 +187,13 @@ class SGDSolver:
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
-            for _ in range(2):  # allocation warmup (inits RCCL comms too)
+            # Allocation + identity warmup (inits RCCL comms too). THREE
+            # iterations before capture: the multi-tensor tables settle at
+            # iter 2 (zero table: built at 0 without the conv dwk scratch,
+            # invalidated at 1 when backward allocates it, rebuilt at 2) and
+            # a table rebuild is a pageable H2D copy -- forbidden inside
+            # capture.
+            for _ in range(3):
                 self._graph_body()
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
