@@ -171,16 +171,23 @@ class DataLayer(_PrefetchingDataLayer):
         shared = bool(dp.shared_file_system) or c.world_size == 1
         if not shared:
             source = f"{source}_{c.rank}"
-        if backend == "LMDB" and not os.path.isfile(source):
-            try:
-                import lmdb  # noqa: F401
-                raise NotImplementedError(
-                    "LMDB directory sources need the lmdb module wiring")
-            except ImportError as e:
-                raise RuntimeError(
-                    f"LMDB backend requested but lmdb module unavailable: {e}. "
-                    "Convert the dataset to PDB with tools/convert_dataset.py")
-        self.db = PDBReader(source)
+        # LMDB environments (reference convert_imageset output) are read
+        # directly by the pure-python walker in data/lmdb_io.py; PDB files
+        # keep the native container path. LevelDB dirs (log/sst format)
+        # are not parsed -- point users at the converter.
+        is_lmdb = (os.path.isdir(source)
+                   and os.path.exists(os.path.join(source, "data.mdb"))) or (
+                  os.path.isfile(source) and source.endswith(".mdb"))
+        if is_lmdb:
+            from ..data.lmdb_io import LmdbReader
+            self.db = LmdbReader(source)
+        elif os.path.isdir(source):
+            raise RuntimeError(
+                f"{source}: directory is not an LMDB environment (no "
+                "data.mdb). LevelDB-format datasets must be converted: "
+                "python -m poseidon_amd.tools.datasets convert ... (PDB)")
+        else:
+            self.db = PDBReader(source)
         self.stride = c.world_size if shared else 1
         self.cursor = c.rank if shared else 0
         rng = np.random.default_rng(c.seed + 131 * c.rank)

@@ -77,6 +77,31 @@ def partition_data(argv=None):
     print(f"{len(db)} records -> {args.num_shards} shards")
 
 
+def convert_db(argv=None):
+    """Convert between the PDB container and LMDB environments in either
+    direction -- lets reference-produced LMDB shards run here unmodified
+    and lets PDB datasets be exported for the reference's own tools."""
+    ap = argparse.ArgumentParser()
+    ap.add_argument("src", help="PDB file or LMDB env dir / data.mdb")
+    ap.add_argument("dst", help="output PDB file (or LMDB dir with --to-lmdb)")
+    ap.add_argument("--to-lmdb", action="store_true")
+    args = ap.parse_args(argv)
+    from ..data.lmdb_io import LmdbReader, LmdbWriter
+    import os
+    src_is_lmdb = os.path.isdir(args.src) or args.src.endswith(".mdb")
+    if args.to_lmdb:
+        db = LmdbReader(args.src) if src_is_lmdb else PDBReader(args.src)
+        with LmdbWriter(args.dst) as w:
+            for i in range(len(db)):
+                w.put(b"%08d" % i, db.get_raw(i))
+    else:
+        db = LmdbReader(args.src) if src_is_lmdb else PDBReader(args.src)
+        with PDBWriter(args.dst) as w:
+            for i in range(len(db)):
+                w.put_raw(db.get_raw(i))
+    print(f"converted {len(db)} records {args.src} -> {args.dst}")
+
+
 def device_query(argv=None):
     """caffe_main device_query equivalent."""
     import torch
@@ -95,6 +120,7 @@ if __name__ == "__main__":
     cmds = {"convert_imageset": convert_imageset,
             "compute_image_mean": compute_image_mean,
             "partition_data": partition_data,
+            "convert_db": convert_db,
             "device_query": device_query}
     if len(sys.argv) < 2 or sys.argv[1] not in cmds:
         print("usage: python -m poseidon_amd.tools.datasets "
