@@ -110,6 +110,7 @@ class Net:
         self._loss_tops: List = []  # (layer_idx, top_idx, weight)
         self._zero_mt = None  # multi-tensor zero table (GPU, built lazily)
         self._repack_mt = None  # multi-tensor conv-weight repack table
+        self._unpack_mt = None  # deferred conv-wgrad unpack table
         self._repack_key = None
         self._loss_marks: Dict[int, List] = {}
         # inter-branch stream parallelism (inception-style nets): built by
@@ -399,6 +400,28 @@ class Net:
         for b in owned:
             b.zero_diff()
 
+    def _flush_deferred_unpacks(self) -> None:
+        """ONE unpack_mt kernel accumulates every conv's khwc wgrad scratch
+        into its NCHW param diff (GoogLeNet: 57 weight_from_khwc launches
+        -> 1). Active only when conv layers ran with defer_unpack (set by
+        the solver in single-GPU mode -- DWBP needs per-layer grads final
+        before the bucketed all-reduce fires)."""
+        convs = [l for l in self.layers
+                 if getattr(l, "_unpack_pending", False)]
+        if not convs:
+            return
+        key = [(id(l._dwk_cache), id(l.blobs[0].diff)) for l in convs]
+        if self._unpack_mt is None or self._unpack_mt[1] != key:
+            self._unpack_mt = (ops.unpack_mt_prepare(
+                [l._dwk_cache for l in convs],
+                [l.blobs[0].diff for l in convs],
+                [l.blobs[0].data.shape[1] for l in convs],
+                [l.blobs[0].data.shape[2] for l in convs],
+                [l.blobs[0].data.shape[3] for l in convs]), key)
+        ops.unpack_mt_run(self._unpack_mt[0])
+        for l in convs:
+            l._unpack_pending = False
+
     def _maybe_mt_repack(self) -> None:
         """One repack_mt kernel refreshes every conv's bf16 khwc shadow +
         dgrad transpose from the fp32 masters (GoogLeNet: 2 launches
@@ -512,6 +535,7 @@ class Net:
             t.diff.fill_(w)
         if self._ms_active():
             self._backward_ms(post_layer_cb)
+            self._flush_deferred_unpacks()
             return
         for i in range(len(self.layers) - 1, -1, -1):
             if not self.layer_need_bwd[i]:
@@ -520,6 +544,7 @@ class Net:
                                     self.bottoms[i])
             if post_layer_cb is not None and self.layers[i].blobs:
                 post_layer_cb(i, self.layers[i])
+        self._flush_deferred_unpacks()
 
     def _backward_ms(self, post_layer_cb) -> None:
         """Backward mirror of _forward_ms: layer i runs on its forward

@@ -49,6 +49,7 @@ class ConvolutionLayer(Layer):
                                 # net-level multi-tensor repack each step
         self._dwk_cache = None  # persistent khwc wgrad scratch (zeroed per
                                 # iter by the net zero table)
+        self._unpack_pending = False  # deferred wgrad unpack flag
         channels = bottom[0].channels
         assert channels % self.group == 0 and self.num_output % self.group == 0
 
@@ -90,6 +91,12 @@ class ConvolutionLayer(Layer):
                  bottom: List[Blob]) -> None:
         w = self.blobs[0].data
         single = len(bottom) == 1  # dwk reuse assumes one wgrad write/iter
+        # Deferred unpack (single-GPU): leave the gradient in the khwc
+        # scratch; Net.backward runs ONE unpack_mt kernel for all convs at
+        # the end (DWBP multi-rank mode needs per-layer grads final, so the
+        # solver only sets defer_unpack when there is no reducer).
+        defer = (single and getattr(self, "defer_unpack", False)
+                 and self._dwk_cache is not None)
         for i, (bo, t) in enumerate(zip(bottom, top)):
             dy = t.diff
             db = self.blobs[1].diff.view(-1) if self.bias_term else None
@@ -98,7 +105,14 @@ class ConvolutionLayer(Layer):
             dwk = ops.conv2d_backward_weight_acc(
                 bo.data, colT, dy, self.blobs[0].diff, db,
                 self.stride, self.pad, self.group,
-                dwk_buf=self._dwk_cache if single else None)
+                dwk_buf=self._dwk_cache if single else None,
+                skip_unpack=defer)
+            if defer:
+                if dwk.data_ptr() != self._dwk_cache.data_ptr():
+                    raise RuntimeError(
+                        f"{self.name}: conv geometry changed under deferred "
+                        "wgrad unpack (dwk buffer was reallocated)")
+                self._unpack_pending = True
             if single and dwk is not None:
                 # persistent khwc wgrad scratch: the Net's zero table zeroes
                 # it each iteration, letting the atomic split-K GEMM skip

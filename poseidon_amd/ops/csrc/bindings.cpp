@@ -506,7 +506,8 @@ Tensor conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
                                   const Tensor& dy, Tensor dw_out,
                                   c10::optional<Tensor> db_out,
                                   int sh, int sw, int ph, int pw, int G,
-                                  const c10::optional<Tensor>& dwk_buf) {
+                                  const c10::optional<Tensor>& dwk_buf,
+                                  bool skip_unpack) {
   check_float_like(dy, "dy");
   auto dy_cl = cl4(dy);
   auto x_cl = cl4(x);
@@ -556,9 +557,10 @@ Tensor conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
                false, false, 1.0f, 0.0f, nullptr, nullptr, false, prez);
     }
   }
-  ps_weight_from_khwc_f32(dwk.data_ptr<float>(), dw_out.data_ptr<float>(),
-                          Co, Cig, kh, kw, /*ld=*/Kgw, /*beta=*/1.0f,
-                          stream());
+  if (!skip_unpack)
+    ps_weight_from_khwc_f32(dwk.data_ptr<float>(), dw_out.data_ptr<float>(),
+                            Co, Cig, kh, kw, /*ld=*/Kgw, /*beta=*/1.0f,
+                            stream());
   if (db_out.has_value()) {
     if (is_bf16(dy_cl))
       ps_colsum_bf16(dy_cl.data_ptr(), db_out->data_ptr<float>(), NP, Co,
@@ -1270,6 +1272,50 @@ void zero_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
              stream());
 }
 
+struct MTUnpackDescHost {
+  const float* dwk;
+  float* dw;
+  int64_t n;
+  int Co, Cig, kh, kw, ldk;
+};
+
+// one table for every deferred conv-wgrad unpack (dwk khwc -> dw NCHW,
+// accumulate); identities are the persistent dwk buffers + param diffs
+std::vector<Tensor> unpack_mt_prepare(std::vector<Tensor> dwks,
+                                      std::vector<Tensor> dws,
+                                      std::vector<int64_t> Cigs,
+                                      std::vector<int64_t> khs,
+                                      std::vector<int64_t> kws) {
+  const int CHUNK = ps_mt_chunk_elts();
+  const size_t nt = dwks.size();
+  std::vector<MTUnpackDescHost> descs(nt);
+  std::vector<MTChunkHost> chunks;
+  for (size_t t = 0; t < nt; ++t) {
+    TORCH_CHECK(dwks[t].is_cuda() && dwks[t].scalar_type() == at::kFloat &&
+                dws[t].is_cuda() && dws[t].is_contiguous() &&
+                dws[t].scalar_type() == at::kFloat,
+                "unpack_mt_prepare: f32 CUDA tensors required");
+    int Co = (int)dwks[t].size(0);
+    int ldk = (int)dwks[t].size(1);
+    int64_t n = dws[t].numel();
+    TORCH_CHECK(n == (int64_t)Co * Cigs[t] * khs[t] * kws[t]);
+    descs[t] = {dwks[t].data_ptr<float>(), dws[t].data_ptr<float>(), n,
+                Co, (int)Cigs[t], (int)khs[t], (int)kws[t], ldk};
+    for (int64_t off = 0; off < n; off += CHUNK)
+      chunks.push_back({(int)t, off});
+  }
+  return {blob_to_dev(descs.data(), nt * sizeof(MTUnpackDescHost), dwks[0]),
+          blob_to_dev(chunks.data(), chunks.size() * sizeof(MTChunkHost),
+                      dwks[0]),
+          at::scalar_tensor((int64_t)chunks.size())};
+}
+
+void unpack_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
+                   int64_t nchunks) {
+  ps_unpack_mt(desc_dev.data_ptr(), chunk_dev.data_ptr(), (int)nchunks,
+               stream());
+}
+
 struct MTRepackDescHost {
   const float* src;
   void* wk;
@@ -1360,6 +1406,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("zero_mt_prepare", &zero_mt_prepare);
   m.def("zero_mt_run", &zero_mt_run);
   m.def("repack_mt_prepare", &repack_mt_prepare);
+  m.def("unpack_mt_prepare", &unpack_mt_prepare);
+  m.def("unpack_mt_run", &unpack_mt_run);
   m.def("repack_mt_run", &repack_mt_run);
   m.def("colT_ld", [](int64_t G, int64_t C, int64_t kh, int64_t kw,
                       int64_t vec) {

@@ -214,6 +214,8 @@ void ps_zero_mt(const void* descs, const void* chunks, int nchunks,
                 hipStream_t);
 void ps_repack_mt(const void* descs, const void* chunks, int nchunks,
                   hipStream_t);
+void ps_unpack_mt(const void* descs, const void* chunks, int nchunks,
+                  hipStream_t);
 void ps_dropout_fwd_f32_offdev(const float*, float*, uint8_t*, int64_t, float,
                                uint64_t, const void*, hipStream_t);
 void ps_dropout_fwd_bf16_offdev(const void*, void*, uint8_t*, int64_t, float,

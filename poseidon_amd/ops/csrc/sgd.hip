@@ -162,6 +162,34 @@ __global__ void repack_mt_k(const MTRepackDesc* __restrict__ descs,
   }
 }
 
+// inverse of repack_mt_k for GRADIENTS: khwc fp32 wgrad scratch -> NCHW
+// fp32 param diff, accumulating (beta=1). One launch covers every conv's
+// unpack at the end of backward (GoogLeNet: 57 weight_from_khwc launches
+// -> 1) -- single-GPU mode only; DWBP needs per-layer grads final.
+struct MTUnpackDesc {
+  const float* dwk;  // fp32 [Co][ldk] khwc
+  float* dw;         // fp32 NCHW [Co][Cig][kh][kw], accumulated into
+  int64_t n;         // Co*Cig*kh*kw
+  int Co, Cig, kh, kw, ldk;
+};
+
+__global__ void unpack_mt_k(const MTUnpackDesc* __restrict__ descs,
+                            const MTChunk* __restrict__ chunks) {
+  const MTChunk ck = chunks[blockIdx.x];
+  const MTUnpackDesc d = descs[ck.t];
+  const int KW = d.kw, KH = d.kh, Cig = d.Cig;
+  for (int64_t i = ck.off + threadIdx.x;
+       i < ck.off + MT_CHUNK && i < d.n; i += 256) {
+    int kkw = (int)(i % KW);
+    int64_t t = i / KW;
+    int kkh = (int)(t % KH); t /= KH;
+    int ci = (int)(t % Cig);
+    int co = (int)(t / Cig);
+    const int kg = (kkh * KW + kkw) * Cig + ci;
+    d.dw[i] += d.dwk[(int64_t)co * d.ldk + kg];
+  }
+}
+
 __global__ void zero_mt_k(const MTZeroDesc* __restrict__ descs,
                           const MTChunk* __restrict__ chunks) {
   const MTChunk ck = chunks[blockIdx.x];
@@ -232,6 +260,13 @@ void ps_repack_mt(const void* descs, const void* chunks, int nchunks,
   if (nchunks <= 0) return;
   repack_mt_k<<<dim3((unsigned)nchunks), 256, 0, s>>>(
       (const MTRepackDesc*)descs, (const MTChunk*)chunks);
+}
+
+void ps_unpack_mt(const void* descs, const void* chunks, int nchunks,
+                  hipStream_t s) {
+  if (nchunks <= 0) return;
+  unpack_mt_k<<<dim3((unsigned)nchunks), 256, 0, s>>>(
+      (const MTUnpackDesc*)descs, (const MTChunk*)chunks);
 }
 
 void ps_zero_mt(const void* descs, const void* chunks, int nchunks,

@@ -116,7 +116,8 @@ def conv2d_backward_input(w: torch.Tensor, dy: torch.Tensor,
 def conv2d_backward_weight_acc(x: torch.Tensor, colT, dy: torch.Tensor,
                                dw: torch.Tensor, db: Optional[torch.Tensor],
                                stride, pad, groups: int,
-                               dwk_buf: Optional[torch.Tensor] = None
+                               dwk_buf: Optional[torch.Tensor] = None,
+                               skip_unpack: bool = False
                                ) -> Optional[torch.Tensor]:
     """Accumulates into dw (NCHW) and db. Returns the khwc dwk scratch used
     on GPU: hand it back as dwk_buf on later iterations (keeping it zeroed
@@ -125,7 +126,7 @@ def conv2d_backward_weight_acc(x: torch.Tensor, colT, dy: torch.Tensor,
     if dy.is_cuda:
         return _ext().conv2d_backward_weight_acc(
             x, colT, dy, dw, db, stride[0], stride[1], pad[0], pad[1],
-            groups, dwk_buf)
+            groups, dwk_buf, skip_unpack)
     dw.add_(torch.nn.grad.conv2d_weight(x, list(dw.shape), dy, stride=stride,
                                         padding=pad, groups=groups))
     if db is not None:
@@ -526,6 +527,17 @@ def repack_mt_prepare(masters, wks, wkTs, Gs):
 def repack_mt_run(mt) -> None:
     d, c, n = mt
     _ext().repack_mt_run(d, c, n)
+
+
+def unpack_mt_prepare(dwks, dws, Cigs, khs, kws):
+    d, c, n = _ext().unpack_mt_prepare(list(dwks), list(dws), list(Cigs),
+                                       list(khs), list(kws))
+    return d, c, int(n)
+
+
+def unpack_mt_run(mt) -> None:
+    d, c, n = mt
+    _ext().unpack_mt_run(d, c, n)
 
 
 def zero_mt_prepare(tensors):

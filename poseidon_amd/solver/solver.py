@@ -77,6 +77,15 @@ class SGDSolver:
             comm.broadcast_params([ps for i, ps in enumerate(self.net.params)
                                    if ps.owner == i])
         self.reducer = comm.GradReducer(self.net) if self.distributed else None
+        if not self.distributed and c.device == "cuda":
+            # single-GPU: conv wgrads stay in their khwc scratch through
+            # backward and ONE unpack_mt kernel folds them into the NCHW
+            # diffs at the end (net._flush_deferred_unpacks). Multi-rank
+            # keeps per-layer unpack: DWBP all-reduces each layer's final
+            # grad as soon as its backward completes.
+            for l in self.net.layers:
+                if l.type_name == "CONVOLUTION":
+                    l.defer_unpack = True
         want_sfb = c.use_sfb if use_sfb is None else use_sfb
         self.sfb = None
         if self.distributed and want_sfb:

@@ -437,3 +437,54 @@ def test_concat_split_channels_fused():
         for w, p in zip(widths, parts):
             close(p, wide.narrow(1, off, w), what=f"split {widths}@{off}")
             off += w
+
+
+def test_deferred_wgrad_unpack_matches_eager():
+    """Single-GPU deferred conv-wgrad unpack (one unpack_mt kernel at end
+    of backward) must produce the same trained weights as the per-layer
+    weight_from_khwc path."""
+    from poseidon_amd.proto import Message, parse_text
+    from poseidon_amd.solver.solver import SGDSolver
+
+    def net_param():
+        return parse_text("NetParameter", """
+            name: "dconv"
+            layers { name: "data" type: DUMMY_DATA top: "data" top: "label"
+                     dummy_data_param { num: 8 channels: 8 height: 10
+                         width: 10 num: 8 channels: 1 height: 1 width: 1
+                         data_filler { type: "gaussian" std: 1.0 }
+                         data_filler { type: "constant" } } }
+            layers { name: "c1" type: CONVOLUTION bottom: "data" top: "c1"
+                     convolution_param { num_output: 16 kernel_size: 3 pad: 1
+                         weight_filler { type: "xavier" }
+                         bias_filler { type: "constant" } } }
+            layers { name: "r1" type: RELU bottom: "c1" top: "c1" }
+            layers { name: "c2" type: CONVOLUTION bottom: "c1" top: "c2"
+                     convolution_param { num_output: 8 kernel_size: 3 pad: 1
+                         weight_filler { type: "xavier" } } }
+            layers { name: "ip" type: INNER_PRODUCT bottom: "c2" top: "ip"
+                     inner_product_param { num_output: 4
+                         weight_filler { type: "xavier" } } }
+            layers { name: "loss" type: SOFTMAX_LOSS bottom: "ip"
+                     bottom: "label" top: "loss" }
+        """)
+
+    results = []
+    for defer in (False, True):
+        pa.init(device="cuda", seed=77)
+        sp = Message("SolverParameter", base_lr=0.05, lr_policy="fixed",
+                     momentum=0.9, weight_decay=0.001, max_iter=10,
+                     display=0, snapshot=0)
+        sp.net_param = net_param()
+        s = SGDSolver(sp, verbose=False)
+        for l in s.net.layers:
+            if l.type_name == "CONVOLUTION":
+                l.defer_unpack = defer
+        s.step(4)
+        torch.cuda.synchronize()
+        results.append({i: ps.blob.data.clone().cpu()
+                        for i, ps in enumerate(s.net.params)
+                        if ps.owner == i})
+    for i in results[0]:
+        assert torch.allclose(results[0][i], results[1][i],
+                              rtol=1e-5, atol=1e-6), f"param {i}"
