@@ -115,6 +115,19 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   g.gather_a = gather_a;
   g.gather_b = gather_b;
   g.relu = relu;
+  // tr16 TN path with split-K: hand it a workspace so the splits store
+  // plain partials + one reduce instead of memset + per-element atomicAdd
+  // chains (VGG wgrad: ~2M contended atomics per GEMM removed)
+  Tensor trws;
+  if (in_bf16 && out_f32) {
+    int64_t trw = ps_gemm_tn_tr_ws_elems(&g);
+    if (trw > 0 && trw * 4 <= (256LL << 20)) {
+      trws = at::empty({trw}, C.options().dtype(at::kFloat));
+      g.ws = trws.data_ptr<float>();
+      ps_gemm_bf16_f32out(&g, stream());
+      return;
+    }
+  }
   // Split-K when the output tile grid cannot fill 256 CUs but K is deep
   // (conv wgrad: M=Cout, N=Kcol, K=N*OH*OW up to ~800k): target ~512
   // workgroups, cap the f32 workspace at 256 MB.

@@ -615,6 +615,20 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
 
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
+// Row-dependent 16 B-chunk XOR for the tr16 LDS image. Without it the
+// ds_read_b64_tr_b16 lane groups put 4 lanes on each dword bank (rows r
+// and r+2 / r+8 alias at row-stride COLS*2 B mod 256 B): measured 6.0
+// LDS bank-conflict cycles per LDS instruction, 58.8%% of wave cycles
+// parked. The XOR spreads the 8 rows a lane group touches across disjoint
+// 16 B bands -> conflict-free reads. Writers permute the SOURCE lane
+// (global_load_lds writes lane-linear), readers apply the same XOR.
+template <int COLS>
+__device__ inline int tr_swz(int row, int chunk) {
+  constexpr int MASK = (COLS / 8) - 1;  // chunks per row
+  const int p = (((row >> 1) & 1) << 1) ^ (((row >> 3) & 1) << 2);
+  return chunk ^ (p & MASK);
+}
+
 template <int COLS, bool GATHER = false>
 __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
                                      int64_t ld, int k0, int c0, int wid,
@@ -628,31 +642,37 @@ __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
   constexpr int CHUNKS = 64 / RPC;
   const int r_in = lane / LPR;
   const int slot = lane % LPR;
-  // gather: the column decode (kg -> kh,kw,cg) is loop-invariant per lane
-  // (slot is fixed); only the row decode (np -> n,oh,ow) runs per chunk.
-  // PMC showed the naive per-chunk gather_addr at 18-20 VALU per MFMA.
-  int kkh = 0, kkw = 0, cg = 0;
+  // gather: the column decode (kg -> kh,kw,cg) depends on the swizzled
+  // source chunk, which takes one of 4 values per lane -- precompute all
+  // 4 decodes so the chunk loop still runs decode-free.
+  int kkh[4], kkw[4], cg[4];
   if (GATHER) {
-    int khw = fdiv_fix(c0 + slot * 8, ga->Cg, ga->inv_Cg, cg);
-    kkh = fdiv_fix(khw, ga->kw, ga->inv_kw, kkw);
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int sl = slot ^ ((p << 1) & (LPR - 1));
+      int khw = fdiv_fix(c0 + sl * 8, ga->Cg, ga->inv_Cg, cg[p]);
+      kkh[p] = fdiv_fix(khw, ga->kw, ga->inv_kw, kkw[p]);
+    }
   }
 #pragma unroll
   for (int ci = wid; ci < CHUNKS; ci += 4) {
     const int row = ci * RPC + r_in;
+    const int pi = ((row >> 1) & 1) | (((row >> 3) & 1) << 1);
+    const int sslot = tr_swz<COLS>(row, slot);
     const __bf16* g2;
     if (GATHER) {
       int ow, oh;
       const int t2 = fdiv_fix(k0 + row, ga->Wo, ga->inv_Wo, ow);
       const int n2 = fdiv_fix(t2, ga->Ho, ga->inv_Ho, oh);
-      const int ih = oh * ga->sh - ga->ph + kkh;
-      const int iw = ow * ga->sw - ga->pw + kkw;
+      const int ih = oh * ga->sh - ga->ph + kkh[pi];
+      const int iw = ow * ga->sw - ga->pw + kkw[pi];
       g2 = (ih < 0 || ih >= ga->H || iw < 0 || iw >= ga->W)
                ? (const __bf16*)ga->zero
                : (const __bf16*)ga->x +
                      (((int64_t)n2 * ga->H + ih) * ga->W + iw) * ga->C +
-                     ga->c0 + cg;
+                     ga->c0 + cg[pi];
     } else {
-      g2 = src + (int64_t)(k0 + row) * ld + c0 + slot * 8;
+      g2 = src + (int64_t)(k0 + row) * ld + c0 + sslot * 8;
     }
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g2,
@@ -663,11 +683,15 @@ __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
 // Issues the two transpose-reads WITHOUT waiting -- the caller batches all
 // fragment reads of a k-step behind one s_waitcnt (tr_wait) so the LDS
 // latency of 16 reads overlaps instead of serializing.
+template <int COLS>
 __device__ inline bf16x8 tr_frag(unsigned lds_base, int kk, int cb, int l,
                                  int ldt) {
-  const unsigned a1 =
-      lds_base + (unsigned)(((kk + (l >> 2)) * ldt + cb + 4 * (l & 3)) * 2);
-  const unsigned a2 = a1 + 4u * ldt * 2u;
+  const int r1 = kk + (l >> 2), r2 = r1 + 4;
+  const int c = cb + 4 * (l & 3);
+  const int c1 = (tr_swz<COLS>(r1, c >> 3) << 3) | (c & 7);
+  const int c2 = (tr_swz<COLS>(r2, c >> 3) << 3) | (c & 7);
+  const unsigned a1 = lds_base + (unsigned)((r1 * ldt + c1) * 2);
+  const unsigned a2 = lds_base + (unsigned)((r2 * ldt + c2) * 2);
   bf16x4 v1, v2;
   // "=&v" early-clobber: insn 1 writes v1 before insn 2 consumes a2
   asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
@@ -692,29 +716,47 @@ __device__ inline void tr_wait(bf16x8 (&a)[NA], bf16x8 (&b)[NB]) {
 }
 
 template <int BM2, int BN2, int WGM2, int WGN2, bool SPLITK,
-          bool GB2 = false>
+          bool GB2 = false, bool WS = false>
 __global__ __launch_bounds__(256)
 void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
                        const __bf16* __restrict__ B, float* __restrict__ C,
                        int M, int N, int K, int64_t ldA, int64_t ldB,
                        int64_t ldC, float alpha, int kchunk,
-                       GatherDesc ga_b = {}) {
+                       GatherDesc ga_b = {}, float* __restrict__ ws = nullptr) {
   constexpr int FM2 = BM2 / WGM2 / 16, FN2 = BN2 / WGN2 / 16;
   static_assert(WGM2 * WGN2 == 4, "4 waves");
   __shared__ __attribute__((aligned(16))) __bf16 a_lds[2][64 * BM2];
   __shared__ __attribute__((aligned(16))) __bf16 b_lds[2][64 * BN2];
-  // T1 XCD swizzle (same as gemm_kernel)
   int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
   {
     const int nbx = gridDim.x, nby = gridDim.y;
-    const int64_t nwg = (int64_t)nbx * nby * gridDim.z;
-    if ((nwg & 7) == 0 && nwg > 8) {
-      int64_t id = ((int64_t)bz * nby + by) * nbx + bx;
-      const int64_t cpx = nwg >> 3;
-      id = (id & 7) * cpx + (id >> 3);
-      bx = (int)(id % nbx);
-      by = (int)((id / nbx) % nby);
-      bz = (int)(id / ((int64_t)nbx * nby));
+    const int64_t G = (int64_t)nby * gridDim.z;  // (by, bz) groups
+    if (SPLITK && (G & 7) == 0 && G >= 8) {
+      // L2-reuse placement: all nbx tiles of one (by, bz) group -- they
+      // stage the SAME A k-slice (bm x kchunk, a few MB) -- land on ONE
+      // XCD so the re-reads hit that XCD's 4 MiB L2 (34.5 TB/s aggregate)
+      // instead of HBM. The hardware dispatcher places linear id l on XCD
+      // l%8, so group g goes to XCD g%8 and its tiles get consecutive
+      // slots there. (The launcher pads gridDim.z until nby*nbz % 8 == 0;
+      // padded bz exits via the k_begin >= k_end guard below.)
+      int64_t l = ((int64_t)blockIdx.z * nby + blockIdx.y) * nbx + blockIdx.x;
+      const int xcd = (int)(l & 7);
+      int64_t j = l >> 3;
+      bx = (int)(j % nbx);
+      int g2 = (int)((j / nbx) * 8 + xcd);
+      by = g2 % nby;
+      bz = g2 / nby;
+    } else {
+      // T1 XCD swizzle (same as gemm_kernel)
+      const int64_t nwg = (int64_t)nbx * nby * gridDim.z;
+      if ((nwg & 7) == 0 && nwg > 8) {
+        int64_t id = ((int64_t)bz * nby + by) * nbx + bx;
+        const int64_t cpx = nwg >> 3;
+        id = (id & 7) * cpx + (id >> 3);
+        bx = (int)(id % nbx);
+        by = (int)((id / nbx) % nby);
+        bz = (int)(id / ((int64_t)nbx * nby));
+      }
     }
   }
   const int tid = threadIdx.x;
@@ -751,22 +793,35 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
       stage_kmaj_tr<BN2, GB2>(b_lds[cur ^ 1], B, ldB, k0 + 64, n0, wid,
                               lane, &ga_b);
     }
+    // DEEP=2: both 32-k halves' transpose-reads go out before ONE wait,
+    // doubling the MFMA burst per s_waitcnt (the 4-MFMA burst cannot hide
+    // 8 ds_read_b64_tr latencies). Register cost doubles the fragment set,
+    // so only tiles with FM2+FN2 <= 4 take it.
+    constexpr int DEEP = (FM2 + FN2 <= 4) ? 2 : 1;
 #pragma unroll
-    for (int kk = 0; kk < 64; kk += 32) {
-      bf16x8 af[FM2], bfr[FN2];
+    for (int kk = 0; kk < 64; kk += 32 * DEEP) {
+      bf16x8 af[DEEP][FM2], bfr[DEEP][FN2];
 #pragma unroll
-      for (int f = 0; f < FM2; ++f)
-        af[f] = tr_frag(ab[cur], kk + q * 8, wm + f * 16, l, BM2);
+      for (int d = 0; d < DEEP; ++d) {
 #pragma unroll
-      for (int f = 0; f < FN2; ++f)
-        bfr[f] = tr_frag(bb[cur], kk + q * 8, wn + f * 16, l, BN2);
-      tr_wait(af, bfr);
+        for (int f = 0; f < FM2; ++f)
+          af[d][f] = tr_frag<BM2>(ab[cur], kk + d * 32 + q * 8, wm + f * 16,
+                                  l, BM2);
 #pragma unroll
-      for (int fm = 0; fm < FM2; ++fm)
+        for (int f = 0; f < FN2; ++f)
+          bfr[d][f] = tr_frag<BN2>(bb[cur], kk + d * 32 + q * 8, wn + f * 16,
+                                   l, BN2);
+      }
+      tr_wait(af[0], bfr[0]);
+      if (DEEP == 2) tr_wait(af[DEEP - 1], bfr[DEEP - 1]);
 #pragma unroll
-        for (int fn = 0; fn < FN2; ++fn)
-          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
+      for (int d = 0; d < DEEP; ++d)
+#pragma unroll
+        for (int fm = 0; fm < FM2; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < FN2; ++fn)
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[d][fm], bfr[d][fn], acc[fm][fn], 0, 0, 0);
     }
     __syncthreads();
     cur ^= 1;
@@ -779,7 +834,12 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
-        if (SPLITK)
+        if (WS)
+          // workspace split-K: plain stores into this split's slice (alpha
+          // applied by splitk_reduce_k) -- no atomic serialization across
+          // the sk blocks sharing an output tile, no pre-zeroed C needed
+          ws[(int64_t)bz * M * N + (int64_t)row * N + col] = acc[fm][fn][r];
+        else if (SPLITK)
           atomicAdd(&C[(int64_t)row * ldC + col], alpha * acc[fm][fn][r]);
         else
           C[(int64_t)row * ldC + col] = alpha * acc[fm][fn][r];
@@ -792,16 +852,18 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
 template <typename T, typename OUT>
 static void gemm_dispatch(const GemmArgs& g, hipStream_t s);
 
-static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
+// Shared eligibility + shape plan for the tr16 TN path (used by both the
+// launcher and the bindings' workspace-size probe so the two can never
+// disagree). Returns false if the generic path must run; fills the
+// interior [M0 x N0], tile (bm, bn) and split-K factor for that interior.
+static bool tn_tr_plan(const GemmArgs& g, int* pM0, int* pN0, int* pbm,
+                       int* pbn, int* psk, int* pkchunk) {
   if (g.a_klast || g.b_klast || g.gather_a) return false;
   if (g.bias || g.relu || g.beta != 0.0f || g.batch > 1) return false;
-  if (g.ws) return false;  // caller committed to workspace split-K layout
   if (g.K % 64) return false;
   if (g.lda % 8) return false;
   if ((uintptr_t)g.A & 15) return false;
   if (g.gather_b) {
-    // gathered B (implicit wgrad): 16 B lane chunks must stay inside one
-    // contiguous cg run of NHWC x, aligned for glds
     const GatherDesc& gb = *g.gather_b;
     if (gb.Cg % 8 || gb.C % 8 || gb.c0 % 8) return false;
   } else {
@@ -809,14 +871,31 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
   }
   const int M0 = g.M & ~15, N0 = g.N & ~63;
   if (M0 == 0 || N0 == 0) return false;
-  if (g.gather_b && (M0 != g.M || N0 != g.N))
-    return false;  // a strip's column offset is not expressible in the
-                   // gather descriptor (kg shift crosses khw boundaries)
+  if (g.gather_b && (M0 != g.M || N0 != g.N)) return false;
+  if ((M0 != g.M || N0 != g.N) && (int64_t)M0 * N0 * g.K < (1LL << 33))
+    return false;
+  int bm = M0 % 64 ? (M0 % 32 ? 16 : 32) : 64;
+  int bn = (N0 % 128 == 0) ? 128 : 64;
+  if (N0 % bn) return false;
+  int64_t tiles = (int64_t)(M0 / bm) * (N0 / bn);
+  if (tiles >= 1024) return false;
+  int sk = 1, kchunk = g.K;
+  if (tiles < 384 && g.K >= 512) {
+    int want = (int)((512 + tiles - 1) / tiles);
+    int maxsk = (g.K + 255) / 256;
+    sk = want < maxsk ? want : maxsk;
+    if (sk < 1) sk = 1;
+    kchunk = ((g.K / sk + 63) / 64) * 64;
+    sk = (g.K + kchunk - 1) / kchunk;
+  }
+  *pM0 = M0; *pN0 = N0; *pbm = bm; *pbn = bn; *psk = sk; *pkchunk = kchunk;
+  return true;
+}
+
+static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
+  int M0, N0, bm, bn, sk, kchunk;
+  if (!tn_tr_plan(g, &M0, &N0, &bm, &bn, &sk, &kchunk)) return false;
   if (M0 != g.M || N0 != g.N) {
-    // edge strips cost 1-2 extra launches: only worth it when the
-    // interior carries real work (GoogLeNet's small inception wgrads
-    // measured slower split)
-    if ((int64_t)M0 * N0 * g.K < (1LL << 33)) return false;
     // edge strips go through the generic kernel; the [M0 x N0] interior
     // through the tr path. Strips: [0,M) x [N0,N) and [M0,M) x [0,N0).
     GemmArgs gi = g;
@@ -827,6 +906,7 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
       gt.N = g.N - N0;
       gt.B = (const void*)((const __bf16*)g.B + N0);
       gt.C = (void*)((float*)g.C + N0);
+      gt.ws = nullptr; gt.splitk = 1;
       gemm_dispatch<__bf16, float>(gt, s);
     }
     if (M0 != g.M) {
@@ -834,34 +914,31 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
       gt.M = g.M - M0; gt.N = N0;
       gt.A = (const void*)((const __bf16*)g.A + M0);
       gt.C = (void*)((float*)g.C + (int64_t)M0 * g.ldc);
+      gt.ws = nullptr; gt.splitk = 1;
       gemm_dispatch<__bf16, float>(gt, s);
     }
     return true;
   }
-  // tile config by shape: thin-M wgrads (inception Cog 16-48) use 16/32-row
-  // tiles; wide-N with N % 128 == 0 halves the barrier count per output
-  int bm = g.M % 64 ? (g.M % 32 ? 16 : 32) : 64;
-  int bn = (g.N % 128 == 0) ? 128 : 64;
-  if (g.N % bn) return false;  // caller's N0 strip guarantees N % 64 == 0
-  int64_t tiles = (int64_t)(g.M / bm) * (g.N / bn);
-  if (tiles >= 1024) return false;  // huge grids: generic path measures better
-  int sk = 1, kchunk = g.K;
-  if (tiles < 384 && g.K >= 512) {
-    int want = (int)((512 + tiles - 1) / tiles);
-    int maxsk = (g.K + 255) / 256;
-    sk = want < maxsk ? want : maxsk;
-    if (sk < 1) sk = 1;
-    kchunk = ((g.K / sk + 63) / 64) * 64;
-    sk = (g.K + kchunk - 1) / kchunk;
+  int skz = sk;
+  if (sk > 1) {
+    const int nby = g.M / bm;
+    while (((int64_t)nby * skz) & 7) ++skz;  // pad: enables L2 clustering
   }
-  dim3 grid(g.N / bn, g.M / bm, sk);
-  if (sk > 1)
+  dim3 grid(g.N / bn, g.M / bm, skz);
+  const bool use_ws = sk > 1 && g.ws != nullptr;
+  if (sk > 1 && !use_ws)
     (void)hipMemsetAsync(g.C, 0, (size_t)g.M * g.N * sizeof(float), s);
   GatherDesc gb = g.gather_b ? *g.gather_b : GatherDesc{};
 #define PS_TR_LAUNCH(BM_, BN_, WGM_, WGN_)                                  \
   do {                                                                      \
     if (g.gather_b) {                                                       \
-      if (sk > 1)                                                           \
+      if (use_ws)                                                           \
+        gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, true, true, true>           \
+            <<<grid, 256, 0, s>>>((const __bf16*)g.A, (const __bf16*)g.B,   \
+                                  (float*)g.C, g.M, g.N, g.K, g.lda,        \
+                                  g.ldb, g.ldc, g.alpha, kchunk, gb,        \
+                                  (float*)g.ws);                            \
+      else if (sk > 1)                                                      \
         gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, true, true>                 \
             <<<grid, 256, 0, s>>>((const __bf16*)g.A, (const __bf16*)g.B,   \
                                   (float*)g.C, g.M, g.N, g.K, g.lda,        \
@@ -871,7 +948,13 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
             <<<grid, 256, 0, s>>>((const __bf16*)g.A, (const __bf16*)g.B,   \
                                   (float*)g.C, g.M, g.N, g.K, g.lda,        \
                                   g.ldb, g.ldc, g.alpha, kchunk, gb);       \
-    } else if (sk > 1)                                                      \
+    } else if (use_ws)                                                      \
+      gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, true, false, true>            \
+          <<<grid, 256, 0, s>>>((const __bf16*)g.A, (const __bf16*)g.B,     \
+                                (float*)g.C, g.M, g.N, g.K, g.lda, g.ldb,   \
+                                g.ldc, g.alpha, kchunk, GatherDesc{},       \
+                                (float*)g.ws);                              \
+    else if (sk > 1)                                                        \
       gemm_tn_tr_kernel<BM_, BN_, WGM_, WGN_, true><<<grid, 256, 0, s>>>(   \
           (const __bf16*)g.A, (const __bf16*)g.B, (float*)g.C, g.M, g.N,    \
           g.K, g.lda, g.ldb, g.ldc, g.alpha, kchunk);                       \
@@ -887,6 +970,14 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
   else if (bn == 128) PS_TR_LAUNCH(64, 128, 2, 2);
   else PS_TR_LAUNCH(64, 64, 2, 2);
 #undef PS_TR_LAUNCH
+  if (use_ws) {
+    int64_t MN = (int64_t)g.M * g.N;
+    int64_t rb = cdiv64(MN, 256);
+    if (rb > 2048) rb = 2048;
+    splitk_reduce_k<float, false><<<dim3((unsigned)rb), 256, 0, s>>>(
+        (const float*)g.ws, (float*)g.C, nullptr, g.M, g.N, g.ldc, sk,
+        g.alpha, 0.0f, false);
+  }
   return true;
 }
 
@@ -980,6 +1071,16 @@ static void gemm_dispatch(const GemmArgs& g, hipStream_t s) {
 }
 
 extern "C" {
+
+// Returns the f32 workspace element count the tr16 TN path wants for this
+// GEMM (sk * interior M0 * N0), or 0 when it would not split / not take
+// the tr path. Callers that provide g->ws of this size get plain-store
+// split-K + a reduce instead of a memset + atomicAdd accumulation.
+int64_t ps_gemm_tn_tr_ws_elems(const GemmArgs* g) {
+  int M0, N0, bm, bn, sk, kchunk;
+  if (!tn_tr_plan(*g, &M0, &N0, &bm, &bn, &sk, &kchunk)) return 0;
+  return sk > 1 ? (int64_t)sk * M0 * N0 : 0;
+}
 
 void ps_gemm_f32(const GemmArgs* g, hipStream_t s) {
   gemm_dispatch<float, float>(*g, s);

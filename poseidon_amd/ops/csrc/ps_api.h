@@ -105,6 +105,7 @@ extern "C" {
 void ps_gemm_f32(const GemmArgs* g, hipStream_t s);
 void ps_gemm_bf16_f32out(const GemmArgs* g, hipStream_t s);
 void ps_gemm_bf16(const GemmArgs* g, hipStream_t s);
+int64_t ps_gemm_tn_tr_ws_elems(const GemmArgs* g);
 
 // elementwise.hip
 void ps_relu_fwd_f32(const float*, float*, int64_t, float, hipStream_t);
