@@ -629,7 +629,7 @@ __device__ inline int tr_swz(int row, int chunk) {
   return chunk ^ (p & MASK);
 }
 
-template <int COLS, bool GATHER = false>
+template <int COLS, bool GATHER = false, bool NT = false>
 __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
                                      int64_t ld, int k0, int c0, int wid,
                                      int lane,
@@ -674,9 +674,12 @@ __device__ inline void stage_kmaj_tr(__bf16* lds, const __bf16* src,
     } else {
       g2 = src + (int64_t)(k0 + row) * ld + c0 + sslot * 8;
     }
+    // NT (aux=2): the B stream is read by exactly ONE block; keep it from
+    // evicting the XCD-L2-resident A slice the clustered tiles share
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g2,
-        (__attribute__((address_space(3))) void*)(lds + ci * 512), 16, 0, 0);
+        (__attribute__((address_space(3))) void*)(lds + ci * 512), 16, 0,
+        NT ? 2 : 0);
   }
 }
 
@@ -775,8 +778,17 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
   const int l = lane & 15, q = lane >> 4;
 
   f32x4 acc[FM2][FN2] = {};
+  // NT-stream B only when ONE row of tiles exists (each B byte read by a
+  // single block): with several by-groups the L3 absorbs B re-reads and
+  // nt would force them all to HBM
+  const bool ntb = SPLITK && gridDim.y == 1;
+  if (ntb)
+    stage_kmaj_tr<BN2, GB2, true>(b_lds[0], B, ldB, k_begin, n0, wid, lane,
+                                  &ga_b);
+  else
+    stage_kmaj_tr<BN2, GB2, false>(b_lds[0], B, ldB, k_begin, n0, wid, lane,
+                                   &ga_b);
   stage_kmaj_tr<BM2>(a_lds[0], A, ldA, k_begin, m0, wid, lane);
-  stage_kmaj_tr<BN2, GB2>(b_lds[0], B, ldB, k_begin, n0, wid, lane, &ga_b);
   __syncthreads();
   unsigned ab[2], bb[2];
 #pragma unroll
@@ -790,8 +802,12 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
   for (int k0 = k_begin; k0 < k_end; k0 += 64) {
     if (k0 + 64 < k_end) {
       stage_kmaj_tr<BM2>(a_lds[cur ^ 1], A, ldA, k0 + 64, m0, wid, lane);
-      stage_kmaj_tr<BN2, GB2>(b_lds[cur ^ 1], B, ldB, k0 + 64, n0, wid,
-                              lane, &ga_b);
+      if (ntb)
+        stage_kmaj_tr<BN2, GB2, true>(b_lds[cur ^ 1], B, ldB, k0 + 64, n0,
+                                      wid, lane, &ga_b);
+      else
+        stage_kmaj_tr<BN2, GB2, false>(b_lds[cur ^ 1], B, ldB, k0 + 64, n0,
+                                       wid, lane, &ga_b);
     }
     // DEEP=2: both 32-k halves' transpose-reads go out before ONE wait,
     // doubling the MFMA burst per s_waitcnt (the 4-MFMA burst cannot hide
@@ -886,6 +902,11 @@ static bool tn_tr_plan(const GemmArgs& g, int* pM0, int* pN0, int* pbm,
     sk = want < maxsk ? want : maxsk;
     if (sk < 1) sk = 1;
     kchunk = ((g.K / sk + 63) / 64) * 64;
+    // A-slice residency: tiles of one (by,bz) group share a bm x kchunk
+    // A slice via their XCD's 4 MiB L2 -- keep it under ~2 MiB so the
+    // NT-streamed B tiles have room to pass through without evicting it
+    const int kcap = ((2 << 20) / (bm * 2) / 64) * 64;
+    if (M0 == bm && kchunk > kcap && g.K > kcap) kchunk = kcap;
     sk = (g.K + kchunk - 1) / kchunk;
   }
   *pM0 = M0; *pN0 = N0; *pbm = bm; *pbn = bn; *psk = sk; *pkchunk = kchunk;
