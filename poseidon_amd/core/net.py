@@ -433,9 +433,13 @@ class Net:
                  if l.type_name == "CONVOLUTION" and l.blobs
                  and l.blobs[0].data.is_cuda
                  and l.blobs[0].data.dtype == torch.float32]
-        if not convs:
+        ips = [l for l in self.layers
+               if l.type_name == "INNER_PRODUCT" and l.blobs
+               and l.blobs[0].data.is_cuda
+               and l.blobs[0].data.dtype == torch.float32]
+        if not convs and not ips:
             return
-        key = [id(l.blobs[0].data) for l in convs]
+        key = [id(l.blobs[0].data) for l in convs + ips]
         if self._repack_mt is None or self._repack_key != key:
             masters, wks, wkTs, Gs = [], [], [], []
             for l in convs:
@@ -454,6 +458,19 @@ class Net:
                 wks.append(wk)
                 wkTs.append(wkT)
                 Gs.append(G)
+            for l in ips:
+                # IP shadows ride the same table: masters viewed (N,K,1,1)
+                # make the khwc layout degenerate to row-major [N][K]; the
+                # transpose output is skipped (empty wkT)
+                w = l.blobs[0].data  # (1,1,N,K) contiguous
+                wk = torch.empty(l.N, l.K, dtype=torch.bfloat16,
+                                 device=w.device)
+                l._wk_cache = wk
+                masters.append(w.view(l.N, l.K, 1, 1))
+                wks.append(wk)
+                wkTs.append(torch.empty(0, dtype=torch.bfloat16,
+                                        device=w.device))
+                Gs.append(1)
             self._repack_mt = ops.repack_mt_prepare(masters, wks, wkTs, Gs)
             self._repack_key = key
         ops.repack_mt_run(self._repack_mt)

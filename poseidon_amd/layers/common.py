@@ -43,6 +43,8 @@ class InnerProductLayer(Layer):
             filler.fill(b, ip.bias_filler if ip.has("bias_filler") else None)
             self.blobs.append(b)
         self.fuse_relu = False  # set by Net's IP+ReLU fusion pass
+        self._wk_cache = None   # bf16 [N,K] shadow, refreshed each step by
+                                # the net-level repack table
         # SFB: when set by the distributed solver, backward defers the dW GEMM
         self.sfb_active = False
         self.sfb_factors: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
@@ -58,7 +60,7 @@ class InnerProductLayer(Layer):
         b = self.blobs[1].data.view(-1) if self.bias_term else None
         top[0].data = ops.linear_forward(
             x, self.blobs[0].data.view(self.N, self.K), b,
-            fuse_relu=self.fuse_relu)
+            fuse_relu=self.fuse_relu, w_shadow=self._wk_cache)
 
     def backward(self, top, propagate_down, bottom) -> None:
         x = bottom[0].data.reshape(self.M, self.K)
@@ -67,7 +69,7 @@ class InnerProductLayer(Layer):
         dx, dw, db = ops.linear_backward(
             x, self.blobs[0].data.view(self.N, self.K), dy,
             need_dx=propagate_down[0], need_dw=need_dw,
-            has_bias=self.bias_term)
+            has_bias=self.bias_term, w_shadow=self._wk_cache)
         if self.sfb_active:
             # Sufficient factors a=dy [M,N], b=x [M,K]; ∇W = aᵀ·b is
             # reconstructed after the all-gather (solver/sfb.py).

@@ -193,10 +193,17 @@ Tensor gemm(const Tensor& A, const Tensor& B, int M, int N, int K,
 // ---------------------------------------------------------------------------
 
 Tensor linear_forward(const Tensor& x, const Tensor& w,
-                      const c10::optional<Tensor>& bias, bool fuse_relu) {
+                      const c10::optional<Tensor>& bias, bool fuse_relu,
+                      const c10::optional<Tensor>& w_shadow) {
   check_float_like(x, "x");
   auto xc = x.contiguous();
-  auto wc = weight_shadow(w, is_bf16(x));
+  // per-step bf16 shadow from the net-level repack table (skips the
+  // per-call f32->bf16 cast of the fc masters: VGG casts 138M params
+  // twice per step otherwise)
+  auto wc = (w_shadow.has_value() && is_bf16(x) &&
+             w_shadow->numel() == w.numel())
+                ? w_shadow->view(w.sizes())
+                : weight_shadow(w, is_bf16(x));
   int M = xc.size(0), K = xc.size(1), N = wc.size(0);
   TORCH_CHECK(wc.size(1) == K, "linear: K mismatch");
   Tensor y = at::empty({M, N}, x.options());
@@ -213,10 +220,14 @@ Tensor linear_forward(const Tensor& x, const Tensor& w,
 
 std::vector<c10::optional<Tensor>> linear_backward(
     const Tensor& x, const Tensor& w, const Tensor& dy,
-    bool need_dx, bool need_dw, bool has_bias) {
+    bool need_dx, bool need_dw, bool has_bias,
+    const c10::optional<Tensor>& w_shadow) {
   check_float_like(dy, "dy");
   auto xc = x.contiguous();
-  auto wc = weight_shadow(w, is_bf16(dy));
+  auto wc = (w_shadow.has_value() && is_bf16(dy) &&
+             w_shadow->numel() == w.numel())
+                ? w_shadow->view(w.sizes())
+                : weight_shadow(w, is_bf16(dy));
   auto dyc = dy.contiguous();
   int M = xc.size(0), K = xc.size(1), N = wc.size(0);
   c10::optional<Tensor> dx, dw, db;
@@ -1349,10 +1360,12 @@ std::vector<Tensor> repack_mt_prepare(std::vector<Tensor> masters,
     const Tensor& m = masters[t];
     TORCH_CHECK(m.is_cuda() && m.is_contiguous() &&
                 m.scalar_type() == at::kFloat && m.dim() == 4 &&
-                wks[t].scalar_type() == at::kBFloat16 &&
-                wkTs[t].scalar_type() == at::kBFloat16,
+                wks[t].scalar_type() == at::kBFloat16,
                 "repack_mt_prepare: f32 NCHW masters + bf16 outputs");
-    descs[t] = {m.data_ptr<float>(), wks[t].data_ptr(), wkTs[t].data_ptr(),
+    const bool has_t = wkTs[t].numel() > 0;
+    TORCH_CHECK(!has_t || wkTs[t].scalar_type() == at::kBFloat16);
+    descs[t] = {m.data_ptr<float>(), wks[t].data_ptr(),
+                has_t ? wkTs[t].data_ptr() : nullptr,
                 m.numel(), (int)m.size(0), (int)m.size(1), (int)m.size(2),
                 (int)m.size(3), (int)Gs[t], (int)wks[t].size(1)};
     for (int64_t off = 0; off < m.numel(); off += CHUNK)

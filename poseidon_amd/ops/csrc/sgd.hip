@@ -157,8 +157,9 @@ __global__ void repack_mt_k(const MTRepackDesc* __restrict__ descs,
     const float v = d.src[i];
     const int kg = (kkh * KW + kkw) * Cig + ci;
     ((__bf16*)d.wk)[(int64_t)co * d.ldk + kg] = (__bf16)v;
-    ((__bf16*)d.wkT)[((int64_t)(co / Cog) * Kg + kg) * Cog + co % Cog] =
-        (__bf16)v;
+    if (d.wkT)  // IP shadows need only the row-major form
+      ((__bf16*)d.wkT)[((int64_t)(co / Cog) * Kg + kg) * Cog + co % Cog] =
+          (__bf16)v;
   }
 }
 

@@ -140,9 +140,10 @@ def conv2d_backward_weight_acc(x: torch.Tensor, colT, dy: torch.Tensor,
 
 def linear_forward(x: torch.Tensor, w: torch.Tensor,
                    b: Optional[torch.Tensor],
-                   fuse_relu: bool = False) -> torch.Tensor:
+                   fuse_relu: bool = False,
+                   w_shadow: Optional[torch.Tensor] = None) -> torch.Tensor:
     if x.is_cuda:
-        return _ext().linear_forward(x, w, b, fuse_relu)
+        return _ext().linear_forward(x, w, b, fuse_relu, w_shadow)
     y = x.matmul(w.t())
     if b is not None:
         y = y + b
@@ -152,9 +153,11 @@ def linear_forward(x: torch.Tensor, w: torch.Tensor,
 
 
 def linear_backward(x: torch.Tensor, w: torch.Tensor, dy: torch.Tensor,
-                    need_dx: bool, need_dw: bool, has_bias: bool):
+                    need_dx: bool, need_dw: bool, has_bias: bool,
+                    w_shadow: Optional[torch.Tensor] = None):
     if dy.is_cuda:
-        return _ext().linear_backward(x, w, dy, need_dx, need_dw, has_bias)
+        return _ext().linear_backward(x, w, dy, need_dx, need_dw, has_bias,
+                                      w_shadow)
     dx = dy.matmul(w) if need_dx else None
     dw = dy.t().matmul(x) if need_dw else None
     db = dy.sum(dim=0) if has_bias else None
