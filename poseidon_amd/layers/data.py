@@ -287,21 +287,93 @@ class ImageDataLayer(_PrefetchingDataLayer):
 
 @register_layer("HDF5_DATA")
 class HDF5DataLayer(Layer):
+    """Reads "data"/"label" datasets from the HDF5 files listed in the
+    source file, cycling through files and rows
+    (hdf5_data_layer.cpp:27-112; contiguous-layout HDF5 parsed by the
+    pure-python data/hdf5_io.py reader -- no libhdf5 in this image)."""
+
+    exact_num_bottom = 0
+
     def layer_setup(self, bottom, top) -> None:
-        raise RuntimeError(
-            "HDF5_DATA requires h5py, which this image does not ship; "
-            "convert to PDB with tools/convert_dataset.py")
+        from ..data.hdf5_io import Hdf5Reader
+        hp = self.param.ensure("hdf5_data_param")
+        self.batch = int(hp.batch_size)
+        with open(hp.source) as f:
+            self.files = [ln.strip() for ln in f if ln.strip()]
+        if not self.files:
+            raise ValueError(f"{hp.source}: empty HDF5 source list")
+        self._file_idx = 0
+        self._row = 0
+        self._load(Hdf5Reader(self.files[0]))
+
+    def _load(self, r) -> None:
+        self._data = r.get("data").astype(np.float32)
+        self._label = r.get("label").astype(np.float32).reshape(-1)
+        if self._data.shape[0] != self._label.shape[0]:
+            raise ValueError("data/label row mismatch")
 
     def reshape(self, bottom, top) -> None:
+        shape = (self.batch,) + tuple(self._data.shape[1:])
+        top[0].reshape(shape if len(shape) == 4 else
+                       shape + (1,) * (4 - len(shape)))
+        if len(top) > 1:
+            top[1].reshape(self.batch)
+
+    def forward(self, bottom, top) -> None:
+        from ..data.hdf5_io import Hdf5Reader
+        rows = []
+        labels = []
+        for _ in range(self.batch):
+            if self._row >= self._data.shape[0]:
+                self._file_idx = (self._file_idx + 1) % len(self.files)
+                self._load(Hdf5Reader(self.files[self._file_idx]))
+                self._row = 0
+            rows.append(self._data[self._row])
+            labels.append(self._label[self._row])
+            self._row += 1
+        c = ctx()
+        batch = np.stack(rows)
+        top[0].data = torch.from_numpy(batch).to(
+            c.torch_device, c.compute_dtype).view(top[0].shape)
+        if len(top) > 1:
+            top[1].data = torch.tensor(labels, dtype=torch.float32,
+                                       device=c.torch_device)
+
+    def backward(self, top, propagate_down, bottom) -> None:
         pass
 
 
 @register_layer("HDF5_OUTPUT")
 class HDF5OutputLayer(Layer):
+    """Accumulates (data, label) bottoms across forwards and writes them
+    as the "data"/"label" datasets of file_name on finalize()
+    (hdf5_output_layer.cpp:16-44)."""
+
+    exact_num_bottom = 2
+    exact_num_top = 0
+
     def layer_setup(self, bottom, top) -> None:
-        raise RuntimeError("HDF5_OUTPUT requires h5py (not available)")
+        op = self.param.ensure("hdf5_output_param")
+        self.file_name = op.file_name
+        self._rows: list = []
+        self._labels: list = []
 
     def reshape(self, bottom, top) -> None:
+        pass
+
+    def forward(self, bottom, top) -> None:
+        self._rows.append(
+            bottom[0].data.detach().to(torch.float32).cpu().numpy().copy())
+        self._labels.append(
+            bottom[1].data.detach().to(torch.float32).cpu().numpy().copy())
+
+    def finalize(self) -> None:
+        from ..data.hdf5_io import Hdf5Writer
+        with Hdf5Writer(self.file_name) as w:
+            w.put("data", np.concatenate(self._rows))
+            w.put("label", np.concatenate(self._labels))
+
+    def backward(self, top, propagate_down, bottom) -> None:
         pass
 
 
