@@ -112,17 +112,12 @@ def test_image_data_layer(tmp_path):
     assert sorted(net.blobs["label"].data.tolist()) == [0.0, 0.0, 1.0, 1.0]
 
 
-def test_hdf5_gating():
-    try:
-        import h5py  # noqa: F401
-        have = True
-    except ImportError:
-        have = False
-    if have:
-        pytest.skip("h5py present; gating path not applicable")
-    with pytest.raises(Exception, match="h5py"):
+def test_hdf5_missing_source_errors():
+    # HDF5 layers are fully implemented (data/hdf5_io.py, no h5py needed);
+    # a missing source list must fail loudly, not silently no-op
+    with pytest.raises(Exception):
         _net("""
             name: "h5"
             layers { name: "d" type: HDF5_DATA top: "data"
-                     hdf5_data_param { source: "x.txt" batch_size: 2 } }
+                     hdf5_data_param { source: "/nonexistent.txt" batch_size: 2 } }
         """)
