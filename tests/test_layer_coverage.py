@@ -1,7 +1,6 @@
 """Coverage for the layer types not exercised elsewhere: ARGMAX, IM2COL,
 IMAGE_DATA, INFOGAIN_LOSS, MULTINOMIAL_LOGISTIC_LOSS, SILENCE, THRESHOLD
-(HDF5_DATA/HDF5_OUTPUT are gated on h5py availability and covered by
-their gating test below)."""
+(HDF5_DATA/HDF5_OUTPUT have their own suite in test_hdf5.py)."""
 
 import numpy as np
 import pytest
