@@ -66,15 +66,19 @@ class InnerProductLayer(Layer):
         x = bottom[0].data.reshape(self.M, self.K)
         dy = top[0].diff.reshape(self.M, self.N)
         need_dw = not self.sfb_active
+        dw_acc = None
+        if need_dw and dy.is_cuda and self.blobs[0].diff.dtype == torch.float32:
+            dw_acc = self.blobs[0].diff.view(self.N, self.K)
         dx, dw, db = ops.linear_backward(
             x, self.blobs[0].data.view(self.N, self.K), dy,
             need_dx=propagate_down[0], need_dw=need_dw,
-            has_bias=self.bias_term, w_shadow=self._wk_cache)
+            has_bias=self.bias_term, w_shadow=self._wk_cache,
+            dw_acc=dw_acc)
         if self.sfb_active:
             # Sufficient factors a=dy [M,N], b=x [M,K]; ∇W = aᵀ·b is
             # reconstructed after the all-gather (solver/sfb.py).
             self.sfb_factors = (dy, x)
-        else:
+        elif dw_acc is None:
             self.blobs[0].diff.view(self.N, self.K).add_(dw)
         if self.bias_term:
             self.blobs[1].diff.view(-1).add_(db)

@@ -121,7 +121,11 @@ __device__ inline void stage_glds(T* lds, const T* __restrict__ src,
     // full-slot permutations also kill bank conflicts but destroy the
     // request coalescer's lane-order contiguity (13x slower at L3-resident
     // sizes); the pair swap keeps 32 B runs contiguous and still cuts
-    // ds_read_b128 conflicts 8-way -> 4-way.
+    // ds_read_b128 conflicts 8-way -> 4-way. (A 2-bit slot swizzle keyed
+    // on row bits 1-2 makes the fragment reads fully conflict-free on
+    // paper, but flipping slot bit 0 reorders 16 B chunks within 32 B
+    // pairs on the WRITE side and measured -5%% end-to-end on AlexNet/VGG
+    // -- the global request coalescer penalty outweighs the LDS win.)
     const int kc = (slot ^ (((row >> 2) & 1) << 1)) * EPB;
     const T* g = GATHER ? gather_addr<T>(*ga, row0 + row, k0 + kc)
                         : src + (int64_t)(row0 + row) * lda + k0 + kc;

@@ -154,10 +154,13 @@ def linear_forward(x: torch.Tensor, w: torch.Tensor,
 
 def linear_backward(x: torch.Tensor, w: torch.Tensor, dy: torch.Tensor,
                     need_dx: bool, need_dw: bool, has_bias: bool,
-                    w_shadow: Optional[torch.Tensor] = None):
+                    w_shadow: Optional[torch.Tensor] = None,
+                    dw_acc: Optional[torch.Tensor] = None):
+    """dw_acc: fp32 [N,K] buffer the weight grad is ACCUMULATED into
+    (beta=1 GEMM); when set, the returned dw is None."""
     if dy.is_cuda:
         return _ext().linear_backward(x, w, dy, need_dx, need_dw, has_bias,
-                                      w_shadow)
+                                      w_shadow, dw_acc)
     dx = dy.matmul(w) if need_dx else None
     dw = dy.t().matmul(x) if need_dw else None
     db = dy.sum(dim=0) if has_bias else None
