@@ -66,8 +66,14 @@ class InnerProductLayer(Layer):
         x = bottom[0].data.reshape(self.M, self.K)
         dy = top[0].diff.reshape(self.M, self.N)
         need_dw = not self.sfb_active
+        # beta=1 in-GEMM grad accumulation measured slightly SLOWER than
+        # the separate add_ on AlexNet fc shapes (same-box A/B); opt in
+        # with PS_DW_ACC=1
         dw_acc = None
-        if need_dw and dy.is_cuda and self.blobs[0].diff.dtype == torch.float32:
+        import os as _os
+        if (need_dw and dy.is_cuda
+                and self.blobs[0].diff.dtype == torch.float32
+                and _os.environ.get("PS_DW_ACC", "0") == "1"):
             dw_acc = self.blobs[0].diff.view(self.N, self.K)
         dx, dw, db = ops.linear_backward(
             x, self.blobs[0].data.view(self.N, self.K), dy,

@@ -782,10 +782,13 @@ void gemm_tn_tr_kernel(const __bf16* __restrict__ A,
   const int l = lane & 15, q = lane >> 4;
 
   f32x4 acc[FM2][FN2] = {};
-  // NT-stream B only when ONE row of tiles exists (each B byte read by a
-  // single block): with several by-groups the L3 absorbs B re-reads and
-  // nt would force them all to HBM
-  const bool ntb = SPLITK && gridDim.y == 1;
+  // NT-stream B only when: one row of tiles (each B byte read once), B is
+  // a materialized operand (a gather re-reads warm x kh*kw-fold -- nt
+  // would push every re-read to HBM), and the stream is too big to be
+  // cache-resident anyway (small colT matrices are L2/L3-HOT from the
+  // im2col that just wrote them; nt measured -4% on AlexNet/GoogLeNet)
+  const bool ntb = SPLITK && gridDim.y == 1 && !GB2 &&
+                   (int64_t)K * N * 2 > (192LL << 20);
   if (ntb)
     stage_kmaj_tr<BN2, GB2, true>(b_lds[0], B, ldB, k_begin, n0, wid, lane,
                                   &ga_b);
