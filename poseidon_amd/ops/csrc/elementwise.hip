@@ -276,18 +276,11 @@ __global__ void colsum_flat_k(const T* __restrict__ in,
 #pragma unroll
     for (int j = 0; j < VEC; ++j) acc[q][j] = 0.f;
   int p = 0;
-  // two outstanding loads per trip: the single-stream form measured
-  // 68.9% wave-parked (pure HBM latency) on VGG bias grads
-  for (; i + stride < nvec; i += 2 * stride) {
-    vecT v = *((const vecT*)in + i);
-    vecT v2 = *((const vecT*)in + i + stride);
-#pragma unroll
-    for (int j = 0; j < VEC; ++j) acc[p][j] += to_f32(v[j]);
-    if (PH > 1 && ++p == PH) p = 0;
-#pragma unroll
-    for (int j = 0; j < VEC; ++j) acc[p][j] += to_f32(v2[j]);
-    if (PH > 1 && ++p == PH) p = 0;
-  }
+  // NOTE r2: a dual-outstanding-load variant and a 4096-block grid were
+  // both tried for the 68.9%-wave-parked profile and REGRESSED (PH=3
+  // launches 77 -> 141 us on AlexNet conv bias: doubling the block count
+  // doubles the terminal atomicAdd traffic onto C addresses, and the
+  // second load stream spills the PH accumulator registers).
   for (; i < nvec; i += stride) {
     vecT v = *((const vecT*)in + i);
 #pragma unroll
@@ -313,7 +306,7 @@ static inline int colsum_gcd(int a, int b) {
 
 static inline dim3 colsum_flat_grid(int64_t nvec) {
   int64_t blocks = cdiv64(nvec, 256);
-  if (blocks > 4096) blocks = 4096;
+  if (blocks > 2048) blocks = 2048;
   return dim3((unsigned)blocks);
 }
 
