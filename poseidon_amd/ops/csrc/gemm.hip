@@ -363,7 +363,7 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
                  float alpha, float beta,
                  float* __restrict__ ws, int kchunk,
                  GatherDesc ga_a = {}, GatherDesc ga_b = {},
-                 bool relu = false) {
+                 bool relu = false, int xcd2d = 0) {
   using TR = GemmTraits<T>;
   constexpr int BK = TR::BK, RS = TR::RS;
   constexpr int FM = BM / WGM / 16, FN = BN / WGN / 16;
@@ -376,7 +376,24 @@ void gemm_kernel(const T* __restrict__ Abase, const T* __restrict__ Bbase,
   // otherwise. z (split-K slice / batch) folds into the flatten so the
   // remap stays a permutation of the whole grid.
   int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
-  {
+  if (xcd2d != 0) {
+    // 2D XCD super-tiling for re-read-bound grids (small K, many tiles --
+    // the fc-wgrad class: fc6 dW is 4096x9216x256, each A panel re-read
+    // by 72 tile columns and each B panel by 32 rows from HBM/L3). Each
+    // XCD owns an ry x cx RECTANGLE of tiles whose A+B panels fit its
+    // 4 MiB L2, so panel re-reads inside the rectangle are L2 hits.
+    // launch side guarantees nby % ry == 0, nbx % cx == 0 and
+    // (nby/ry)*(nbx/cx) == 8; grid.z is 1 for this class.
+    const int ry = xcd2d >> 16, cx = xcd2d & 0xFFFF;
+    const int nbx = gridDim.x;
+    const int regs_x = nbx / cx;
+    int64_t l = ((int64_t)bz * gridDim.y + by) * nbx + bx;
+    const int k = (int)(l & 7);
+    int64_t j = l >> 3;
+    bx = (k % regs_x) * cx + (int)(j % cx);
+    by = (k / regs_x) * ry + (int)(j / cx);
+    bz = 0;
+  } else {
     const int nbx = gridDim.x, nby = gridDim.y;
     const int64_t nwg = (int64_t)nbx * nby * gridDim.z;
     if ((nwg & 7) == 0 && nwg > 8) {
@@ -1026,6 +1043,28 @@ static void launch_tile(const GemmArgs& g, hipStream_t s) {
     constexpr int TBK = GemmTraits<T>::BK;
     kchunk = cdiv(cdiv(g.K, g.splitk), TBK) * TBK;
   }
+  // 2D XCD clustering: re-read-bound shallow-K grids with full tiles
+  int xcd2d = 0;
+  if (!sk && g.batch == 1 && g.K <= 1024 && g.M % BM == 0 && g.N % BN == 0) {
+    const int nby = g.M / BM, nbx = g.N / BN;
+    if ((int64_t)nby * nbx >= 512) {
+      // pick (ry, cx): ry | nby, cx | nbx, (nby/ry)*(nbx/cx) == 8,
+      // minimizing the L2 working set ry*BM + cx*BN (K common factor)
+      int64_t best = -1;
+      for (int gy = 1; gy <= 8; gy <<= 1) {
+        const int gx = 8 / gy;
+        if (nby % gy || nbx % gx) continue;
+        const int ry = nby / gy, cx = nbx / gx;
+        const int64_t ws_bytes =
+            ((int64_t)ry * BM + (int64_t)cx * BN) * g.K * (int64_t)sizeof(T);
+        if (ws_bytes > (3LL << 20)) continue;  // must fit the 4 MiB L2
+        if (best < 0 || ws_bytes < best) {
+          best = ws_bytes;
+          xcd2d = (ry << 16) | cx;
+        }
+      }
+    }
+  }
   if (sk) {
     gemm_kernel<T, OUT, BM, BN, WGM, WGN, AK, BK_, HB, true, GA, GB>
         <<<grid, block, 0, s>>>(
@@ -1049,7 +1088,7 @@ static void launch_tile(const GemmArgs& g, hipStream_t s) {
         <<<grid, block, 0, s>>>(
             (const T*)g.A, (const T*)g.B, (OUT*)g.C, g.bias, g.M, g.N, g.K,
             g.lda, g.ldb, g.ldc, g.strideA, g.strideB, g.strideC, g.alpha,
-            g.beta, nullptr, 0, da, db, g.relu);
+            g.beta, nullptr, 0, da, db, g.relu, xcd2d);
   }
 }
 
