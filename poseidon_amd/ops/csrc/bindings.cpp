@@ -556,9 +556,11 @@ Tensor conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
   // colT may carry zero pad columns (ps_colT_ld): run the GEMM over the
   // padded width too (zero B columns -> zero dwk columns, skipped below).
   // Implicit wgrads pad the same way via the gather's kg_max bound, which
-  // makes them tr16-eligible (N % 64 == 0, no edge strips).
-  const int Kgw = implicit ? (G == 1 ? (Kg + 63) & ~63 : Kg)
-                           : (int)(Kcol / G);
+  // makes them tr16-eligible (N % 64 == 0, no edge strips) -- grouped
+  // convs included (per-group dwk slices stay contiguous at c_off =
+  // grp*Cog*Kgw; AlexNet conv2 Kg=1200 was stuck on the register-staged
+  // gather TN path at 403 us/group without this).
+  const int Kgw = implicit ? (Kg + 63) & ~63 : (int)(Kcol / G);
 
   // fp32 gradient accumulation regardless of activation dtype
   const bool prez = dwk_buf.has_value() && dwk_buf->numel() == (int64_t)Co * Kgw;
