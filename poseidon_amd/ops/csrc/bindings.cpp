@@ -18,6 +18,7 @@
 #include <algorithm>
 #include <atomic>
 #include <cmath>
+#include <cstring>
 
 #include "ps_api.h"
 
@@ -115,11 +116,19 @@ void run_gemm(const Tensor& A, const Tensor& B, Tensor& C,
   g.gather_a = gather_a;
   g.gather_b = gather_b;
   g.relu = relu;
+  g.c_prezeroed = c_prezeroed;
   // tr16 TN path with split-K: hand it a workspace so the splits store
   // plain partials + one reduce instead of memset + per-element atomicAdd
   // chains (VGG wgrad: ~2M contended atomics per GEMM removed)
   Tensor trws;
-  if (in_bf16 && out_f32) {
+  // workspace split-K measured a net LOSS end-to-end vs the atomic form
+  // (GoogLeNet -4%, VGG -2%: the reduce launch + ws traffic outweigh the
+  // atomic contention); PS_TR_WS=1 re-enables it for experiments
+  static const bool tr_ws_on = [] {
+    const char* e = getenv("PS_TR_WS");
+    return e != nullptr && strcmp(e, "1") == 0;
+  }();
+  if (tr_ws_on && in_bf16 && out_f32) {
     int64_t trw = ps_gemm_tn_tr_ws_elems(&g);
     if (trw > 0 && trw * 4 <= (256LL << 20)) {
       trws = at::empty({trw}, C.options().dtype(at::kFloat));

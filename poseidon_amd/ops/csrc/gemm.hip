@@ -971,7 +971,7 @@ static bool try_gemm_tn_tr(const GemmArgs& g, hipStream_t s) {
   }
   dim3 grid(g.N / bn, g.M / bm, skz);
   const bool use_ws = sk > 1 && g.ws != nullptr;
-  if (sk > 1 && !use_ws)
+  if (sk > 1 && !use_ws && !g.c_prezeroed)
     (void)hipMemsetAsync(g.C, 0, (size_t)g.M * g.N * sizeof(float), s);
   GatherDesc gb = g.gather_b ? *g.gather_b : GatherDesc{};
 #define PS_TR_LAUNCH(BM_, BN_, WGM_, WGN_)                                  \

@@ -46,6 +46,9 @@ struct GemmArgs {
   const GatherDesc* gather_b;
   // fused epilogue: y = max(y, 0) (conv/IP + in-place ReLU pairs)
   bool relu;
+  // caller guarantees C is already zero (net-level zero_mt): atomic
+  // split-K paths skip their per-launch memset
+  bool c_prezeroed;
 };
 
 struct PoolGeom {
