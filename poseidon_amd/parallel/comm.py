@@ -52,8 +52,13 @@ def init_distributed(backend: Optional[str] = None) -> bool:
         # (before capture begins) instead of lazily inside it.
         os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "0")
         kw["device_id"] = c.torch_device
-    dist.init_process_group(backend=backend, rank=c.rank,
-                            world_size=c.world_size, **kw)
+    try:
+        dist.init_process_group(backend=backend, rank=c.rank,
+                                world_size=c.world_size, **kw)
+    except TypeError:
+        # older/variant torch builds without the device_id kwarg
+        dist.init_process_group(backend=backend, rank=c.rank,
+                                world_size=c.world_size)
     return True
 
 
