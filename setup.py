@@ -38,6 +38,12 @@ _ld = []
 if os.environ.get("PS_SANITIZE") == "address":
     _cxx += ["-fsanitize=address", "-fno-omit-frame-pointer", "-g"]
     _ld += ["-fsanitize=address"]
+elif os.environ.get("PS_SANITIZE") == "thread":
+    # TSAN for the host-side binding code (the HIP device code is outside
+    # TSAN's scope; python-thread handoffs are covered by the prefetch
+    # stress test instead, tests/test_prefetch_stress.py)
+    _cxx += ["-fsanitize=thread", "-fno-omit-frame-pointer", "-g"]
+    _ld += ["-fsanitize=thread"]
 
 setup(
     name="poseidon_amd_hip",
