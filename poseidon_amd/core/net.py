@@ -111,6 +111,8 @@ class Net:
         self._zero_mt = None  # multi-tensor zero table (GPU, built lazily)
         self._repack_mt = None  # multi-tensor conv-weight repack table
         self._unpack_mt = None  # deferred conv-wgrad unpack table
+        self._colsum_mt = None   # deferred bias-colsum table
+        self._colsum_prev_key = None
         self._repack_key = None
         self._loss_marks: Dict[int, List] = {}
         # inter-branch stream parallelism (inception-style nets): built by
@@ -421,6 +423,32 @@ class Net:
         ops.unpack_mt_run(self._unpack_mt[0])
         for l in convs:
             l._unpack_pending = False
+        # deferred conv bias grads: one colsum_mt launch when the dy
+        # identities held steady for a full iteration (the persistent
+        # dgrad/split caches make them steady from iteration 2; eager
+        # per-layer colsums cover the settling iterations so capture
+        # never records a table upload)
+        pend = [(l, l._pending_colsum) for l in self.layers
+                if getattr(l, "_pending_colsum", None) is not None]
+        if pend:
+            key = [(dy.data_ptr(), db.data_ptr(), tuple(dy.shape))
+                   for _, (dy, db) in pend]
+            if self._colsum_mt is not None and self._colsum_mt[1] == key:
+                ops.colsum_mt_run(self._colsum_mt[0],
+                                  pend[0][1][0].dtype == torch.bfloat16)
+            elif key == self._colsum_prev_key:
+                self._colsum_mt = (ops.colsum_mt_prepare(
+                    [dy for _, (dy, db) in pend],
+                    [db for _, (dy, db) in pend]), key)
+                ops.colsum_mt_run(self._colsum_mt[0],
+                                  pend[0][1][0].dtype == torch.bfloat16)
+            else:
+                self._colsum_mt = None
+                for _, (dy, db) in pend:
+                    ops.colsum_acc(dy, db)
+            self._colsum_prev_key = key
+            for l, _ in pend:
+                l._pending_colsum = None
 
     def _maybe_mt_repack(self) -> None:
         """One repack_mt kernel refreshes every conv's bf16 khwc shadow +

@@ -115,8 +115,16 @@ class ConcatLayer(Layer):
     def backward(self, top, propagate_down, bottom) -> None:
         chan4 = self.dim == 1 and len(bottom[0].shape) == 4
         if chan4 and all(propagate_down):
+            cache = getattr(self, "_split_cache", None)
             parts = ops.split_channels(top[0].diff,
-                                       [b.shape[1] for b in bottom])
+                                       [b.shape[1] for b in bottom],
+                                       outs_cache=cache)
+            if parts and parts[0].is_cuda:
+                if (cache is None or len(cache) != len(parts) or any(
+                        c.data_ptr() != p.data_ptr()
+                        for c, p in zip(cache, parts))):
+                    self._split_cache = parts
+                parts = self._split_cache
             for b, d in zip(bottom, parts):
                 b.diff = d
             return

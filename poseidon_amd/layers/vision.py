@@ -50,6 +50,9 @@ class ConvolutionLayer(Layer):
         self._dwk_cache = None  # persistent khwc wgrad scratch (zeroed per
                                 # iter by the net zero table)
         self._unpack_pending = False  # deferred wgrad unpack flag
+        self._dx_cache = {}     # persistent dgrad outputs (stable dy
+                                # identities for downstream batching)
+        self._pending_colsum = None  # (dy, db) for the net-level batch
         channels = bottom[0].channels
         assert channels % self.group == 0 and self.num_output % self.group == 0
 
@@ -97,6 +100,7 @@ class ConvolutionLayer(Layer):
         # solver only sets defer_unpack when there is no reducer).
         defer = (single and getattr(self, "defer_unpack", False)
                  and self._dwk_cache is not None)
+        defer_db = defer and self.bias_term
         for i, (bo, t) in enumerate(zip(bottom, top)):
             dy = t.diff
             db = self.blobs[1].diff.view(-1) if self.bias_term else None
@@ -106,7 +110,10 @@ class ConvolutionLayer(Layer):
                 bo.data, colT, dy, self.blobs[0].diff, db,
                 self.stride, self.pad, self.group,
                 dwk_buf=self._dwk_cache if single else None,
-                skip_unpack=defer)
+                skip_unpack=defer, skip_db=defer_db)
+            if defer_db:
+                # bias grad joins the ONE net-level colsum_mt launch
+                self._pending_colsum = (dy, db)
             if defer:
                 if dwk.data_ptr() != self._dwk_cache.data_ptr():
                     raise RuntimeError(
@@ -126,9 +133,14 @@ class ConvolutionLayer(Layer):
                         or cur.numel() != dwk.numel()):
                     self._dwk_cache = dwk
             if propagate_down[i]:
-                bo.diff = ops.conv2d_backward_input(
+                dx = ops.conv2d_backward_input(
                     w, dy, bo.shape, self.stride, self.pad, self.group,
-                    wkT_cache=wkT)
+                    wkT_cache=wkT, dx_out=self._dx_cache.get(i))
+                cur = self._dx_cache.get(i)
+                if dx.is_cuda and (cur is None
+                                   or cur.data_ptr() != dx.data_ptr()):
+                    self._dx_cache[i] = dx
+                bo.diff = self._dx_cache.get(i, dx) if dx.is_cuda else dx
         self._colT = []
 
     def extra_zero_buffers(self) -> List[torch.Tensor]:
@@ -201,9 +213,16 @@ class PoolingLayer(Layer):
             bottom[0].diff = ops.pool_ave_backward(
                 dy, bottom[0].shape, self.kernel, self.stride, self.pad)
         else:  # MAX and STOCHASTIC both scatter via stored mask
-            bottom[0].diff = ops.pool_max_backward(
+            dx = ops.pool_max_backward(
                 dy, self._mask, bottom[0].shape, self.kernel, self.stride,
-                self.pad)
+                self.pad, dx_out=getattr(self, "_dx_cache", None))
+            if dx.is_cuda:
+                cur = getattr(self, "_dx_cache", None)
+                if cur is None or cur.data_ptr() != dx.data_ptr():
+                    self._dx_cache = dx
+                bottom[0].diff = self._dx_cache
+            else:
+                bottom[0].diff = dx
 
 
 @register_layer("LRN")

@@ -454,13 +454,21 @@ std::vector<Tensor> conv2d_forward_ex(const Tensor& x, const Tensor& w,
 Tensor conv2d_backward_input(const Tensor& w, const Tensor& dy,
                              std::vector<int64_t> x_shape, int sh, int sw,
                              int ph, int pw, int G,
-                             const c10::optional<Tensor>& wkT_cache) {
+                             const c10::optional<Tensor>& wkT_cache,
+                             const c10::optional<Tensor>& dx_out) {
   check_float_like(dy, "dy");
   const bool bf16 = is_bf16(dy);
   auto dy_cl = cl4(dy);
   int Co = w.size(0), kh = w.size(2), kw = w.size(3);
-  Tensor dx = at::empty({x_shape[0], x_shape[1], x_shape[2], x_shape[3]},
-                        dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  // persistent dx (layer cache): keeps the consumer's dy identity stable
+  // across iterations so net-level multi-tensor batching (bias colsum)
+  // can key on it
+  Tensor dx = (dx_out.has_value() && dx_out->sizes() ==
+               at::IntArrayRef(x_shape) && dx_out->scalar_type() ==
+               dy.scalar_type())
+      ? *dx_out
+      : at::empty({x_shape[0], x_shape[1], x_shape[2], x_shape[3]},
+                  dy.options().memory_format(at::MemoryFormat::ChannelsLast));
   ConvGeom g;
   g.N = x_shape[0]; g.C = x_shape[1]; g.H = x_shape[2]; g.W = x_shape[3];
   g.kh = kh; g.kw = kw; g.sh = sh; g.sw = sw; g.ph = ph; g.pw = pw; g.G = G;
@@ -551,7 +559,7 @@ Tensor conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
                                   c10::optional<Tensor> db_out,
                                   int sh, int sw, int ph, int pw, int G,
                                   const c10::optional<Tensor>& dwk_buf,
-                                  bool skip_unpack) {
+                                  bool skip_unpack, bool skip_db) {
   check_float_like(dy, "dy");
   auto dy_cl = cl4(dy);
   auto x_cl = cl4(x);
@@ -607,7 +615,7 @@ Tensor conv2d_backward_weight_acc(const Tensor& x, const Tensor& colT,
     ps_weight_from_khwc_f32(dwk.data_ptr<float>(), dw_out.data_ptr<float>(),
                             Co, Cig, kh, kw, /*ld=*/Kgw, /*beta=*/1.0f,
                             stream());
-  if (db_out.has_value()) {
+  if (db_out.has_value() && !skip_db) {
     if (is_bf16(dy_cl))
       ps_colsum_bf16(dy_cl.data_ptr(), db_out->data_ptr<float>(), NP, Co,
                      stream());
@@ -677,7 +685,9 @@ Tensor concat_channels(std::vector<Tensor> inputs) {
 // split the wide NHWC tensor into per-range narrow tensors (concat backward
 // / slice forward), up to 4 ranges per launch
 std::vector<Tensor> split_channels(const Tensor& x,
-                                   std::vector<int64_t> sizes) {
+                                   std::vector<int64_t> sizes,
+                                   const c10::optional<std::vector<Tensor>>&
+                                       outs_cache) {
   auto xc = cl4(x);
   int64_t N = xc.size(0), H = xc.size(2), W = xc.size(3);
   int64_t rows = N * H * W;
@@ -686,9 +696,22 @@ std::vector<Tensor> split_channels(const Tensor& x,
   const int V = bf16 ? 8 : 4;
   std::vector<Tensor> outs;
   outs.reserve(sizes.size());
-  for (int64_t c : sizes)
+  const bool reuse = outs_cache.has_value() &&
+      outs_cache->size() == sizes.size() &&
+      std::all_of(outs_cache->begin(), outs_cache->end(),
+                  [&](const Tensor& t) {
+                    return t.scalar_type() == x.scalar_type();
+                  });
+  for (size_t j = 0; j < sizes.size(); ++j) {
+    int64_t c = sizes[j];
+    if (reuse && (*outs_cache)[j].sizes() ==
+        at::IntArrayRef({N, c, H, W})) {
+      outs.push_back((*outs_cache)[j]);
+      continue;
+    }
     outs.push_back(at::empty({N, c, H, W},
         x.options().memory_format(at::MemoryFormat::ChannelsLast)));
+  }
   bool aligned = true;
   for (int64_t c : sizes) aligned = aligned && (c % V == 0);
   if (aligned) {
@@ -782,15 +805,20 @@ std::vector<Tensor> pool_max_forward(const Tensor& x, int kh, int kw, int sh,
 
 Tensor pool_max_backward(const Tensor& dy, const Tensor& mask,
                          std::vector<int64_t> x_shape, int kh, int kw, int sh,
-                         int sw, int ph, int pw) {
+                         int sw, int ph, int pw,
+                         const c10::optional<Tensor>& dx_out) {
   auto dy_cl = cl4(dy);
   auto mask_cl = cl4(mask);
   PoolGeom g;
   g.N = x_shape[0]; g.C = x_shape[1]; g.H = x_shape[2]; g.W = x_shape[3];
   g.kh = kh; g.kw = kw; g.sh = sh; g.sw = sw; g.ph = ph; g.pw = pw;
   g.Ho = dy_cl.size(2); g.Wo = dy_cl.size(3);
-  Tensor dx = at::empty({g.N, g.C, g.H, g.W},
-                        dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  Tensor dx = (dx_out.has_value() &&
+               dx_out->sizes() == at::IntArrayRef(x_shape) &&
+               dx_out->scalar_type() == dy.scalar_type())
+      ? *dx_out
+      : at::empty({g.N, g.C, g.H, g.W},
+                  dy.options().memory_format(at::MemoryFormat::ChannelsLast));
   if (is_bf16(dy))
     ps_maxpool_bwd_bf16(dy_cl.data_ptr(), mask_cl.data_ptr<uint8_t>(),
                         dx.data_ptr(), &g, stream());
@@ -1318,6 +1346,51 @@ void zero_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
              stream());
 }
 
+// layout mirror of ps::ColsumDesc (elementwise.hip)
+struct ColsumDescHost {
+  const void* dy;
+  float* db;
+  int64_t R;
+  int64_t rows_per;  // rows per chunk (block)
+  int C;
+};
+
+// one launch sums every deferred conv bias gradient: descs keyed on the
+// (now identity-stable) activation-grad buffers. Chunk.off = start row.
+std::vector<Tensor> colsum_mt_prepare(std::vector<Tensor> dys,
+                                      std::vector<Tensor> dbs) {
+  std::vector<ColsumDescHost> descs(dys.size());
+  std::vector<MTChunkHost> chunks;
+  for (size_t t = 0; t < dys.size(); ++t) {
+    const Tensor& d = dys[t];
+    // NO cl4() here: a layout copy would leave the desc pointing at a
+    // temporary. The conv top diffs are channels-last by construction.
+    TORCH_CHECK(d.is_cuda() && d.dim() == 4 &&
+                d.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                dbs[t].scalar_type() == at::kFloat,
+                "colsum_mt: channels-last CUDA dy required");
+    int C = (int)d.size(1);
+    int64_t R = d.numel() / C;
+    // ~256K elements per block: coarse enough that terminal atomics on
+    // the C-wide bias vectors stay rare
+    int64_t rows_per = std::max<int64_t>(4, (262144 / C + 3) & ~3LL);
+    descs[t] = {d.data_ptr(), dbs[t].data_ptr<float>(), R, rows_per, C};
+    for (int64_t r = 0; r < R; r += rows_per)
+      chunks.push_back({(int)t, r});
+  }
+  return {blob_to_dev(descs.data(), descs.size() * sizeof(ColsumDescHost),
+                      dys[0]),
+          blob_to_dev(chunks.data(), chunks.size() * sizeof(MTChunkHost),
+                      dys[0]),
+          at::scalar_tensor((int64_t)chunks.size())};
+}
+
+void colsum_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
+                   int64_t nchunks, bool bf16) {
+  ps_colsum_mt(desc_dev.data_ptr(), chunk_dev.data_ptr(), (int)nchunks,
+               bf16 ? 1 : 0, stream());
+}
+
 struct MTUnpackDescHost {
   const float* dwk;
   float* dw;
@@ -1406,6 +1479,19 @@ void repack_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
                stream());
 }
 
+// standalone bias colsum (eager fallback for the net-level batch)
+void colsum_acc(const Tensor& dy, Tensor db) {
+  TORCH_CHECK(dy.dim() == 4 &&
+              dy.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int C = (int)dy.size(1);
+  int64_t R = dy.numel() / C;
+  if (is_bf16(dy))
+    ps_colsum_bf16(dy.data_ptr(), db.data_ptr<float>(), R, C, stream());
+  else
+    ps_colsum_f32(dy.data_ptr<float>(), db.data_ptr<float>(), R, C,
+                  stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1454,6 +1540,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("zero_mt_prepare", &zero_mt_prepare);
   m.def("zero_mt_run", &zero_mt_run);
   m.def("repack_mt_prepare", &repack_mt_prepare);
+  m.def("colsum_mt_prepare", &colsum_mt_prepare);
+  m.def("colsum_mt_run", &colsum_mt_run);
+  m.def("colsum_acc", &colsum_acc);
   m.def("unpack_mt_prepare", &unpack_mt_prepare);
   m.def("unpack_mt_run", &unpack_mt_run);
   m.def("repack_mt_run", &repack_mt_run);

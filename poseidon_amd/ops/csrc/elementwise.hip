@@ -318,6 +318,73 @@ static inline dim3 colsum_flat_grid(int64_t nvec) {
 #define PS_EW_LAUNCH(name, ...)                                         \
   name<<<ew_grid(n), 256, 0, s>>>(__VA_ARGS__)
 
+// ---------------------------------------------------------------------------
+// Multi-tensor bias colsum: one launch covers every deferred conv bias
+// gradient (GoogLeNet: ~57 per-layer colsum launches -> 1). Each block
+// owns a row slab of one tensor (chunk.off = start row), partial-reduces
+// in LDS like colsum_k, then atomicAdds its C partials into db.
+// ---------------------------------------------------------------------------
+struct ColsumDesc {
+  const void* dy;
+  float* db;
+  int64_t R;
+  int64_t rows_per;
+  int C;
+};
+struct ColsumChunk {
+  int t;
+  int64_t off;
+};
+
+template <typename T>
+__global__ void colsum_mt_k(const ColsumDesc* __restrict__ descs,
+                            const ColsumChunk* __restrict__ chunks) {
+  typedef T vec2 __attribute__((ext_vector_type(2)));
+  __shared__ float part[4][128];
+  const ColsumChunk ck = chunks[blockIdx.x];
+  const ColsumDesc d = descs[ck.t];
+  const T* in = (const T*)d.dy;
+  const int C = d.C;
+  const int lane = threadIdx.x & 63;
+  const int rg = threadIdx.x >> 6;
+  const int64_t r0 = ck.off;
+  const int64_t r1 = min(d.R, ck.off + d.rows_per);
+  const bool v2 = (C % 2) == 0;
+  const int span = v2 ? 128 : 64;
+  for (int c0 = 0; c0 < C; c0 += span) {
+    float acc0 = 0.f, acc1 = 0.f;
+    if (v2) {
+      int c = c0 + lane * 2;
+      if (c < C) {
+        for (int64_t r = r0 + rg; r < r1; r += 4) {
+          vec2 v = *reinterpret_cast<const vec2*>(&in[r * C + c]);
+          acc0 += to_f32(v[0]);
+          acc1 += to_f32(v[1]);
+        }
+      }
+      part[rg][lane * 2] = acc0;
+      part[rg][lane * 2 + 1] = acc1;
+    } else {
+      int c = c0 + lane;
+      if (c < C)
+        for (int64_t r = r0 + rg; r < r1; r += 4)
+          acc0 += to_f32(in[r * C + c]);
+      part[rg][lane] = acc0;
+    }
+    __syncthreads();
+    if (rg < 2) {
+      int idx = rg * 64 + lane;
+      int c = c0 + idx;
+      if (idx < span && c < C) {
+        float v = part[0][idx] + part[1][idx] + part[2][idx] + part[3][idx];
+        atomicAdd(&d.db[c], v);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+
 extern "C" {
 
 #define PS_DEF_UNARY(opname, kern)                                            \
@@ -481,6 +548,17 @@ void ps_colsum_bf16(const void* in, float* out, int64_t R, int C, hipStream_t s)
     PS_COLSUM_FLAT(__bf16, 8, (const __bf16*)in);
   else
     colsum_k<__bf16><<<colsum_grid(R, C), 256, 0, s>>>((const __bf16*)in, out, R, C);
+}
+
+void ps_colsum_mt(const void* descs, const void* chunks, int nchunks,
+                  int bf16, hipStream_t s) {
+  if (nchunks <= 0) return;
+  if (bf16)
+    colsum_mt_k<__bf16><<<dim3((unsigned)nchunks), 256, 0, s>>>(
+        (const ColsumDesc*)descs, (const ColsumChunk*)chunks);
+  else
+    colsum_mt_k<float><<<dim3((unsigned)nchunks), 256, 0, s>>>(
+        (const ColsumDesc*)descs, (const ColsumChunk*)chunks);
 }
 
 }  // extern "C"

@@ -104,11 +104,11 @@ def conv2d_forward_ex(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor
 
 def conv2d_backward_input(w: torch.Tensor, dy: torch.Tensor,
                           x_shape, stride, pad, groups: int,
-                          wkT_cache=None) -> torch.Tensor:
+                          wkT_cache=None, dx_out=None) -> torch.Tensor:
     if dy.is_cuda:
         return _ext().conv2d_backward_input(w, dy, list(x_shape), stride[0],
                                             stride[1], pad[0], pad[1], groups,
-                                            wkT_cache)
+                                            wkT_cache, dx_out)
     return torch.nn.grad.conv2d_input(list(x_shape), w, dy, stride=stride,
                                       padding=pad, groups=groups)
 
@@ -117,7 +117,8 @@ def conv2d_backward_weight_acc(x: torch.Tensor, colT, dy: torch.Tensor,
                                dw: torch.Tensor, db: Optional[torch.Tensor],
                                stride, pad, groups: int,
                                dwk_buf: Optional[torch.Tensor] = None,
-                               skip_unpack: bool = False
+                               skip_unpack: bool = False,
+                               skip_db: bool = False
                                ) -> Optional[torch.Tensor]:
     """Accumulates into dw (NCHW) and db. Returns the khwc dwk scratch used
     on GPU: hand it back as dwk_buf on later iterations (keeping it zeroed
@@ -126,7 +127,7 @@ def conv2d_backward_weight_acc(x: torch.Tensor, colT, dy: torch.Tensor,
     if dy.is_cuda:
         return _ext().conv2d_backward_weight_acc(
             x, colT, dy, dw, db, stride[0], stride[1], pad[0], pad[1],
-            groups, dwk_buf, skip_unpack)
+            groups, dwk_buf, skip_unpack, skip_db)
     dw.add_(torch.nn.grad.conv2d_weight(x, list(dw.shape), dy, stride=stride,
                                         padding=pad, groups=groups))
     if db is not None:
@@ -209,10 +210,11 @@ def pool_max_forward(x, k, s, p):
     return y, mask
 
 
-def pool_max_backward(dy, mask, x_shape, k=None, s_=None, p=None):
+def pool_max_backward(dy, mask, x_shape, k=None, s_=None, p=None,
+                      dx_out=None):
     if dy.is_cuda:
         return _ext().pool_max_backward(dy, mask, list(x_shape), k[0], k[1],
-                                        s_[0], s_[1], p[0], p[1])
+                                        s_[0], s_[1], p[0], p[1], dx_out)
     N, C, H, W = x_shape
     dx = torch.zeros(N, C, H * W, dtype=dy.dtype, device=dy.device)
     dx.scatter_add_(2, mask.view(N, C, -1).long(), dy.view(N, C, -1))
@@ -546,6 +548,20 @@ def unpack_mt_run(mt) -> None:
     _ext().unpack_mt_run(d, c, n)
 
 
+def colsum_mt_prepare(dys, dbs):
+    d, c, n = _ext().colsum_mt_prepare(list(dys), list(dbs))
+    return d, c, int(n)
+
+
+def colsum_mt_run(mt, bf16: bool) -> None:
+    d, c, n = mt
+    _ext().colsum_mt_run(d, c, n, bf16)
+
+
+def colsum_acc(dy, db) -> None:
+    _ext().colsum_acc(dy, db)
+
+
 def zero_mt_prepare(tensors):
     d, c, n = _ext().zero_mt_prepare(list(tensors))
     return d, c, int(n.item())
@@ -595,11 +611,13 @@ def slice_channels(x, c_off: int, c_len: int):
     return x.narrow(1, c_off, c_len).contiguous()
 
 
-def split_channels(x, sizes):
+def split_channels(x, sizes, outs_cache=None):
     """All channel ranges in one go: up to 4 ranges per kernel launch
-    (concat backward / slice forward over inception joins)."""
+    (concat backward / slice forward over inception joins). outs_cache:
+    persistent output tensors (keeps downstream grad identities stable
+    for net-level batching)."""
     if x.is_cuda and x.dim() == 4:
-        return _ext().split_channels(x, [int(s) for s in sizes])
+        return _ext().split_channels(x, [int(s) for s in sizes], outs_cache)
     out, off = [], 0
     for s in sizes:
         out.append(x.narrow(1, off, int(s)).contiguous())

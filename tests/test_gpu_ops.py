@@ -480,7 +480,7 @@ def test_deferred_wgrad_unpack_matches_eager():
         for l in s.net.layers:
             if l.type_name == "CONVOLUTION":
                 l.defer_unpack = defer
-        s.step(4)
+        s.step(7)  # >=3 settling iters + several batched-colsum iters
         torch.cuda.synchronize()
         results.append({i: ps.blob.data.clone().cpu()
                         for i, ps in enumerate(s.net.params)
