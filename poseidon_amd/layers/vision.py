@@ -100,9 +100,14 @@ class ConvolutionLayer(Layer):
         # solver only sets defer_unpack when there is no reducer).
         defer = (single and getattr(self, "defer_unpack", False)
                  and self._dwk_cache is not None)
-        defer_db = defer and self.bias_term
         for i, (bo, t) in enumerate(zip(bottom, top)):
             dy = t.diff
+            # batched colsum requires the CL layout the kernels produce; a
+            # dy reshaped from an IP/flatten consumer stays on the in-call
+            # path
+            defer_db = (defer and self.bias_term and dy.is_cuda
+                        and dy.dim() == 4 and dy.is_contiguous(
+                            memory_format=torch.channels_last))
             db = self.blobs[1].diff.view(-1) if self.bias_term else None
             cache = self._colT[i]
             colT, wkT = cache if isinstance(cache, tuple) else (cache, None)

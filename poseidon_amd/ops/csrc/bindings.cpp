@@ -1481,14 +1481,13 @@ void repack_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
 
 // standalone bias colsum (eager fallback for the net-level batch)
 void colsum_acc(const Tensor& dy, Tensor db) {
-  TORCH_CHECK(dy.dim() == 4 &&
-              dy.is_contiguous(at::MemoryFormat::ChannelsLast));
-  int C = (int)dy.size(1);
-  int64_t R = dy.numel() / C;
-  if (is_bf16(dy))
-    ps_colsum_bf16(dy.data_ptr(), db.data_ptr<float>(), R, C, stream());
+  auto d = dy.dim() == 4 ? cl4(dy) : dy.contiguous();
+  int C = (int)(dy.dim() == 4 ? d.size(1) : d.size(-1));
+  int64_t R = d.numel() / C;
+  if (is_bf16(d))
+    ps_colsum_bf16(d.data_ptr(), db.data_ptr<float>(), R, C, stream());
   else
-    ps_colsum_f32(dy.data_ptr<float>(), db.data_ptr<float>(), R, C,
+    ps_colsum_f32(d.data_ptr<float>(), db.data_ptr<float>(), R, C,
                   stream());
 }
 
