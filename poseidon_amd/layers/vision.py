@@ -53,6 +53,7 @@ class ConvolutionLayer(Layer):
         self._dx_cache = {}     # persistent dgrad outputs (stable dy
                                 # identities for downstream batching)
         self._pending_colsum = None  # (dy, db) for the net-level batch
+        self._last_dy_ptr = None     # stability tracker for the batch
         channels = bottom[0].channels
         assert channels % self.group == 0 and self.num_output % self.group == 0
 
@@ -102,12 +103,17 @@ class ConvolutionLayer(Layer):
                  and self._dwk_cache is not None)
         for i, (bo, t) in enumerate(zip(bottom, top)):
             dy = t.diff
-            # batched colsum requires the CL layout the kernels produce; a
-            # dy reshaped from an IP/flatten consumer stays on the in-call
-            # path
-            defer_db = (defer and self.bias_term and dy.is_cuda
-                        and dy.dim() == 4 and dy.is_contiguous(
-                            memory_format=torch.channels_last))
+            # Defer the bias colsum into the net-level batch ONLY when this
+            # layer's dy identity repeated from last iteration (producers
+            # with persistent grad buffers): an unstable dy would force a
+            # table rebuild (H2D) every step. Unstable ones keep the
+            # in-call colsum, which also overlaps on the backward streams.
+            ptr = (dy.data_ptr() if dy.is_cuda and dy.dim() == 4
+                   and dy.is_contiguous(memory_format=torch.channels_last)
+                   else None)
+            defer_db = (defer and self.bias_term and ptr is not None
+                        and ptr == self._last_dy_ptr)
+            self._last_dy_ptr = ptr
             db = self.blobs[1].diff.view(-1) if self.bias_term else None
             cache = self._colT[i]
             colT, wkT = cache if isinstance(cache, tuple) else (cache, None)
