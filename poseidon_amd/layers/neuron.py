@@ -38,9 +38,12 @@ class ReLULayer(NeuronLayer):
 
     def backward(self, top, propagate_down, bottom) -> None:
         if propagate_down[0]:
-            # in-place safe: sign(bottom.data)==sign(x) after overwrite
+            # in-place safe: sign(bottom.data)==sign(x) after overwrite.
+            # The GPU masks dy IN ITS OWN BUFFER (single-consumer DAG via
+            # insert_splits): the diff tensor identity survives, so
+            # downstream convs can batch their bias colsums.
             bottom[0].diff = ops.relu_backward(bottom[0].data, top[0].diff,
-                                               self.slope)
+                                               self.slope, in_place=True)
 
 
 @register_layer("SIGMOID")

@@ -1074,11 +1074,17 @@ Tensor relu_forward(const Tensor& x, double slope) {
   return y;
 }
 
-Tensor relu_backward(const Tensor& x, const Tensor& dy, double slope) {
+Tensor relu_backward(const Tensor& x, const Tensor& dy, double slope,
+                     bool in_place) {
   auto xc = any_contig(x);
   auto dyc = xc.dim() == 4 && xc.is_contiguous(at::MemoryFormat::ChannelsLast)
                  ? cl4(dy) : dy.contiguous();
-  Tensor dx = at::empty_like(xc);
+  // in-place (dx == dy): elementwise same-index, safe; keeps the grad
+  // buffer identity stable for the net-level colsum batch (a fresh dx
+  // every iteration churned every fused-relu conv's dy pointer)
+  Tensor dx = (in_place && dyc.is_alias_of(dy) &&
+               dyc.data_ptr() == dy.data_ptr())
+                  ? dyc : at::empty_like(xc);
   if (is_bf16(x))
     ps_relu_bwd_bf16(xc.data_ptr(), dyc.data_ptr(), dx.data_ptr(), xc.numel(),
                      (float)slope, stream());

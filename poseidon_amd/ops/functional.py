@@ -381,9 +381,12 @@ def relu_forward(x, negative_slope: float = 0.0):
     return F.leaky_relu(x, negative_slope)
 
 
-def relu_backward(x, dy, negative_slope: float = 0.0):
+def relu_backward(x, dy, negative_slope: float = 0.0,
+                  in_place: bool = False):
+    """in_place: write dx into dy's buffer (elementwise same-index safe).
+    Keeps the grad identity stable for net-level batching."""
     if x.is_cuda:
-        return _ext().relu_backward(x, dy, negative_slope)
+        return _ext().relu_backward(x, dy, negative_slope, in_place)
     return torch.where(x > 0, dy, dy * negative_slope)
 
 
