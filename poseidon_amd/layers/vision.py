@@ -108,7 +108,9 @@ class ConvolutionLayer(Layer):
             # with persistent grad buffers): an unstable dy would force a
             # table rebuild (H2D) every step. Unstable ones keep the
             # in-call colsum, which also overlaps on the backward streams.
+            vecw = 8 if dy.dtype == torch.bfloat16 else 4
             ptr = (dy.data_ptr() if dy.is_cuda and dy.dim() == 4
+                   and dy.shape[1] % vecw == 0
                    and dy.is_contiguous(memory_format=torch.channels_last)
                    else None)
             defer_db = (defer and self.bias_term and ptr is not None

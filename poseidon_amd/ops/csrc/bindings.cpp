@@ -1371,10 +1371,12 @@ std::vector<Tensor> colsum_mt_prepare(std::vector<Tensor> dys,
     const Tensor& d = dys[t];
     // NO cl4() here: a layout copy would leave the desc pointing at a
     // temporary. The conv top diffs are channels-last by construction.
+    const int vec = 16 / (int)d.element_size();
     TORCH_CHECK(d.is_cuda() && d.dim() == 4 &&
                 d.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                d.size(1) % vec == 0 &&
                 dbs[t].scalar_type() == at::kFloat,
-                "colsum_mt: channels-last CUDA dy required");
+                "colsum_mt: channels-last CUDA dy with C % 16B required");
     int C = (int)d.size(1);
     int64_t R = d.numel() / C;
     // ~256K elements per block: coarse enough that terminal atomics on

@@ -336,46 +336,44 @@ struct ColsumChunk {
   int64_t off;
 };
 
+// 16 B vector loads, one 64*VEC-column band (covers every conv bias width
+// up to 512 bf16 / 256 f32 in a single slab pass -- the 4 B banded form
+// re-read each slab per 128-column band and measured 3x SLOWER than the
+// per-layer launches it replaced). Requires C % VEC == 0 (prepare checks).
 template <typename T>
 __global__ void colsum_mt_k(const ColsumDesc* __restrict__ descs,
                             const ColsumChunk* __restrict__ chunks) {
-  typedef T vec2 __attribute__((ext_vector_type(2)));
-  __shared__ float part[4][128];
+  constexpr int VEC = 16 / (int)sizeof(T);
+  typedef T vecT __attribute__((ext_vector_type(VEC)));
+  __shared__ float part[4][64 * VEC];
   const ColsumChunk ck = chunks[blockIdx.x];
   const ColsumDesc d = descs[ck.t];
   const T* in = (const T*)d.dy;
   const int C = d.C;
+  const int CV = C / VEC;
   const int lane = threadIdx.x & 63;
   const int rg = threadIdx.x >> 6;
   const int64_t r0 = ck.off;
   const int64_t r1 = min(d.R, ck.off + d.rows_per);
-  const bool v2 = (C % 2) == 0;
-  const int span = v2 ? 128 : 64;
+  const int span = 64 * VEC;
   for (int c0 = 0; c0 < C; c0 += span) {
-    float acc0 = 0.f, acc1 = 0.f;
-    if (v2) {
-      int c = c0 + lane * 2;
-      if (c < C) {
-        for (int64_t r = r0 + rg; r < r1; r += 4) {
-          vec2 v = *reinterpret_cast<const vec2*>(&in[r * C + c]);
-          acc0 += to_f32(v[0]);
-          acc1 += to_f32(v[1]);
-        }
+    float acc[VEC];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) acc[j] = 0.f;
+    const int cv = c0 / VEC + lane;  // this lane's vector chunk
+    if (cv < CV) {
+      for (int64_t r = r0 + rg; r < r1; r += 4) {
+        vecT v = *(((const vecT*)&in[r * C]) + cv);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) acc[j] += to_f32(v[j]);
       }
-      part[rg][lane * 2] = acc0;
-      part[rg][lane * 2 + 1] = acc1;
-    } else {
-      int c = c0 + lane;
-      if (c < C)
-        for (int64_t r = r0 + rg; r < r1; r += 4)
-          acc0 += to_f32(in[r * C + c]);
-      part[rg][lane] = acc0;
     }
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) part[rg][lane * VEC + j] = acc[j];
     __syncthreads();
-    if (rg < 2) {
-      int idx = rg * 64 + lane;
+    for (int idx = threadIdx.x; idx < span; idx += 256) {
       int c = c0 + idx;
-      if (idx < span && c < C) {
+      if (c < C) {
         float v = part[0][idx] + part[1][idx] + part[2][idx] + part[3][idx];
         atomicAdd(&d.db[c], v);
       }
