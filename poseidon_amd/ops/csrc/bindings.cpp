@@ -1386,17 +1386,20 @@ std::vector<Tensor> colsum_mt_prepare(std::vector<Tensor> dys,
     for (int64_t r = 0; r < R; r += rows_per)
       chunks.push_back({(int)t, r});
   }
+  int max_c = 0;
+  for (auto& de : descs) max_c = std::max(max_c, de.C);
   return {blob_to_dev(descs.data(), descs.size() * sizeof(ColsumDescHost),
                       dys[0]),
           blob_to_dev(chunks.data(), chunks.size() * sizeof(MTChunkHost),
                       dys[0]),
-          at::scalar_tensor((int64_t)chunks.size())};
+          at::scalar_tensor((int64_t)chunks.size()),
+          at::scalar_tensor((int64_t)max_c)};
 }
 
 void colsum_mt_run(const Tensor& desc_dev, const Tensor& chunk_dev,
-                   int64_t nchunks, bool bf16) {
+                   int64_t nchunks, bool bf16, int64_t max_c) {
   ps_colsum_mt(desc_dev.data_ptr(), chunk_dev.data_ptr(), (int)nchunks,
-               bf16 ? 1 : 0, stream());
+               bf16 ? 1 : 0, (int)max_c, stream());
 }
 
 struct MTUnpackDescHost {

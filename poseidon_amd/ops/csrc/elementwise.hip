@@ -320,10 +320,12 @@ static inline dim3 colsum_flat_grid(int64_t nvec) {
 
 // ---------------------------------------------------------------------------
 // Multi-tensor bias colsum: one launch covers every deferred conv bias
-// gradient (GoogLeNet: ~57 per-layer colsum launches -> 1). Each block
-// owns a row slab of one tensor (chunk.off = start row), partial-reduces
-// in LDS like colsum_k, then atomicAdds its C partials into db.
-// ---------------------------------------------------------------------------
+// gradient. 16 B vector loads; thread t owns column-chunk (t % CV) and
+// row-phase (t / CV), so ALL 256 threads stream the slab even for C=16
+// reduce layers (a lane-per-chunk form stranded 62/64 lanes there and a
+// 4 B banded form re-read slabs per band: both measured ~3x SLOWER than
+// the launches they replaced). Flush via LDS atomics once per block,
+// then one global atomicAdd per column. Dynamic LDS = max C floats.
 struct ColsumDesc {
   const void* dy;
   float* db;
@@ -336,52 +338,56 @@ struct ColsumChunk {
   int64_t off;
 };
 
-// 16 B vector loads, one 64*VEC-column band (covers every conv bias width
-// up to 512 bf16 / 256 f32 in a single slab pass -- the 4 B banded form
-// re-read each slab per 128-column band and measured 3x SLOWER than the
-// per-layer launches it replaced). Requires C % VEC == 0 (prepare checks).
 template <typename T>
 __global__ void colsum_mt_k(const ColsumDesc* __restrict__ descs,
                             const ColsumChunk* __restrict__ chunks) {
   constexpr int VEC = 16 / (int)sizeof(T);
   typedef T vecT __attribute__((ext_vector_type(VEC)));
-  __shared__ float part[4][64 * VEC];
+  extern __shared__ float part[];  // max C floats
   const ColsumChunk ck = chunks[blockIdx.x];
   const ColsumDesc d = descs[ck.t];
   const T* in = (const T*)d.dy;
   const int C = d.C;
   const int CV = C / VEC;
-  const int lane = threadIdx.x & 63;
-  const int rg = threadIdx.x >> 6;
-  const int64_t r0 = ck.off;
+  for (int c = threadIdx.x; c < C; c += 256) part[c] = 0.f;
+  __syncthreads();
+  const int nph = 256 / CV;  // row phases; threads >= nph*CV idle
+  const int t = (int)threadIdx.x;
   const int64_t r1 = min(d.R, ck.off + d.rows_per);
-  const int span = 64 * VEC;
-  for (int c0 = 0; c0 < C; c0 += span) {
-    float acc[VEC];
+  if (nph == 0) {
+    // C > 2048 elements: single-phase strided over column chunks
+    for (int cv = t; cv < CV; cv += 256) {
+      float acc[VEC];
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) acc[j] = 0.f;
-    const int cv = c0 / VEC + lane;  // this lane's vector chunk
-    if (cv < CV) {
-      for (int64_t r = r0 + rg; r < r1; r += 4) {
+      for (int j = 0; j < VEC; ++j) acc[j] = 0.f;
+      for (int64_t r = ck.off; r < r1; ++r) {
         vecT v = *(((const vecT*)&in[r * C]) + cv);
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += to_f32(v[j]);
       }
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        atomicAdd(&part[cv * VEC + j], acc[j]);
+    }
+  } else if (t < nph * CV) {
+    const int cv = t % CV;
+    const int ph = t / CV;
+    float acc[VEC];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) acc[j] = 0.f;
+    for (int64_t r = ck.off + ph; r < r1; r += nph) {
+      vecT v = *(((const vecT*)&in[r * C]) + cv);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) acc[j] += to_f32(v[j]);
     }
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) part[rg][lane * VEC + j] = acc[j];
-    __syncthreads();
-    for (int idx = threadIdx.x; idx < span; idx += 256) {
-      int c = c0 + idx;
-      if (c < C) {
-        float v = part[0][idx] + part[1][idx] + part[2][idx] + part[3][idx];
-        atomicAdd(&d.db[c], v);
-      }
-    }
-    __syncthreads();
+    for (int j = 0; j < VEC; ++j)
+      atomicAdd(&part[cv * VEC + j], acc[j]);
   }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += 256)
+    atomicAdd(&d.db[c], part[c]);
 }
-
 
 extern "C" {
 
@@ -549,13 +555,14 @@ void ps_colsum_bf16(const void* in, float* out, int64_t R, int C, hipStream_t s)
 }
 
 void ps_colsum_mt(const void* descs, const void* chunks, int nchunks,
-                  int bf16, hipStream_t s) {
+                  int bf16, int max_c, hipStream_t s) {
   if (nchunks <= 0) return;
+  const size_t lds = (size_t)max_c * sizeof(float);
   if (bf16)
-    colsum_mt_k<__bf16><<<dim3((unsigned)nchunks), 256, 0, s>>>(
+    colsum_mt_k<__bf16><<<dim3((unsigned)nchunks), 256, lds, s>>>(
         (const ColsumDesc*)descs, (const ColsumChunk*)chunks);
   else
-    colsum_mt_k<float><<<dim3((unsigned)nchunks), 256, 0, s>>>(
+    colsum_mt_k<float><<<dim3((unsigned)nchunks), 256, lds, s>>>(
         (const ColsumDesc*)descs, (const ColsumChunk*)chunks);
 }
 

@@ -137,7 +137,7 @@ void ps_dropout_fwd_bf16(const void*, void*, uint8_t*, int64_t, float,
 void ps_dropout_bwd_bf16(const void*, const uint8_t*, void*, int64_t, float,
                          hipStream_t);
 void ps_colsum_bf16(const void*, float*, int64_t, int, hipStream_t);
-void ps_colsum_mt(const void*, const void*, int, int, hipStream_t);
+void ps_colsum_mt(const void*, const void*, int, int, int, hipStream_t);
 void ps_threshold_fwd_f32(const float*, float*, int64_t, float, hipStream_t);
 void ps_threshold_fwd_bf16(const void*, void*, int64_t, float, hipStream_t);
 void ps_eltwise_max_fwd_f32(const float*, const float*, float*, uint8_t*,

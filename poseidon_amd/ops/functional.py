@@ -552,13 +552,13 @@ def unpack_mt_run(mt) -> None:
 
 
 def colsum_mt_prepare(dys, dbs):
-    d, c, n = _ext().colsum_mt_prepare(list(dys), list(dbs))
-    return d, c, int(n)
+    d, c, n, mc = _ext().colsum_mt_prepare(list(dys), list(dbs))
+    return d, c, int(n), int(mc)
 
 
 def colsum_mt_run(mt, bf16: bool) -> None:
-    d, c, n = mt
-    _ext().colsum_mt_run(d, c, n, bf16)
+    d, c, n, mc = mt
+    _ext().colsum_mt_run(d, c, n, bf16, mc)
 
 
 def colsum_acc(dy, db) -> None:
