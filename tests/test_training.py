@@ -245,3 +245,48 @@ def test_netoutputs_flattens_vector_outputs(tmp_path):
     solver.write_net_outputs(str(out))
     header = (tmp_path / "vec.netoutputs").read_text().splitlines()[0]
     assert header.startswith("iter,time,loss,") and "sm_5" in header
+
+
+def test_lenet_cpu_single_process_config1():
+    """BASELINE.json config 1: LeNet on MNIST-shaped data, CPU solver,
+    single process (plumbing slice). Synthetic two-class 28x28 digits
+    stand in for MNIST (no datasets in the image); the zoo LeNet must
+    train end-to-end and beat chance decisively."""
+    import numpy as np
+    from poseidon_amd.models import zoo
+    pa.init(device="cpu", seed=19)
+    npb = zoo.build_net("lenet", batch=32)
+    # swap the data layer for MEMORY_DATA so we can feed synthetic digits
+    from poseidon_amd.proto import Message
+    np2 = Message("NetParameter", name="lenet_cpu")
+    for lp in npb.layers:
+        if lp.enum_name("type") in ("DUMMY_DATA", "DATA"):
+            d = np2.add("layers", name="data", type="MEMORY_DATA")
+            d.top.append("data")
+            d.top.append("label")
+            mp = d.ensure("memory_data_param")
+            mp.batch_size = 32
+            mp.channels, mp.height, mp.width = 1, 28, 28
+        else:
+            np2.layers.append(lp)
+    sp = _solver_param(base_lr=0.01, max_iter=10**9)
+    sp.net_param = np2
+    solver = SGDSolver(sp, verbose=False)
+    g = torch.Generator().manual_seed(4)
+    n = 256
+    labels = torch.randint(0, 10, (n,), generator=g)
+    data = torch.zeros(n, 1, 28, 28)
+    for i in range(n):  # class k = bright kth row band + noise
+        r = int(labels[i]) * 2 + 4
+        data[i, 0, r:r + 2, 4:24] = 1.0
+    data += 0.1 * torch.randn(n, 1, 28, 28, generator=g)
+    solver.net.layers[0].add_data(data, labels.float())
+    first = float(solver.net.forward())
+    solver.step(120)
+    solver.net.forward()
+    acc = float(solver.net.blobs["accuracy"].data) \
+        if "accuracy" in solver.net.blobs else None
+    loss = float(solver.net.blobs["loss"].data)
+    assert loss < first * 0.3, (first, loss)
+    if acc is not None:
+        assert acc > 0.8, acc
