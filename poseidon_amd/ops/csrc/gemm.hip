@@ -112,6 +112,19 @@ __device__ inline void stage_glds(T* lds, const T* __restrict__ src,
   constexpr int CHUNKS = ROWS / RPC;
   const int r_in = lane / LPR;
   const int slot = lane % LPR;
+  // GATHER: the column decode (k -> kh,kw,cg) depends only on the lane's
+  // slot and the row-swizzle bit -- hoist both variants out of the chunk
+  // loop (the full per-chunk gather_addr decode measured 8.6 VALU per
+  // MFMA on the implicit conv forward)
+  int g_kkh[2], g_kkw[2], g_cg[2];
+  if (GATHER) {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      const int kcp = (slot ^ (p << 1)) * EPB;
+      int khw = fdiv_fix(k0 + kcp, ga->Cg, ga->inv_Cg, g_cg[p]);
+      g_kkh[p] = fdiv_fix(khw, ga->kw, ga->inv_kw, g_kkw[p]);
+    }
+  }
 #pragma unroll
   for (int ci = wid; ci < CHUNKS; ci += 4) {
     const int row = ci * RPC + r_in;
@@ -126,9 +139,24 @@ __device__ inline void stage_glds(T* lds, const T* __restrict__ src,
     // paper, but flipping slot bit 0 reorders 16 B chunks within 32 B
     // pairs on the WRITE side and measured -5%% end-to-end on AlexNet/VGG
     // -- the global request coalescer penalty outweighs the LDS win.)
-    const int kc = (slot ^ (((row >> 2) & 1) << 1)) * EPB;
-    const T* g = GATHER ? gather_addr<T>(*ga, row0 + row, k0 + kc)
-                        : src + (int64_t)(row0 + row) * lda + k0 + kc;
+    const int swb = (row >> 2) & 1;
+    const int kc = (slot ^ (swb << 1)) * EPB;
+    const T* g;
+    if (GATHER) {
+      int ow, oh;
+      const int t2 = fdiv_fix(row0 + row, ga->Wo, ga->inv_Wo, ow);
+      const int n2 = fdiv_fix(t2, ga->Ho, ga->inv_Ho, oh);
+      const int ih = oh * ga->sh - ga->ph + g_kkh[swb];
+      const int iw = ow * ga->sw - ga->pw + g_kkw[swb];
+      const bool oob = (k0 + kc >= ga->kg_max) || ih < 0 || ih >= ga->H ||
+                       iw < 0 || iw >= ga->W;
+      g = oob ? (const T*)ga->zero
+              : (const T*)ga->x +
+                    (((int64_t)n2 * ga->H + ih) * ga->W + iw) * ga->C +
+                    ga->c0 + g_cg[swb];
+    } else {
+      g = src + (int64_t)(row0 + row) * lda + k0 + kc;
+    }
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g,
         (__attribute__((address_space(3))) void*)(lds + ci * (1024 / (int)sizeof(T))),
