@@ -171,10 +171,10 @@ class DataLayer(_PrefetchingDataLayer):
         shared = bool(dp.shared_file_system) or c.world_size == 1
         if not shared:
             source = f"{source}_{c.rank}"
-        # LMDB environments (reference convert_imageset output) are read
-        # directly by the pure-python walker in data/lmdb_io.py; PDB files
-        # keep the native container path. LevelDB dirs (log/sst format)
-        # are not parsed -- point users at the converter.
+        # Reference-produced environments are read directly by the
+        # pure-python walkers: LMDB (data/lmdb_io.py) and LevelDB
+        # incl. snappy blocks + log replay (data/leveldb_io.py); PDB
+        # files keep the native container path.
         is_lmdb = (os.path.isdir(source)
                    and os.path.exists(os.path.join(source, "data.mdb"))) or (
                   os.path.isfile(source) and source.endswith(".mdb"))
@@ -182,10 +182,8 @@ class DataLayer(_PrefetchingDataLayer):
             from ..data.lmdb_io import LmdbReader
             self.db = LmdbReader(source)
         elif os.path.isdir(source):
-            raise RuntimeError(
-                f"{source}: directory is not an LMDB environment (no "
-                "data.mdb). LevelDB-format datasets must be converted: "
-                "python -m poseidon_amd.tools.datasets convert ... (PDB)")
+            from ..data.leveldb_io import LevelDbReader
+            self.db = LevelDbReader(source)
         else:
             self.db = PDBReader(source)
         self.stride = c.world_size if shared else 1

@@ -87,15 +87,24 @@ def convert_db(argv=None):
     ap.add_argument("--to-lmdb", action="store_true")
     args = ap.parse_args(argv)
     from ..data.lmdb_io import LmdbReader, LmdbWriter
+    from ..data.leveldb_io import LevelDbReader
     import os
-    src_is_lmdb = os.path.isdir(args.src) or args.src.endswith(".mdb")
+
+    def open_src(p):
+        if os.path.isdir(p):
+            if os.path.exists(os.path.join(p, "data.mdb")):
+                return LmdbReader(p)
+            return LevelDbReader(p)  # reference create_*.sh default
+        if p.endswith(".mdb"):
+            return LmdbReader(p)
+        return PDBReader(p)
+
+    db = open_src(args.src)
     if args.to_lmdb:
-        db = LmdbReader(args.src) if src_is_lmdb else PDBReader(args.src)
         with LmdbWriter(args.dst) as w:
             for i in range(len(db)):
                 w.put(b"%08d" % i, db.get_raw(i))
     else:
-        db = LmdbReader(args.src) if src_is_lmdb else PDBReader(args.src)
         with PDBWriter(args.dst) as w:
             for i in range(len(db)):
                 w.put_raw(db.get_raw(i))
