@@ -245,6 +245,7 @@ class Net:
                 self.layer_need_bwd[li] = False
 
         self._fuse_relu_epilogues()
+        self._fuse_relu_bwd_into_concats()
         for (li, ti, w) in self._loss_tops:
             self._loss_marks.setdefault(li, []).append((ti, w))
         # Loss layers whose loss top feeds NO consumer get the seeded weight
@@ -256,6 +257,49 @@ class Net:
             tname = p.layers[li].top[ti] if ti < len(p.layers[li].top) else None
             if tname is not None and tname not in consumed:
                 self.layers[li].seeded_loss_weight = w
+
+    def _fuse_relu_bwd_into_concats(self) -> None:
+        """GoogLeNet inception branches end relu -> concat: the ReLU's
+        backward (dy *= x>0) folds into the concat's backward scatter
+        (split_channels relu_masks), deleting one elementwise kernel per
+        branch per step (36 of GoogLeNet's 65 relu_bwd launches). GPU,
+        in-place slope-0 ReLUs, channel concat with vector-aligned widths
+        and all-propagate bottoms only; PS_NO_CONCAT_MASK=1 disables (used
+        by the equivalence test)."""
+        import os
+        if ctx().device != "cuda" or os.environ.get("PS_NO_CONCAT_MASK"):
+            return
+        p = self.param
+        producer_layer: Dict[str, int] = {}
+        for li, lp in enumerate(p.layers):
+            for t in lp.top:
+                producer_layer[t] = li
+        vecw = 8 if ctx().compute_dtype == torch.bfloat16 else 4
+        for ci, lp in enumerate(p.layers):
+            if lp.enum_name("type") != "CONCAT":
+                continue
+            concat = self.layers[ci]
+            if getattr(concat, "dim", 1) != 1:
+                continue
+            if not all(self.bottom_need_bwd[ci]):
+                continue
+            if any(b.channels % vecw for b in self.bottoms[ci]):
+                continue  # masked path needs the aligned fused kernel
+            mask_idx = set()
+            for j, bname in enumerate(lp.bottom):
+                pj = producer_layer.get(bname)
+                if pj is None:
+                    continue
+                pl = p.layers[pj]
+                relu = self.layers[pj]
+                if (pl.enum_name("type") == "RELU"
+                        and pl.top[0] == pl.bottom[0]  # in-place
+                        and getattr(relu, "slope", 1.0) == 0.0
+                        and relu.loss_weights[0] == 0.0):
+                    mask_idx.add(j)
+                    relu.bwd_fused_into_consumer = True
+            if mask_idx:
+                concat._mask_bottoms = mask_idx
 
     def _build_stream_schedule(self, n_streams: int = 4) -> None:
         """Dataflow schedule for inter-branch stream parallelism.

@@ -116,9 +116,17 @@ class ConcatLayer(Layer):
         chan4 = self.dim == 1 and len(bottom[0].shape) == 4
         if chan4 and all(propagate_down):
             cache = getattr(self, "_split_cache", None)
+            # fused consumer-ReLU backward: bottoms flagged by the net pass
+            # get their gradient zeroed where the (post-relu) activation is
+            # zero, right in the scatter -- the ReLU layer skips backward
+            mask_idx = getattr(self, "_mask_bottoms", None)
+            masks = None
+            if mask_idx:
+                masks = [bottom[j].data if j in mask_idx else None
+                         for j in range(len(bottom))]
             parts = ops.split_channels(top[0].diff,
                                        [b.shape[1] for b in bottom],
-                                       outs_cache=cache)
+                                       outs_cache=cache, relu_masks=masks)
             if parts and parts[0].is_cuda:
                 if (cache is None or len(cache) != len(parts) or any(
                         c.data_ptr() != p.data_ptr()

@@ -37,6 +37,8 @@ class ReLULayer(NeuronLayer):
         top[0].data = ops.relu_forward(bottom[0].data, self.slope)
 
     def backward(self, top, propagate_down, bottom) -> None:
+        if getattr(self, "bwd_fused_into_consumer", False):
+            return  # consumer CONCAT's split already applied the mask
         if propagate_down[0]:
             # in-place safe: sign(bottom.data)==sign(x) after overwrite.
             # The GPU masks dy IN ITS OWN BUFFER (single-consumer DAG via

@@ -661,10 +661,10 @@ Tensor concat_channels(std::vector<Tensor> inputs) {
       }
       if (bf16)
         ps_chan_concat4_bf16(y.data_ptr(), ptrs, c_end, n, c_begin, rows,
-                             C_out, 0, stream());
+                             C_out, 0, nullptr, stream());
       else
         ps_chan_concat4_f32(y.data_ptr<float>(), ptrs, c_end, n, c_begin,
-                            rows, C_out, 0, stream());
+                            rows, C_out, 0, nullptr, stream());
     }
     return y;
   }
@@ -687,7 +687,9 @@ Tensor concat_channels(std::vector<Tensor> inputs) {
 std::vector<Tensor> split_channels(const Tensor& x,
                                    std::vector<int64_t> sizes,
                                    const c10::optional<std::vector<Tensor>>&
-                                       outs_cache) {
+                                       outs_cache,
+                                   const c10::optional<std::vector<Tensor>>&
+                                       relu_masks) {
   auto xc = cl4(x);
   int64_t N = xc.size(0), H = xc.size(2), W = xc.size(3);
   int64_t rows = N * H * W;
@@ -718,20 +720,33 @@ std::vector<Tensor> split_channels(const Tensor& x,
     int off = 0;
     for (size_t i0 = 0; i0 < outs.size(); i0 += 4) {
       void* ptrs[4];
+      const void* masks[4];
+      bool any_mask = false;
       int c_end[4];
       int n = (int)std::min<size_t>(4, outs.size() - i0);
       int c_begin = off;
       for (int j = 0; j < n; ++j) {
         ptrs[j] = outs[i0 + j].data_ptr();
+        masks[j] = nullptr;
+        if (relu_masks.has_value() && i0 + j < relu_masks->size() &&
+            (*relu_masks)[i0 + j].defined()) {
+          const Tensor& mk = (*relu_masks)[i0 + j];
+          TORCH_CHECK(mk.scalar_type() == x.scalar_type() &&
+                      mk.numel() == outs[i0 + j].numel(),
+                      "split mask dtype/shape mismatch");
+          masks[j] = mk.data_ptr();
+          any_mask = true;
+        }
         off += (int)sizes[i0 + j];
         c_end[j] = off;
       }
       if (bf16)
         ps_chan_concat4_bf16(xc.data_ptr(), ptrs, c_end, n, c_begin, rows,
-                             C_in, 1, stream());
+                             C_in, 1, any_mask ? masks : nullptr, stream());
       else
         ps_chan_concat4_f32(xc.data_ptr<float>(), ptrs, c_end, n, c_begin,
-                            rows, C_in, 1, stream());
+                            rows, C_in, 1, any_mask ? masks : nullptr,
+                            stream());
     }
   } else {
     int off = 0;

@@ -433,12 +433,16 @@ __global__ void chan_copy_k(const T* __restrict__ in, T* __restrict__ out,
 // over the <=4 channel ranges picks the narrow tensor it pairs with.
 struct Chan4 {
   void* p[4];    // narrow tensors (src for concat, dst for split)
+  const void* m[4];  // optional relu masks (post-relu activations): split
+                     // outputs are zeroed where the activation is zero --
+                     // the consumer ReLU's backward fused into the scatter
   int c_end[4];  // exclusive channel end of each range in the wide tensor
   int c_begin;   // first channel this launch covers
   int n;         // live entries
 };
 
-template <typename T, int V, bool GATHER>  // GATHER: wide->narrow (split)
+template <typename T, int V, bool GATHER,  // GATHER: wide->narrow (split)
+          bool MASKED = false>
 __global__ void chan_concat4_k(T* __restrict__ wide, Chan4 t, int64_t rows,
                                int C_wide) {
   typedef T vec_t __attribute__((ext_vector_type(V)));
@@ -456,10 +460,20 @@ __global__ void chan_concat4_k(T* __restrict__ wide, Chan4 t, int64_t rows,
     const int Cn = t.c_end[j] - c0;
     T* nar = (T*)t.p[j] + r * Cn + (c - c0);
     vec_t* wp = reinterpret_cast<vec_t*>(&wide[r * C_wide + c]);
-    if (GATHER)
-      *reinterpret_cast<vec_t*>(nar) = *wp;
-    else
+    if (GATHER) {
+      vec_t v = *wp;
+      if (MASKED && t.m[j]) {
+        // post-relu activations are never negative: x == 0 <=> clamped
+        const vec_t x = *reinterpret_cast<const vec_t*>(
+            (const T*)t.m[j] + r * Cn + (c - c0));
+#pragma unroll
+        for (int e = 0; e < V; ++e)
+          if (to_f32(x[e]) == 0.0f) v[e] = (T)0.0f;
+      }
+      *reinterpret_cast<vec_t*>(nar) = v;
+    } else {
       *wp = *reinterpret_cast<const vec_t*>(nar);
+    }
   }
 }
 
@@ -485,12 +499,13 @@ extern "C" {
 // Requirements checked by caller: every range width and boundary divisible
 // by V. gather=1 splits wide->narrow, 0 concats narrow->wide.
 static inline Chan4 chan4_pack(void* const* ptrs, const int* c_end, int n,
-                               int c_begin) {
+                               int c_begin, const void* const* masks) {
   Chan4 t;
   t.n = n;
   t.c_begin = c_begin;
   for (int i = 0; i < 4; ++i) {
     t.p[i] = i < n ? const_cast<void*>(ptrs[i]) : nullptr;
+    t.m[i] = (masks && i < n) ? masks[i] : nullptr;
     t.c_end[i] = i < n ? c_end[i] : (n ? c_end[n - 1] : 0);
   }
   return t;
@@ -498,10 +513,14 @@ static inline Chan4 chan4_pack(void* const* ptrs, const int* c_end, int n,
 
 void ps_chan_concat4_f32(float* wide, void* const* ptrs, const int* c_end,
                          int n, int c_begin, int64_t rows, int C_wide,
-                         int gather, hipStream_t s) {
-  Chan4 t = chan4_pack(ptrs, c_end, n, c_begin);
+                         int gather, const void* const* masks,
+                         hipStream_t s) {
+  Chan4 t = chan4_pack(ptrs, c_end, n, c_begin, masks);
   int64_t work = rows * ((c_end[n - 1] - c_begin) / 4);
-  if (gather)
+  if (gather && masks)
+    chan_concat4_k<float, 4, true, true>
+        <<<ew_grid(work), 256, 0, s>>>(wide, t, rows, C_wide);
+  else if (gather)
     chan_concat4_k<float, 4, true>
         <<<ew_grid(work), 256, 0, s>>>(wide, t, rows, C_wide);
   else
@@ -510,10 +529,14 @@ void ps_chan_concat4_f32(float* wide, void* const* ptrs, const int* c_end,
 }
 void ps_chan_concat4_bf16(void* wide, void* const* ptrs, const int* c_end,
                           int n, int c_begin, int64_t rows, int C_wide,
-                          int gather, hipStream_t s) {
-  Chan4 t = chan4_pack(ptrs, c_end, n, c_begin);
+                          int gather, const void* const* masks,
+                          hipStream_t s) {
+  Chan4 t = chan4_pack(ptrs, c_end, n, c_begin, masks);
   int64_t work = rows * ((c_end[n - 1] - c_begin) / 8);
-  if (gather)
+  if (gather && masks)
+    chan_concat4_k<__bf16, 8, true, true>
+        <<<ew_grid(work), 256, 0, s>>>((__bf16*)wide, t, rows, C_wide);
+  else if (gather)
     chan_concat4_k<__bf16, 8, true>
         <<<ew_grid(work), 256, 0, s>>>((__bf16*)wide, t, rows, C_wide);
   else

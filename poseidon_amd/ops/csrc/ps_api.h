@@ -228,9 +228,9 @@ void ps_dropout_fwd_bf16_offdev(const void*, void*, uint8_t*, int64_t, float,
 
 // im2col.hip
 void ps_chan_concat4_f32(float*, void* const*, const int*, int, int, int64_t,
-                         int, int, hipStream_t);
+                         int, int, const void* const*, hipStream_t);
 void ps_chan_concat4_bf16(void*, void* const*, const int*, int, int, int64_t,
-                          int, int, hipStream_t);
+                          int, int, const void* const*, hipStream_t);
 void ps_chan_copy_f32(const float*, float*, int64_t, int, int, int, hipStream_t);
 void ps_chan_copy_bf16(const void*, void*, int64_t, int, int, int, hipStream_t);
 void ps_chan_slice_f32(const float*, float*, int64_t, int, int, int, hipStream_t);

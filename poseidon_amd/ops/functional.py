@@ -614,13 +614,16 @@ def slice_channels(x, c_off: int, c_len: int):
     return x.narrow(1, c_off, c_len).contiguous()
 
 
-def split_channels(x, sizes, outs_cache=None):
+def split_channels(x, sizes, outs_cache=None, relu_masks=None):
     """All channel ranges in one go: up to 4 ranges per kernel launch
     (concat backward / slice forward over inception joins). outs_cache:
     persistent output tensors (keeps downstream grad identities stable
-    for net-level batching)."""
+    for net-level batching). relu_masks: per-output post-relu activation
+    tensors -- outputs are zeroed where the activation is zero, fusing
+    the consumer ReLU's backward into the scatter."""
     if x.is_cuda and x.dim() == 4:
-        return _ext().split_channels(x, [int(s) for s in sizes], outs_cache)
+        return _ext().split_channels(x, [int(s) for s in sizes], outs_cache,
+                                     relu_masks)
     out, off = [], 0
     for s in sizes:
         out.append(x.narrow(1, off, int(s)).contiguous())

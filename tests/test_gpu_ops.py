@@ -488,3 +488,62 @@ def test_deferred_wgrad_unpack_matches_eager():
     for i in results[0]:
         assert torch.allclose(results[0][i], results[1][i],
                               rtol=1e-5, atol=1e-6), f"param {i}"
+
+
+def test_concat_fused_relu_bwd_matches_eager():
+    """Inception-style branches (conv+relu -> concat): the ReLU backward
+    fused into the concat scatter must train identically to the eager
+    per-layer relu_bwd (PS_NO_CONCAT_MASK=1 baseline)."""
+    import os
+    from poseidon_amd.proto import Message, parse_text
+    from poseidon_amd.solver.solver import SGDSolver
+
+    def net_param():
+        return parse_text("NetParameter", """
+            name: "mini_inception"
+            layers { name: "data" type: DUMMY_DATA top: "data" top: "label"
+                     dummy_data_param { num: 8 channels: 16 height: 8
+                         width: 8 num: 8 channels: 1 height: 1 width: 1
+                         data_filler { type: "gaussian" std: 1.0 }
+                         data_filler { type: "constant" } } }
+            layers { name: "b1" type: CONVOLUTION bottom: "data" top: "b1"
+                     convolution_param { num_output: 16 kernel_size: 1
+                         weight_filler { type: "xavier" } } }
+            layers { name: "rb1" type: RELU bottom: "b1" top: "b1" }
+            layers { name: "b2" type: CONVOLUTION bottom: "data" top: "b2"
+                     convolution_param { num_output: 24 kernel_size: 3 pad: 1
+                         weight_filler { type: "xavier" } } }
+            layers { name: "rb2" type: RELU bottom: "b2" top: "b2" }
+            layers { name: "cc" type: CONCAT bottom: "b1" bottom: "b2"
+                     top: "cc" }
+            layers { name: "ip" type: INNER_PRODUCT bottom: "cc" top: "ip"
+                     inner_product_param { num_output: 4
+                         weight_filler { type: "xavier" } } }
+            layers { name: "loss" type: SOFTMAX_LOSS bottom: "ip"
+                     bottom: "label" top: "loss" }
+        """)
+
+    results = []
+    for disable in ("1", ""):
+        if disable:
+            os.environ["PS_NO_CONCAT_MASK"] = disable
+        else:
+            os.environ.pop("PS_NO_CONCAT_MASK", None)
+        pa.init(device="cuda", seed=99)
+        sp = Message("SolverParameter", base_lr=0.05, lr_policy="fixed",
+                     momentum=0.9, weight_decay=0.001, max_iter=10,
+                     display=0, snapshot=0)
+        sp.net_param = net_param()
+        s = SGDSolver(sp, verbose=False)
+        cc = next(l for l in s.net.layers if l.name == "cc")
+        if not disable:
+            assert getattr(cc, "_mask_bottoms", None) == {0, 1}
+        s.step(5)
+        torch.cuda.synchronize()
+        results.append({i: ps.blob.data.clone().cpu()
+                        for i, ps in enumerate(s.net.params)
+                        if ps.owner == i})
+    os.environ.pop("PS_NO_CONCAT_MASK", None)
+    for i in results[0]:
+        assert torch.allclose(results[0][i], results[1][i],
+                              rtol=1e-5, atol=1e-6), f"param {i}"
