@@ -105,11 +105,12 @@ def _varint32(data: bytes, i: int) -> Tuple[int, int]:
 # SSTable reading
 # ---------------------------------------------------------------------------
 
-def _read_block(raw: bytes, offset: int, size: int) -> bytes:
+def _read_block(raw, offset: int, size: int) -> bytes:
     """BlockHandle points at block contents; a 5-byte trailer follows:
-    1 byte compression type (0 raw, 1 snappy) + 4 byte crc."""
+    1 byte compression type (0 raw, 1 snappy) + 4 byte crc. `raw` may be
+    bytes or an mmap."""
     ctype = raw[offset + size]
-    body = raw[offset:offset + size]
+    body = bytes(raw[offset:offset + size])
     if ctype == 0:
         return body
     if ctype == 1:
@@ -225,27 +226,48 @@ def _iter_log(path: str) -> Iterator[Tuple[bytes, int, int, bytes]]:
 class LevelDbReader:
     """Read-only merge over every table + log file of an environment dir.
     Records exposed positionally in key order (Caffe writes zero-padded
-    decimal keys, so key order == insertion order)."""
+    decimal keys, so key order == insertion order).
 
-    def __init__(self, path: str):
+    Memory model: the index pass stores (file, block, entry) LOCATORS for
+    table records -- values are fetched lazily per block with a small LRU
+    of decompressed blocks, so an ImageNet-sized (tens of GB) LevelDB
+    only costs its key index in RAM. Log records (normally a tail of
+    recent writes) are held in memory."""
+
+    def __init__(self, path: str, block_cache: int = 32):
         self.path = path
         if not os.path.isdir(path):
             raise ValueError(f"{path}: LevelDB environments are directories")
-        best: Dict[bytes, Tuple[int, int, bytes]] = {}
+        # key -> (seq, type, locator); locator = bytes (log value) or
+        # (file_idx, block_off, block_size, entry_idx)
+        best: Dict[bytes, Tuple[int, int, object]] = {}
+        self._files: List[str] = []
+        self._mms: List[object] = []
+        from collections import OrderedDict
+        self._cache: "OrderedDict" = OrderedDict()
+        self._cache_max = block_cache
         n_files = 0
         for name in sorted(os.listdir(path)):
             full = os.path.join(path, name)
             try:
                 if name.endswith((".ldb", ".sst")):
-                    src = _iter_table(full)
+                    fi = len(self._files)
+                    self._files.append(full)
+                    f = open(full, "rb")
+                    import mmap as _mmap
+                    self._mms.append(_mmap.mmap(f.fileno(), 0,
+                                                access=_mmap.ACCESS_READ))
+                    for key, seq, typ, loc in self._index_table(fi):
+                        cur = best.get(key)
+                        if cur is None or seq >= cur[0]:
+                            best[key] = (seq, typ, loc)
                 elif name.endswith(".log"):
-                    src = _iter_log(full)
+                    for key, seq, typ, val in _iter_log(full):
+                        cur = best.get(key)
+                        if cur is None or seq >= cur[0]:
+                            best[key] = (seq, typ, val)
                 else:
                     continue
-                for key, seq, typ, val in src:
-                    cur = best.get(key)
-                    if cur is None or seq >= cur[0]:
-                        best[key] = (seq, typ, val)
                 n_files += 1
             except ValueError as e:
                 raise ValueError(f"{full}: {e}") from e
@@ -253,7 +275,42 @@ class LevelDbReader:
             raise ValueError(
                 f"{path}: no .ldb/.sst/.log files (not a LevelDB dir?)")
         self._keys = sorted(k for k, (s, t, v) in best.items() if t == 1)
-        self._vals = {k: best[k][2] for k in self._keys}
+        self._loc = {k: best[k][2] for k in self._keys}
+
+    def _index_table(self, fi: int):
+        raw = self._mms[fi]
+        if len(raw) < 48:
+            return
+        footer = bytes(raw[-48:])
+        if struct.unpack_from("<Q", footer, 40)[0] != MAGIC:
+            raise ValueError("bad sstable magic")
+        i = 0
+        _, i = _varint32(footer, i)
+        _, i = _varint32(footer, i)
+        idx_off, i = _varint32(footer, i)
+        idx_size, i = _varint32(footer, i)
+        index = _read_block(raw, idx_off, idx_size)
+        for _, handle in _block_entries(index):
+            off, j = _varint32(handle, 0)
+            size, j = _varint32(handle, j)
+            block = self._block(fi, off, size)
+            for ei, (ikey, _val) in enumerate(_block_entries(block)):
+                if len(ikey) < 8:
+                    continue
+                tag = struct.unpack_from("<Q", ikey, len(ikey) - 8)[0]
+                yield ikey[:-8], tag >> 8, tag & 0xFF, (fi, off, size, ei)
+
+    def _block(self, fi: int, off: int, size: int) -> bytes:
+        ck = (fi, off)
+        blk = self._cache.get(ck)
+        if blk is None:
+            blk = _read_block(self._mms[fi], off, size)
+            self._cache[ck] = blk
+            if len(self._cache) > self._cache_max:
+                self._cache.popitem(last=False)
+        else:
+            self._cache.move_to_end(ck)
+        return blk
 
     def __len__(self) -> int:
         return len(self._keys)
@@ -261,19 +318,30 @@ class LevelDbReader:
     def key(self, i: int) -> bytes:
         return self._keys[i]
 
+    def _fetch(self, loc) -> bytes:
+        if isinstance(loc, bytes):
+            return loc  # log-resident value
+        fi, off, size, ei = loc
+        block = self._block(fi, off, size)
+        for j, (_k, v) in enumerate(_block_entries(block)):
+            if j == ei:
+                return v
+        raise ValueError("entry index out of range")
+
     def get_raw(self, i: int) -> bytes:
-        return self._vals[self._keys[i]]
+        return self._fetch(self._loc[self._keys[i]])
 
     def get(self, i: int):
         from ..proto import Message
         return Message.decode("Datum", self.get_raw(i))
 
     def get_by_key(self, key: bytes) -> Optional[bytes]:
-        return self._vals.get(key)
+        loc = self._loc.get(key)
+        return None if loc is None else self._fetch(loc)
 
     def __iter__(self) -> Iterator[Tuple[bytes, bytes]]:
         for k in self._keys:
-            yield k, self._vals[k]
+            yield k, self._fetch(self._loc[k])
 
 
 # ---------------------------------------------------------------------------

@@ -1403,6 +1403,7 @@ std::vector<Tensor> colsum_mt_prepare(std::vector<Tensor> dys,
   }
   int max_c = 0;
   for (auto& de : descs) max_c = std::max(max_c, de.C);
+  TORCH_CHECK(max_c <= 8192, "colsum_mt: C exceeds the LDS budget");
   return {blob_to_dev(descs.data(), descs.size() * sizeof(ColsumDescHost),
                       dys[0]),
           blob_to_dev(chunks.data(), chunks.size() * sizeof(MTChunkHost),
