@@ -188,7 +188,12 @@ class LmdbWriter:
         self.psize = psize
         # mdb: values >= nodemax go to overflow pages
         self.nodemax = ((psize - PAGEHDRSZ) // 2) & ~1
-        self._pages: List[bytes] = []      # data pages, pgno = index + 2
+        # pages stream straight to disk (ImageNet-scale conversions must
+        # not buffer the dataset); the two meta pages are back-patched at
+        # close. Page allocation is strictly sequential.
+        self._f = open(path, "wb")
+        self._f.write(b"\x00" * (2 * psize))  # meta placeholders
+        self._next_pgno = 2
         self._leaf: List[Tuple[bytes, Tuple[int, bytes]]] = []  # pending nodes
         self._leaf_fill = 0
         self._leaf_firsts: List[Tuple[bytes, int]] = []  # (first key, pgno)
@@ -197,7 +202,7 @@ class LmdbWriter:
         self._last_key: Optional[bytes] = None
 
     def _new_pgno(self) -> int:
-        return len(self._pages) + 2
+        return self._next_pgno
 
     def _emit_page(self, flags: int, nodes: List[Tuple[bytes, bytes]],
                    pgno: int) -> None:
@@ -217,9 +222,7 @@ class LmdbWriter:
         struct.pack_into("<QHHHH", page, 0, pgno, 0, flags, lower, upper)
         struct.pack_into(f"<{nk}H", page, PAGEHDRSZ, *ptrs)
         page[upper:upper + len(body)] = body
-        while len(self._pages) + 2 <= pgno:
-            self._pages.append(b"")  # placeholder (pre-reserved pgno)
-        self._pages[pgno - 2] = bytes(page)
+        self._write_page(pgno, bytes(page))
 
     def put(self, key: bytes, value: bytes) -> None:
         if isinstance(key, str):
@@ -255,13 +258,17 @@ class LmdbWriter:
         self._entries += 1
 
     def _new_pgno_reserve(self, n: int) -> int:
-        pgno = self._new_pgno()
-        self._pages.extend([b""] * n)
+        pgno = self._next_pgno
+        self._next_pgno += n
         return pgno
 
+    def _write_page(self, pgno: int, data: bytes) -> None:
+        assert len(data) % self.psize == 0
+        self._f.seek(pgno * self.psize)
+        self._f.write(data)
+
     def _set_pages(self, pgno: int, data: bytes, npages: int) -> None:
-        for j in range(npages):
-            self._pages[pgno - 2 + j] = data[j * self.psize:(j + 1) * self.psize]
+        self._write_page(pgno, data)
 
     def _emit_leaf(self) -> None:
         pgno = self._new_pgno_reserve(1)
@@ -314,7 +321,7 @@ class LmdbWriter:
             depth += 1
             root = level[0][1]
         leaf_pages = len(self._leaf_firsts)
-        last_pg = len(self._pages) + 1
+        last_pg = self._next_pgno - 1
         # meta pages: page 0 stale (txnid 0), page 1 live (txnid 1)
         def meta(pgno: int, txnid: int) -> bytes:
             page = bytearray(self.psize)
@@ -338,11 +345,10 @@ class LmdbWriter:
             struct.pack_into("<QQ", page, m + 24 + 96, last_pg, txnid)
             return bytes(page)
 
-        with open(self.path, "wb") as f:
-            f.write(meta(0, 0))
-            f.write(meta(1, 1))
-            for p in self._pages:
-                f.write(p if p else b"\x00" * self.psize)
+        self._f.seek(0)
+        self._f.write(meta(0, 0))
+        self._f.write(meta(1, 1))
+        self._f.close()
 
     def __enter__(self):
         return self
