@@ -173,3 +173,20 @@ def test_leveldb_reads_snappy_compressed_table(tmp_path):
     assert len(r) == 50
     for i in range(50):
         assert r.get_raw(i) == bytes([i]) * 100
+
+
+def test_convert_db_from_leveldb(tmp_path):
+    """convert_db ingests a LevelDB environment directly (reference
+    create_*.sh output) into PDB."""
+    from poseidon_amd.tools.datasets import convert_db
+    from poseidon_amd.data.pdb import PDBReader
+    env = str(tmp_path / "src_leveldb")
+    with LevelDbWriter(env) as w:
+        for i in range(20):
+            w.put(b"%08d" % i, bytes([i]) * (30 + i))
+    dst = str(tmp_path / "out.pdb")
+    convert_db([env, dst])
+    r = PDBReader(dst)
+    assert len(r) == 20
+    for i in range(20):
+        assert r.get_raw(i) == bytes([i]) * (30 + i)

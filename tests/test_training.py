@@ -290,3 +290,15 @@ def test_lenet_cpu_single_process_config1():
     assert loss < first * 0.3, (first, loss)
     if acc is not None:
         assert acc > 0.8, acc
+
+
+def test_graph_capture_refused_for_non_sgd():
+    """enable_graph is SGD-only by design (Nesterov/AdaGrad keep their
+    per-param fused kernels eager)."""
+    from poseidon_amd.solver.solver import NesterovSolver
+    pa.init(device="cpu", seed=3)
+    sp = _solver_param()
+    sp.solver_type = "NESTEROV"
+    sp.net_param = _separable_net_param()
+    s = NesterovSolver(sp, verbose=False)
+    assert s.enable_graph() is False
